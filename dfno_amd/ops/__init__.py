@@ -1,0 +1,2 @@
+from .pointwise import linear_nd, add_gelu, gelu
+from .spectral import spectral_conv

@@ -1,0 +1,94 @@
+"""Fused corner-block complex spectral contraction (the FNO hot spot).
+
+Replaces the reference's per-corner sliced complex einsum over a freshly
+zeroed clone of the full truncated spectrum
+(``y = 0*x.clone(); y[sl] = torch.einsum(eqn, x[sl], w)``,
+/root/reference/dfno/dfno.py:269-271).
+
+The contraction is, per kept frequency point f:
+    y[b, o, f] = sum_i x[b, i, f] * w[i, o, f]
+i.e. a width x width complex matvec batched over frequencies.  At batch 1 and
+width ~20 this is HBM-bandwidth-bound on the *weight* stream (intensity
+~1 flop/byte), so the native kernel is a vectorized streaming VALU kernel
+(float4 = 2 complex per load) with the corner-box gather/scatter fused into
+its addressing — the truncated spectrum is touched exactly once and no
+per-corner slices are materialized.  (An MFMA formulation does not pay here:
+each weight element is used `batch` times only, far below the fp32 ridge.)
+
+Autograd: grad_x[b,i,f] = sum_o conj(w[i,o,f]) gy[b,o,f] (same kernel,
+conjugate-transposed weight); grad_w[i,o,f] = sum_b conj(x[b,i,f]) gy[b,o,f]
+(library einsum over the corner views).
+"""
+
+from __future__ import annotations
+
+from typing import List, Sequence, Tuple
+
+import torch
+
+from .. import _ext
+
+__all__ = ["spectral_conv"]
+
+
+def _corner_slices(bounds: Sequence[Tuple[int, int]]):
+    return (slice(None), slice(None)) + tuple(slice(a, b) for a, b in bounds)
+
+
+class _SpectralConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, bounds_list, out_channels: int, *weights: torch.Tensor):
+        B, I = x.shape[0], x.shape[1]
+        fdims = list(x.shape[2:])
+        y = torch.zeros((B, out_channels, *fdims), dtype=x.dtype, device=x.device)
+        if x.is_cuda and x.numel() > 0:
+            ext = _ext.get(required=True)
+            xc = x.contiguous()
+            for w, bounds in zip(weights, bounds_list):
+                starts = [a for a, _ in bounds]
+                ext.spectral_corner_fwd(xc, w.contiguous(), y, starts)
+            x_saved = xc
+        else:
+            x_saved = x
+            for w, bounds in zip(weights, bounds_list):
+                sl = _corner_slices(bounds)
+                y[sl] = torch.einsum("bi...,io...->bo...", x[sl], w)
+        ctx.save_for_backward(x_saved, *weights)
+        ctx.bounds_list = bounds_list
+        return y
+
+    @staticmethod
+    def backward(ctx, gy: torch.Tensor):
+        x = ctx.saved_tensors[0]
+        weights = ctx.saved_tensors[1:]
+        bounds_list = ctx.bounds_list
+        gy = gy.contiguous()
+        gx = torch.zeros_like(x)
+        gws = []
+        if gy.is_cuda and gy.numel() > 0:
+            ext = _ext.get(required=True)
+            for w, bounds in zip(weights, bounds_list):
+                starts = [a for a, _ in bounds]
+                ext.spectral_corner_bwd_x(gy, w.contiguous(), gx, starts)
+                sl = _corner_slices(bounds)
+                gws.append(torch.einsum("bo...,bi...->io...", gy[sl], x[sl].conj()))
+        else:
+            for w, bounds in zip(weights, bounds_list):
+                sl = _corner_slices(bounds)
+                gx[sl] = torch.einsum("bo...,io...->bi...", gy[sl], w.conj())
+                gws.append(torch.einsum("bo...,bi...->io...", gy[sl], x[sl].conj()))
+        return (gx, None, None, *gws)
+
+
+def spectral_conv(x: torch.Tensor, weights: List[torch.Tensor],
+                  bounds_list: List[List[Tuple[int, int]]],
+                  out_channels: int = None) -> torch.Tensor:
+    """Apply the corner-block spectral contraction.
+
+    x: [B, I, *F] complex; weights[c]: [I, O, *box_c] complex;
+    bounds_list[c]: per-frequency-dim (start, stop) of corner c within F.
+    Returns [B, O, *F] with zeros outside the corner boxes.
+    """
+    if out_channels is None:
+        out_channels = weights[0].shape[1] if weights else x.shape[1]
+    return _SpectralConvFn.apply(x, bounds_list, out_channels, *weights)
