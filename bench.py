@@ -1,0 +1,146 @@
+"""Driver benchmark contract: flagship two-phase 3D FNO training step.
+
+Measures sec/batch (forward + distributed relative-Lp loss + backward +
+Adam step) of the 3D two-phase-flow FNO (BASELINE.md flagship: width 20,
+modes (12,12,12,8), batch 1, T_out 30) on a fixed 64^3 global grid, strong
+scaling over 1/2/4/8 GPUs of one node (partitions 1x1x1 / 2x1x1 / 2x2x1 /
+2x2x2 over the spatial axes), synthetic data, random-init weights, fp32
+(the reference's compute dtype; complex64 spectral path).
+
+Launch:
+  python bench.py --gpus 1 --steps K --warmup W          (serial)
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N ...       (one rank per GPU, RCCL)
+
+Rank 0 prints exactly one JSON line per the driver contract.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+import dfno_amd as dfno
+from dfno_amd.partition import init_distributed, is_distributed, world_rank, world_size
+from dfno_amd.partition import compute_distribution_info
+
+
+PARTITIONS = {
+    1: (1, 1, 1, 1, 1, 1),
+    2: (1, 1, 2, 1, 1, 1),
+    4: (1, 1, 2, 2, 1, 1),
+    8: (1, 1, 2, 2, 2, 1),
+}
+
+# flagship config (BASELINE.md / BASELINE.json): 3D two-phase FNO, fixed grid
+GLOBAL_SHAPE = [1, 2, 64, 64, 64, 1]   # (batch, channels, x, y, z, t_in)
+OUT_T = 30
+WIDTH = 20
+MODES = (12, 12, 12, 8)
+NUM_BLOCKS = 4
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--width", type=int, default=WIDTH)
+    p.add_argument("--num-blocks", type=int, default=NUM_BLOCKS)
+    args = p.parse_args()
+
+    n = args.gpus
+    assert n in PARTITIONS, f"--gpus must be one of {sorted(PARTITIONS)}"
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    init_distributed()
+    if is_distributed():
+        assert world_size() == n, f"world size {world_size()} != --gpus {n}"
+
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        local_rank = int(os.environ.get("LOCAL_RANK", 0))
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
+        device = torch.device("cuda", torch.cuda.current_device())
+    else:
+        device = torch.device("cpu")
+
+    P_world, P_x, P_0 = dfno.create_standard_partitions(PARTITIONS[n])
+
+    torch.manual_seed(1234 + max(P_x.rank, 0))
+    model = dfno.DistributedFNONd(P_x, GLOBAL_SHAPE, OUT_T, args.width, MODES,
+                                  num_blocks=args.num_blocks, device=device,
+                                  dtype=torch.float32)
+    criterion = dfno.DistributedRelativeLpLoss(P_x)
+    optimizer = torch.optim.Adam(model.parameters(), lr=1e-3)
+
+    # synthetic local shards of the global tensors
+    info_x = compute_distribution_info(P_x, GLOBAL_SHAPE)
+    out_shape = [GLOBAL_SHAPE[0], 1, *GLOBAL_SHAPE[2:-1], OUT_T]
+    info_y = compute_distribution_info(P_x, out_shape)
+    x = torch.rand(*info_x["shape"], device=device)
+    y_true = torch.rand(*info_y["shape"], device=device)
+
+    def step():
+        optimizer.zero_grad(set_to_none=True)
+        y = model(x)
+        loss = criterion(y, y_true)
+        loss.backward()
+        optimizer.step()
+        return loss
+
+    dt_comm_acc = 0.0
+    for _ in range(args.warmup):
+        step()
+
+    P_x.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(args.steps):
+        step()
+        dt_comm_acc += model.dt_comm
+    if use_cuda:
+        torch.cuda.synchronize()
+    P_x.barrier()
+    t1 = time.time()
+
+    elapsed = t1 - t0
+    # max over ranks
+    elapsed = P_x.allreduce_scalar(elapsed, op="max")
+    sec_per_batch = elapsed / args.steps
+
+    if max(P_x.rank, 0) == 0:
+        print(json.dumps({
+            "metric": "sec/batch",
+            "value": sec_per_batch,
+            "unit": "s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": sec_per_batch * 1e3,
+            "higher_is_better": False,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic (random input/target shards, random-init weights)",
+            "dt_comm_per_step": dt_comm_acc / args.steps,
+            "config": {
+                "model": "dfno-two-phase-3d",
+                "global_batch": GLOBAL_SHAPE[0],
+                "grid": GLOBAL_SHAPE[2:5],
+                "t_in": GLOBAL_SHAPE[5],
+                "t_out": OUT_T,
+                "width": args.width,
+                "modes": list(MODES),
+                "num_blocks": args.num_blocks,
+                "parallelism": "spatial-model-parallel " + "x".join(map(str, PARTITIONS[n])),
+                "step": "fwd+relative-Lp-loss+bwd+adam",
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,19 @@
+// Python bindings for the dfno_amd gfx950 HIP kernels.
+#include <torch/extension.h>
+
+#include "kernels.h"
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "dfno_amd MI355X (gfx950) native kernels";
+  m.def("channel_mix_fwd", &channel_mix_fwd,
+        "fused channel linear + bias + gelu: returns (y, z)");
+  m.def("channel_mix_fwd_t", &channel_mix_fwd_t,
+        "transposed channel contraction (grad-x)");
+  m.def("gelu_fwd", &gelu_fwd, "exact gelu");
+  m.def("gelu_bwd", &gelu_bwd, "gelu backward (gy, z) -> gz");
+  m.def("add_gelu_fwd", &add_gelu_fwd, "fused residual add + gelu: returns (y, z)");
+  m.def("spectral_corner_fwd", &spectral_corner_fwd,
+        "corner-block complex spectral contraction (accumulate into y box)");
+  m.def("spectral_corner_bwd_x", &spectral_corner_bwd_x,
+        "corner-block spectral contraction adjoint wrt x");
+}

@@ -1,0 +1,26 @@
+// Host-visible declarations of the gfx950 HIP kernels (implemented in
+// pointwise.hip / spectral.hip, bound in bindings.cpp).
+#pragma once
+
+#include <torch/extension.h>
+#include <vector>
+
+// fused channel-contraction linear: y[b,o,s] = act(sum_i W[o,i] x[b,i,s] + b[o])
+// returns {y, z} where z is the pre-activation (z == y when act == false).
+std::vector<at::Tensor> channel_mix_fwd(const at::Tensor& x, const at::Tensor& W,
+                                        const at::Tensor& b, bool act);
+
+// transposed contraction for grad-x: gx[b,i,s] = sum_o W[o,i] gz[b,o,s]
+at::Tensor channel_mix_fwd_t(const at::Tensor& gz, const at::Tensor& W);
+
+at::Tensor gelu_fwd(const at::Tensor& x);
+at::Tensor gelu_bwd(const at::Tensor& gy, const at::Tensor& z);
+std::vector<at::Tensor> add_gelu_fwd(const at::Tensor& a, const at::Tensor& b);
+
+// corner-block spectral contraction on the truncated complex spectrum:
+//   y[b,o,f] += sum_i x[b,i,f] * w[i,o,f_box]  for f in the corner box
+void spectral_corner_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
+                         std::vector<int64_t> starts);
+// gx[b,i,f] += sum_o conj(w[i,o,f_box]) * gy[b,o,f]
+void spectral_corner_bwd_x(const at::Tensor& gy, const at::Tensor& w, at::Tensor& gx,
+                           std::vector<int64_t> starts);

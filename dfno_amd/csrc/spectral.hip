@@ -1,0 +1,233 @@
+// gfx950 (CDNA4) fused corner-block complex spectral contraction.
+//
+// y[b,o,f] = sum_i x[b,i,f] * w[i,o,f_box] for every frequency f inside a
+// corner box of the truncated spectrum (see ops/spectral.py header for the
+// roofline argument: weight-stream bandwidth-bound at batch ~1, so this is a
+// streaming VALU kernel, not an MFMA one).
+//
+// The corner gather/scatter is fused into the addressing: threads walk the
+// box in linear order (coalesced in the innermost frequency dim for both the
+// weight [i][o][box] stream and the spectrum [b][c][F] stream), so no
+// per-corner slice of the spectrum is ever materialized — unlike the
+// reference's y[sl] = einsum(x[sl], w) over a zeroed clone
+// (/root/reference/dfno/dfno.py:269-271).
+//
+// Parallelism: one thread owns one box element for a tile of output channels
+// (OTILE accumulators); the o-tiling multiplies thread count so small
+// per-rank corner shards still fill the 256-CU chip, at the price of
+// re-reading the (I-element) x column once per tile — noise next to the
+// I*O-element weight column.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include "kernels.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kMaxDims = 6;
+
+struct BoxGeom {
+  // box extents and the global strides (in complex elements) of the
+  // spectrum's frequency dims, plus the global offset of the box origin.
+  long box[kMaxDims];
+  long fstride[kMaxDims];
+  long origin;   // sum(starts[d] * fstride[d])
+  int nd;
+  long nelem;    // prod(box)
+};
+
+template <typename T>
+struct Cplx {
+  T re, im;
+};
+
+// acc += a * b (complex)
+template <typename T>
+__device__ __forceinline__ void cmac(T& ar, T& ai, T br, T bi, T cr, T ci) {
+  ar += br * cr - bi * ci;
+  ai += br * ci + bi * cr;
+}
+
+// acc += conj(a) * b
+template <typename T>
+__device__ __forceinline__ void cmac_conj(T& ar, T& ai, T br, T bi, T cr, T ci) {
+  ar += br * cr + bi * ci;
+  ai += br * ci - bi * cr;
+}
+
+__device__ __forceinline__ long box_to_global(long e, const BoxGeom& g) {
+  long off = g.origin;
+  // innermost dim last in box[]; decompose right-to-left
+  for (int d = g.nd - 1; d >= 0; --d) {
+    long c = e % g.box[d];
+    e /= g.box[d];
+    off += c * g.fstride[d];
+  }
+  return off;
+}
+
+// FWD: y[b, o, f] = sum_i x[b, i, f] w[i, o, e]
+// CONJT (bwd-x): gx[b, i, f] = sum_o conj(w[i, o, e]) gy[b, o, f]
+//   (same code with the roles of the w indices swapped and conjugation)
+template <typename T, int OTILE, bool CONJT>
+__global__ __launch_bounds__(kBlock) void spectral_corner_kernel(
+    const T* __restrict__ x,   // [B, CI, Ftot] complex interleaved
+    const T* __restrict__ w,   // [I, O, E] complex interleaved
+    T* __restrict__ y,         // [B, CO, Ftot] complex interleaved
+    BoxGeom g, int B, int I, int O, long Ftot) {
+  // CONJT: contraction index is o (CI = O), output index is i (CO = I)
+  const int n_out = CONJT ? I : O;
+  const int n_in = CONJT ? O : I;
+  const long ntiles = (n_out + OTILE - 1) / OTILE;
+  const long total = g.nelem * ntiles * B;
+
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < total; t += stride) {
+    long e = t % g.nelem;
+    long rest = t / g.nelem;
+    int tile = (int)(rest % ntiles);
+    int b = (int)(rest / ntiles);
+
+    long f = box_to_global(e, g);
+    int o0 = tile * OTILE;
+    int olim = min(OTILE, n_out - o0);
+
+    T accr[OTILE], acci[OTILE];
+#pragma unroll
+    for (int k = 0; k < OTILE; ++k) { accr[k] = T(0); acci[k] = T(0); }
+
+    const T* xb = x + 2 * (((long)b * n_in) * Ftot + f);
+    for (int i = 0; i < n_in; ++i) {
+      T xr = xb[2 * (long)i * Ftot];
+      T xi = xb[2 * (long)i * Ftot + 1];
+      if (CONJT) {
+        // w index: [out=i_out][in=i(=o)][e]; here i iterates o (the weight's
+        // second index), output k iterates the weight's first index.
+        for (int k = 0; k < olim; ++k) {
+          long widx = 2 * ((((long)(o0 + k)) * O + i) * g.nelem + e);
+          cmac_conj(accr[k], acci[k], w[widx], w[widx + 1], xr, xi);
+        }
+      } else {
+        const T* wb = w + 2 * (((long)i * O + o0) * g.nelem + e);
+        for (int k = 0; k < olim; ++k) {
+          cmac(accr[k], acci[k], xr, xi, wb[2 * (long)k * g.nelem],
+               wb[2 * (long)k * g.nelem + 1]);
+        }
+      }
+    }
+
+    T* yb = y + 2 * (((long)b * n_out + o0) * Ftot + f);
+    for (int k = 0; k < olim; ++k) {
+      yb[2 * (long)k * Ftot] = accr[k];
+      yb[2 * (long)k * Ftot + 1] = acci[k];
+    }
+  }
+}
+
+int grid_for_s(long work) {
+  long g = (work + kBlock - 1) / kBlock;
+  long cap = 256L * 16;
+  if (g > cap) g = cap;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+BoxGeom make_geom(const at::Tensor& x, const at::Tensor& w,
+                  const std::vector<int64_t>& starts) {
+  BoxGeom g{};
+  int nd = (int)x.dim() - 2;
+  TORCH_CHECK(nd >= 1 && nd <= kMaxDims, "spectral: bad ndim");
+  TORCH_CHECK((int)w.dim() - 2 == nd, "spectral: w ndim mismatch");
+  TORCH_CHECK((int)starts.size() == nd, "spectral: starts size mismatch");
+  g.nd = nd;
+  g.nelem = 1;
+  // frequency strides of the spectrum (contiguous layout)
+  long stride = 1;
+  long fs[kMaxDims];
+  for (int d = nd - 1; d >= 0; --d) {
+    fs[d] = stride;
+    stride *= x.size(d + 2);
+  }
+  g.origin = 0;
+  for (int d = 0; d < nd; ++d) {
+    g.box[d] = w.size(d + 2);
+    g.fstride[d] = fs[d];
+    g.origin += starts[d] * fs[d];
+    g.nelem *= g.box[d];
+  }
+  return g;
+}
+
+void check_c(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == at::kComplexFloat || t.scalar_type() == at::kComplexDouble,
+              name, " must be complex64/complex128");
+}
+
+}  // namespace
+
+void spectral_corner_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
+                         std::vector<int64_t> starts) {
+  check_c(x, "x"); check_c(w, "w"); check_c(y, "y");
+  int B = (int)x.size(0), I = (int)x.size(1), O = (int)y.size(1);
+  TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "spectral: w shape");
+  long Ftot = 1;
+  for (int d = 2; d < x.dim(); ++d) Ftot *= x.size(d);
+  BoxGeom g = make_geom(x, w, starts);
+  if (g.nelem == 0 || B == 0) return;
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  constexpr int OT = 8;
+  long ntiles = (O + OT - 1) / OT;
+  int grid = grid_for_s(g.nelem * ntiles * B);
+
+  if (x.scalar_type() == at::kComplexFloat) {
+    hipLaunchKernelGGL((spectral_corner_kernel<float, OT, false>), dim3(grid),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const float*>(x.data_ptr()),
+                       reinterpret_cast<const float*>(w.data_ptr()),
+                       reinterpret_cast<float*>(y.data_ptr()), g, B, I, O, Ftot);
+  } else {
+    hipLaunchKernelGGL((spectral_corner_kernel<double, OT, false>), dim3(grid),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const double*>(x.data_ptr()),
+                       reinterpret_cast<const double*>(w.data_ptr()),
+                       reinterpret_cast<double*>(y.data_ptr()), g, B, I, O, Ftot);
+  }
+}
+
+void spectral_corner_bwd_x(const at::Tensor& gy, const at::Tensor& w, at::Tensor& gx,
+                           std::vector<int64_t> starts) {
+  check_c(gy, "gy"); check_c(w, "w"); check_c(gx, "gx");
+  int B = (int)gy.size(0), O = (int)gy.size(1), I = (int)gx.size(1);
+  TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "spectral: w shape");
+  long Ftot = 1;
+  for (int d = 2; d < gy.dim(); ++d) Ftot *= gy.size(d);
+  BoxGeom g = make_geom(gy, w, starts);
+  if (g.nelem == 0 || B == 0) return;
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  constexpr int OT = 8;
+  long ntiles = (I + OT - 1) / OT;
+  int grid = grid_for_s(g.nelem * ntiles * B);
+
+  if (gy.scalar_type() == at::kComplexFloat) {
+    hipLaunchKernelGGL((spectral_corner_kernel<float, OT, true>), dim3(grid),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const float*>(gy.data_ptr()),
+                       reinterpret_cast<const float*>(w.data_ptr()),
+                       reinterpret_cast<float*>(gx.data_ptr()), g, B, I, O, Ftot);
+  } else {
+    hipLaunchKernelGGL((spectral_corner_kernel<double, OT, true>), dim3(grid),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const double*>(gy.data_ptr()),
+                       reinterpret_cast<const double*>(w.data_ptr()),
+                       reinterpret_cast<double*>(gx.data_ptr()), g, B, I, O, Ftot);
+  }
+}
