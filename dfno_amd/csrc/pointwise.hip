@@ -247,13 +247,12 @@ __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
 // VEC s-positions and loops over all O.
 // ---------------------------------------------------------------------------
 
-template <typename T, bool ACT>
+template <typename T, bool ACT, int TS>
 __global__ __launch_bounds__(kBlock) void channel_mix_lds_kernel(
     const T* __restrict__ x, const T* __restrict__ W, const T* __restrict__ bias,
     T* __restrict__ y, T* __restrict__ z,
     int B, int I, int O, long S, bool wt, bool has_bias, bool write_z) {
-  // tile of TS = kBlock spatial positions staged in LDS as [I][TS]
-  constexpr int TS = kBlock;
+  // tile of TS spatial positions staged in LDS as [I][TS]
   extern __shared__ __align__(16) char smem_raw[];
   T* xt = reinterpret_cast<T*>(smem_raw);  // [I][TS]
 
@@ -272,8 +271,7 @@ __global__ __launch_bounds__(kBlock) void channel_mix_lds_kernel(
     }
     __syncthreads();
 
-    int s = threadIdx.x;
-    if (s < ts) {
+    for (int s = threadIdx.x; s < ts; s += blockDim.x) {
       T* yb = y + ((long)b * O) * S + s0 + s;
       T* zb = write_z ? z + ((long)b * O) * S + s0 + s : nullptr;
       for (int o = 0; o < O; ++o) {
@@ -325,18 +323,31 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
   else if (O <= 8) { CMIX_DISPATCH(channel_mix_ores_kernel, 8) }
   else if (O <= 16) { CMIX_DISPATCH(channel_mix_ores_kernel, 16) }
   else {
-    size_t smem_lds = sizeof(T) * (size_t)I * kBlock;
-    TORCH_CHECK(smem_lds <= 160 * 1024, "channel_mix: I too large for LDS tile");
+    // LDS-staged generic path; pick the largest tile that keeps the x tile
+    // within 64 KiB (>= 2 blocks/CU of LDS headroom)
     int grid2 = grid_for((long)B * ((S + kBlock - 1) / kBlock) * kBlock, kBlock);
-    if (act) {
-      hipLaunchKernelGGL((channel_mix_lds_kernel<T, true>), dim3(grid2), dim3(kBlock),
-                         smem_lds, stream, x, W, bias, y, z, B, I, O, S, wt,
-                         has_bias, write_z);
-    } else {
-      hipLaunchKernelGGL((channel_mix_lds_kernel<T, false>), dim3(grid2), dim3(kBlock),
-                         smem_lds, stream, x, W, bias, y, z, B, I, O, S, wt,
-                         has_bias, write_z);
+#define CMIX_LDS(TS)                                                            \
+  {                                                                             \
+    size_t smem_lds = sizeof(T) * (size_t)I * TS;                               \
+    if (act) {                                                                  \
+      hipLaunchKernelGGL((channel_mix_lds_kernel<T, true, TS>), dim3(grid2),    \
+                         dim3(kBlock), smem_lds, stream, x, W, bias, y, z, B,   \
+                         I, O, S, wt, has_bias, write_z);                       \
+    } else {                                                                    \
+      hipLaunchKernelGGL((channel_mix_lds_kernel<T, false, TS>), dim3(grid2),   \
+                         dim3(kBlock), smem_lds, stream, x, W, bias, y, z, B,   \
+                         I, O, S, wt, has_bias, write_z);                       \
+    }                                                                           \
+  }
+    size_t per_s = sizeof(T) * (size_t)I;
+    if (per_s * 256 <= 64 * 1024) { CMIX_LDS(256) }
+    else if (per_s * 128 <= 64 * 1024) { CMIX_LDS(128) }
+    else if (per_s * 64 <= 64 * 1024) { CMIX_LDS(64) }
+    else {
+      TORCH_CHECK(per_s * 32 <= 160 * 1024, "channel_mix: I too large for LDS tile");
+      CMIX_LDS(32)
     }
+#undef CMIX_LDS
   }
 #undef CMIX_DISPATCH
 }
