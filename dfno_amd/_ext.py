@@ -27,6 +27,7 @@ SOURCES = [
     _CSRC / "bindings.cpp",
     _CSRC / "pointwise.hip",
     _CSRC / "spectral.hip",
+    _CSRC / "proj_head.hip",
 ]
 
 

@@ -331,6 +331,22 @@ class _RepartitionPlan:
         self.sends.sort(key=lambda t: t[0])
         self.recvs.sort(key=lambda t: t[0])
 
+        # Identity detection: this rank keeps its whole block and exchanges
+        # nothing -> the repartition is a no-op here (e.g. P_x == P_m for the
+        # two-phase partition (1,1,1,N,1,1)).  Skipping it avoids a full
+        # tensor copy per call on the hot path.
+        self.is_identity = False
+        if not self.sends and not self.recvs:
+            if self.out_shape is None:
+                self.is_identity = not P_src.active  # zero-volume in, out
+            elif self.local_copy is not None:
+                ssl, dsl = self.local_copy
+                full = all(
+                    s.start == 0 and s.stop == e and d.start == 0 and d.stop == e
+                    for s, d, e in zip(ssl, dsl, self.out_shape))
+                src_shape = [b - a for a, b in src_bounds] if src_bounds else None
+                self.is_identity = full and src_shape == self.out_shape
+
 
 _PLAN_REGISTRY: Dict[Tuple, _RepartitionPlan] = {}
 
@@ -464,6 +480,9 @@ class Repartition(torch.nn.Module):
             g = self._gshape
         else:
             g = self._infer_gshape(x)
+        plan = _get_plan(self.P_src, self.P_dst, g)
+        if plan.is_identity:
+            return x
         return _RepartitionFn.apply(x, self, g)
 
 

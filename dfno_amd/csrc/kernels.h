@@ -17,6 +17,15 @@ at::Tensor gelu_fwd(const at::Tensor& x);
 at::Tensor gelu_bwd(const at::Tensor& gy, const at::Tensor& z);
 std::vector<at::Tensor> add_gelu_fwd(const at::Tensor& a, const at::Tensor& b);
 
+// fused projection head: out = W4 @ gelu(W3 @ x + b3) + b4 (no intermediates)
+at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
+                         const at::Tensor& b3, const at::Tensor& W4,
+                         const at::Tensor& b4);
+// returns {gz3, gx, gb3, gW4, gb4}; grad-W3 = gz3 @ x^T (library GEMM)
+std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
+                                      const at::Tensor& W3, const at::Tensor& b3,
+                                      const at::Tensor& W4);
+
 // corner-block spectral contraction on the truncated complex spectrum:
 //   y[b,o,f] += sum_i x[b,i,f] * w[i,o,f_box]  for f in the corner box
 void spectral_corner_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,

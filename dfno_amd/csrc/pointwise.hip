@@ -114,7 +114,7 @@ __global__ void add_gelu_kernel(const T* __restrict__ a, const T* __restrict__ b
 // W (O x I or transposed) staged in LDS; reads are wave-uniform broadcasts.
 // ---------------------------------------------------------------------------
 
-template <typename T, int IMAX, int VEC, bool ACT>
+template <typename T, int IMAX, int VEC, bool ACT, bool VECTOR>
 __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
     const T* __restrict__ x, const T* __restrict__ W, const T* __restrict__ bias,
     T* __restrict__ y, T* __restrict__ z,
@@ -139,13 +139,21 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
 
     T xr[IMAX][VEC];
     const T* xb = x + ((long)b * I) * S + s;
+    if constexpr (VECTOR && std::is_same<T, float>::value) {
 #pragma unroll 4
-    for (int i = 0; i < I; ++i) {
-      if (full) {
+      for (int i = 0; i < I; ++i) {
+        const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
+        xr[i][0] = v.x; xr[i][1] = v.y; xr[i][2] = v.z; xr[i][3] = v.w;
+      }
+    } else {
+#pragma unroll 4
+      for (int i = 0; i < I; ++i) {
+        if (full) {
 #pragma unroll
-        for (int k = 0; k < VEC; ++k) xr[i][k] = xb[(long)i * S + k];
-      } else {
-        for (int k = 0; k < nv; ++k) xr[i][k] = xb[(long)i * S + k];
+          for (int k = 0; k < VEC; ++k) xr[i][k] = xb[(long)i * S + k];
+        } else {
+          for (int k = 0; k < nv; ++k) xr[i][k] = xb[(long)i * S + k];
+        }
       }
     }
 
@@ -161,7 +169,19 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
 #pragma unroll
         for (int k = 0; k < VEC; ++k) acc[k] += wv * xr[i][k];
       }
-      if (full) {
+      if constexpr (VECTOR && std::is_same<T, float>::value) {
+        if (write_z)
+          *reinterpret_cast<float4*>(zb + (long)o * S) =
+              make_float4(acc[0], acc[1], acc[2], acc[3]);
+        if (ACT) {
+          *reinterpret_cast<float4*>(yb + (long)o * S) =
+              make_float4(gelu_erf(acc[0]), gelu_erf(acc[1]),
+                          gelu_erf(acc[2]), gelu_erf(acc[3]));
+        } else {
+          *reinterpret_cast<float4*>(yb + (long)o * S) =
+              make_float4(acc[0], acc[1], acc[2], acc[3]);
+        }
+      } else if (full) {
         if (write_z) {
 #pragma unroll
           for (int k = 0; k < VEC; ++k) zb[(long)o * S + k] = acc[k];
@@ -184,7 +204,7 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
 // Used for the projection head 128 -> 1 and for grad-x of the lift 1 -> C.
 // ---------------------------------------------------------------------------
 
-template <typename T, int OMAX, int VEC, bool ACT>
+template <typename T, int OMAX, int VEC, bool ACT, bool VECTOR>
 __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
     const T* __restrict__ x, const T* __restrict__ W, const T* __restrict__ bias,
     T* __restrict__ y, T* __restrict__ z,
@@ -217,7 +237,10 @@ __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
     const T* xb = x + ((long)b * I) * S + s;
     for (int i = 0; i < I; ++i) {
       T xv[VEC];
-      if (full) {
+      if constexpr (VECTOR && std::is_same<T, float>::value) {
+        const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
+        xv[0] = v.x; xv[1] = v.y; xv[2] = v.z; xv[3] = v.w;
+      } else if (full) {
 #pragma unroll
         for (int k = 0; k < VEC; ++k) xv[k] = xb[(long)i * S + k];
       } else {
@@ -233,10 +256,24 @@ __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
     T* yb = y + ((long)b * O) * S + s;
     T* zb = write_z ? z + ((long)b * O) * S + s : nullptr;
     for (int o = 0; o < O; ++o) {
-      int lim = full ? VEC : nv;
-      for (int k = 0; k < lim; ++k) {
-        if (write_z) zb[(long)o * S + k] = acc[o][k];
-        yb[(long)o * S + k] = ACT ? gelu_erf(acc[o][k]) : acc[o][k];
+      if constexpr (VECTOR && std::is_same<T, float>::value) {
+        if (write_z)
+          *reinterpret_cast<float4*>(zb + (long)o * S) =
+              make_float4(acc[o][0], acc[o][1], acc[o][2], acc[o][3]);
+        if (ACT) {
+          *reinterpret_cast<float4*>(yb + (long)o * S) =
+              make_float4(gelu_erf(acc[o][0]), gelu_erf(acc[o][1]),
+                          gelu_erf(acc[o][2]), gelu_erf(acc[o][3]));
+        } else {
+          *reinterpret_cast<float4*>(yb + (long)o * S) =
+              make_float4(acc[o][0], acc[o][1], acc[o][2], acc[o][3]);
+        }
+      } else {
+        int lim = full ? VEC : nv;
+        for (int k = 0; k < lim; ++k) {
+          if (write_z) zb[(long)o * S + k] = acc[o][k];
+          yb[(long)o * S + k] = ACT ? gelu_erf(acc[o][k]) : acc[o][k];
+        }
       }
     }
   }
@@ -252,9 +289,14 @@ __global__ __launch_bounds__(kBlock) void channel_mix_lds_kernel(
     const T* __restrict__ x, const T* __restrict__ W, const T* __restrict__ bias,
     T* __restrict__ y, T* __restrict__ z,
     int B, int I, int O, long S, bool wt, bool has_bias, bool write_z) {
-  // tile of TS spatial positions staged in LDS as [I][TS]
+  // tile of TS spatial positions staged in LDS as [I][TS], then W, then bias
   extern __shared__ __align__(16) char smem_raw[];
   T* xt = reinterpret_cast<T*>(smem_raw);  // [I][TS]
+  T* Wl = xt + (size_t)I * TS;             // [O*I]
+  T* bl = Wl + (size_t)O * I;              // [O]
+  for (int k = threadIdx.x; k < O * I; k += blockDim.x) Wl[k] = W[k];
+  if (has_bias)
+    for (int k = threadIdx.x; k < O; k += blockDim.x) bl[k] = bias[k];
 
   long ntiles = (S + TS - 1) / TS;
   for (long tile = blockIdx.x; tile < (long)B * ntiles; tile += gridDim.x) {
@@ -275,9 +317,9 @@ __global__ __launch_bounds__(kBlock) void channel_mix_lds_kernel(
       T* yb = y + ((long)b * O) * S + s0 + s;
       T* zb = write_z ? z + ((long)b * O) * S + s0 + s : nullptr;
       for (int o = 0; o < O; ++o) {
-        T acc = has_bias ? bias[o] : T(0);
+        T acc = has_bias ? bl[o] : T(0);
         for (int i = 0; i < I; ++i) {
-          T wv = wt ? W[(size_t)i * O + o] : W[(size_t)o * I + i];
+          T wv = wt ? Wl[(size_t)i * O + o] : Wl[(size_t)o * I + i];
           acc += wv * xt[i * TS + s];
         }
         if (write_z) zb[(long)o * S] = acc;
@@ -296,6 +338,14 @@ int grid_for(long work, int block) {
 }
 
 template <typename T>
+bool can_vectorize(const T* x, const T* y, const T* z, long S, bool write_z) {
+  if (!std::is_same<T, float>::value) return false;
+  if (S % 4 != 0) return false;
+  auto aligned = [](const void* p) { return (reinterpret_cast<uintptr_t>(p) & 15) == 0; };
+  return aligned(x) && aligned(y) && (!write_z || aligned(z));
+}
+
+template <typename T>
 void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
                         int B, int I, int O, long S, bool wt, bool has_bias,
                         bool act, bool write_z, hipStream_t stream) {
@@ -303,16 +353,19 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
   long nchunk_work = (long)B * ((S + 3) / 4);
   int grid = grid_for(nchunk_work, kBlock);
   size_t smem = sizeof(T) * ((size_t)O * I + O);
+  const bool vec = can_vectorize(x, y, z, S, write_z);
 
+#define CMIX_LAUNCH(KERNEL, CAP, A, V)                                          \
+  hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V>), dim3(grid), dim3(kBlock),       \
+                     smem, stream, x, W, bias, y, z, B, I, O, S, wt,            \
+                     has_bias, write_z);
 #define CMIX_DISPATCH(KERNEL, CAP)                                              \
   if (act) {                                                                    \
-    hipLaunchKernelGGL((KERNEL<T, CAP, 4, true>), dim3(grid), dim3(kBlock),     \
-                       smem, stream, x, W, bias, y, z, B, I, O, S, wt,          \
-                       has_bias, write_z);                                      \
+    if (vec) { CMIX_LAUNCH(KERNEL, CAP, true, true) }                           \
+    else { CMIX_LAUNCH(KERNEL, CAP, true, false) }                              \
   } else {                                                                      \
-    hipLaunchKernelGGL((KERNEL<T, CAP, 4, false>), dim3(grid), dim3(kBlock),    \
-                       smem, stream, x, W, bias, y, z, B, I, O, S, wt,          \
-                       has_bias, write_z);                                      \
+    if (vec) { CMIX_LAUNCH(KERNEL, CAP, false, true) }                          \
+    else { CMIX_LAUNCH(KERNEL, CAP, false, false) }                             \
   }
 
   if (I <= 8) { CMIX_DISPATCH(channel_mix_xres_kernel, 8) }
@@ -322,13 +375,15 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
   else if (O <= 4) { CMIX_DISPATCH(channel_mix_ores_kernel, 4) }
   else if (O <= 8) { CMIX_DISPATCH(channel_mix_ores_kernel, 8) }
   else if (O <= 16) { CMIX_DISPATCH(channel_mix_ores_kernel, 16) }
+  else if (O <= 24 && std::is_same<T, float>::value) { CMIX_DISPATCH(channel_mix_ores_kernel, 24) }
+  else if (O <= 32 && std::is_same<T, float>::value) { CMIX_DISPATCH(channel_mix_ores_kernel, 32) }
   else {
     // LDS-staged generic path; pick the largest tile that keeps the x tile
     // within 64 KiB (>= 2 blocks/CU of LDS headroom)
     int grid2 = grid_for((long)B * ((S + kBlock - 1) / kBlock) * kBlock, kBlock);
 #define CMIX_LDS(TS)                                                            \
   {                                                                             \
-    size_t smem_lds = sizeof(T) * (size_t)I * TS;                               \
+    size_t smem_lds = sizeof(T) * ((size_t)I * TS + (size_t)O * I + O);         \
     if (act) {                                                                  \
       hipLaunchKernelGGL((channel_mix_lds_kernel<T, true, TS>), dim3(grid2),    \
                          dim3(kBlock), smem_lds, stream, x, W, bias, y, z, B,   \
@@ -340,11 +395,13 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
     }                                                                           \
   }
     size_t per_s = sizeof(T) * (size_t)I;
-    if (per_s * 256 <= 64 * 1024) { CMIX_LDS(256) }
-    else if (per_s * 128 <= 64 * 1024) { CMIX_LDS(128) }
-    else if (per_s * 64 <= 64 * 1024) { CMIX_LDS(64) }
+    size_t wbytes = sizeof(T) * ((size_t)O * I + O);
+    if (per_s * 256 + wbytes <= 64 * 1024) { CMIX_LDS(256) }
+    else if (per_s * 128 + wbytes <= 96 * 1024) { CMIX_LDS(128) }
+    else if (per_s * 64 + wbytes <= 128 * 1024) { CMIX_LDS(64) }
     else {
-      TORCH_CHECK(per_s * 32 <= 160 * 1024, "channel_mix: I too large for LDS tile");
+      TORCH_CHECK(per_s * 32 + wbytes <= 160 * 1024,
+                  "channel_mix: I/O too large for LDS tile");
       CMIX_LDS(32)
     }
 #undef CMIX_LDS

@@ -1,0 +1,378 @@
+// gfx950 fused FNO projection head: out = W4 @ gelu(W3 @ x + b3) + b4.
+//
+// The reference computes this as two BroadcastedLinear einsums with a
+// separate GELU between (/root/reference/dfno/dfno.py:348-352), which at the
+// flagship config materializes a [1,128,64^3*30] fp32 intermediate (~4 GB)
+// three times over (einsum out, bias add, gelu out).  On MI355X that is
+// ~25 GB of pointless HBM traffic per forward (and ~2x in backward).
+//
+// Here the whole head is one pass: each thread holds its x column (I<=32
+// channels) in registers, walks the M<=512 hidden channels recomputing
+// z3 = W3 x + b3 on the fly (weights in LDS, wave-uniform broadcast reads),
+// and accumulates the O2<=8 outputs.  Forward traffic = read x + write out.
+// Backward recomputes z3 from x (cheap: x is I channels) in one kernel that
+// produces grad-x, grad-b3, grad-W4, grad-b4 directly (wave-shuffle
+// reductions + one LDS/global atomic per block) and materializes gz3 for the
+// library-GEMM grad-W3 reduction.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include "kernels.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+template <typename T>
+__device__ __forceinline__ T gelu_erf_(T z) {
+  return T(0.5) * z * (T(1.0) + erf(z * T(0.7071067811865476)));
+}
+
+template <typename T>
+__device__ __forceinline__ T gelu_grad_erf_(T z) {
+  return T(0.5) * (T(1.0) + erf(z * T(0.7071067811865476))) +
+         z * exp(T(-0.5) * z * z) * T(0.3989422804014327);
+}
+
+// butterfly sum over the 64-lane wave
+template <typename T>
+__device__ __forceinline__ T wave_sum(T v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR>
+__global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ W3, const T* __restrict__ b3,
+    const T* __restrict__ W4, const T* __restrict__ b4, T* __restrict__ out,
+    int B, int I, int M, int O2, long S) {
+  extern __shared__ __align__(16) char smem_raw[];
+  T* W3l = reinterpret_cast<T*>(smem_raw);   // [M*I]
+  T* b3l = W3l + (size_t)M * I;              // [M]
+  T* W4l = b3l + M;                          // [O2*M]
+  T* b4l = W4l + (size_t)O2 * M;             // [O2]
+  for (int k = threadIdx.x; k < M * I; k += blockDim.x) W3l[k] = W3[k];
+  for (int k = threadIdx.x; k < M; k += blockDim.x) b3l[k] = b3[k];
+  for (int k = threadIdx.x; k < O2 * M; k += blockDim.x) W4l[k] = W4[k];
+  for (int k = threadIdx.x; k < O2; k += blockDim.x) b4l[k] = b4[k];
+  __syncthreads();
+
+  long nchunks = (S + VEC - 1) / VEC;
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < (long)B * nchunks; t += stride) {
+    int b = (int)(t / nchunks);
+    long s = (t % nchunks) * VEC;
+    bool full = (s + VEC) <= S;
+    int nv = full ? VEC : (int)(S - s);
+
+    T xr[IMAX][VEC];
+    const T* xb = x + ((long)b * I) * S + s;
+    if constexpr (VECTOR && std::is_same<T, float>::value) {
+#pragma unroll 4
+      for (int i = 0; i < I; ++i) {
+        const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
+        xr[i][0] = v.x; xr[i][1] = v.y; xr[i][2] = v.z; xr[i][3] = v.w;
+      }
+    } else {
+      for (int i = 0; i < I; ++i)
+        for (int k = 0; k < nv; ++k) xr[i][k] = xb[(long)i * S + k];
+    }
+
+    T acc[O2MAX][VEC];
+    for (int o = 0; o < O2; ++o)
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) acc[o][k] = b4l[o];
+
+    for (int j = 0; j < M; ++j) {
+      T zk[VEC];
+      T bj = b3l[j];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) zk[k] = bj;
+      for (int i = 0; i < I; ++i) {
+        T wv = W3l[(size_t)j * I + i];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) zk[k] += wv * xr[i][k];
+      }
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) zk[k] = gelu_erf_(zk[k]);
+      for (int o = 0; o < O2; ++o) {
+        T w4 = W4l[(size_t)o * M + j];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) acc[o][k] += w4 * zk[k];
+      }
+    }
+
+    T* ob = out + ((long)b * O2) * S + s;
+    if constexpr (VECTOR && std::is_same<T, float>::value) {
+      for (int o = 0; o < O2; ++o)
+        *reinterpret_cast<float4*>(ob + (long)o * S) =
+            make_float4(acc[o][0], acc[o][1], acc[o][2], acc[o][3]);
+    } else {
+      for (int o = 0; o < O2; ++o)
+        for (int k = 0; k < nv; ++k) ob[(long)o * S + k] = acc[o][k];
+    }
+  }
+}
+
+// Backward: one pass producing gx, gb3, gW4, gb4 and materializing gz3
+// (grad wrt z3 = pre-gelu hidden) for the grad-W3 library GEMM.
+template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR>
+__global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
+    const T* __restrict__ gy, const T* __restrict__ x,
+    const T* __restrict__ W3, const T* __restrict__ b3,
+    const T* __restrict__ W4,
+    T* __restrict__ gz3, T* __restrict__ gx,
+    T* __restrict__ gb3, T* __restrict__ gW4, T* __restrict__ gb4,
+    int B, int I, int M, int O2, long S) {
+  extern __shared__ __align__(16) char smem_raw[];
+  T* W3l = reinterpret_cast<T*>(smem_raw);   // [M*I]
+  T* b3l = W3l + (size_t)M * I;              // [M]
+  T* W4l = b3l + M;                          // [O2*M]
+  // per-wave partial accumulators for gb3 [4][M] and gW4 [4][O2*M]
+  T* gb3w = W4l + (size_t)O2 * M;            // [4*M]
+  T* gW4w = gb3w + 4 * (size_t)M;            // [4*O2*M]
+  for (int k = threadIdx.x; k < M * I; k += blockDim.x) W3l[k] = W3[k];
+  for (int k = threadIdx.x; k < M; k += blockDim.x) b3l[k] = b3[k];
+  for (int k = threadIdx.x; k < O2 * M; k += blockDim.x) W4l[k] = W4[k];
+  for (int k = threadIdx.x; k < 4 * M; k += blockDim.x) gb3w[k] = T(0);
+  for (int k = threadIdx.x; k < 4 * O2 * M; k += blockDim.x) gW4w[k] = T(0);
+  __syncthreads();
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+
+  T gb4acc[O2MAX];
+#pragma unroll
+  for (int o = 0; o < O2MAX; ++o) gb4acc[o] = T(0);
+
+  long nchunks = (S + VEC - 1) / VEC;
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < (long)B * nchunks; t += stride) {
+    int b = (int)(t / nchunks);
+    long s = (t % nchunks) * VEC;
+    bool full = (s + VEC) <= S;
+    int nv = full ? VEC : (int)(S - s);
+
+    T xr[IMAX][VEC];
+    const T* xb = x + ((long)b * I) * S + s;
+    T gyv[O2MAX][VEC];
+    const T* gyb = gy + ((long)b * O2) * S + s;
+    if constexpr (VECTOR && std::is_same<T, float>::value) {
+#pragma unroll 4
+      for (int i = 0; i < I; ++i) {
+        const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
+        xr[i][0] = v.x; xr[i][1] = v.y; xr[i][2] = v.z; xr[i][3] = v.w;
+      }
+      for (int o = 0; o < O2; ++o) {
+        const float4 v = *reinterpret_cast<const float4*>(gyb + (long)o * S);
+        gyv[o][0] = v.x; gyv[o][1] = v.y; gyv[o][2] = v.z; gyv[o][3] = v.w;
+      }
+    } else {
+      for (int i = 0; i < I; ++i)
+        for (int k = 0; k < nv; ++k) xr[i][k] = xb[(long)i * S + k];
+      for (int o = 0; o < O2; ++o)
+        for (int k = 0; k < nv; ++k) gyv[o][k] = gyb[(long)o * S + k];
+      for (int o = 0; o < O2; ++o)
+        for (int k = nv; k < VEC; ++k) gyv[o][k] = T(0);
+    }
+
+    for (int o = 0; o < O2; ++o)
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) gb4acc[o] += (full || k < nv) ? gyv[o][k] : T(0);
+
+    T gxa[IMAX][VEC];
+    for (int i = 0; i < I; ++i)
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) gxa[i][k] = T(0);
+
+    T* gz3b = gz3 + ((long)b * M) * S + s;
+    for (int j = 0; j < M; ++j) {
+      // recompute z3 and gelu pieces
+      T zk[VEC];
+      T bj = b3l[j];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) zk[k] = bj;
+      for (int i = 0; i < I; ++i) {
+        T wv = W3l[(size_t)j * I + i];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) zk[k] += wv * xr[i][k];
+      }
+      T gk[VEC], dgk[VEC];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) {
+        gk[k] = gelu_erf_(zk[k]);
+        dgk[k] = gelu_grad_erf_(zk[k]);
+      }
+      // gz3_j = (sum_o W4[o,j] gy[o]) * gelu'(z3_j)
+      T gzk[VEC];
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) gzk[k] = T(0);
+      for (int o = 0; o < O2; ++o) {
+        T w4 = W4l[(size_t)o * M + j];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) gzk[k] += w4 * gyv[o][k];
+      }
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) gzk[k] *= dgk[k];
+
+      // store gz3 and accumulate gx
+      if constexpr (VECTOR && std::is_same<T, float>::value) {
+        *reinterpret_cast<float4*>(gz3b + (long)j * S) =
+            make_float4(gzk[0], gzk[1], gzk[2], gzk[3]);
+      } else {
+        for (int k = 0; k < nv; ++k) gz3b[(long)j * S + k] = gzk[k];
+      }
+      for (int i = 0; i < I; ++i) {
+        T wv = W3l[(size_t)j * I + i];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) gxa[i][k] += wv * gzk[k];
+      }
+
+      // wave-reduced gb3[j] and gW4[o][j] partials
+      T pb = T(0);
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) pb += (full || k < nv) ? gzk[k] : T(0);
+      pb = wave_sum(pb);
+      if (lane == 0) gb3w[wave * M + j] += pb;
+      for (int o = 0; o < O2; ++o) {
+        T pw = T(0);
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) pw += (full || k < nv) ? gyv[o][k] * gk[k] : T(0);
+        pw = wave_sum(pw);
+        if (lane == 0) gW4w[(wave * O2 + o) * M + j] += pw;
+      }
+    }
+
+    T* gxb = gx + ((long)b * I) * S + s;
+    if constexpr (VECTOR && std::is_same<T, float>::value) {
+      for (int i = 0; i < I; ++i)
+        *reinterpret_cast<float4*>(gxb + (long)i * S) =
+            make_float4(gxa[i][0], gxa[i][1], gxa[i][2], gxa[i][3]);
+    } else {
+      for (int i = 0; i < I; ++i)
+        for (int k = 0; k < nv; ++k) gxb[(long)i * S + k] = gxa[i][k];
+    }
+  }
+
+  // block-level flush: per-wave partials -> global atomics
+  for (int o = 0; o < O2; ++o) {
+    T v = wave_sum(gb4acc[o]);
+    if (lane == 0 && v != T(0)) atomicAdd(&gb4[o], v);
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < M; k += blockDim.x) {
+    T v = gb3w[k] + gb3w[M + k] + gb3w[2 * M + k] + gb3w[3 * M + k];
+    if (v != T(0)) atomicAdd(&gb3[k], v);
+  }
+  for (int k = threadIdx.x; k < O2 * M; k += blockDim.x) {
+    T v = gW4w[k] + gW4w[O2 * M + k] + gW4w[2 * O2 * M + k] + gW4w[3 * O2 * M + k];
+    if (v != T(0)) atomicAdd(&gW4[k], v);
+  }
+}
+
+int grid_for_p(long work) {
+  long g = (work + kBlock - 1) / kBlock;
+  long cap = 256L * 8;
+  if (g > cap) g = cap;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+void check_pf(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == at::kFloat || t.scalar_type() == at::kDouble,
+              name, " must be float32/float64");
+}
+
+template <typename T>
+bool vec_ok(long S, std::initializer_list<const void*> ptrs) {
+  if (!std::is_same<T, float>::value) return false;
+  if (S % 4 != 0) return false;
+  for (auto p : ptrs)
+    if ((reinterpret_cast<uintptr_t>(p) & 15) != 0) return false;
+  return true;
+}
+
+}  // namespace
+
+at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
+                         const at::Tensor& b3, const at::Tensor& W4,
+                         const at::Tensor& b4) {
+  check_pf(x, "x"); check_pf(W3, "W3"); check_pf(b3, "b3");
+  check_pf(W4, "W4"); check_pf(b4, "b4");
+  TORCH_CHECK(x.dim() == 3, "x must be [B,I,S]");
+  int B = (int)x.size(0), I = (int)x.size(1);
+  long S = x.size(2);
+  int M = (int)W3.size(0), O2 = (int)W4.size(0);
+  TORCH_CHECK((int)W3.size(1) == I && (int)W4.size(1) == M, "proj_head shapes");
+  TORCH_CHECK(I <= 32 && O2 <= 8 && M <= 512, "proj_head: unsupported dims");
+
+  auto out = at::empty({B, O2, S}, x.options());
+  if (x.numel() == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int grid = grid_for_p((long)B * ((S + 3) / 4));
+
+#define PH_LAUNCH_F(V)                                                        \
+    hipLaunchKernelGGL((proj_head_fwd_kernel<scalar_t, 32, 8, 4, V>),         \
+                       dim3(grid), dim3(kBlock), smem, stream,                \
+                       x.data_ptr<scalar_t>(), W3.data_ptr<scalar_t>(),       \
+                       b3.data_ptr<scalar_t>(), W4.data_ptr<scalar_t>(),      \
+                       b4.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),     \
+                       B, I, M, O2, S);
+  AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "proj_head_fwd", [&] {
+    size_t smem = sizeof(scalar_t) * ((size_t)M * I + M + (size_t)O2 * M + O2);
+    bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), out.data_ptr()});
+    if (vec) { PH_LAUNCH_F(true) } else { PH_LAUNCH_F(false) }
+  });
+#undef PH_LAUNCH_F
+  return out;
+}
+
+std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
+                                      const at::Tensor& W3, const at::Tensor& b3,
+                                      const at::Tensor& W4) {
+  check_pf(gy, "gy"); check_pf(x, "x"); check_pf(W3, "W3");
+  check_pf(b3, "b3"); check_pf(W4, "W4");
+  int B = (int)x.size(0), I = (int)x.size(1);
+  long S = x.size(2);
+  int M = (int)W3.size(0), O2 = (int)W4.size(0);
+  TORCH_CHECK(I <= 32 && O2 <= 8 && M <= 512, "proj_head: unsupported dims");
+
+  auto gz3 = at::empty({B, M, S}, x.options());
+  auto gx = at::empty_like(x);
+  auto gb3 = at::zeros({M}, x.options());
+  auto gW4 = at::zeros({O2, M}, x.options());
+  auto gb4 = at::zeros({O2}, x.options());
+  if (x.numel() == 0) return {gz3, gx, gb3, gW4, gb4};
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int grid = grid_for_p((long)B * ((S + 3) / 4));
+
+#define PH_LAUNCH_B(V)                                                        \
+    hipLaunchKernelGGL((proj_head_bwd_kernel<scalar_t, 32, 8, 4, V>),         \
+                       dim3(grid), dim3(kBlock), smem, stream,                \
+                       gy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),       \
+                       W3.data_ptr<scalar_t>(), b3.data_ptr<scalar_t>(),      \
+                       W4.data_ptr<scalar_t>(), gz3.data_ptr<scalar_t>(),     \
+                       gx.data_ptr<scalar_t>(), gb3.data_ptr<scalar_t>(),     \
+                       gW4.data_ptr<scalar_t>(), gb4.data_ptr<scalar_t>(),    \
+                       B, I, M, O2, S);
+  AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "proj_head_bwd", [&] {
+    size_t smem = sizeof(scalar_t) *
+        ((size_t)M * I + M + (size_t)O2 * M + 4 * (size_t)M + 4 * (size_t)O2 * M);
+    TORCH_CHECK(smem <= 160 * 1024, "proj_head_bwd: LDS overflow");
+    bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), gy.data_ptr(), gz3.data_ptr(),
+                                    gx.data_ptr()});
+    if (vec) { PH_LAUNCH_B(true) } else { PH_LAUNCH_B(false) }
+  });
+#undef PH_LAUNCH_B
+  return {gz3, gx, gb3, gW4, gb4};
+}

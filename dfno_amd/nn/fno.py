@@ -82,10 +82,32 @@ class DistributedFNONd(nn.Module):
             x = block(x)
             self.dt_comm += block.dt_comm
 
-        x = self.linear3(x, activation="gelu")
-        self.dt_comm += self.linear3.dt_comm
-        x = self.linear4(x)
-        self.dt_comm += self.linear4.dt_comm
+        x = self._projection(x)
+        return x
+
+    def _projection(self, x: torch.Tensor) -> torch.Tensor:
+        """width -> 128 -> gelu -> 1 head.  On GPU this is a single fused
+        kernel (ops.proj_head) instead of two linears with a GELU pass."""
+        from ..ops import proj_head, proj_head_supported
+        import time as _time
+
+        l3, l4 = self.linear3, self.linear4
+        supported = (x.is_cuda and x.dtype in (torch.float32, torch.float64)
+                     and l3.in_features <= 32 and l3.out_features <= 512
+                     and l4.out_features <= 8)
+        if supported:
+            t0 = _time.time()
+            W3 = l3.W_bcast(l3.W)
+            b3 = l3.b_bcast(l3.b)
+            W4 = l4.W_bcast(l4.W)
+            b4 = l4.b_bcast(l4.b)
+            self.dt_comm += _time.time() - t0
+            return proj_head(x, W3, b3, W4, b4)
+
+        x = l3(x, activation="gelu")
+        self.dt_comm += l3.dt_comm
+        x = l4(x)
+        self.dt_comm += l4.dt_comm
         return x
 
 

@@ -1,2 +1,3 @@
 from .pointwise import linear_nd, add_gelu, gelu
 from .spectral import spectral_conv
+from .projhead import proj_head, proj_head_supported

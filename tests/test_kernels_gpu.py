@@ -189,3 +189,40 @@ def test_op_spectral_autograd_gpu():
         assert torch.allclose(x.grad, xr.grad, rtol=tt, atol=tt)
         for w, w2 in zip(ws, wr):
             assert torch.allclose(w.grad, w2.grad, rtol=tt, atol=tt)
+
+
+# ---------------------------------------------------------------------------
+# fused projection head
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dtype,tt", [(torch.float32, 3e-4), (torch.float64, 1e-10)])
+@pytest.mark.parametrize("B,I,M,O2,S", [
+    (1, 20, 128, 1, 4096),   # flagship head shape
+    (2, 12, 64, 2, 1001),    # odd S (scalar path), multi-batch, O2=2
+])
+def test_proj_head(dtype, tt, B, I, M, O2, S):
+    from dfno_amd.ops import proj_head
+    torch.manual_seed(8)
+    x = torch.randn(B, I, S, device="cuda", dtype=dtype, requires_grad=True)
+    W3 = (torch.randn(M, I, device="cuda", dtype=dtype) / I).requires_grad_(True)
+    b3 = torch.randn(M, device="cuda", dtype=dtype).requires_grad_(True)
+    W4 = (torch.randn(O2, M, device="cuda", dtype=dtype) / M).requires_grad_(True)
+    b4 = torch.randn(O2, device="cuda", dtype=dtype).requires_grad_(True)
+
+    y = proj_head(x, W3, b3, W4, b4)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    xr = x.detach().clone().requires_grad_(True)
+    W3r = W3.detach().clone().requires_grad_(True)
+    b3r = b3.detach().clone().requires_grad_(True)
+    W4r = W4.detach().clone().requires_grad_(True)
+    b4r = b4.detach().clone().requires_grad_(True)
+    h = F.gelu(torch.einsum("mi,bis->bms", W3r, xr) + b3r.view(1, -1, 1))
+    yr = torch.einsum("om,bms->bos", W4r, h) + b4r.view(1, -1, 1)
+    yr.backward(gy)
+
+    assert torch.allclose(y, yr, rtol=tt, atol=tt), f"fwd max {(y-yr).abs().max()}"
+    for a, b in [(x, xr), (W3, W3r), (b3, b3r), (W4, W4r), (b4, b4r)]:
+        assert torch.allclose(a.grad, b.grad, rtol=tt * 30, atol=tt * 30), \
+            f"grad max {(a.grad - b.grad).abs().max()}"
