@@ -140,20 +140,21 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
     T xr[IMAX][VEC];
     const T* xb = x + ((long)b * I) * S + s;
     if constexpr (VECTOR && std::is_same<T, float>::value) {
-#pragma unroll 4
-      for (int i = 0; i < I; ++i) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
         const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
         xr[i][0] = v.x; xr[i][1] = v.y; xr[i][2] = v.z; xr[i][3] = v.w;
+              }
       }
     } else {
-#pragma unroll 4
-      for (int i = 0; i < I; ++i) {
-        if (full) {
 #pragma unroll
-          for (int k = 0; k < VEC; ++k) xr[i][k] = xb[(long)i * S + k];
-        } else {
-          for (int k = 0; k < nv; ++k) xr[i][k] = xb[(long)i * S + k];
-        }
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          xr[i][k] = (full || k < nv) ? xb[(long)i * S + k] : T(0);
+              }
       }
     }
 
@@ -164,10 +165,13 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
       T bv = has_bias ? bl[o] : T(0);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) acc[k] = bv;
-      for (int i = 0; i < I; ++i) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
         T wv = wt ? Wl[(size_t)i * O + o] : Wl[(size_t)o * I + i];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) acc[k] += wv * xr[i][k];
+              }
       }
       if constexpr (VECTOR && std::is_same<T, float>::value) {
         if (write_z)
@@ -181,16 +185,10 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
           *reinterpret_cast<float4*>(yb + (long)o * S) =
               make_float4(acc[0], acc[1], acc[2], acc[3]);
         }
-      } else if (full) {
-        if (write_z) {
-#pragma unroll
-          for (int k = 0; k < VEC; ++k) zb[(long)o * S + k] = acc[k];
-        }
-#pragma unroll
-        for (int k = 0; k < VEC; ++k)
-          yb[(long)o * S + k] = ACT ? gelu_erf(acc[k]) : acc[k];
       } else {
-        for (int k = 0; k < nv; ++k) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          if (!full && k >= nv) break;
           if (write_z) zb[(long)o * S + k] = acc[k];
           yb[(long)o * S + k] = ACT ? gelu_erf(acc[k]) : acc[k];
         }
@@ -228,10 +226,13 @@ __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
     int nv = full ? VEC : (int)(S - s);
 
     T acc[OMAX][VEC];
-    for (int o = 0; o < O; ++o) {
+#pragma unroll
+    for (int o = 0; o < OMAX; ++o) {
+      if (o < O) {
       T bv = has_bias ? bl[o] : T(0);
 #pragma unroll
       for (int k = 0; k < VEC; ++k) acc[o][k] = bv;
+          }
     }
 
     const T* xb = x + ((long)b * I) * S + s;
@@ -240,22 +241,26 @@ __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
       if constexpr (VECTOR && std::is_same<T, float>::value) {
         const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
         xv[0] = v.x; xv[1] = v.y; xv[2] = v.z; xv[3] = v.w;
-      } else if (full) {
-#pragma unroll
-        for (int k = 0; k < VEC; ++k) xv[k] = xb[(long)i * S + k];
       } else {
-        for (int k = 0; k < nv; ++k) xv[k] = xb[(long)i * S + k];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          xv[k] = (full || k < nv) ? xb[(long)i * S + k] : T(0);
       }
-      for (int o = 0; o < O; ++o) {
+#pragma unroll
+      for (int o = 0; o < OMAX; ++o) {
+        if (o < O) {
         T wv = wt ? Wl[(size_t)i * O + o] : Wl[(size_t)o * I + i];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) acc[o][k] += wv * xv[k];
+              }
       }
     }
 
     T* yb = y + ((long)b * O) * S + s;
     T* zb = write_z ? z + ((long)b * O) * S + s : nullptr;
-    for (int o = 0; o < O; ++o) {
+#pragma unroll
+    for (int o = 0; o < OMAX; ++o) {
+      if (o < O) {
       if constexpr (VECTOR && std::is_same<T, float>::value) {
         if (write_z)
           *reinterpret_cast<float4*>(zb + (long)o * S) =
@@ -269,12 +274,14 @@ __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
               make_float4(acc[o][0], acc[o][1], acc[o][2], acc[o][3]);
         }
       } else {
-        int lim = full ? VEC : nv;
-        for (int k = 0; k < lim; ++k) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          if (!full && k >= nv) break;
           if (write_z) zb[(long)o * S + k] = acc[o][k];
           yb[(long)o * S + k] = ACT ? gelu_erf(acc[o][k]) : acc[o][k];
         }
       }
+          }
     }
   }
 }

@@ -73,48 +73,73 @@ __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
     T xr[IMAX][VEC];
     const T* xb = x + ((long)b * I) * S + s;
     if constexpr (VECTOR && std::is_same<T, float>::value) {
-#pragma unroll 4
-      for (int i = 0; i < I; ++i) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
         const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
         xr[i][0] = v.x; xr[i][1] = v.y; xr[i][2] = v.z; xr[i][3] = v.w;
+              }
       }
     } else {
-      for (int i = 0; i < I; ++i)
-        for (int k = 0; k < nv; ++k) xr[i][k] = xb[(long)i * S + k];
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          xr[i][k] = (full || k < nv) ? xb[(long)i * S + k] : T(0);
+              }
+      }
     }
 
     T acc[O2MAX][VEC];
-    for (int o = 0; o < O2; ++o)
+#pragma unroll
+    for (int o = 0; o < O2MAX; ++o) {
+      if (o < O2) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) acc[o][k] = b4l[o];
+          }
+    }
 
     for (int j = 0; j < M; ++j) {
       T zk[VEC];
       T bj = b3l[j];
 #pragma unroll
       for (int k = 0; k < VEC; ++k) zk[k] = bj;
-      for (int i = 0; i < I; ++i) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
         T wv = W3l[(size_t)j * I + i];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) zk[k] += wv * xr[i][k];
+              }
       }
 #pragma unroll
       for (int k = 0; k < VEC; ++k) zk[k] = gelu_erf_(zk[k]);
-      for (int o = 0; o < O2; ++o) {
+#pragma unroll
+      for (int o = 0; o < O2MAX; ++o) {
+        if (o < O2) {
         T w4 = W4l[(size_t)o * M + j];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) acc[o][k] += w4 * zk[k];
+              }
       }
     }
 
     T* ob = out + ((long)b * O2) * S + s;
-    if constexpr (VECTOR && std::is_same<T, float>::value) {
-      for (int o = 0; o < O2; ++o)
+#pragma unroll
+    for (int o = 0; o < O2MAX; ++o) {
+      if (o < O2) {
+      if constexpr (VECTOR && std::is_same<T, float>::value) {
         *reinterpret_cast<float4*>(ob + (long)o * S) =
             make_float4(acc[o][0], acc[o][1], acc[o][2], acc[o][3]);
-    } else {
-      for (int o = 0; o < O2; ++o)
-        for (int k = 0; k < nv; ++k) ob[(long)o * S + k] = acc[o][k];
+      } else {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          if (!full && k >= nv) break;
+          ob[(long)o * S + k] = acc[o][k];
+        }
+      }
+          }
     }
   }
 }
@@ -165,32 +190,55 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
     T gyv[O2MAX][VEC];
     const T* gyb = gy + ((long)b * O2) * S + s;
     if constexpr (VECTOR && std::is_same<T, float>::value) {
-#pragma unroll 4
-      for (int i = 0; i < I; ++i) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
         const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
         xr[i][0] = v.x; xr[i][1] = v.y; xr[i][2] = v.z; xr[i][3] = v.w;
+              }
       }
-      for (int o = 0; o < O2; ++o) {
+#pragma unroll
+      for (int o = 0; o < O2MAX; ++o) {
+        if (o < O2) {
         const float4 v = *reinterpret_cast<const float4*>(gyb + (long)o * S);
         gyv[o][0] = v.x; gyv[o][1] = v.y; gyv[o][2] = v.z; gyv[o][3] = v.w;
+              }
       }
     } else {
-      for (int i = 0; i < I; ++i)
-        for (int k = 0; k < nv; ++k) xr[i][k] = xb[(long)i * S + k];
-      for (int o = 0; o < O2; ++o)
-        for (int k = 0; k < nv; ++k) gyv[o][k] = gyb[(long)o * S + k];
-      for (int o = 0; o < O2; ++o)
-        for (int k = nv; k < VEC; ++k) gyv[o][k] = T(0);
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          xr[i][k] = (full || k < nv) ? xb[(long)i * S + k] : T(0);
+              }
+      }
+#pragma unroll
+      for (int o = 0; o < O2MAX; ++o) {
+        if (o < O2) {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k)
+          gyv[o][k] = (full || k < nv) ? gyb[(long)o * S + k] : T(0);
+              }
+      }
     }
 
-    for (int o = 0; o < O2; ++o)
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) gb4acc[o] += (full || k < nv) ? gyv[o][k] : T(0);
+    for (int o = 0; o < O2MAX; ++o) {
+      if (o < O2) {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) gb4acc[o] += gyv[o][k];
+          }
+    }
 
     T gxa[IMAX][VEC];
-    for (int i = 0; i < I; ++i)
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) {
+      if (i < I) {
 #pragma unroll
       for (int k = 0; k < VEC; ++k) gxa[i][k] = T(0);
+          }
+    }
 
     T* gz3b = gz3 + ((long)b * M) * S + s;
     for (int j = 0; j < M; ++j) {
@@ -199,10 +247,13 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
       T bj = b3l[j];
 #pragma unroll
       for (int k = 0; k < VEC; ++k) zk[k] = bj;
-      for (int i = 0; i < I; ++i) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
         T wv = W3l[(size_t)j * I + i];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) zk[k] += wv * xr[i][k];
+              }
       }
       T gk[VEC], dgk[VEC];
 #pragma unroll
@@ -214,10 +265,13 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
       T gzk[VEC];
 #pragma unroll
       for (int k = 0; k < VEC; ++k) gzk[k] = T(0);
-      for (int o = 0; o < O2; ++o) {
+#pragma unroll
+      for (int o = 0; o < O2MAX; ++o) {
+        if (o < O2) {
         T w4 = W4l[(size_t)o * M + j];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) gzk[k] += w4 * gyv[o][k];
+              }
       }
 #pragma unroll
       for (int k = 0; k < VEC; ++k) gzk[k] *= dgk[k];
@@ -227,12 +281,19 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
         *reinterpret_cast<float4*>(gz3b + (long)j * S) =
             make_float4(gzk[0], gzk[1], gzk[2], gzk[3]);
       } else {
-        for (int k = 0; k < nv; ++k) gz3b[(long)j * S + k] = gzk[k];
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          if (!full && k >= nv) break;
+          gz3b[(long)j * S + k] = gzk[k];
+        }
       }
-      for (int i = 0; i < I; ++i) {
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
         T wv = W3l[(size_t)j * I + i];
 #pragma unroll
         for (int k = 0; k < VEC; ++k) gxa[i][k] += wv * gzk[k];
+              }
       }
 
       // wave-reduced gb3[j] and gW4[o][j] partials
@@ -241,30 +302,43 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
       for (int k = 0; k < VEC; ++k) pb += (full || k < nv) ? gzk[k] : T(0);
       pb = wave_sum(pb);
       if (lane == 0) gb3w[wave * M + j] += pb;
-      for (int o = 0; o < O2; ++o) {
+#pragma unroll
+      for (int o = 0; o < O2MAX; ++o) {
+        if (o < O2) {
         T pw = T(0);
 #pragma unroll
         for (int k = 0; k < VEC; ++k) pw += (full || k < nv) ? gyv[o][k] * gk[k] : T(0);
         pw = wave_sum(pw);
         if (lane == 0) gW4w[(wave * O2 + o) * M + j] += pw;
+              }
       }
     }
 
     T* gxb = gx + ((long)b * I) * S + s;
-    if constexpr (VECTOR && std::is_same<T, float>::value) {
-      for (int i = 0; i < I; ++i)
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) {
+      if (i < I) {
+      if constexpr (VECTOR && std::is_same<T, float>::value) {
         *reinterpret_cast<float4*>(gxb + (long)i * S) =
             make_float4(gxa[i][0], gxa[i][1], gxa[i][2], gxa[i][3]);
-    } else {
-      for (int i = 0; i < I; ++i)
-        for (int k = 0; k < nv; ++k) gxb[(long)i * S + k] = gxa[i][k];
+      } else {
+#pragma unroll
+        for (int k = 0; k < VEC; ++k) {
+          if (!full && k >= nv) break;
+          gxb[(long)i * S + k] = gxa[i][k];
+        }
+      }
+          }
     }
   }
 
   // block-level flush: per-wave partials -> global atomics
-  for (int o = 0; o < O2; ++o) {
+#pragma unroll
+  for (int o = 0; o < O2MAX; ++o) {
+    if (o < O2) {
     T v = wave_sum(gb4acc[o]);
     if (lane == 0 && v != T(0)) atomicAdd(&gb4[o], v);
+      }
   }
   __syncthreads();
   for (int k = threadIdx.x; k < M; k += blockDim.x) {
@@ -321,7 +395,7 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
   int grid = grid_for_p((long)B * ((S + 3) / 4));
 
 #define PH_LAUNCH_F(V)                                                        \
-    hipLaunchKernelGGL((proj_head_fwd_kernel<scalar_t, 32, 8, 4, V>),         \
+    hipLaunchKernelGGL((proj_head_fwd_kernel<scalar_t, IM, OM, 4, V>),        \
                        dim3(grid), dim3(kBlock), smem, stream,                \
                        x.data_ptr<scalar_t>(), W3.data_ptr<scalar_t>(),       \
                        b3.data_ptr<scalar_t>(), W4.data_ptr<scalar_t>(),      \
@@ -330,7 +404,13 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "proj_head_fwd", [&] {
     size_t smem = sizeof(scalar_t) * ((size_t)M * I + M + (size_t)O2 * M + O2);
     bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), out.data_ptr()});
-    if (vec) { PH_LAUNCH_F(true) } else { PH_LAUNCH_F(false) }
+    if (I <= 24 && O2 <= 2) {
+      constexpr int IM = 24, OM = 2;
+      if (vec) { PH_LAUNCH_F(true) } else { PH_LAUNCH_F(false) }
+    } else {
+      constexpr int IM = 32, OM = 8;
+      if (vec) { PH_LAUNCH_F(true) } else { PH_LAUNCH_F(false) }
+    }
   });
 #undef PH_LAUNCH_F
   return out;
@@ -357,7 +437,7 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
   int grid = grid_for_p((long)B * ((S + 3) / 4));
 
 #define PH_LAUNCH_B(V)                                                        \
-    hipLaunchKernelGGL((proj_head_bwd_kernel<scalar_t, 32, 8, 4, V>),         \
+    hipLaunchKernelGGL((proj_head_bwd_kernel<scalar_t, IM, OM, 4, V>),        \
                        dim3(grid), dim3(kBlock), smem, stream,                \
                        gy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),       \
                        W3.data_ptr<scalar_t>(), b3.data_ptr<scalar_t>(),      \
@@ -371,7 +451,13 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
     TORCH_CHECK(smem <= 160 * 1024, "proj_head_bwd: LDS overflow");
     bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), gy.data_ptr(), gz3.data_ptr(),
                                     gx.data_ptr()});
-    if (vec) { PH_LAUNCH_B(true) } else { PH_LAUNCH_B(false) }
+    if (I <= 24 && O2 <= 2) {
+      constexpr int IM = 24, OM = 2;
+      if (vec) { PH_LAUNCH_B(true) } else { PH_LAUNCH_B(false) }
+    } else {
+      constexpr int IM = 32, OM = 8;
+      if (vec) { PH_LAUNCH_B(true) } else { PH_LAUNCH_B(false) }
+    }
   });
 #undef PH_LAUNCH_B
   return {gz3, gx, gb3, gW4, gb4};

@@ -108,23 +108,32 @@ __global__ __launch_bounds__(kBlock) void spectral_corner_kernel(
       if (CONJT) {
         // w index: [out=i_out][in=i(=o)][e]; here i iterates o (the weight's
         // second index), output k iterates the weight's first index.
-        for (int k = 0; k < olim; ++k) {
+#pragma unroll
+        for (int k = 0; k < OTILE; ++k) {
+          if (k < olim) {
           long widx = 2 * ((((long)(o0 + k)) * O + i) * g.nelem + e);
           cmac_conj(accr[k], acci[k], w[widx], w[widx + 1], xr, xi);
+                  }
         }
       } else {
         const T* wb = w + 2 * (((long)i * O + o0) * g.nelem + e);
-        for (int k = 0; k < olim; ++k) {
+#pragma unroll
+        for (int k = 0; k < OTILE; ++k) {
+          if (k < olim) {
           cmac(accr[k], acci[k], xr, xi, wb[2 * (long)k * g.nelem],
                wb[2 * (long)k * g.nelem + 1]);
+                  }
         }
       }
     }
 
     T* yb = y + 2 * (((long)b * n_out + o0) * Ftot + f);
-    for (int k = 0; k < olim; ++k) {
+#pragma unroll
+    for (int k = 0; k < OTILE; ++k) {
+      if (k < olim) {
       yb[2 * (long)k * Ftot] = accr[k];
       yb[2 * (long)k * Ftot + 1] = acci[k];
+          }
     }
   }
 }
