@@ -9,12 +9,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused channel linear + bias + gelu: returns (y, z)");
   m.def("channel_mix_fwd_t", &channel_mix_fwd_t,
         "transposed channel contraction (grad-x)");
+  m.def("channel_mix_bwd_w", &channel_mix_bwd_w,
+        "split-s grad-W (+grad-bias) reduction");
   m.def("gelu_fwd", &gelu_fwd, "exact gelu");
   m.def("gelu_bwd", &gelu_bwd, "gelu backward (gy, z) -> gz");
   m.def("add_gelu_fwd", &add_gelu_fwd, "fused residual add + gelu: returns (y, z)");
   m.def("proj_head_fwd", &proj_head_fwd, "fused linear->gelu->linear head");
   m.def("proj_head_bwd", &proj_head_bwd,
-        "fused head backward: returns (gz3, gx, gb3, gW4, gb4)");
+        "fused head backward: returns (gz3, gb3, gW4, gb4)");
   m.def("spectral_corner_fwd", &spectral_corner_fwd,
         "corner-block complex spectral contraction (accumulate into y box)");
   m.def("spectral_corner_bwd_x", &spectral_corner_bwd_x,

@@ -21,10 +21,16 @@ std::vector<at::Tensor> add_gelu_fwd(const at::Tensor& a, const at::Tensor& b);
 at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
                          const at::Tensor& b3, const at::Tensor& W4,
                          const at::Tensor& b4);
-// returns {gz3, gx, gb3, gW4, gb4}; grad-W3 = gz3 @ x^T (library GEMM)
+// returns {gz3, gb3, gW4, gb4}; grad-x = W3^T gz3 (channel_mix_fwd_t) and
+// grad-W3 = gz3 @ x^T (channel_mix_bwd_w)
 std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& W3, const at::Tensor& b3,
                                       const at::Tensor& W4);
+
+// split-s outer-product reduction: gW[o,i] = sum_{b,s} gz[b,o,s] x[b,i,s]
+// (+ gb[o] = sum gz when want_bias); requires I <= 32.
+std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
+                                          bool want_bias);
 
 // corner-block spectral contraction on the truncated complex spectrum:
 //   y[b,o,f] += sum_i x[b,i,f] * w[i,o,f_box]  for f in the corner box

@@ -524,3 +524,139 @@ std::vector<at::Tensor> add_gelu_fwd(const at::Tensor& a, const at::Tensor& b) {
   });
   return {y, z};
 }
+
+// ---------------------------------------------------------------------------
+// grad-W outer-product reduction: gW[o,i] = sum_{b,s} A[b,o,s] B[b,i,s]
+// (optionally gb[o] = sum A[b,o,s]).
+//
+// rocBLAS picks a no-split-K schedule for these M,N<=128 / K~10^7 GEMMs
+// (observed: ~2 workgroups on 256 CUs, 3.2 ms for a 1.3 GB reduction); here
+// the s axis is split across blocks (grid ordered s-chunk-major so the
+// o-tiles sweeping one s-chunk hit L2/L3 on the B re-reads), each wave owns
+// one A row with the <=32 B rows accumulated in registers, and partials are
+// wave-shuffle-reduced then atomically added into gW once per block.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <typename T>
+__device__ __forceinline__ T gw_wave_sum(T v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <typename T, int ICAP, bool BIAS, bool VECTOR>
+__global__ __launch_bounds__(kBlock) void gw_outer_kernel(
+    const T* __restrict__ A, const T* __restrict__ Bm,
+    T* __restrict__ gW, T* __restrict__ gb,
+    int B, int O, int I, long S, int n_schunk) {
+  // grid: blockIdx.x = schunk * o_tiles + o_tile (s-chunk-major)
+  const int o_tiles = (O + 3) / 4;
+  const int schunk = blockIdx.x / o_tiles;
+  const int o_tile = blockIdx.x % o_tiles;
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const int o = o_tile * 4 + wave;
+
+  // chunk boundaries on 256-element grain so the float4 lanes stay aligned
+  long chunk_sz = ((S + n_schunk - 1) / n_schunk + 255) & ~255L;
+  long s0 = (long)schunk * chunk_sz;
+  long s1 = min(S, s0 + chunk_sz);
+
+  T acc[ICAP];
+#pragma unroll
+  for (int i = 0; i < ICAP; ++i) acc[i] = T(0);
+  T bacc = T(0);
+
+  if (o < O) {
+    for (int b = 0; b < B; ++b) {
+      const T* Ab = A + ((long)b * O + o) * S;
+      const T* Bb = Bm + ((long)b * I) * S;
+      if constexpr (VECTOR && std::is_same<T, float>::value) {
+        const long tail0 = s0 + ((s1 - s0) / (64 * 4)) * (64 * 4);
+        for (long s = s0 + (long)lane * 4; s < tail0; s += 64 * 4) {
+          const float4 av = *reinterpret_cast<const float4*>(Ab + s);
+          if (BIAS) bacc += av.x + av.y + av.z + av.w;
+#pragma unroll
+          for (int i = 0; i < ICAP; ++i) {
+            if (i < I) {
+              const float4 bv = *reinterpret_cast<const float4*>(Bb + (long)i * S + s);
+              acc[i] += av.x * bv.x + av.y * bv.y + av.z * bv.z + av.w * bv.w;
+            }
+          }
+        }
+        // tail (at most 255 elements at the end of the last chunk)
+        for (long s = tail0 + lane; s < s1; s += 64) {
+          T av = Ab[s];
+          if (BIAS) bacc += av;
+#pragma unroll
+          for (int i = 0; i < ICAP; ++i)
+            if (i < I) acc[i] += av * Bb[(long)i * S + s];
+        }
+      } else {
+        for (long s = s0 + lane; s < s1; s += 64) {
+          T av = Ab[s];
+          if (BIAS) bacc += av;
+#pragma unroll
+          for (int i = 0; i < ICAP; ++i)
+            if (i < I) acc[i] += av * Bb[(long)i * S + s];
+        }
+      }
+    }
+  }
+
+  // flush: wave-reduce each acc[i], lane 0 atomically accumulates
+#pragma unroll
+  for (int i = 0; i < ICAP; ++i) {
+    if (i < I) {
+      T v = gw_wave_sum(acc[i]);
+      if (lane == 0 && o < O && v != T(0)) atomicAdd(&gW[(size_t)o * I + i], v);
+    }
+  }
+  if (BIAS) {
+    T v = gw_wave_sum(bacc);
+    if (lane == 0 && o < O && v != T(0)) atomicAdd(&gb[o], v);
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
+                                          bool want_bias) {
+  check_f(gz, "gz"); check_f(x, "x");
+  TORCH_CHECK(gz.dim() == 3 && x.dim() == 3, "gz/x must be [B,*,S]");
+  int B = (int)gz.size(0), O = (int)gz.size(1), I = (int)x.size(1);
+  long S = gz.size(2);
+  TORCH_CHECK(x.size(0) == B && x.size(2) == S, "shape mismatch");
+  TORCH_CHECK(I <= 32, "channel_mix_bwd_w: I must be <= 32");
+
+  auto gW = at::zeros({O, I}, gz.options());
+  auto gb = want_bias ? at::zeros({O}, gz.options())
+                      : at::empty({0}, gz.options());
+  if (gz.numel() == 0) return {gW, gb};
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int o_tiles = (O + 3) / 4;
+  int n_schunk = (int)std::max(1L, std::min((long)(4096 / o_tiles), S / (64 * 16)));
+  int grid = n_schunk * o_tiles;
+
+#define GW_LAUNCH(BIAS_, V)                                                     \
+  hipLaunchKernelGGL((gw_outer_kernel<scalar_t, 32, BIAS_, V>), dim3(grid),     \
+                     dim3(kBlock), 0, stream, gz.data_ptr<scalar_t>(),          \
+                     x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),           \
+                     want_bias ? gb.data_ptr<scalar_t>() : nullptr,             \
+                     B, O, I, S, n_schunk);
+  AT_DISPATCH_FLOATING_TYPES(gz.scalar_type(), "channel_mix_bwd_w", [&] {
+    bool vec = std::is_same<scalar_t, float>::value && (S % 4 == 0) &&
+               ((reinterpret_cast<uintptr_t>(gz.data_ptr()) & 15) == 0) &&
+               ((reinterpret_cast<uintptr_t>(x.data_ptr()) & 15) == 0);
+    if (want_bias) {
+      if (vec) { GW_LAUNCH(true, true) } else { GW_LAUNCH(true, false) }
+    } else {
+      if (vec) { GW_LAUNCH(false, true) } else { GW_LAUNCH(false, false) }
+    }
+  });
+#undef GW_LAUNCH
+  return {gW, gb};
+}

@@ -151,7 +151,7 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ W3, const T* __restrict__ b3,
     const T* __restrict__ W4,
-    T* __restrict__ gz3, T* __restrict__ gx,
+    T* __restrict__ gz3,
     T* __restrict__ gb3, T* __restrict__ gW4, T* __restrict__ gb4,
     int B, int I, int M, int O2, long S) {
   extern __shared__ __align__(16) char smem_raw[];
@@ -231,14 +231,6 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
           }
     }
 
-    T gxa[IMAX][VEC];
-#pragma unroll
-    for (int i = 0; i < IMAX; ++i) {
-      if (i < I) {
-#pragma unroll
-      for (int k = 0; k < VEC; ++k) gxa[i][k] = T(0);
-          }
-    }
 
     T* gz3b = gz3 + ((long)b * M) * S + s;
     for (int j = 0; j < M; ++j) {
@@ -276,24 +268,15 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
 #pragma unroll
       for (int k = 0; k < VEC; ++k) gzk[k] *= dgk[k];
 
-      // store gz3 and accumulate gx
+      // store gz3 (grad-x and grad-W3 are computed from it afterwards)
       if constexpr (VECTOR && std::is_same<T, float>::value) {
         *reinterpret_cast<float4*>(gz3b + (long)j * S) =
             make_float4(gzk[0], gzk[1], gzk[2], gzk[3]);
       } else {
 #pragma unroll
         for (int k = 0; k < VEC; ++k) {
-          if (!full && k >= nv) break;
-          gz3b[(long)j * S + k] = gzk[k];
+          if (full || k < nv) gz3b[(long)j * S + k] = gzk[k];
         }
-      }
-#pragma unroll
-      for (int i = 0; i < IMAX; ++i) {
-        if (i < I) {
-        T wv = W3l[(size_t)j * I + i];
-#pragma unroll
-        for (int k = 0; k < VEC; ++k) gxa[i][k] += wv * gzk[k];
-              }
       }
 
       // wave-reduced gb3[j] and gW4[o][j] partials
@@ -314,22 +297,6 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
       }
     }
 
-    T* gxb = gx + ((long)b * I) * S + s;
-#pragma unroll
-    for (int i = 0; i < IMAX; ++i) {
-      if (i < I) {
-      if constexpr (VECTOR && std::is_same<T, float>::value) {
-        *reinterpret_cast<float4*>(gxb + (long)i * S) =
-            make_float4(gxa[i][0], gxa[i][1], gxa[i][2], gxa[i][3]);
-      } else {
-#pragma unroll
-        for (int k = 0; k < VEC; ++k) {
-          if (!full && k >= nv) break;
-          gxb[(long)i * S + k] = gxa[i][k];
-        }
-      }
-          }
-    }
   }
 
   // block-level flush: per-wave partials -> global atomics
@@ -427,11 +394,10 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
   TORCH_CHECK(I <= 32 && O2 <= 8 && M <= 512, "proj_head: unsupported dims");
 
   auto gz3 = at::empty({B, M, S}, x.options());
-  auto gx = at::empty_like(x);
   auto gb3 = at::zeros({M}, x.options());
   auto gW4 = at::zeros({O2, M}, x.options());
   auto gb4 = at::zeros({O2}, x.options());
-  if (x.numel() == 0) return {gz3, gx, gb3, gW4, gb4};
+  if (x.numel() == 0) return {gz3, gb3, gW4, gb4};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_p((long)B * ((S + 3) / 4));
@@ -442,15 +408,14 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
                        gy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),       \
                        W3.data_ptr<scalar_t>(), b3.data_ptr<scalar_t>(),      \
                        W4.data_ptr<scalar_t>(), gz3.data_ptr<scalar_t>(),     \
-                       gx.data_ptr<scalar_t>(), gb3.data_ptr<scalar_t>(),     \
+                       gb3.data_ptr<scalar_t>(),                              \
                        gW4.data_ptr<scalar_t>(), gb4.data_ptr<scalar_t>(),    \
                        B, I, M, O2, S);
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "proj_head_bwd", [&] {
     size_t smem = sizeof(scalar_t) *
         ((size_t)M * I + M + (size_t)O2 * M + 4 * (size_t)M + 4 * (size_t)O2 * M);
     TORCH_CHECK(smem <= 160 * 1024, "proj_head_bwd: LDS overflow");
-    bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), gy.data_ptr(), gz3.data_ptr(),
-                                    gx.data_ptr()});
+    bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), gy.data_ptr(), gz3.data_ptr()});
     if (I <= 24 && O2 <= 2) {
       constexpr int IM = 24, OM = 2;
       if (vec) { PH_LAUNCH_B(true) } else { PH_LAUNCH_B(false) }
@@ -460,5 +425,5 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
     }
   });
 #undef PH_LAUNCH_B
-  return {gz3, gx, gb3, gW4, gb4};
+  return {gz3, gb3, gW4, gb4};
 }

@@ -83,9 +83,15 @@ class _ChannelMixFn(torch.autograd.Function):
             gx = ext.channel_mix_fwd_t(gz, W)  # sum_o W[o,i] gz[b,o,s]
         else:
             gx = torch.einsum("oi,bos->bis", W, gz)
-        # grad W / b: GEMM-shaped reductions (library GEMM)
-        gW = torch.einsum("bos,bis->oi", gz, x3)
-        gb = gz.sum(dim=(0, 2)) if ctx.has_bias else None
+        # grad W / b
+        if gy.is_cuda and x3.shape[1] <= 32:
+            ext = _ext.get(required=True)
+            gW, gb = ext.channel_mix_bwd_w(gz.contiguous(), x3, ctx.has_bias)
+            if not ctx.has_bias:
+                gb = None
+        else:
+            gW = torch.einsum("bos,bis->oi", gz, x3)
+            gb = gz.sum(dim=(0, 2)) if ctx.has_bias else None
         return gx.reshape(ctx.x_shape), gW, gb, None
 
 

@@ -37,10 +37,10 @@ class _ProjHeadFn(torch.autograd.Function):
     def backward(ctx, gy):
         x3, W3, b3, W4 = ctx.saved_tensors
         ext = _ext.get(required=True)
-        gz3, gx, gb3, gW4, gb4 = ext.proj_head_bwd(
+        gz3, gb3, gW4, gb4 = ext.proj_head_bwd(
             gy.contiguous(), x3, W3.contiguous(), b3.contiguous(), W4.contiguous())
-        # grad-W3 is a [M,S]x[S,I] reduction: library GEMM (rocBLAS)
-        gW3 = torch.einsum("bms,bis->mi", gz3, x3)
+        gx = ext.channel_mix_fwd_t(gz3, W3.contiguous())   # W3^T @ gz3
+        gW3, _ = ext.channel_mix_bwd_w(gz3, x3, False)     # gz3 @ x^T
         return gx.reshape(ctx.x_shape), gW3, gb3, gW4, gb4
 
 
