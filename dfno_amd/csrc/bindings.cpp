@@ -21,4 +21,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "corner-block complex spectral contraction (accumulate into y box)");
   m.def("spectral_corner_bwd_x", &spectral_corner_bwd_x,
         "corner-block spectral contraction adjoint wrt x");
+  m.def("spectral_corners_fwd", &spectral_corners_fwd,
+        "all corner boxes in one launch");
+  m.def("spectral_corners_bwd_x", &spectral_corners_bwd_x,
+        "all corner boxes adjoint wrt x in one launch");
 }

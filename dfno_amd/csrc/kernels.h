@@ -39,3 +39,9 @@ void spectral_corner_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y
 // gx[b,i,f] += sum_o conj(w[i,o,f_box]) * gy[b,o,f]
 void spectral_corner_bwd_x(const at::Tensor& gy, const at::Tensor& w, at::Tensor& gx,
                            std::vector<int64_t> starts);
+
+// single-launch variants covering all corner boxes of one spectral conv
+void spectral_corners_fwd(const at::Tensor& x, std::vector<at::Tensor> ws,
+                          at::Tensor& y, std::vector<std::vector<int64_t>> starts);
+void spectral_corners_bwd_x(const at::Tensor& gy, std::vector<at::Tensor> ws,
+                            at::Tensor& gx, std::vector<std::vector<int64_t>> starts);

@@ -546,77 +546,108 @@ __device__ __forceinline__ T gw_wave_sum(T v) {
   return v;
 }
 
-template <typename T, int ICAP, bool BIAS, bool VECTOR>
+template <typename T, int ICAP, int OW, bool BIAS, bool VECTOR>
 __global__ __launch_bounds__(kBlock) void gw_outer_kernel(
     const T* __restrict__ A, const T* __restrict__ Bm,
     T* __restrict__ gW, T* __restrict__ gb,
     int B, int O, int I, long S, int n_schunk) {
-  // grid: blockIdx.x = schunk * o_tiles + o_tile (s-chunk-major)
-  const int o_tiles = (O + 3) / 4;
+  // grid: blockIdx.x = schunk * o_tiles + o_tile (s-chunk-major); each wave
+  // owns OW consecutive output rows (cuts B re-reads by OW and raises the
+  // fma:load ratio).
+  const int o_tiles = (O + 4 * OW - 1) / (4 * OW);
   const int schunk = blockIdx.x / o_tiles;
   const int o_tile = blockIdx.x % o_tiles;
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
-  const int o = o_tile * 4 + wave;
+  const int o0 = (o_tile * 4 + wave) * OW;
 
   // chunk boundaries on 256-element grain so the float4 lanes stay aligned
   long chunk_sz = ((S + n_schunk - 1) / n_schunk + 255) & ~255L;
   long s0 = (long)schunk * chunk_sz;
   long s1 = min(S, s0 + chunk_sz);
 
-  T acc[ICAP];
+  T acc[OW][ICAP];
 #pragma unroll
-  for (int i = 0; i < ICAP; ++i) acc[i] = T(0);
-  T bacc = T(0);
+  for (int w = 0; w < OW; ++w)
+#pragma unroll
+    for (int i = 0; i < ICAP; ++i) acc[w][i] = T(0);
+  T bacc[OW];
+#pragma unroll
+  for (int w = 0; w < OW; ++w) bacc[w] = T(0);
 
-  if (o < O) {
+  if (o0 < O) {
     for (int b = 0; b < B; ++b) {
-      const T* Ab = A + ((long)b * O + o) * S;
+      const T* Ab = A + ((long)b * O + o0) * S;
       const T* Bb = Bm + ((long)b * I) * S;
       if constexpr (VECTOR && std::is_same<T, float>::value) {
         const long tail0 = s0 + ((s1 - s0) / (64 * 4)) * (64 * 4);
         for (long s = s0 + (long)lane * 4; s < tail0; s += 64 * 4) {
-          const float4 av = *reinterpret_cast<const float4*>(Ab + s);
-          if (BIAS) bacc += av.x + av.y + av.z + av.w;
+          float4 av[OW];
+#pragma unroll
+          for (int w = 0; w < OW; ++w) {
+            if (o0 + w < O) {
+              av[w] = *reinterpret_cast<const float4*>(Ab + (long)w * S + s);
+              if (BIAS) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
+            }
+          }
 #pragma unroll
           for (int i = 0; i < ICAP; ++i) {
             if (i < I) {
               const float4 bv = *reinterpret_cast<const float4*>(Bb + (long)i * S + s);
-              acc[i] += av.x * bv.x + av.y * bv.y + av.z * bv.z + av.w * bv.w;
+#pragma unroll
+              for (int w = 0; w < OW; ++w) {
+                if (o0 + w < O)
+                  acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
+                               av[w].z * bv.z + av[w].w * bv.w;
+              }
             }
           }
         }
-        // tail (at most 255 elements at the end of the last chunk)
         for (long s = tail0 + lane; s < s1; s += 64) {
-          T av = Ab[s];
-          if (BIAS) bacc += av;
 #pragma unroll
-          for (int i = 0; i < ICAP; ++i)
-            if (i < I) acc[i] += av * Bb[(long)i * S + s];
+          for (int w = 0; w < OW; ++w) {
+            if (o0 + w < O) {
+              T av = Ab[(long)w * S + s];
+              if (BIAS) bacc[w] += av;
+#pragma unroll
+              for (int i = 0; i < ICAP; ++i)
+                if (i < I) acc[w][i] += av * Bb[(long)i * S + s];
+            }
+          }
         }
       } else {
         for (long s = s0 + lane; s < s1; s += 64) {
-          T av = Ab[s];
-          if (BIAS) bacc += av;
 #pragma unroll
-          for (int i = 0; i < ICAP; ++i)
-            if (i < I) acc[i] += av * Bb[(long)i * S + s];
+          for (int w = 0; w < OW; ++w) {
+            if (o0 + w < O) {
+              T av = Ab[(long)w * S + s];
+              if (BIAS) bacc[w] += av;
+#pragma unroll
+              for (int i = 0; i < ICAP; ++i)
+                if (i < I) acc[w][i] += av * Bb[(long)i * S + s];
+            }
+          }
         }
       }
     }
   }
 
-  // flush: wave-reduce each acc[i], lane 0 atomically accumulates
+  // flush: wave-reduce each acc, lane 0 atomically accumulates
 #pragma unroll
-  for (int i = 0; i < ICAP; ++i) {
-    if (i < I) {
-      T v = gw_wave_sum(acc[i]);
-      if (lane == 0 && o < O && v != T(0)) atomicAdd(&gW[(size_t)o * I + i], v);
+  for (int w = 0; w < OW; ++w) {
+    if (o0 + w < O) {
+#pragma unroll
+      for (int i = 0; i < ICAP; ++i) {
+        if (i < I) {
+          T v = gw_wave_sum(acc[w][i]);
+          if (lane == 0 && v != T(0)) atomicAdd(&gW[(size_t)(o0 + w) * I + i], v);
+        }
+      }
+      if (BIAS) {
+        T v = gw_wave_sum(bacc[w]);
+        if (lane == 0 && v != T(0)) atomicAdd(&gb[o0 + w], v);
+      }
     }
-  }
-  if (BIAS) {
-    T v = gw_wave_sum(bacc);
-    if (lane == 0 && o < O && v != T(0)) atomicAdd(&gb[o], v);
   }
 }
 
@@ -637,26 +668,40 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
   if (gz.numel() == 0) return {gW, gb};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  int o_tiles = (O + 3) / 4;
+  // OW=4 when O is large enough; ICAP sized to I
+  const int OW = (O >= 8) ? 4 : 1;
+  int o_tiles = (O + 4 * OW - 1) / (4 * OW);
   int n_schunk = (int)std::max(1L, std::min((long)(4096 / o_tiles), S / (64 * 16)));
   int grid = n_schunk * o_tiles;
 
-#define GW_LAUNCH(BIAS_, V)                                                     \
-  hipLaunchKernelGGL((gw_outer_kernel<scalar_t, 32, BIAS_, V>), dim3(grid),     \
-                     dim3(kBlock), 0, stream, gz.data_ptr<scalar_t>(),          \
+#define GW_LAUNCH(ICAP_, OW_, BIAS_, V)                                         \
+  hipLaunchKernelGGL((gw_outer_kernel<scalar_t, ICAP_, OW_, BIAS_, V>),         \
+                     dim3(grid), dim3(kBlock), 0, stream,                       \
+                     gz.data_ptr<scalar_t>(),                                   \
                      x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),           \
                      want_bias ? gb.data_ptr<scalar_t>() : nullptr,             \
                      B, O, I, S, n_schunk);
+#define GW_DISPATCH2(ICAP_, OW_)                                                \
+    if (want_bias) {                                                            \
+      if (vec) { GW_LAUNCH(ICAP_, OW_, true, true) }                            \
+      else { GW_LAUNCH(ICAP_, OW_, true, false) }                               \
+    } else {                                                                    \
+      if (vec) { GW_LAUNCH(ICAP_, OW_, false, true) }                           \
+      else { GW_LAUNCH(ICAP_, OW_, false, false) }                              \
+    }
   AT_DISPATCH_FLOATING_TYPES(gz.scalar_type(), "channel_mix_bwd_w", [&] {
     bool vec = std::is_same<scalar_t, float>::value && (S % 4 == 0) &&
                ((reinterpret_cast<uintptr_t>(gz.data_ptr()) & 15) == 0) &&
                ((reinterpret_cast<uintptr_t>(x.data_ptr()) & 15) == 0);
-    if (want_bias) {
-      if (vec) { GW_LAUNCH(true, true) } else { GW_LAUNCH(true, false) }
+    if (OW == 4) {
+      if (I <= 8) { GW_DISPATCH2(8, 4) }
+      else if (I <= 24) { GW_DISPATCH2(24, 4) }
+      else { GW_DISPATCH2(32, 4) }
     } else {
-      if (vec) { GW_LAUNCH(false, true) } else { GW_LAUNCH(false, false) }
+      GW_DISPATCH2(32, 1)
     }
   });
+#undef GW_DISPATCH2
 #undef GW_LAUNCH
   return {gW, gb};
 }

@@ -138,6 +138,83 @@ __global__ __launch_bounds__(kBlock) void spectral_corner_kernel(
   }
 }
 
+// -------- multi-corner single-launch variant --------------------------------
+constexpr int kMaxCorners = 8;
+
+template <typename T>
+struct MultiGeom {
+  BoxGeom g[kMaxCorners];
+  const T* w[kMaxCorners];
+  long cum[kMaxCorners + 1];  // cumulative work (nelem * ntiles) per corner
+  int ncorners;
+};
+
+template <typename T, int OTILE, bool CONJT>
+__global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
+    const T* __restrict__ x, MultiGeom<T> mg, T* __restrict__ y,
+    int B, int I, int O, long Ftot) {
+  const int n_out = CONJT ? I : O;
+  const int n_in = CONJT ? O : I;
+  const long ntiles = (n_out + OTILE - 1) / OTILE;
+  const long total = mg.cum[mg.ncorners] * B;
+
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < total; t += stride) {
+    long work = t % mg.cum[mg.ncorners];
+    int b = (int)(t / mg.cum[mg.ncorners]);
+    int c = 0;
+    while (work >= mg.cum[c + 1]) ++c;
+    work -= mg.cum[c];
+    const BoxGeom& g = mg.g[c];
+    const T* w = mg.w[c];
+    long e = work % g.nelem;
+    int tile = (int)(work / g.nelem);
+
+    long f = box_to_global(e, g);
+    int o0 = tile * OTILE;
+    int olim = min(OTILE, n_out - o0);
+
+    T accr[OTILE], acci[OTILE];
+#pragma unroll
+    for (int k = 0; k < OTILE; ++k) { accr[k] = T(0); acci[k] = T(0); }
+
+    const T* xb = x + 2 * (((long)b * n_in) * Ftot + f);
+    for (int i = 0; i < n_in; ++i) {
+      T xr = xb[2 * (long)i * Ftot];
+      T xi = xb[2 * (long)i * Ftot + 1];
+      if (CONJT) {
+#pragma unroll
+        for (int k = 0; k < OTILE; ++k) {
+          if (k < olim) {
+            long widx = 2 * ((((long)(o0 + k)) * O + i) * g.nelem + e);
+            cmac_conj(accr[k], acci[k], w[widx], w[widx + 1], xr, xi);
+          }
+        }
+      } else {
+        const T* wb = w + 2 * (((long)i * O + o0) * g.nelem + e);
+#pragma unroll
+        for (int k = 0; k < OTILE; ++k) {
+          if (k < olim) {
+            cmac(accr[k], acci[k], xr, xi, wb[2 * (long)k * g.nelem],
+                 wb[2 * (long)k * g.nelem + 1]);
+          }
+        }
+      }
+    }
+
+    T* yb = y + 2 * (((long)b * n_out + o0) * Ftot + f);
+#pragma unroll
+    for (int k = 0; k < OTILE; ++k) {
+      if (k < olim) {
+        yb[2 * (long)k * Ftot] = accr[k];
+        yb[2 * (long)k * Ftot + 1] = acci[k];
+      }
+    }
+  }
+}
+
 int grid_for_s(long work) {
   long g = (work + kBlock - 1) / kBlock;
   long cap = 256L * 16;
@@ -238,5 +315,79 @@ void spectral_corner_bwd_x(const at::Tensor& gy, const at::Tensor& w, at::Tensor
                        reinterpret_cast<const double*>(gy.data_ptr()),
                        reinterpret_cast<const double*>(w.data_ptr()),
                        reinterpret_cast<double*>(gx.data_ptr()), g, B, I, O, Ftot);
+  }
+}
+
+// single-launch multi-corner entries ----------------------------------------
+
+template <typename T, bool CONJT>
+static void launch_corners(const at::Tensor& x,
+                           const std::vector<at::Tensor>& ws,
+                           at::Tensor& y,
+                           const std::vector<std::vector<int64_t>>& starts,
+                           int B, int I, int O, long Ftot) {
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  constexpr int OT = 8;
+  const int n_out = CONJT ? I : O;
+  long ntiles = (n_out + OT - 1) / OT;
+
+  size_t idx = 0;
+  while (idx < ws.size()) {
+    MultiGeom<T> mg{};
+    mg.ncorners = 0;
+    mg.cum[0] = 0;
+    while (idx < ws.size() && mg.ncorners < kMaxCorners) {
+      BoxGeom g = make_geom(x, ws[idx], starts[idx]);
+      if (g.nelem == 0) { ++idx; continue; }
+      int c = mg.ncorners++;
+      mg.g[c] = g;
+      mg.w[c] = reinterpret_cast<const T*>(ws[idx].data_ptr());
+      mg.cum[c + 1] = mg.cum[c] + g.nelem * ntiles;
+      ++idx;
+    }
+    if (mg.ncorners == 0) continue;
+    int grid = grid_for_s(mg.cum[mg.ncorners] * B);
+    hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT>), dim3(grid),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const T*>(x.data_ptr()), mg,
+                       reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
+  }
+}
+
+void spectral_corners_fwd(const at::Tensor& x, std::vector<at::Tensor> ws,
+                          at::Tensor& y, std::vector<std::vector<int64_t>> starts) {
+  check_c(x, "x"); check_c(y, "y");
+  TORCH_CHECK(ws.size() == starts.size(), "ws/starts size mismatch");
+  int B = (int)x.size(0), I = (int)x.size(1), O = (int)y.size(1);
+  long Ftot = 1;
+  for (int d = 2; d < x.dim(); ++d) Ftot *= x.size(d);
+  for (auto& w : ws) {
+    check_c(w, "w");
+    TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "spectral: w shape");
+  }
+  if (B == 0 || ws.empty()) return;
+  if (x.scalar_type() == at::kComplexFloat) {
+    launch_corners<float, false>(x, ws, y, starts, B, I, O, Ftot);
+  } else {
+    launch_corners<double, false>(x, ws, y, starts, B, I, O, Ftot);
+  }
+}
+
+void spectral_corners_bwd_x(const at::Tensor& gy, std::vector<at::Tensor> ws,
+                            at::Tensor& gx, std::vector<std::vector<int64_t>> starts) {
+  check_c(gy, "gy"); check_c(gx, "gx");
+  TORCH_CHECK(ws.size() == starts.size(), "ws/starts size mismatch");
+  int B = (int)gy.size(0), O = (int)gy.size(1), I = (int)gx.size(1);
+  long Ftot = 1;
+  for (int d = 2; d < gy.dim(); ++d) Ftot *= gy.size(d);
+  for (auto& w : ws) {
+    check_c(w, "w");
+    TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "spectral: w shape");
+  }
+  if (B == 0 || ws.empty()) return;
+  if (gy.scalar_type() == at::kComplexFloat) {
+    launch_corners<float, true>(gy, ws, gx, starts, B, I, O, Ftot);
+  } else {
+    launch_corners<double, true>(gy, ws, gx, starts, B, I, O, Ftot);
   }
 }

@@ -44,9 +44,9 @@ class _SpectralConvFn(torch.autograd.Function):
         if x.is_cuda and x.numel() > 0:
             ext = _ext.get(required=True)
             xc = x.contiguous()
-            for w, bounds in zip(weights, bounds_list):
-                starts = [a for a, _ in bounds]
-                ext.spectral_corner_fwd(xc, w.contiguous(), y, starts)
+            ext.spectral_corners_fwd(
+                xc, [w.contiguous() for w in weights], y,
+                [[a for a, _ in bounds] for bounds in bounds_list])
             x_saved = xc
         else:
             x_saved = x
@@ -67,9 +67,10 @@ class _SpectralConvFn(torch.autograd.Function):
         gws = []
         if gy.is_cuda and gy.numel() > 0:
             ext = _ext.get(required=True)
+            ext.spectral_corners_bwd_x(
+                gy, [w.contiguous() for w in weights], gx,
+                [[a for a, _ in bounds] for bounds in bounds_list])
             for w, bounds in zip(weights, bounds_list):
-                starts = [a for a, _ in bounds]
-                ext.spectral_corner_bwd_x(gy, w.contiguous(), gx, starts)
                 sl = _corner_slices(bounds)
                 gws.append(torch.einsum("bo...,bi...->io...", gy[sl], x[sl].conj()))
         else:
