@@ -651,6 +651,122 @@ __global__ __launch_bounds__(kBlock) void gw_outer_kernel(
   }
 }
 
+
+// LDS-staged variant: the B tile (the I<=32 narrow operand) is staged into
+// LDS once per 256-element s-tile and shared by all 4 waves x OW rows of the
+// block — without this, co-resident blocks evict the shared rows from L1/L2
+// and the measured traffic is ~4x algorithmic (0.76 TB/s vs 5.3 peak).
+template <typename T, int ICAP, int OW>
+__global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
+    const T* __restrict__ A, const T* __restrict__ Bm,
+    T* __restrict__ gW, T* __restrict__ gb,
+    int B, int O, int I, long S, int n_schunk, bool want_bias) {
+  constexpr int TS = 256;  // floats per s-tile
+  __shared__ float btile[ICAP * TS];
+
+  const int o_tiles = (O + 4 * OW - 1) / (4 * OW);
+  const int schunk = blockIdx.x / o_tiles;
+  const int o_tile = blockIdx.x % o_tiles;
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+  const int o0 = (o_tile * 4 + wave) * OW;
+
+  long chunk_sz = ((S + n_schunk - 1) / n_schunk + (TS - 1)) & ~((long)TS - 1);
+  long s0 = (long)schunk * chunk_sz;
+  long s1 = min(S, s0 + chunk_sz);
+
+  float acc[OW][ICAP];
+#pragma unroll
+  for (int w = 0; w < OW; ++w)
+#pragma unroll
+    for (int i = 0; i < ICAP; ++i) acc[w][i] = 0.f;
+  float bacc[OW];
+#pragma unroll
+  for (int w = 0; w < OW; ++w) bacc[w] = 0.f;
+
+  for (int b = 0; b < B; ++b) {
+    const float* Ab = reinterpret_cast<const float*>(A) + ((long)b * O + o0) * S;
+    const float* Bb = reinterpret_cast<const float*>(Bm) + ((long)b * I) * S;
+    for (long st = s0; st < s1; st += TS) {
+      const bool fullt = (st + TS) <= s1;
+      __syncthreads();
+      // cooperative stage of B[0:I][st:st+TS] into LDS
+      if (fullt) {
+        for (int idx = threadIdx.x * 4; idx < I * TS; idx += kBlock * 4) {
+          int row = idx / TS, col = idx % TS;
+          *reinterpret_cast<float4*>(&btile[row * TS + col]) =
+              *reinterpret_cast<const float4*>(Bb + (long)row * S + st + col);
+        }
+      } else {
+        int ts = (int)(s1 - st);
+        for (int idx = threadIdx.x; idx < I * TS; idx += kBlock) {
+          int row = idx / TS, col = idx % TS;
+          btile[idx] = (col < ts) ? Bb[(long)row * S + st + col] : 0.f;
+        }
+      }
+      __syncthreads();
+
+      if (o0 < O) {
+        if (fullt) {
+          float4 av[OW];
+#pragma unroll
+          for (int w = 0; w < OW; ++w) {
+            if (o0 + w < O) {
+              av[w] = *reinterpret_cast<const float4*>(Ab + (long)w * S + st + lane * 4);
+              if (want_bias) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
+            }
+          }
+#pragma unroll
+          for (int i = 0; i < ICAP; ++i) {
+            if (i < I) {
+              const float4 bv = *reinterpret_cast<const float4*>(&btile[i * TS + lane * 4]);
+#pragma unroll
+              for (int w = 0; w < OW; ++w) {
+                if (o0 + w < O)
+                  acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
+                               av[w].z * bv.z + av[w].w * bv.w;
+              }
+            }
+          }
+        } else {
+          int ts = (int)(s1 - st);
+          for (int c = lane; c < ts; c += 64) {
+#pragma unroll
+            for (int w = 0; w < OW; ++w) {
+              if (o0 + w < O) {
+                float av = Ab[(long)w * S + st + c];
+                if (want_bias) bacc[w] += av;
+#pragma unroll
+                for (int i = 0; i < ICAP; ++i)
+                  if (i < I) acc[w][i] += av * btile[i * TS + c];
+              }
+            }
+          }
+        }
+      }
+    }
+  }
+
+#pragma unroll
+  for (int w = 0; w < OW; ++w) {
+    if (o0 + w < O) {
+#pragma unroll
+      for (int i = 0; i < ICAP; ++i) {
+        if (i < I) {
+          float v = gw_wave_sum(acc[w][i]);
+          if (lane == 0 && v != 0.f)
+            atomicAdd(reinterpret_cast<float*>(&gW[(size_t)(o0 + w) * I + i]), v);
+        }
+      }
+      if (want_bias) {
+        float v = gw_wave_sum(bacc[w]);
+        if (lane == 0 && v != 0.f)
+          atomicAdd(reinterpret_cast<float*>(&gb[o0 + w]), v);
+      }
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
@@ -689,11 +805,21 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
       if (vec) { GW_LAUNCH(ICAP_, OW_, false, true) }                           \
       else { GW_LAUNCH(ICAP_, OW_, false, false) }                              \
     }
+#define GW_LDS(ICAP_)                                                           \
+      hipLaunchKernelGGL((gw_outer_lds_kernel<scalar_t, ICAP_, 4>), dim3(grid), \
+                         dim3(kBlock), 0, stream, gz.data_ptr<scalar_t>(),      \
+                         x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),       \
+                         want_bias ? gb.data_ptr<scalar_t>() : nullptr,         \
+                         B, O, I, S, n_schunk, want_bias);
   AT_DISPATCH_FLOATING_TYPES(gz.scalar_type(), "channel_mix_bwd_w", [&] {
     bool vec = std::is_same<scalar_t, float>::value && (S % 4 == 0) &&
                ((reinterpret_cast<uintptr_t>(gz.data_ptr()) & 15) == 0) &&
                ((reinterpret_cast<uintptr_t>(x.data_ptr()) & 15) == 0);
-    if (OW == 4) {
+    if (vec && OW == 4) {
+      if (I <= 8) { GW_LDS(8) }
+      else if (I <= 24) { GW_LDS(24) }
+      else { GW_LDS(32) }
+    } else if (OW == 4) {
       if (I <= 8) { GW_DISPATCH2(8, 4) }
       else if (I <= 24) { GW_DISPATCH2(24, 4) }
       else { GW_DISPATCH2(32, 4) }
@@ -701,6 +827,7 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
       GW_DISPATCH2(32, 1)
     }
   });
+#undef GW_LDS
 #undef GW_DISPATCH2
 #undef GW_LAUNCH
   return {gW, gb};
