@@ -661,7 +661,8 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
     const T* __restrict__ A, const T* __restrict__ Bm,
     T* __restrict__ gW, T* __restrict__ gb,
     int B, int O, int I, long S, int n_schunk, bool want_bias) {
-  constexpr int TS = 256;  // floats per s-tile
+  // s-tile sized so the 2 barriers/tile amortize over ~4 compute steps
+  constexpr int TS = (ICAP <= 8) ? 2048 : (ICAP <= 24 ? 1024 : 512);
   __shared__ float btile[ICAP * TS];
 
   const int o_tiles = (O + 4 * OW - 1) / (4 * OW);
@@ -708,23 +709,25 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
 
       if (o0 < O) {
         if (fullt) {
-          float4 av[OW];
+          for (int c0 = lane * 4; c0 < TS; c0 += 64 * 4) {
+            float4 av[OW];
 #pragma unroll
-          for (int w = 0; w < OW; ++w) {
-            if (o0 + w < O) {
-              av[w] = *reinterpret_cast<const float4*>(Ab + (long)w * S + st + lane * 4);
-              if (want_bias) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
+            for (int w = 0; w < OW; ++w) {
+              if (o0 + w < O) {
+                av[w] = *reinterpret_cast<const float4*>(Ab + (long)w * S + st + c0);
+                if (want_bias) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
+              }
             }
-          }
 #pragma unroll
-          for (int i = 0; i < ICAP; ++i) {
-            if (i < I) {
-              const float4 bv = *reinterpret_cast<const float4*>(&btile[i * TS + lane * 4]);
+            for (int i = 0; i < ICAP; ++i) {
+              if (i < I) {
+                const float4 bv = *reinterpret_cast<const float4*>(&btile[i * TS + c0]);
 #pragma unroll
-              for (int w = 0; w < OW; ++w) {
-                if (o0 + w < O)
-                  acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
-                               av[w].z * bv.z + av[w].w * bv.w;
+                for (int w = 0; w < OW; ++w) {
+                  if (o0 + w < O)
+                    acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
+                                 av[w].z * bv.z + av[w].w * bv.w;
+                }
               }
             }
           }
