@@ -653,17 +653,21 @@ __global__ __launch_bounds__(kBlock) void gw_outer_kernel(
 
 
 // LDS-staged variant: the B tile (the I<=32 narrow operand) is staged into
-// LDS once per 256-element s-tile and shared by all 4 waves x OW rows of the
-// block — without this, co-resident blocks evict the shared rows from L1/L2
-// and the measured traffic is ~4x algorithmic (0.76 TB/s vs 5.3 peak).
+// a double-buffered LDS ring by global_load_lds DMA (no VGPR round trip, no
+// spill) and shared by the block's waves/OW rows; the next tile's DMA is
+// issued before computing the current one, so HBM latency hides under the
+// fma phase.  __syncthreads() drains the outstanding DMA (hipcc emits
+// vmcnt(0) in the barrier when glds is in flight) and doubles as the ring
+// hand-off.  Without the staging, co-resident blocks evict the shared rows
+// from L1/L2 (~4x algorithmic traffic); without the pipelining the
+// load->write pairs serialize (SQ_WAIT_ANY 76%).
 template <typename T, int ICAP, int OW>
 __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
     const T* __restrict__ A, const T* __restrict__ Bm,
     T* __restrict__ gW, T* __restrict__ gb,
     int B, int O, int I, long S, int n_schunk, bool want_bias) {
-  // s-tile sized so the 2 barriers/tile amortize over ~4 compute steps
-  constexpr int TS = (ICAP <= 8) ? 2048 : (ICAP <= 24 ? 1024 : 512);
-  __shared__ float btile[ICAP * TS];
+  constexpr int TS = 256;  // floats per s-tile
+  __shared__ float btile[2][ICAP * TS];
 
   const int o_tiles = (O + 4 * OW - 1) / (4 * OW);
   const int schunk = blockIdx.x / o_tiles;
@@ -675,6 +679,7 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
   long chunk_sz = ((S + n_schunk - 1) / n_schunk + (TS - 1)) & ~((long)TS - 1);
   long s0 = (long)schunk * chunk_sz;
   long s1 = min(S, s0 + chunk_sz);
+  long full_end = (s0 < s1) ? s0 + ((s1 - s0) / TS) * TS : s0;
 
   float acc[OW][ICAP];
 #pragma unroll
@@ -685,65 +690,70 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
 #pragma unroll
   for (int w = 0; w < OW; ++w) bacc[w] = 0.f;
 
+  const int tot_slots = I * (TS / 4);   // float4 slots per tile
+
   for (int b = 0; b < B; ++b) {
     const float* Ab = reinterpret_cast<const float*>(A) + ((long)b * O + o0) * S;
     const float* Bb = reinterpret_cast<const float*>(Bm) + ((long)b * I) * S;
-    for (long st = s0; st < s1; st += TS) {
-      const bool fullt = (st + TS) <= s1;
-      __syncthreads();
-      // cooperative stage of B[0:I][st:st+TS] into LDS
-      if (fullt) {
-        for (int idx = threadIdx.x * 4; idx < I * TS; idx += kBlock * 4) {
-          int row = idx / TS, col = idx % TS;
-          *reinterpret_cast<float4*>(&btile[row * TS + col]) =
-              *reinterpret_cast<const float4*>(Bb + (long)row * S + st + col);
-        }
-      } else {
-        int ts = (int)(s1 - st);
-        for (int idx = threadIdx.x; idx < I * TS; idx += kBlock) {
-          int row = idx / TS, col = idx % TS;
-          btile[idx] = (col < ts) ? Bb[(long)row * S + st + col] : 0.f;
-        }
-      }
-      __syncthreads();
 
-      if (o0 < O) {
-        if (fullt) {
-          for (int c0 = lane * 4; c0 < TS; c0 += 64 * 4) {
-            float4 av[OW];
+    // one tile's DMA: wave-cooperative, lane-linear LDS image
+    auto issue_tile = [&](int buf, long st) {
+      for (int sb = wave * 64; sb < tot_slots; sb += kBlock) {
+        int slot = sb + lane;
+        int fo = slot * 4;
+        int row = fo / TS, col = fo % TS;
+        const float* gsrc = Bb + (long)row * S + st + col;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)gsrc,
+            (__attribute__((address_space(3))) void*)&btile[buf][sb * 4],
+            16, 0, 0);
+      }
+    };
+
+    if (s0 < full_end) {
+      issue_tile(0, s0);
+      __syncthreads();   // drain tile-0 DMA
+      int cur = 0;
+      for (long st = s0; st < full_end; st += TS, cur ^= 1) {
+        if (st + TS < full_end) issue_tile(cur ^ 1, st + TS);
+        if (o0 < O) {
+          float4 av[OW];
 #pragma unroll
-            for (int w = 0; w < OW; ++w) {
-              if (o0 + w < O) {
-                av[w] = *reinterpret_cast<const float4*>(Ab + (long)w * S + st + c0);
-                if (want_bias) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
-              }
+          for (int w = 0; w < OW; ++w) {
+            if (o0 + w < O) {
+              av[w] = *reinterpret_cast<const float4*>(Ab + (long)w * S + st + lane * 4);
+              if (want_bias) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
             }
+          }
 #pragma unroll
-            for (int i = 0; i < ICAP; ++i) {
-              if (i < I) {
-                const float4 bv = *reinterpret_cast<const float4*>(&btile[i * TS + c0]);
+          for (int i = 0; i < ICAP; ++i) {
+            if (i < I) {
+              const float4 bv =
+                  *reinterpret_cast<const float4*>(&btile[cur][i * TS + lane * 4]);
 #pragma unroll
-                for (int w = 0; w < OW; ++w) {
-                  if (o0 + w < O)
-                    acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
-                                 av[w].z * bv.z + av[w].w * bv.w;
-                }
+              for (int w = 0; w < OW; ++w) {
+                if (o0 + w < O)
+                  acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
+                               av[w].z * bv.z + av[w].w * bv.w;
               }
             }
           }
-        } else {
-          int ts = (int)(s1 - st);
-          for (int c = lane; c < ts; c += 64) {
+        }
+        __syncthreads();  // drain next-tile DMA + ring hand-off
+      }
+    }
+
+    // ragged tail of the last (partial) tile, direct from global
+    if (o0 < O) {
+      for (long s = full_end + lane; s < s1; s += 64) {
 #pragma unroll
-            for (int w = 0; w < OW; ++w) {
-              if (o0 + w < O) {
-                float av = Ab[(long)w * S + st + c];
-                if (want_bias) bacc[w] += av;
+        for (int w = 0; w < OW; ++w) {
+          if (o0 + w < O) {
+            float av = Ab[(long)w * S + s];
+            if (want_bias) bacc[w] += av;
 #pragma unroll
-                for (int i = 0; i < ICAP; ++i)
-                  if (i < I) acc[w][i] += av * btile[i * TS + c];
-              }
-            }
+            for (int i = 0; i < ICAP; ++i)
+              if (i < I) acc[w][i] += av * Bb[(long)i * S + s];
           }
         }
       }
@@ -787,8 +797,12 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
   if (gz.numel() == 0) return {gW, gb};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  // OW=4 when O is large enough; ICAP sized to I
-  const int OW = (O >= 8) ? 4 : 1;
+  bool is_f32 = gz.scalar_type() == at::kFloat;
+  bool vec = is_f32 && (S % 4 == 0) &&
+             ((reinterpret_cast<uintptr_t>(gz.data_ptr()) & 15) == 0) &&
+             ((reinterpret_cast<uintptr_t>(x.data_ptr()) & 15) == 0);
+  // rows per wave: wider cuts B re-reads but costs registers/occupancy
+  const int OW = (O >= 8) ? ((I <= 8) ? 4 : 2) : 1;
   int o_tiles = (O + 4 * OW - 1) / (4 * OW);
   int n_schunk = (int)std::max(1L, std::min((long)(4096 / o_tiles), S / (64 * 16)));
   int grid = n_schunk * o_tiles;
@@ -808,27 +822,21 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
       if (vec) { GW_LAUNCH(ICAP_, OW_, false, true) }                           \
       else { GW_LAUNCH(ICAP_, OW_, false, false) }                              \
     }
-#define GW_LDS(ICAP_)                                                           \
-      hipLaunchKernelGGL((gw_outer_lds_kernel<scalar_t, ICAP_, 4>), dim3(grid), \
-                         dim3(kBlock), 0, stream, gz.data_ptr<scalar_t>(),      \
+#define GW_LDS(ICAP_, OW_)                                                      \
+      hipLaunchKernelGGL((gw_outer_lds_kernel<scalar_t, ICAP_, OW_>),           \
+                         dim3(grid), dim3(kBlock), 0, stream,                   \
+                         gz.data_ptr<scalar_t>(),                               \
                          x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),       \
                          want_bias ? gb.data_ptr<scalar_t>() : nullptr,         \
                          B, O, I, S, n_schunk, want_bias);
   AT_DISPATCH_FLOATING_TYPES(gz.scalar_type(), "channel_mix_bwd_w", [&] {
-    bool vec = std::is_same<scalar_t, float>::value && (S % 4 == 0) &&
-               ((reinterpret_cast<uintptr_t>(gz.data_ptr()) & 15) == 0) &&
-               ((reinterpret_cast<uintptr_t>(x.data_ptr()) & 15) == 0);
-    if (vec && OW == 4) {
-      if (I <= 8) { GW_LDS(8) }
-      else if (I <= 24) { GW_LDS(24) }
-      else { GW_LDS(32) }
-    } else if (OW == 4) {
-      if (I <= 8) { GW_DISPATCH2(8, 4) }
-      else if (I <= 24) { GW_DISPATCH2(24, 4) }
-      else { GW_DISPATCH2(32, 4) }
-    } else {
-      GW_DISPATCH2(32, 1)
-    }
+    if (vec && OW == 4) { GW_LDS(8, 4) }
+    else if (vec && OW == 2) {
+      if (I <= 24) { GW_LDS(24, 2) } else { GW_LDS(32, 2) }
+    } else if (OW == 4) { GW_DISPATCH2(8, 4) }
+    else if (OW == 2) {
+      if (I <= 24) { GW_DISPATCH2(24, 2) } else { GW_DISPATCH2(32, 2) }
+    } else { GW_DISPATCH2(32, 1) }
   });
 #undef GW_LDS
 #undef GW_DISPATCH2
