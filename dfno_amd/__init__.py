@@ -39,6 +39,7 @@ from .nn import (
 )
 from .utils import (
     alphabet,
+    generate_batch_indices,
     get_env,
     get_gpu_memory,
     profile_gpu_memory,

@@ -15,10 +15,12 @@ import time
 from contextlib import nullcontext
 from typing import List
 
+import numpy as np
 import torch
 
 __all__ = [
     "alphabet",
+    "generate_batch_indices",
     "get_env",
     "get_gpu_memory",
     "profile_gpu_memory",
@@ -93,6 +95,24 @@ def profile_gpu_memory(outfile, dt: float = 1.0):
             f.write(f"{time.time() - t0}, " + ", ".join(str(m) for m in muv) + "\n")
             f.flush()
             time.sleep(dt)
+
+
+def generate_batch_indices(P, n: int, batch_size: int, shuffle: bool = False,
+                           seed: int = 0):
+    """Yield (start, stop) index ranges covering ``n`` samples.
+
+    Restores the helper the reference calls but never defines
+    (experiment_navier_stokes.py:130,157 — SURVEY.md 2.5).  The range order
+    is derived from a fixed seed so every rank of ``P`` iterates the same
+    batches (ranks are seeded differently for weights, so relying on the
+    global torch RNG would desynchronize the collectives).
+    """
+    bounds = [(a, min(a + batch_size, n)) for a in range(0, n, batch_size)]
+    if shuffle:
+        rng = np.random.RandomState(seed)
+        order = rng.permutation(len(bounds))
+        bounds = [bounds[i] for i in order]
+    return bounds
 
 
 def unit_gaussian_normalize(x):
