@@ -23,6 +23,7 @@ def main():
     p.add_argument("--num-blocks", type=int, default=4)
     p.add_argument("--grid", type=int, default=64)
     p.add_argument("--out", type=str, default=None)
+    p.add_argument("--shapes", action="store_true")
     args = p.parse_args()
 
     device = torch.device("cuda" if torch.cuda.is_available() else "cpu")
@@ -52,12 +53,17 @@ def main():
 
     from torch.profiler import profile, ProfilerActivity
 
-    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA]) as prof:
+    with profile(activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                 record_shapes=args.shapes) as prof:
         for _ in range(args.steps):
             step()
         torch.cuda.synchronize()
 
-    table = prof.key_averages().table(sort_by="self_cuda_time_total", row_limit=40)
+    if args.shapes:
+        table = prof.key_averages(group_by_input_shape=True).table(
+            sort_by="self_cuda_time_total", row_limit=60)
+    else:
+        table = prof.key_averages().table(sort_by="self_cuda_time_total", row_limit=40)
     print(table)
     if args.out:
         with open(args.out, "w") as f:
