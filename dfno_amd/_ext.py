@@ -28,6 +28,7 @@ SOURCES = [
     _CSRC / "pointwise.hip",
     _CSRC / "spectral.hip",
     _CSRC / "proj_head.hip",
+    _CSRC / "dft.hip",
 ]
 
 

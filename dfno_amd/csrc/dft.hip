@@ -1,0 +1,395 @@
+// gfx950 fused truncated-spectrum DFT kernels (see ops/fft.py for the
+// semantics and the roofline argument).
+//
+// For FNO mode counts (m ~ 8..32 kept of N <= 64) a truncated naive DFT is
+// N*m complex MACs per line — FLOP-comparable to a full FFT — and lets one
+// kernel fuse transform + truncation/zero-padding + 1/n scale while reading
+// the tensor once along its NATIVE strides (no transposes, no hipFFT
+// staging copies, no separate cat/zeros passes, no autograd mirror copies).
+//
+// Geometry: tensor viewed as [outer, L, inner] around the transform dim.
+//  * inner > 1 (the spatial dims): thread <-> (outer, inner) line; loads of
+//    x[o, j, :] are lane-consecutive in `inner` -> fully coalesced.
+//  * inner == 1 (the trailing rfft/irfft time dim): each thread streams its
+//    own contiguous line; a wave's combined footprint is a contiguous span,
+//    so L1/L2 serve the per-thread scalar loads after the first touch, and
+//    per-line outputs are >= 64B contiguous runs.
+//
+// Twiddles: one table of w^r = exp(-2*pi*i*r/N), r in [0, N), built in LDS
+// at block start (sign applied on read); (j*k) mod N tracked incrementally
+// per kept mode in registers (unrolled + predicated, no scratch).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include "kernels.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kMaxN = 64;
+
+template <typename T>
+__device__ __forceinline__ void sincos_t(T a, T* s, T* c);
+template <>
+__device__ __forceinline__ void sincos_t<float>(float a, float* s, float* c) {
+  sincosf(a, s, c);
+}
+template <>
+__device__ __forceinline__ void sincos_t<double>(double a, double* s, double* c) {
+  *s = sin(a); *c = cos(a);
+}
+
+// fill the LDS twiddle table: w[r] = exp(-2 pi i r / N)
+template <typename T>
+__device__ __forceinline__ void fill_twiddle(T* wr, T* wi, int N) {
+  for (int r = threadIdx.x; r < N; r += blockDim.x) {
+    T s, c;
+    sincos_t<T>(T(-2.0) * T(M_PI) * T(r) / T(N), &s, &c);
+    wr[r] = c;
+    wi[r] = s;
+  }
+  __syncthreads();
+}
+
+// kept-mode k value for index ki
+__device__ __forceinline__ int kept_k(int ki, int m_lo, int N, int m_hi) {
+  return (ki < m_lo) ? ki : (N - m_hi + (ki - m_lo));
+}
+
+// ---------------------------------------------------------------------------
+// C2C analysis: out[o, ki, i] = scale * sum_j in[o, j, i] * w^{-j k(ki)}
+// (interleaved complex; SIGN=-1 fixed: analysis always uses the table sign)
+// ---------------------------------------------------------------------------
+
+template <typename T, int MCAP>
+__global__ __launch_bounds__(kBlock) void dft_c2c_analysis_kernel(
+    const T* __restrict__ in, T* __restrict__ out,
+    long outer, int N, long inner, int m_lo, int m_hi, T scale) {
+  __shared__ T wr[kMaxN], wi[kMaxN];
+  fill_twiddle(wr, wi, N);
+  const int m = m_lo + m_hi;
+
+  long total = outer * inner;
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = t0; t < total; t += stride) {
+    long o = t / inner;
+    long i = t % inner;
+    const T* src = in + 2 * (o * N * inner + i);
+
+    T ar[MCAP], ai[MCAP];
+    int rk[MCAP], kk[MCAP];
+#pragma unroll
+    for (int ki = 0; ki < MCAP; ++ki) {
+      if (ki < m) {
+        ar[ki] = T(0); ai[ki] = T(0);
+        rk[ki] = 0;
+        kk[ki] = kept_k(ki, m_lo, N, m_hi);
+      }
+    }
+    for (int j = 0; j < N; ++j) {
+      const T xr = src[2 * j * inner];
+      const T xi = src[2 * j * inner + 1];
+#pragma unroll
+      for (int ki = 0; ki < MCAP; ++ki) {
+        if (ki < m) {
+          const T cr = wr[rk[ki]], ci = wi[rk[ki]];
+          ar[ki] += xr * cr - xi * ci;
+          ai[ki] += xr * ci + xi * cr;
+          rk[ki] += kk[ki];
+          if (rk[ki] >= N) rk[ki] -= N;
+        }
+      }
+    }
+    T* dst = out + 2 * (o * m * inner + i);
+#pragma unroll
+    for (int ki = 0; ki < MCAP; ++ki) {
+      if (ki < m) {
+        dst[2 * ki * inner] = scale * ar[ki];
+        dst[2 * ki * inner + 1] = scale * ai[ki];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// C2C synthesis: out[o, j, i] = scale * sum_ki in[o, ki, i] * w^{+j k(ki)}
+// (the kept modes are the inputs; output is the full length N)
+// ---------------------------------------------------------------------------
+
+template <typename T, int MCAP>
+__global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
+    const T* __restrict__ in, T* __restrict__ out,
+    long outer, int N, long inner, int m_lo, int m_hi, T scale) {
+  __shared__ T wr[kMaxN], wi[kMaxN];
+  fill_twiddle(wr, wi, N);
+  const int m = m_lo + m_hi;
+
+  long total = outer * inner;
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = t0; t < total; t += stride) {
+    long o = t / inner;
+    long i = t % inner;
+    const T* src = in + 2 * (o * m * inner + i);
+
+    // stage the m kept inputs in registers (scaled once)
+    T yr[MCAP], yi[MCAP];
+    int rk[MCAP], kk[MCAP];
+#pragma unroll
+    for (int ki = 0; ki < MCAP; ++ki) {
+      if (ki < m) {
+        yr[ki] = scale * src[2 * ki * inner];
+        yi[ki] = scale * src[2 * ki * inner + 1];
+        rk[ki] = 0;
+        kk[ki] = kept_k(ki, m_lo, N, m_hi);
+      }
+    }
+    T* dst = out + 2 * (o * N * inner + i);
+    for (int j = 0; j < N; ++j) {
+      T sr = T(0), si = T(0);
+#pragma unroll
+      for (int ki = 0; ki < MCAP; ++ki) {
+        if (ki < m) {
+          // +jk: conjugate of the table entry
+          const T cr = wr[rk[ki]], ci = -wi[rk[ki]];
+          sr += yr[ki] * cr - yi[ki] * ci;
+          si += yr[ki] * ci + yi[ki] * cr;
+          rk[ki] += kk[ki];
+          if (rk[ki] >= N) rk[ki] -= N;
+        }
+      }
+      dst[2 * j * inner] = sr;
+      dst[2 * j * inner + 1] = si;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// R2C (last dim): out[l, k] = fac_k * scale * sum_j in[l, j] * w^{-jk}
+// fac_k = 1 (plain rfft_trunc) or the irfft-adjoint factors {1, 2, ..,
+// 2, (1 at Nyquist)} with the DC/Nyquist imaginary parts zeroed.
+// ---------------------------------------------------------------------------
+
+template <typename T, int MCAP>
+__global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
+    const T* __restrict__ in, T* __restrict__ out,
+    long lines, int N, int m, T scale, bool factors) {
+  __shared__ T wr[kMaxN], wi[kMaxN];
+  fill_twiddle(wr, wi, N);
+
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long l = t0; l < lines; l += stride) {
+    const T* src = in + l * N;
+    T ar[MCAP], ai[MCAP];
+    int rk[MCAP];
+#pragma unroll
+    for (int k = 0; k < MCAP; ++k) {
+      if (k < m) { ar[k] = T(0); ai[k] = T(0); rk[k] = 0; }
+    }
+    for (int j = 0; j < N; ++j) {
+      const T x = src[j];
+#pragma unroll
+      for (int k = 0; k < MCAP; ++k) {
+        if (k < m) {
+          ar[k] += x * wr[rk[k]];
+          ai[k] += x * wi[rk[k]];
+          rk[k] += k;
+          if (rk[k] >= N) rk[k] -= N;
+        }
+      }
+    }
+    T* dst = out + 2 * l * m;
+#pragma unroll
+    for (int k = 0; k < MCAP; ++k) {
+      if (k < m) {
+        T f = T(1);
+        bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
+        if (factors && !edge) f = T(2);
+        dst[2 * k] = f * scale * ar[k];
+        dst[2 * k + 1] = (factors && edge) ? T(0) : f * scale * ai[k];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// C2R (last dim): out[l, j] = scale * Re( sum_k fac_k * in[l, k] * w^{+jk} )
+// fac_k = {1, 2, .., 2, (1 at Nyquist)} for pad_irfft (scale = 1/N), or all
+// ones for the rfft_trunc adjoint (scale = 1).
+// ---------------------------------------------------------------------------
+
+template <typename T, int MCAP>
+__global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
+    const T* __restrict__ in, T* __restrict__ out,
+    long lines, int N, int m, T scale, bool factors) {
+  __shared__ T wr[kMaxN], wi[kMaxN];
+  fill_twiddle(wr, wi, N);
+
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long l = t0; l < lines; l += stride) {
+    const T* src = in + 2 * l * m;
+    T yr[MCAP], yi[MCAP];
+    int rk[MCAP];
+#pragma unroll
+    for (int k = 0; k < MCAP; ++k) {
+      if (k < m) {
+        T f = scale;
+        if (factors && !(k == 0 || (N % 2 == 0 && 2 * k == N))) f = T(2) * scale;
+        yr[k] = f * src[2 * k];
+        yi[k] = f * src[2 * k + 1];
+        if (factors && (k == 0 || (N % 2 == 0 && 2 * k == N))) yi[k] = T(0);
+        rk[k] = 0;
+      }
+    }
+    T* dst = out + l * N;
+    for (int j = 0; j < N; ++j) {
+      T s = T(0);
+#pragma unroll
+      for (int k = 0; k < MCAP; ++k) {
+        if (k < m) {
+          // Re(y * conj-sign(+jk)): w^{+jk} = (wr, -wi)
+          s += yr[k] * wr[rk[k]] + yi[k] * wi[rk[k]];
+          rk[k] += k;
+          if (rk[k] >= N) rk[k] -= N;
+        }
+      }
+      dst[j] = s;
+    }
+  }
+}
+
+int grid_for_d(long work) {
+  long g = (work + kBlock - 1) / kBlock;
+  long cap = 256L * 16;
+  if (g > cap) g = cap;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+void split_dims(const at::Tensor& x, int dim, long& outer, long& inner) {
+  outer = 1;
+  for (int d = 0; d < dim; ++d) outer *= x.size(d);
+  inner = 1;
+  for (int d = dim + 1; d < x.dim(); ++d) inner *= x.size(d);
+}
+
+#define DFT_MDISPATCH(KERNEL, ...)                                             \
+  if (m <= 8) { hipLaunchKernelGGL((KERNEL<scalar_t, 8>), __VA_ARGS__); }      \
+  else if (m <= 16) { hipLaunchKernelGGL((KERNEL<scalar_t, 16>), __VA_ARGS__); } \
+  else { TORCH_CHECK(m <= 32, "dft: m > 32 unsupported natively");             \
+         hipLaunchKernelGGL((KERNEL<scalar_t, 32>), __VA_ARGS__); }
+
+}  // namespace
+
+at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
+                   int64_t m_lo, int64_t m_hi, bool analysis, double scale) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "dft_c2c: contiguous GPU input");
+  TORCH_CHECK(x.scalar_type() == at::kComplexFloat || x.scalar_type() == at::kComplexDouble,
+              "dft_c2c: complex input");
+  TORCH_CHECK(n <= kMaxN, "dft_c2c: N too large");
+  const int m = (int)(m_lo + m_hi);
+  TORCH_CHECK(m <= 64 && m <= n, "dft_c2c: bad mode count");
+  TORCH_CHECK(x.size(dim) == (analysis ? n : m), "dft_c2c: dim extent mismatch");
+
+  long outer, inner;
+  split_dims(x, (int)dim, outer, inner);
+  auto sizes = x.sizes().vec();
+  sizes[dim] = analysis ? m : n;
+  auto out = at::empty(sizes, x.options());
+  if (x.numel() == 0) return out;
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int grid = grid_for_d(outer * inner);
+
+  AT_DISPATCH_FLOATING_TYPES(c10::toRealValueType(x.scalar_type()), "dft_c2c", [&] {
+    auto inp = reinterpret_cast<const scalar_t*>(x.data_ptr());
+    auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
+    if (analysis) {
+      DFT_MDISPATCH(dft_c2c_analysis_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                    inp, op, outer, (int)n, inner, (int)m_lo, (int)m_hi,
+                    (scalar_t)scale)
+    } else {
+      DFT_MDISPATCH(dft_c2c_synthesis_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                    inp, op, outer, (int)n, inner, (int)m_lo, (int)m_hi,
+                    (scalar_t)scale)
+    }
+  });
+  return out;
+}
+
+static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
+                               double scale, bool factors) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "dft_r2c: contiguous GPU input");
+  TORCH_CHECK(dim == x.dim() - 1, "dft_r2c: last-dim only");
+  const int N = (int)x.size(dim);
+  TORCH_CHECK(N <= kMaxN && m <= N, "dft_r2c: bad sizes");
+  long lines = x.numel() / std::max(N, 1);
+
+  auto sizes = x.sizes().vec();
+  sizes[dim] = m;
+  auto out = at::empty(sizes, x.options().dtype(
+      x.scalar_type() == at::kFloat ? at::kComplexFloat : at::kComplexDouble));
+  if (x.numel() == 0) return out;
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int grid = grid_for_d(lines);
+  AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "dft_r2c", [&] {
+    auto inp = x.data_ptr<scalar_t>();
+    auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
+    DFT_MDISPATCH(dft_r2c_last_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                  inp, op, lines, N, (int)m, (scalar_t)scale, factors)
+  });
+  return out;
+}
+
+static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
+                               double scale, bool factors) {
+  TORCH_CHECK(y.is_cuda() && y.is_contiguous(), "dft_c2r: contiguous GPU input");
+  TORCH_CHECK(dim == y.dim() - 1, "dft_c2r: last-dim only");
+  TORCH_CHECK(y.scalar_type() == at::kComplexFloat || y.scalar_type() == at::kComplexDouble,
+              "dft_c2r: complex input");
+  const int m = (int)y.size(dim);
+  const int N = (int)n_out;
+  TORCH_CHECK(N <= kMaxN && m <= N, "dft_c2r: bad sizes");
+  long lines = y.numel() / std::max(m, 1);
+
+  auto sizes = y.sizes().vec();
+  sizes[dim] = N;
+  auto out = at::empty(sizes, y.options().dtype(
+      y.scalar_type() == at::kComplexFloat ? at::kFloat : at::kDouble));
+  if (y.numel() == 0) return out;
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int grid = grid_for_d(lines);
+  AT_DISPATCH_FLOATING_TYPES(out.scalar_type(), "dft_c2r", [&] {
+    auto inp = reinterpret_cast<const scalar_t*>(y.data_ptr());
+    auto op = out.data_ptr<scalar_t>();
+    DFT_MDISPATCH(dft_c2r_last_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                  inp, op, lines, N, (int)m, (scalar_t)scale, factors)
+  });
+  return out;
+}
+
+at::Tensor dft_rfft_trunc(const at::Tensor& x, int64_t dim, int64_t m) {
+  return dft_r2c_impl(x, dim, m, 1.0, /*factors=*/false);
+}
+
+at::Tensor dft_rfft_trunc_adj(const at::Tensor& gy, int64_t dim, int64_t n) {
+  // gx_j = Re(sum_k gY_k w^{+jk}) — c2r with unit factors / unit scale
+  return dft_c2r_impl(gy, dim, n, 1.0, /*factors=*/false);
+}
+
+at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m) {
+  TORCH_CHECK(y.size(dim) == m, "dft_pad_irfft: m mismatch");
+  return dft_c2r_impl(y, dim, n_out, 1.0 / (double)n_out, /*factors=*/true);
+}
+
+at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m) {
+  const int N = (int)gx.size(dim);
+  return dft_r2c_impl(gx, dim, m, 1.0 / (double)N, /*factors=*/true);
+}

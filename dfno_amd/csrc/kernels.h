@@ -32,6 +32,14 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
 std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
                                           bool want_bias);
 
+// fused truncated-spectrum DFTs (see ops/fft.py):
+at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
+                   int64_t m_lo, int64_t m_hi, bool analysis, double scale);
+at::Tensor dft_rfft_trunc(const at::Tensor& x, int64_t dim, int64_t m);
+at::Tensor dft_rfft_trunc_adj(const at::Tensor& gy, int64_t dim, int64_t n);
+at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m);
+at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m);
+
 // corner-block spectral contraction on the truncated complex spectrum:
 //   y[b,o,f] += sum_i x[b,i,f] * w[i,o,f_box]  for f in the corner box
 void spectral_corner_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,

@@ -30,7 +30,7 @@ import torch.nn as nn
 
 from ..comm import Repartition
 from ..partition import Partition, compute_distribution_info
-from ..ops import spectral_conv, add_gelu
+from ..ops import spectral_conv, add_gelu, rfft_trunc, fft_trunc, pad_ifft, pad_irfft
 from .linear import BroadcastedLinear
 
 __all__ = ["DistributedFNOBlock"]
@@ -190,60 +190,11 @@ class DistributedFNOBlock(nn.Module):
             pieces.append(y[tuple(sl)])
         return torch.cat(pieces, dim=dim)
 
-    # ---- layout-rolled FFT chains ----------------------------------------
-    # Every 1-D FFT runs with its transform dim rolled to the LAST position
-    # (a movedim view): torch/hipFFT then takes the contiguous fast path (one
-    # internal gather instead of a copy-in + copy-out per middle-dim
-    # transform), and restrict/zeropad act on the last dim.  The permutation
-    # is undone with a single cheap pass once the spectrum is truncated
-    # (17 MB vs the 500+ MB the old per-dim copies touched).  Bookkeeping:
-    # ``order[i]`` = which canonical dim sits at position i.
-
-    @staticmethod
-    def _roll_last(x: torch.Tensor, order: List[int], dim: int):
-        pos = order.index(dim)
-        x = x.movedim(pos, -1)
-        order = order[:pos] + order[pos + 1:] + [dim]
-        return x, order
-
-    @staticmethod
-    def _unroll(x: torch.Tensor, order: List[int]) -> torch.Tensor:
-        if order == sorted(order):
-            return x.contiguous()
-        inv = [order.index(d) for d in sorted(order)]
-        return x.permute(inv).contiguous()
-
-    def _restrict_last(self, x: torch.Tensor, dim: int) -> torch.Tensor:
-        pieces = []
-        if dim in self.restrict_prefixes:
-            pieces.append(x[..., : self.restrict_prefixes[dim]])
-        if dim in self.restrict_suffixes:
-            pieces.append(x[..., -self.restrict_suffixes[dim]:])
-        if len(pieces) == 1:
-            return pieces[0]
-        return torch.cat(pieces, dim=-1)
-
-    def _zeropad_last(self, y: torch.Tensor, dim: int, target_last: int,
-                      scale: float = 1.0) -> torch.Tensor:
-        """Inverse of _restrict_last; optionally folds the ifft 1/n scale
-        (applied pre-transform: DFT linearity) into the same pass."""
-        pad = target_last - y.shape[-1]
-        if scale != 1.0:
-            y = y * scale
-        if pad < 1:
-            return y
-        lo = self.restrict_prefixes.get(dim, 0)
-        hi = self.restrict_suffixes.get(dim, 0)
-        pieces = []
-        if lo:
-            pieces.append(y[..., :lo])
-        pieces.append(torch.zeros((*y.shape[:-1], pad), dtype=y.dtype,
-                                  layout=y.layout, device=y.device))
-        if hi:
-            pieces.append(y[..., -hi:])
-        return torch.cat(pieces, dim=-1)
-
-    # ---- forward (reference dfno.py:241-291, layout-rolled) ---------------
+    # ---- forward (reference dfno.py:241-291) ------------------------------
+    # Each transform runs through the fused truncated-DFT ops (ops/fft.py):
+    # transform + mode truncation / zero-padding + inverse scale in a single
+    # strided-native kernel per dim (hand-written gfx950 DFT on GPU, a
+    # torch.fft composition on CPU).
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         self.dt_comm = 0.0
 
@@ -253,20 +204,16 @@ class DistributedFNOBlock(nn.Module):
         x = self.R1(x)
         self.dt_comm += time.time() - t0
 
-        nd = x.dim() if x.numel() else len(self.in_shape)
-        saved_last: Dict[int, int] = {}   # pre-restrict extent of each dim
+        saved_last: Dict[int, int] = {}   # pre-truncation extent per dim
         outermost = self.dim_m[-1]
         if x.numel() > 0:
-            order = list(range(nd))
-            x = torch.fft.rfft(x, dim=-1)           # outermost == last dim
-            saved_last[outermost] = x.shape[-1]
-            x = self._restrict_last(x, outermost)
+            n_t = x.shape[outermost]
+            saved_last[outermost] = n_t // 2 + 1
+            x = rfft_trunc(x, outermost, self.restrict_prefixes[outermost])
             for dim in reversed(self.dim_m[:-1]):
-                x, order = self._roll_last(x, order, dim)
-                x = torch.fft.fft(x, dim=-1)
-                saved_last[dim] = x.shape[-1]
-                x = self._restrict_last(x, dim)
-            x = self._unroll(x, order)
+                saved_last[dim] = x.shape[dim]
+                x = fft_trunc(x, dim, self.restrict_prefixes[dim],
+                              self.restrict_suffixes.get(dim, 0))
         else:
             x = torch.empty(0, dtype=self.dtype_complex, device=x.device)
 
@@ -275,25 +222,17 @@ class DistributedFNOBlock(nn.Module):
         self.dt_comm += time.time() - t0
 
         if x.numel() > 0:
-            order = list(range(nd))
             for dim in reversed(self.dim_y):
-                x, order = self._roll_last(x, order, dim)
-                x = torch.fft.fft(x, dim=-1)
-                saved_last[dim] = x.shape[-1]
-                x = self._restrict_last(x, dim)
-            x = self._unroll(x, order)
+                saved_last[dim] = x.shape[dim]
+                x = fft_trunc(x, dim, self.restrict_prefixes[dim],
+                              self.restrict_suffixes.get(dim, 0))
 
             y = spectral_conv(x, list(self.weights), self.corner_bounds, self.width)
 
-            order = list(range(nd))
             for dim in self.dim_y:
-                y, order = self._roll_last(y, order, dim)
-                n = saved_last[dim]
-                y = self._zeropad_last(y, dim, n, scale=1.0 / n)
-                # 1/n prescale folded into the zeropad pass; norm="forward"
-                # makes torch.fft.ifft apply no scale of its own
-                y = torch.fft.ifft(y, dim=-1, norm="forward")
-            y = self._unroll(y, order)
+                y = pad_ifft(y, dim, saved_last[dim],
+                             self.restrict_prefixes[dim],
+                             self.restrict_suffixes.get(dim, 0))
         else:
             y = x
 
@@ -302,17 +241,12 @@ class DistributedFNOBlock(nn.Module):
         self.dt_comm += time.time() - t0
 
         if y.numel() > 0:
-            order = list(range(nd))
             for dim in self.dim_m[:-1]:
-                y, order = self._roll_last(y, order, dim)
-                n = saved_last[dim]
-                y = self._zeropad_last(y, dim, n, scale=1.0 / n)
-                y = torch.fft.ifft(y, dim=-1, norm="forward")
-            # trailing rfft dim: zeropad then irfft on the last position
-            y, order = self._roll_last(y, order, outermost)
-            y = self._zeropad_last(y, outermost, saved_last[outermost])
-            y = torch.fft.irfft(y, n=self.in_shape[-1], dim=-1)
-            y = self._unroll(y, order)
+                y = pad_ifft(y, dim, saved_last[dim],
+                             self.restrict_prefixes[dim],
+                             self.restrict_suffixes.get(dim, 0))
+            y = pad_irfft(y, outermost, saved_last[outermost],
+                          self.in_shape[-1], self.restrict_prefixes[outermost])
         else:
             y = torch.empty(0, dtype=self.dtype, device=y.device)
 

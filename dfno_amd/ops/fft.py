@@ -1,0 +1,176 @@
+"""Fused truncated-spectrum transforms for the FNO pencil chains.
+
+Four linear ops cover every transform in DistributedFNOBlock:
+
+  rfft_trunc(x, dim, m)            real -> kept low modes of the half-spectrum
+  fft_trunc(x, dim, m_lo, m_hi)    complex -> kept low+high modes
+  pad_ifft(y, dim, n, m_lo, m_hi)  kept modes -> full inverse (1/n folded in)
+  pad_irfft(y, dim, n_half, n_out, m)  kept low half-spectrum modes -> real
+
+On MI355X these run as single-pass hand-written DFT kernels (csrc/dft.hip):
+for the kept-mode counts of an FNO (m ~ 8..24 of N ~ 30..64) a truncated
+naive DFT costs N*m complex MACs per line — comparable FLOPs to a full FFT —
+while fusing the mode truncation / zero-padding / 1/n scale into the
+transform and reading the tensor ONCE along its native strides.  The
+torch/hipFFT path this replaces pays a DtoD staging copy per transform,
+permute copies for middle dims, separate cat/zeros passes for the
+truncation, and their autograd mirrors (~25 ms/step at the flagship config).
+
+Fallback (CPU, N > 64, or unsupported dtype) composes differentiable
+torch.fft ops with identical semantics.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from .. import _ext
+
+__all__ = ["rfft_trunc", "fft_trunc", "pad_ifft", "pad_irfft"]
+
+_MAX_N = 64
+
+
+def _native_ok(x: torch.Tensor, n: int, m: int) -> bool:
+    return (x.is_cuda and n <= _MAX_N and m <= 32
+            and x.dtype in (torch.float32, torch.complex64,
+                            torch.float64, torch.complex128))
+
+
+# ---------------------------------------------------------------------------
+# torch fallback compositions (differentiable; also the semantic reference)
+# ---------------------------------------------------------------------------
+
+def _t_rfft_trunc(x, dim, m):
+    X = torch.fft.rfft(x, dim=dim)
+    return X.narrow(dim, 0, m)
+
+
+def _t_fft_trunc(x, dim, m_lo, m_hi):
+    X = torch.fft.fft(x, dim=dim)
+    lo = X.narrow(dim, 0, m_lo)
+    if m_hi == 0:
+        return lo.contiguous()
+    hi = X.narrow(dim, X.shape[dim] - m_hi, m_hi)
+    return torch.cat([lo, hi], dim=dim)
+
+
+def _pad_modes(y, dim, n, m_lo, m_hi):
+    pad = n - y.shape[dim]
+    if pad < 1:
+        return y
+    shape = list(y.shape)
+    shape[dim] = pad
+    z = torch.zeros(shape, dtype=y.dtype, device=y.device, layout=y.layout)
+    pieces = []
+    if m_lo:
+        pieces.append(y.narrow(dim, 0, m_lo))
+    pieces.append(z)
+    if m_hi:
+        pieces.append(y.narrow(dim, y.shape[dim] - m_hi, m_hi))
+    return torch.cat(pieces, dim=dim)
+
+
+def _t_pad_ifft(y, dim, n, m_lo, m_hi):
+    return torch.fft.ifft(_pad_modes(y, dim, n, m_lo, m_hi), dim=dim)
+
+
+def _t_pad_irfft(y, dim, n_half, n_out, m):
+    return torch.fft.irfft(_pad_modes(y, dim, n_half, m, 0), n=n_out, dim=dim)
+
+
+# ---------------------------------------------------------------------------
+# native autograd Functions
+# ---------------------------------------------------------------------------
+
+class _RfftTruncFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, dim, m):
+        ext = _ext.get(required=True)
+        ctx.dim, ctx.m, ctx.n = dim, m, x.shape[dim]
+        return ext.dft_rfft_trunc(x.contiguous(), dim, m)
+
+    @staticmethod
+    def backward(ctx, gy):
+        ext = _ext.get(required=True)
+        gx = ext.dft_rfft_trunc_adj(gy.contiguous(), ctx.dim, ctx.n)
+        return gx, None, None
+
+
+class _FftTruncFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, dim, m_lo, m_hi):
+        ext = _ext.get(required=True)
+        ctx.dim, ctx.m_lo, ctx.m_hi, ctx.n = dim, m_lo, m_hi, x.shape[dim]
+        return ext.dft_c2c(x.contiguous(), dim, ctx.n, m_lo, m_hi, True, 1.0)
+
+    @staticmethod
+    def backward(ctx, gy):
+        # adjoint of (truncate . DFT) = conj-DFT of the zero-extended grad:
+        # gx_j = sum_{k in kept} gY_k w^{+jk}  (inverse sign, no 1/n)
+        ext = _ext.get(required=True)
+        gx = ext.dft_c2c(gy.contiguous(), ctx.dim, ctx.n, ctx.m_lo, ctx.m_hi,
+                         False, 1.0)
+        return gx, None, None, None
+
+
+class _PadIfftFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y, dim, n, m_lo, m_hi):
+        ext = _ext.get(required=True)
+        ctx.dim, ctx.n, ctx.m_lo, ctx.m_hi = dim, n, m_lo, m_hi
+        return ext.dft_c2c(y.contiguous(), dim, n, m_lo, m_hi, False, 1.0 / n)
+
+    @staticmethod
+    def backward(ctx, gx):
+        # adjoint: gY_k = (1/n) sum_j gx_j w^{-jk}, k in kept
+        ext = _ext.get(required=True)
+        gy = ext.dft_c2c(gx.contiguous(), ctx.dim, ctx.n, ctx.m_lo, ctx.m_hi,
+                         True, 1.0 / ctx.n)
+        return gy, None, None, None, None
+
+
+class _PadIrfftFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y, dim, n_half, n_out, m):
+        ext = _ext.get(required=True)
+        ctx.dim, ctx.n_half, ctx.n_out, ctx.m = dim, n_half, n_out, m
+        return ext.dft_pad_irfft(y.contiguous(), dim, n_out, m)
+
+    @staticmethod
+    def backward(ctx, gx):
+        ext = _ext.get(required=True)
+        gy = ext.dft_pad_irfft_adj(gx.contiguous(), ctx.dim, ctx.m)
+        return gy, None, None, None, None
+
+
+# ---------------------------------------------------------------------------
+# public API
+# ---------------------------------------------------------------------------
+
+def rfft_trunc(x, dim, m):
+    d = dim % x.dim()
+    if _native_ok(x, x.shape[d], m) and d == x.dim() - 1:
+        return _RfftTruncFn.apply(x, d, m)
+    return _t_rfft_trunc(x, dim, m)
+
+
+def fft_trunc(x, dim, m_lo, m_hi):
+    d = dim % x.dim()
+    if _native_ok(x, x.shape[d], m_lo + m_hi):
+        return _FftTruncFn.apply(x, d, m_lo, m_hi)
+    return _t_fft_trunc(x, dim, m_lo, m_hi)
+
+
+def pad_ifft(y, dim, n, m_lo, m_hi):
+    d = dim % y.dim()
+    if _native_ok(y, n, m_lo + m_hi):
+        return _PadIfftFn.apply(y, d, n, m_lo, m_hi)
+    return _t_pad_ifft(y, dim, n, m_lo, m_hi)
+
+
+def pad_irfft(y, dim, n_half, n_out, m):
+    d = dim % y.dim()
+    if _native_ok(y, n_out, m) and d == y.dim() - 1:
+        return _PadIrfftFn.apply(y, d, n_half, n_out, m)
+    return _t_pad_irfft(y, dim, n_half, n_out, m)

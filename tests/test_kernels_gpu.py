@@ -226,3 +226,82 @@ def test_proj_head(dtype, tt, B, I, M, O2, S):
     for a, b in [(x, xr), (W3, W3r), (b3, b3r), (W4, W4r), (b4, b4r)]:
         assert torch.allclose(a.grad, b.grad, rtol=tt * 30, atol=tt * 30), \
             f"grad max {(a.grad - b.grad).abs().max()}"
+
+
+# ---------------------------------------------------------------------------
+# fused truncated-spectrum DFTs vs torch.fft compositions
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dtype,tt", [(torch.float32, 2e-4), (torch.float64, 1e-11)])
+@pytest.mark.parametrize("shape,dim,N,m", [
+    ((1, 20, 9, 9, 6, 30), 5, 30, 8),    # flagship t-dim rfft
+    ((2, 4, 7, 15), 3, 15, 5),           # odd N
+])
+def test_dft_rfft_trunc(dtype, tt, shape, dim, N, m):
+    from dfno_amd.ops.fft import rfft_trunc, _t_rfft_trunc
+    torch.manual_seed(10)
+    x = torch.randn(*shape, device="cuda", dtype=dtype, requires_grad=True)
+    y = rfft_trunc(x, dim, m)
+    xr = x.detach().clone().requires_grad_(True)
+    yr = _t_rfft_trunc(xr, dim, m)
+    assert torch.allclose(y, yr, rtol=tt, atol=tt * 10), f"fwd {(y-yr).abs().max()}"
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g)
+    assert torch.allclose(x.grad, xr.grad, rtol=tt, atol=tt * 10), \
+        f"bwd {(x.grad-xr.grad).abs().max()}"
+
+
+@pytest.mark.parametrize("dtype,tt", [(torch.complex64, 3e-4), (torch.complex128, 1e-11)])
+@pytest.mark.parametrize("shape,dim,mlo,mhi", [
+    ((1, 20, 64, 24, 8), 2, 12, 12),     # z-dim with inner>1
+    ((1, 6, 10, 5), 1, 3, 2),
+    ((2, 5, 12), 2, 4, 3),               # last dim c2c
+])
+def test_dft_fft_trunc_and_pad_ifft(dtype, tt, shape, dim, mlo, mhi):
+    from dfno_amd.ops.fft import fft_trunc, pad_ifft, _t_fft_trunc, _t_pad_ifft
+    torch.manual_seed(11)
+    n = shape[dim]
+    x = torch.randn(*shape, device="cuda", dtype=dtype, requires_grad=True)
+    y = fft_trunc(x, dim, mlo, mhi)
+    xr = x.detach().clone().requires_grad_(True)
+    yr = _t_fft_trunc(xr, dim, mlo, mhi)
+    assert torch.allclose(y, yr, rtol=tt, atol=tt * 10), f"fwd {(y-yr).abs().max()}"
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g)
+    assert torch.allclose(x.grad, xr.grad, rtol=tt, atol=tt * 10), \
+        f"bwd {(x.grad-xr.grad).abs().max()}"
+
+    # inverse path
+    z = torch.randn_like(y).requires_grad_(True)
+    w = pad_ifft(z, dim, n, mlo, mhi)
+    zr = z.detach().clone().requires_grad_(True)
+    wr = _t_pad_ifft(zr, dim, n, mlo, mhi)
+    assert torch.allclose(w, wr, rtol=tt, atol=tt * 10), f"ifft fwd {(w-wr).abs().max()}"
+    g2 = torch.randn_like(w)
+    w.backward(g2)
+    wr.backward(g2)
+    assert torch.allclose(z.grad, zr.grad, rtol=tt, atol=tt * 10), \
+        f"ifft bwd {(z.grad-zr.grad).abs().max()}"
+
+
+@pytest.mark.parametrize("dtype,tt", [(torch.complex64, 2e-4), (torch.complex128, 1e-11)])
+@pytest.mark.parametrize("shape,n_half,n_out,m", [
+    ((1, 20, 9, 9, 6, 8), 16, 30, 8),    # flagship irfft
+    ((2, 4, 3), 9, 17, 3),               # odd n_out
+    ((2, 4, 8), 8, 14, 8),               # m hits the Nyquist bin (even n)
+])
+def test_dft_pad_irfft(dtype, tt, shape, n_half, n_out, m):
+    from dfno_amd.ops.fft import pad_irfft, _t_pad_irfft
+    torch.manual_seed(12)
+    y = torch.randn(*shape, device="cuda", dtype=dtype, requires_grad=True)
+    x = pad_irfft(y, -1, n_half, n_out, m)
+    yr = y.detach().clone().requires_grad_(True)
+    xr = _t_pad_irfft(yr, -1, n_half, n_out, m)
+    assert torch.allclose(x, xr, rtol=tt, atol=tt * 10), f"fwd {(x-xr).abs().max()}"
+    g = torch.randn_like(x)
+    x.backward(g)
+    xr.backward(g)
+    assert torch.allclose(y.grad, yr.grad, rtol=tt, atol=tt * 10), \
+        f"bwd {(y.grad-yr.grad).abs().max()}"
