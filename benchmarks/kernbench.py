@@ -112,4 +112,36 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    if "--dft" not in sys.argv:
+        main()
+
+
+def dft_bench():
+    """DFT kernels at flagship shapes."""
+    # t-dim rfft: [1,20,64,64,64,30] -> m 8
+    x = torch.randn(1, 20, 64, 64, 64, 30, device="cuda")
+    dt = bench(lambda: ext.dft_rfft_trunc(x, 5, 8))
+    report("dft_rfft_trunc t30->8", dt, (30 + 16) * 5242880 * 4)
+    # z-dim c2c: [1,20,64,64,64,8] c64 -> 24
+    xz = torch.randn(1, 20, 64, 64, 64, 8, device="cuda", dtype=torch.complex64)
+    dt = bench(lambda: ext.dft_c2c(xz, 4, 64, 12, 12, True, 1.0))
+    report("dft_c2c z64->24 analysis", dt, (64 + 24) * 8 * 20 * 64 * 64 * 8)
+    dtS = bench(lambda: ext.dft_c2c(ext.dft_c2c(xz, 4, 64, 12, 12, True, 1.0), 4, 64, 12, 12, False, 1.0/64))
+    report("dft_c2c z analysis+synthesis", dtS, 2 * (64 + 24) * 8 * 20 * 64 * 64 * 8)
+    # y-dim c2c on truncated: [1,20,64,64,24,8]
+    xy = torch.randn(1, 20, 64, 64, 24, 8, device="cuda", dtype=torch.complex64)
+    dt = bench(lambda: ext.dft_c2c(xy, 3, 64, 12, 12, True, 1.0))
+    report("dft_c2c y64->24 analysis", dt, (64 + 24) * 8 * 20 * 64 * 24 * 8)
+    # irfft
+    yt = torch.randn(1, 20, 64, 64, 64, 8, device="cuda", dtype=torch.complex64)
+    dt = bench(lambda: ext.dft_pad_irfft(yt, 5, 30, 8))
+    report("dft_pad_irfft 8->30", dt, (16 + 30) * 5242880 * 4)
+    # torch reference: full rfft
+    dt = bench(lambda: torch.fft.rfft(x, dim=5))
+    report("torch rfft t30 (no trunc)", dt, (30 + 32) * 5242880 * 4)
+    dt = bench(lambda: torch.fft.fft(xz, dim=4))
+    report("torch fft z64 (no trunc)", dt, 2 * 64 * 8 * 20 * 64 * 64 * 8)
+
+
+if "__main__" == __name__ and "--dft" in sys.argv:
+    dft_bench()

@@ -41,16 +41,17 @@ __device__ __forceinline__ void sincos_t<double>(double a, double* s, double* c)
   *s = sin(a); *c = cos(a);
 }
 
-// fill the LDS twiddle table: w[r] = exp(-2 pi i r / N)
+// Per-mode twiddle recurrence: each kept mode ki keeps its current factor
+// w^{j k(ki)} in two registers and multiplies by a per-mode step each j —
+// pure independent fma chains that pipeline fully (an LDS table costs a
+// dependent ds_read per MAC and stalls ~10x).  Drift over N <= 64 steps is
+// O(N eps), far below the fp32/fp64 tolerances used here.
+
 template <typename T>
-__device__ __forceinline__ void fill_twiddle(T* wr, T* wi, int N) {
-  for (int r = threadIdx.x; r < N; r += blockDim.x) {
-    T s, c;
-    sincos_t<T>(T(-2.0) * T(M_PI) * T(r) / T(N), &s, &c);
-    wr[r] = c;
-    wi[r] = s;
-  }
-  __syncthreads();
+__device__ __forceinline__ void cmul_acc(T& cr, T& ci, T sr, T si) {
+  T nr = cr * sr - ci * si;
+  T ni = cr * si + ci * sr;
+  cr = nr; ci = ni;
 }
 
 // kept-mode k value for index ki
@@ -60,16 +61,21 @@ __device__ __forceinline__ int kept_k(int ki, int m_lo, int N, int m_hi) {
 
 // ---------------------------------------------------------------------------
 // C2C analysis: out[o, ki, i] = scale * sum_j in[o, j, i] * w^{-j k(ki)}
-// (interleaved complex; SIGN=-1 fixed: analysis always uses the table sign)
 // ---------------------------------------------------------------------------
 
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_c2c_analysis_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long outer, int N, long inner, int m_lo, int m_hi, T scale) {
-  __shared__ T wr[kMaxN], wi[kMaxN];
-  fill_twiddle(wr, wi, N);
   const int m = m_lo + m_hi;
+  T str[MCAP], sti[MCAP];
+#pragma unroll
+  for (int ki = 0; ki < MCAP; ++ki) {
+    if (ki < m) {
+      int k = kept_k(ki, m_lo, N, m_hi);
+      sincos_t<T>(T(-2.0) * T(M_PI) * T(k) / T(N), &sti[ki], &str[ki]);
+    }
+  }
 
   long total = outer * inner;
   long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -79,15 +85,10 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis_kernel(
     long i = t % inner;
     const T* src = in + 2 * (o * N * inner + i);
 
-    T ar[MCAP], ai[MCAP];
-    int rk[MCAP], kk[MCAP];
+    T ar[MCAP], ai[MCAP], cr[MCAP], ci[MCAP];
 #pragma unroll
     for (int ki = 0; ki < MCAP; ++ki) {
-      if (ki < m) {
-        ar[ki] = T(0); ai[ki] = T(0);
-        rk[ki] = 0;
-        kk[ki] = kept_k(ki, m_lo, N, m_hi);
-      }
+      if (ki < m) { ar[ki] = T(0); ai[ki] = T(0); cr[ki] = T(1); ci[ki] = T(0); }
     }
     for (int j = 0; j < N; ++j) {
       const T xr = src[2 * j * inner];
@@ -95,11 +96,9 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis_kernel(
 #pragma unroll
       for (int ki = 0; ki < MCAP; ++ki) {
         if (ki < m) {
-          const T cr = wr[rk[ki]], ci = wi[rk[ki]];
-          ar[ki] += xr * cr - xi * ci;
-          ai[ki] += xr * ci + xi * cr;
-          rk[ki] += kk[ki];
-          if (rk[ki] >= N) rk[ki] -= N;
+          ar[ki] += xr * cr[ki] - xi * ci[ki];
+          ai[ki] += xr * ci[ki] + xi * cr[ki];
+          cmul_acc(cr[ki], ci[ki], str[ki], sti[ki]);
         }
       }
     }
@@ -116,16 +115,21 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis_kernel(
 
 // ---------------------------------------------------------------------------
 // C2C synthesis: out[o, j, i] = scale * sum_ki in[o, ki, i] * w^{+j k(ki)}
-// (the kept modes are the inputs; output is the full length N)
 // ---------------------------------------------------------------------------
 
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long outer, int N, long inner, int m_lo, int m_hi, T scale) {
-  __shared__ T wr[kMaxN], wi[kMaxN];
-  fill_twiddle(wr, wi, N);
   const int m = m_lo + m_hi;
+  T str[MCAP], sti[MCAP];
+#pragma unroll
+  for (int ki = 0; ki < MCAP; ++ki) {
+    if (ki < m) {
+      int k = kept_k(ki, m_lo, N, m_hi);
+      sincos_t<T>(T(2.0) * T(M_PI) * T(k) / T(N), &sti[ki], &str[ki]);
+    }
+  }
 
   long total = outer * inner;
   long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -135,16 +139,13 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
     long i = t % inner;
     const T* src = in + 2 * (o * m * inner + i);
 
-    // stage the m kept inputs in registers (scaled once)
-    T yr[MCAP], yi[MCAP];
-    int rk[MCAP], kk[MCAP];
+    T yr[MCAP], yi[MCAP], cr[MCAP], ci[MCAP];
 #pragma unroll
     for (int ki = 0; ki < MCAP; ++ki) {
       if (ki < m) {
         yr[ki] = scale * src[2 * ki * inner];
         yi[ki] = scale * src[2 * ki * inner + 1];
-        rk[ki] = 0;
-        kk[ki] = kept_k(ki, m_lo, N, m_hi);
+        cr[ki] = T(1); ci[ki] = T(0);
       }
     }
     T* dst = out + 2 * (o * N * inner + i);
@@ -153,12 +154,9 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
 #pragma unroll
       for (int ki = 0; ki < MCAP; ++ki) {
         if (ki < m) {
-          // +jk: conjugate of the table entry
-          const T cr = wr[rk[ki]], ci = -wi[rk[ki]];
-          sr += yr[ki] * cr - yi[ki] * ci;
-          si += yr[ki] * ci + yi[ki] * cr;
-          rk[ki] += kk[ki];
-          if (rk[ki] >= N) rk[ki] -= N;
+          sr += yr[ki] * cr[ki] - yi[ki] * ci[ki];
+          si += yr[ki] * ci[ki] + yi[ki] * cr[ki];
+          cmul_acc(cr[ki], ci[ki], str[ki], sti[ki]);
         }
       }
       dst[2 * j * inner] = sr;
@@ -169,36 +167,36 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
 
 // ---------------------------------------------------------------------------
 // R2C (last dim): out[l, k] = fac_k * scale * sum_j in[l, j] * w^{-jk}
-// fac_k = 1 (plain rfft_trunc) or the irfft-adjoint factors {1, 2, ..,
-// 2, (1 at Nyquist)} with the DC/Nyquist imaginary parts zeroed.
 // ---------------------------------------------------------------------------
 
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long lines, int N, int m, T scale, bool factors) {
-  __shared__ T wr[kMaxN], wi[kMaxN];
-  fill_twiddle(wr, wi, N);
+  T str[MCAP], sti[MCAP];
+#pragma unroll
+  for (int k = 0; k < MCAP; ++k) {
+    if (k < m)
+      sincos_t<T>(T(-2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
+  }
 
   long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long l = t0; l < lines; l += stride) {
     const T* src = in + l * N;
-    T ar[MCAP], ai[MCAP];
-    int rk[MCAP];
+    T ar[MCAP], ai[MCAP], cr[MCAP], ci[MCAP];
 #pragma unroll
     for (int k = 0; k < MCAP; ++k) {
-      if (k < m) { ar[k] = T(0); ai[k] = T(0); rk[k] = 0; }
+      if (k < m) { ar[k] = T(0); ai[k] = T(0); cr[k] = T(1); ci[k] = T(0); }
     }
     for (int j = 0; j < N; ++j) {
       const T x = src[j];
 #pragma unroll
       for (int k = 0; k < MCAP; ++k) {
         if (k < m) {
-          ar[k] += x * wr[rk[k]];
-          ai[k] += x * wi[rk[k]];
-          rk[k] += k;
-          if (rk[k] >= N) rk[k] -= N;
+          ar[k] += x * cr[k];
+          ai[k] += x * ci[k];
+          cmul_acc(cr[k], ci[k], str[k], sti[k]);
         }
       }
     }
@@ -218,47 +216,48 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
 
 // ---------------------------------------------------------------------------
 // C2R (last dim): out[l, j] = scale * Re( sum_k fac_k * in[l, k] * w^{+jk} )
-// fac_k = {1, 2, .., 2, (1 at Nyquist)} for pad_irfft (scale = 1/N), or all
-// ones for the rfft_trunc adjoint (scale = 1).
 // ---------------------------------------------------------------------------
 
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long lines, int N, int m, T scale, bool factors) {
-  __shared__ T wr[kMaxN], wi[kMaxN];
-  fill_twiddle(wr, wi, N);
+  T str[MCAP], sti[MCAP];
+#pragma unroll
+  for (int k = 0; k < MCAP; ++k) {
+    if (k < m)
+      sincos_t<T>(T(2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
+  }
 
   long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long l = t0; l < lines; l += stride) {
     const T* src = in + 2 * l * m;
-    T yr[MCAP], yi[MCAP];
-    int rk[MCAP];
+    T yr[MCAP], yi[MCAP], cr[MCAP], ci[MCAP];
 #pragma unroll
     for (int k = 0; k < MCAP; ++k) {
       if (k < m) {
         T f = scale;
-        if (factors && !(k == 0 || (N % 2 == 0 && 2 * k == N))) f = T(2) * scale;
+        bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
+        if (factors && !edge) f = T(2) * scale;
         yr[k] = f * src[2 * k];
         yi[k] = f * src[2 * k + 1];
-        if (factors && (k == 0 || (N % 2 == 0 && 2 * k == N))) yi[k] = T(0);
-        rk[k] = 0;
+        if (factors && edge) yi[k] = T(0);
+        cr[k] = T(1); ci[k] = T(0);
       }
     }
     T* dst = out + l * N;
     for (int j = 0; j < N; ++j) {
-      T s = T(0);
+      T sacc = T(0);
 #pragma unroll
       for (int k = 0; k < MCAP; ++k) {
         if (k < m) {
-          // Re(y * conj-sign(+jk)): w^{+jk} = (wr, -wi)
-          s += yr[k] * wr[rk[k]] + yi[k] * wi[rk[k]];
-          rk[k] += k;
-          if (rk[k] >= N) rk[k] -= N;
+          // Re(y * w^{+jk}) with w^{+jk} = (cr, ci)
+          sacc += yr[k] * cr[k] - yi[k] * ci[k];
+          cmul_acc(cr[k], ci[k], str[k], sti[k]);
         }
       }
-      dst[j] = s;
+      dst[j] = sacc;
     }
   }
 }
