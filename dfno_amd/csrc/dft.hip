@@ -280,6 +280,7 @@ void split_dims(const at::Tensor& x, int dim, long& outer, long& inner) {
 #define DFT_MDISPATCH(KERNEL, ...)                                             \
   if (m <= 8) { hipLaunchKernelGGL((KERNEL<scalar_t, 8>), __VA_ARGS__); }      \
   else if (m <= 16) { hipLaunchKernelGGL((KERNEL<scalar_t, 16>), __VA_ARGS__); } \
+  else if (m <= 24) { hipLaunchKernelGGL((KERNEL<scalar_t, 24>), __VA_ARGS__); } \
   else { TORCH_CHECK(m <= 32, "dft: m > 32 unsupported natively");             \
          hipLaunchKernelGGL((KERNEL<scalar_t, 32>), __VA_ARGS__); }
 
