@@ -169,10 +169,15 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
 // R2C (last dim): out[l, k] = fac_k * scale * sum_j in[l, j] * w^{-jk}
 // ---------------------------------------------------------------------------
 
+// One 256-line tile is cooperatively staged into LDS with coalesced loads
+// (per-thread direct line reads thrash L1: 16 waves x line-span > 32 KiB,
+// measured SQ_WAIT_ANY = 72%); threads then stream their line from LDS.
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long lines, int N, int m, T scale, bool factors) {
+  extern __shared__ __align__(16) char smem_raw[];
+  T* tile = reinterpret_cast<T*>(smem_raw);   // [kBlock * N]
   T str[MCAP], sti[MCAP];
 #pragma unroll
   for (int k = 0; k < MCAP; ++k) {
@@ -180,35 +185,42 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
       sincos_t<T>(T(-2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
   }
 
-  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (long l = t0; l < lines; l += stride) {
-    const T* src = in + l * N;
-    T ar[MCAP], ai[MCAP], cr[MCAP], ci[MCAP];
+  long ntiles = (lines + kBlock - 1) / kBlock;
+  for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x) {
+    long l0 = tb * kBlock;
+    int nl = (int)min((long)kBlock, lines - l0);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+      tile[idx] = in[l0 * N + idx];
+    __syncthreads();
+    if ((int)threadIdx.x < nl) {
+      const T* src = tile + threadIdx.x * N;
+      T ar[MCAP], ai[MCAP], cr[MCAP], ci[MCAP];
 #pragma unroll
-    for (int k = 0; k < MCAP; ++k) {
-      if (k < m) { ar[k] = T(0); ai[k] = T(0); cr[k] = T(1); ci[k] = T(0); }
-    }
-    for (int j = 0; j < N; ++j) {
-      const T x = src[j];
+      for (int k = 0; k < MCAP; ++k) {
+        if (k < m) { ar[k] = T(0); ai[k] = T(0); cr[k] = T(1); ci[k] = T(0); }
+      }
+      for (int j = 0; j < N; ++j) {
+        const T x = src[j];
+#pragma unroll
+        for (int k = 0; k < MCAP; ++k) {
+          if (k < m) {
+            ar[k] += x * cr[k];
+            ai[k] += x * ci[k];
+            cmul_acc(cr[k], ci[k], str[k], sti[k]);
+          }
+        }
+      }
+      T* dst = out + 2 * (l0 + threadIdx.x) * m;
 #pragma unroll
       for (int k = 0; k < MCAP; ++k) {
         if (k < m) {
-          ar[k] += x * cr[k];
-          ai[k] += x * ci[k];
-          cmul_acc(cr[k], ci[k], str[k], sti[k]);
+          T f = T(1);
+          bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
+          if (factors && !edge) f = T(2);
+          dst[2 * k] = f * scale * ar[k];
+          dst[2 * k + 1] = (factors && edge) ? T(0) : f * scale * ai[k];
         }
-      }
-    }
-    T* dst = out + 2 * l * m;
-#pragma unroll
-    for (int k = 0; k < MCAP; ++k) {
-      if (k < m) {
-        T f = T(1);
-        bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
-        if (factors && !edge) f = T(2);
-        dst[2 * k] = f * scale * ar[k];
-        dst[2 * k + 1] = (factors && edge) ? T(0) : f * scale * ai[k];
       }
     }
   }
@@ -218,10 +230,14 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
 // C2R (last dim): out[l, j] = scale * Re( sum_k fac_k * in[l, k] * w^{+jk} )
 // ---------------------------------------------------------------------------
 
+// Output lines are written through an LDS tile and stored cooperatively
+// (coalesced); inputs are 64-128B contiguous per line and read directly.
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long lines, int N, int m, T scale, bool factors) {
+  extern __shared__ __align__(16) char smem_raw[];
+  T* tile = reinterpret_cast<T*>(smem_raw);   // [kBlock * N]
   T str[MCAP], sti[MCAP];
 #pragma unroll
   for (int k = 0; k < MCAP; ++k) {
@@ -229,36 +245,43 @@ __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
       sincos_t<T>(T(2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
   }
 
-  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (long l = t0; l < lines; l += stride) {
-    const T* src = in + 2 * l * m;
-    T yr[MCAP], yi[MCAP], cr[MCAP], ci[MCAP];
-#pragma unroll
-    for (int k = 0; k < MCAP; ++k) {
-      if (k < m) {
-        T f = scale;
-        bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
-        if (factors && !edge) f = T(2) * scale;
-        yr[k] = f * src[2 * k];
-        yi[k] = f * src[2 * k + 1];
-        if (factors && edge) yi[k] = T(0);
-        cr[k] = T(1); ci[k] = T(0);
-      }
-    }
-    T* dst = out + l * N;
-    for (int j = 0; j < N; ++j) {
-      T sacc = T(0);
+  long ntiles = (lines + kBlock - 1) / kBlock;
+  for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x) {
+    long l0 = tb * kBlock;
+    int nl = (int)min((long)kBlock, lines - l0);
+    __syncthreads();
+    if ((int)threadIdx.x < nl) {
+      const T* src = in + 2 * (l0 + threadIdx.x) * m;
+      T yr[MCAP], yi[MCAP], cr[MCAP], ci[MCAP];
 #pragma unroll
       for (int k = 0; k < MCAP; ++k) {
         if (k < m) {
-          // Re(y * w^{+jk}) with w^{+jk} = (cr, ci)
-          sacc += yr[k] * cr[k] - yi[k] * ci[k];
-          cmul_acc(cr[k], ci[k], str[k], sti[k]);
+          T f = scale;
+          bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
+          if (factors && !edge) f = T(2) * scale;
+          yr[k] = f * src[2 * k];
+          yi[k] = f * src[2 * k + 1];
+          if (factors && edge) yi[k] = T(0);
+          cr[k] = T(1); ci[k] = T(0);
         }
       }
-      dst[j] = sacc;
+      T* dst = tile + threadIdx.x * N;
+      for (int j = 0; j < N; ++j) {
+        T sacc = T(0);
+#pragma unroll
+        for (int k = 0; k < MCAP; ++k) {
+          if (k < m) {
+            // Re(y * w^{+jk}) with w^{+jk} = (cr, ci)
+            sacc += yr[k] * cr[k] - yi[k] * ci[k];
+            cmul_acc(cr[k], ci[k], str[k], sti[k]);
+          }
+        }
+        dst[j] = sacc;
+      }
     }
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+      out[l0 * N + idx] = tile[idx];
   }
 }
 
@@ -337,11 +360,13 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
   if (x.numel() == 0) return out;
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  int grid = grid_for_d(lines);
+  long ntiles = (lines + kBlock - 1) / kBlock;
+  int grid = (int)std::min(ntiles, 4096L);
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "dft_r2c", [&] {
+    size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = x.data_ptr<scalar_t>();
     auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
-    DFT_MDISPATCH(dft_r2c_last_kernel, dim3(grid), dim3(kBlock), 0, stream,
+    DFT_MDISPATCH(dft_r2c_last_kernel, dim3(grid), dim3(kBlock), smem, stream,
                   inp, op, lines, N, (int)m, (scalar_t)scale, factors)
   });
   return out;
@@ -365,11 +390,13 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
   if (y.numel() == 0) return out;
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  int grid = grid_for_d(lines);
+  long ntiles = (lines + kBlock - 1) / kBlock;
+  int grid = (int)std::min(ntiles, 4096L);
   AT_DISPATCH_FLOATING_TYPES(out.scalar_type(), "dft_c2r", [&] {
+    size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = reinterpret_cast<const scalar_t*>(y.data_ptr());
     auto op = out.data_ptr<scalar_t>();
-    DFT_MDISPATCH(dft_c2r_last_kernel, dim3(grid), dim3(kBlock), 0, stream,
+    DFT_MDISPATCH(dft_c2r_last_kernel, dim3(grid), dim3(kBlock), smem, stream,
                   inp, op, lines, N, (int)m, (scalar_t)scale, factors)
   });
   return out;
