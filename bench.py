@@ -74,7 +74,8 @@ def main():
                                   num_blocks=args.num_blocks, device=device,
                                   dtype=torch.float32)
     criterion = dfno.DistributedRelativeLpLoss(P_x)
-    optimizer = torch.optim.Adam(model.parameters(), lr=1e-3)
+    from dfno_amd.optim import Adam as FusedAdam
+    optimizer = FusedAdam(model.parameters(), lr=1e-3)
 
     # synthetic local shards of the global tensors
     info_x = compute_distribution_info(P_x, GLOBAL_SHAPE)

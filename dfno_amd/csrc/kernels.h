@@ -32,6 +32,11 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
 std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
                                           bool want_bias);
 
+// fused Adam step on flat real views
+void adam_step_(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
+                double lr, double beta1, double beta2, double eps,
+                double weight_decay, int64_t step);
+
 // fused truncated-spectrum DFTs (see ops/fft.py):
 at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
                    int64_t m_lo, int64_t m_hi, bool analysis, double scale);

@@ -843,3 +843,91 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
 #undef GW_LAUNCH
   return {gW, gb};
 }
+
+// ---------------------------------------------------------------------------
+// fused Adam step (flat real view; complex params are elementwise-identical
+// to their real views under torch's Adam). One vec4 pass over p/g/m/v.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+template <typename T, bool VECTOR>
+__global__ void adam_step_kernel(T* __restrict__ p, const T* __restrict__ g,
+                                 T* __restrict__ m, T* __restrict__ v,
+                                 long n, T lr, T b1, T b2, T eps, T wd,
+                                 T c1, T c2) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  if constexpr (VECTOR && std::is_same<T, float>::value) {
+    for (long i = i0; i * 4 < n; i += stride) {
+      float4 pv = *reinterpret_cast<float4*>(p + i * 4);
+      const float4 gv = *reinterpret_cast<const float4*>(g + i * 4);
+      float4 mv = *reinterpret_cast<float4*>(m + i * 4);
+      float4 vv = *reinterpret_cast<float4*>(v + i * 4);
+      float pr[4] = {pv.x, pv.y, pv.z, pv.w};
+      float gr[4] = {gv.x, gv.y, gv.z, gv.w};
+      float mr[4] = {mv.x, mv.y, mv.z, mv.w};
+      float vr[4] = {vv.x, vv.y, vv.z, vv.w};
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        float gg = gr[k] + wd * pr[k];
+        mr[k] = b1 * mr[k] + (1.f - b1) * gg;
+        vr[k] = b2 * vr[k] + (1.f - b2) * gg * gg;
+        float mh = mr[k] / c1;
+        float vh = vr[k] / c2;
+        pr[k] -= lr * mh / (sqrtf(vh) + eps);
+      }
+      *reinterpret_cast<float4*>(p + i * 4) = make_float4(pr[0], pr[1], pr[2], pr[3]);
+      *reinterpret_cast<float4*>(m + i * 4) = make_float4(mr[0], mr[1], mr[2], mr[3]);
+      *reinterpret_cast<float4*>(v + i * 4) = make_float4(vr[0], vr[1], vr[2], vr[3]);
+    }
+  } else {
+    for (long i = i0; i < n; i += stride) {
+      T gg = g[i] + wd * p[i];
+      T mi = b1 * m[i] + (T(1) - b1) * gg;
+      T vi = b2 * v[i] + (T(1) - b2) * gg * gg;
+      m[i] = mi;
+      v[i] = vi;
+      T mh = mi / c1;
+      T vh = vi / c2;
+      p[i] -= lr * mh / (sqrt(vh) + eps);
+    }
+  }
+}
+
+}  // namespace
+
+void adam_step_(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
+                double lr, double beta1, double beta2, double eps,
+                double weight_decay, int64_t step) {
+  check_f(p, "p"); check_f(g, "g"); check_f(m, "m"); check_f(v, "v");
+  long n = p.numel();
+  TORCH_CHECK(g.numel() == n && m.numel() == n && v.numel() == n, "adam: size mismatch");
+  if (n == 0) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  double c1 = 1.0 - std::pow(beta1, (double)step);
+  double c2 = 1.0 - std::pow(beta2, (double)step);
+  AT_DISPATCH_FLOATING_TYPES(p.scalar_type(), "adam_step_", [&] {
+    bool vec = std::is_same<scalar_t, float>::value && (n % 4 == 0) &&
+               ((reinterpret_cast<uintptr_t>(p.data_ptr()) & 15) == 0) &&
+               ((reinterpret_cast<uintptr_t>(g.data_ptr()) & 15) == 0) &&
+               ((reinterpret_cast<uintptr_t>(m.data_ptr()) & 15) == 0) &&
+               ((reinterpret_cast<uintptr_t>(v.data_ptr()) & 15) == 0);
+    int grid = grid_for(vec ? (n + 3) / 4 : n, kBlock);
+    if (vec) {
+      hipLaunchKernelGGL((adam_step_kernel<scalar_t, true>), dim3(grid), dim3(kBlock),
+                         0, stream, p.data_ptr<scalar_t>(), g.data_ptr<scalar_t>(),
+                         m.data_ptr<scalar_t>(), v.data_ptr<scalar_t>(), n,
+                         (scalar_t)lr, (scalar_t)beta1, (scalar_t)beta2,
+                         (scalar_t)eps, (scalar_t)weight_decay,
+                         (scalar_t)c1, (scalar_t)c2);
+    } else {
+      hipLaunchKernelGGL((adam_step_kernel<scalar_t, false>), dim3(grid), dim3(kBlock),
+                         0, stream, p.data_ptr<scalar_t>(), g.data_ptr<scalar_t>(),
+                         m.data_ptr<scalar_t>(), v.data_ptr<scalar_t>(), n,
+                         (scalar_t)lr, (scalar_t)beta1, (scalar_t)beta2,
+                         (scalar_t)eps, (scalar_t)weight_decay,
+                         (scalar_t)c1, (scalar_t)c2);
+    }
+  });
+}

@@ -1,0 +1,73 @@
+"""Fused Adam for MI355X.
+
+``dfno_amd.optim.Adam`` subclasses ``torch.optim.Adam`` with an identical
+state layout (``step``/``exp_avg``/``exp_avg_sq``) and semantics; on GPU
+fp32/complex64 parameters the update runs as one fused HIP kernel per
+parameter over flat real views (complex Adam is elementwise-identical on
+the real view), ~3x less optimizer overhead than the foreach path at the
+flagship config.  Any other parameter falls back to the stock step.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import _ext
+
+__all__ = ["Adam"]
+
+
+class Adam(torch.optim.Adam):
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+
+        ext = _ext.get(required=False) if torch.cuda.is_available() else None
+        fallback_groups = []
+        for group in self.param_groups:
+            lr = group["lr"]
+            beta1, beta2 = group["betas"]
+            eps = group["eps"]
+            wd = group["weight_decay"]
+            if group.get("amsgrad", False) or ext is None:
+                fallback_groups.append(group)
+                continue
+            slow = []
+            for p in group["params"]:
+                if p.grad is None or p.numel() == 0:
+                    continue
+                if not (p.is_cuda and p.dtype in (torch.float32, torch.complex64,
+                                                  torch.float64, torch.complex128)
+                        and not p.grad.is_sparse):
+                    slow.append(p)
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = torch.tensor(0.0)
+                    state["exp_avg"] = torch.zeros_like(p)
+                    state["exp_avg_sq"] = torch.zeros_like(p)
+                state["step"] += 1
+                pv = torch.view_as_real(p) if p.is_complex() else p
+                gv = torch.view_as_real(p.grad) if p.grad.is_complex() else p.grad
+                mv = state["exp_avg"]
+                mv = torch.view_as_real(mv) if mv.is_complex() else mv
+                vv = state["exp_avg_sq"]
+                vv = torch.view_as_real(vv) if vv.is_complex() else vv
+                ext.adam_step_(pv.reshape(-1), gv.reshape(-1).contiguous(),
+                               mv.reshape(-1), vv.reshape(-1),
+                               lr, beta1, beta2, eps, wd,
+                               int(state["step"].item()))
+            if slow:
+                fallback_groups.append({**group, "params": slow})
+
+        if fallback_groups:
+            saved = self.param_groups
+            try:
+                self.param_groups = fallback_groups
+                super().step()
+            finally:
+                self.param_groups = saved
+        return loss

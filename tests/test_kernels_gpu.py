@@ -305,3 +305,38 @@ def test_dft_pad_irfft(dtype, tt, shape, n_half, n_out, m):
     xr.backward(g)
     assert torch.allclose(y.grad, yr.grad, rtol=tt, atol=tt * 10), \
         f"bwd {(y.grad-yr.grad).abs().max()}"
+
+
+# ---------------------------------------------------------------------------
+# fused Adam vs torch.optim.Adam
+# ---------------------------------------------------------------------------
+
+def test_fused_adam_matches_torch():
+    from dfno_amd.optim import Adam as FusedAdam
+    torch.manual_seed(20)
+
+    def make_params():
+        return [torch.randn(1000, device="cuda", requires_grad=True),
+                torch.randn(20, 20, 123, device="cuda", dtype=torch.complex64,
+                            requires_grad=True),
+                torch.randn(7, device="cuda", dtype=torch.float64, requires_grad=True)]
+
+    p1 = make_params()
+    torch.manual_seed(20)
+    p2 = make_params()
+    for a, b in zip(p1, p2):
+        assert torch.equal(a.detach(), b.detach())
+
+    o1 = FusedAdam(p1, lr=1e-2, weight_decay=1e-4)
+    o2 = torch.optim.Adam(p2, lr=1e-2, weight_decay=1e-4)
+    for it in range(5):
+        torch.manual_seed(100 + it)
+        for a, b in zip(p1, p2):
+            g = torch.randn_like(a)
+            a.grad = g.clone()
+            b.grad = g.clone()
+        o1.step()
+        o2.step()
+    for a, b in zip(p1, p2):
+        assert torch.allclose(a.detach(), b.detach(), rtol=1e-5, atol=1e-6), \
+            f"max {(a.detach()-b.detach()).abs().max()}"
