@@ -126,16 +126,18 @@ class Partition:
             self.index = np.asarray([-1] * self.dim, dtype=int)
         self._group = None
         self._group_built = False
+        # Sub-world groups are created EAGERLY here: dist.new_group is
+        # collective over the WORLD, and Partition construction is the only
+        # point guaranteed to run SPMD on every rank (at first use, inactive
+        # ranks would skip the call and deadlock the members).
+        if is_distributed() and 1 < self.size < world_size():
+            self._group = _get_group(self.ranks)
+            self._group_built = True
 
     # -- communicator ------------------------------------------------------
     @property
     def group(self):
-        """The torch.distributed process group for this partition's ranks.
-
-        NOTE: first access is collective over the *world* (dist.new_group), so
-        partitions must be constructed SPMD on every rank — which the model
-        code guarantees (same constructors run everywhere).
-        """
+        """The torch.distributed process group for this partition's ranks."""
         if not self._group_built:
             self._group = _get_group(self.ranks)
             self._group_built = True
