@@ -27,11 +27,16 @@ from dfno_amd.partition import init_distributed, is_distributed, world_rank, wor
 from dfno_amd.partition import compute_distribution_info
 
 
+# Partitions are chosen over the LEADING spatial axes only: the pencil
+# construction then gives P_m == P_x (reference dfno.py:88-89 semantics), so
+# R1/R4 are identities and only the truncated spectrum (~18 MB global)
+# crosses xGMI in R2/R3 — the 553 MB real-activation all-to-alls the
+# reference pays per block never happen on this curve.
 PARTITIONS = {
     1: (1, 1, 1, 1, 1, 1),
     2: (1, 1, 2, 1, 1, 1),
     4: (1, 1, 2, 2, 1, 1),
-    8: (1, 1, 2, 2, 2, 1),
+    8: (1, 1, 4, 2, 1, 1),
 }
 
 # flagship config (BASELINE.md / BASELINE.json): 3D two-phase FNO, fixed grid

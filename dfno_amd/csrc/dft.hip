@@ -361,8 +361,21 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
     long l0 = tb * kBlock;
     int nl = (int)min((long)kBlock, lines - l0);
     __syncthreads();
-    for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
-      tile[idx] = in[l0 * N + idx];
+    if constexpr (std::is_same<T, float>::value) {
+      const long base = l0 * N;
+      if ((nl * N) % 4 == 0 && (base % 4 == 0) &&
+          ((reinterpret_cast<uintptr_t>(in) & 15) == 0)) {
+        for (int idx = threadIdx.x * 4; idx < nl * N; idx += kBlock * 4)
+          *reinterpret_cast<float4*>(&tile[idx]) =
+              *reinterpret_cast<const float4*>(in + base + idx);
+      } else {
+        for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+          tile[idx] = in[base + idx];
+      }
+    } else {
+      for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+        tile[idx] = in[l0 * N + idx];
+    }
     __syncthreads();
     if ((int)threadIdx.x < nl) {
       const T* src = tile + threadIdx.x * N;
@@ -451,8 +464,21 @@ __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
       }
     }
     __syncthreads();
-    for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
-      out[l0 * N + idx] = tile[idx];
+    if constexpr (std::is_same<T, float>::value) {
+      const long base = l0 * N;
+      if ((nl * N) % 4 == 0 && (base % 4 == 0) &&
+          ((reinterpret_cast<uintptr_t>(out) & 15) == 0)) {
+        for (int idx = threadIdx.x * 4; idx < nl * N; idx += kBlock * 4)
+          *reinterpret_cast<float4*>(out + base + idx) =
+              *reinterpret_cast<const float4*>(&tile[idx]);
+      } else {
+        for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+          out[base + idx] = tile[idx];
+      }
+    } else {
+      for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+        out[l0 * N + idx] = tile[idx];
+    }
   }
 }
 
