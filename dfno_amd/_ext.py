@@ -29,6 +29,7 @@ SOURCES = [
     _CSRC / "spectral.hip",
     _CSRC / "proj_head.hip",
     _CSRC / "dft.hip",
+    _CSRC / "lift_head.hip",
 ]
 
 

@@ -17,6 +17,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("proj_head_fwd", &proj_head_fwd, "fused linear->gelu->linear head");
   m.def("proj_head_bwd", &proj_head_bwd,
         "fused head backward: returns (gz3, gb3, gW4, gb4)");
+  m.def("lift_head_fwd", &lift_head_fwd, "fused time-lift + channel-lift + gelus");
+  m.def("lift_head_bwd", &lift_head_bwd,
+        "lift head backward: (gx, gW1, gb1, gW2, gb2)");
   m.def("adam_step_", &adam_step_, "fused Adam step (in-place)");
   m.def("dft_c2c", &dft_c2c, "truncated/padded complex DFT along a dim");
   m.def("dft_rfft_trunc", &dft_rfft_trunc, "real->kept-low-modes DFT (last dim)");

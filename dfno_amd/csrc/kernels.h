@@ -32,6 +32,14 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
 std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
                                           bool want_bias);
 
+// fused lift head (T_in == 1): y = gelu(W2 @ gelu(W1 *t x + b1) + b2)
+at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
+                         const at::Tensor& b1, const at::Tensor& W2,
+                         const at::Tensor& b2);
+std::vector<at::Tensor> lift_head_bwd(const at::Tensor& gy, const at::Tensor& x,
+                                      const at::Tensor& W1, const at::Tensor& b1,
+                                      const at::Tensor& W2, const at::Tensor& b2);
+
 // fused Adam step on flat real views
 void adam_step_(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
                 double lr, double beta1, double beta2, double eps,

@@ -1,0 +1,315 @@
+// gfx950 fused FNO lift head: y = gelu(W2 @ gelu(W1 @t x + b1) + b2).
+//
+// The reference runs this as a time-lift einsum (T_in -> T_out along the
+// trailing dim), a GELU pass, a channel-lift einsum (C_in -> width) and
+// another GELU (/root/reference/dfno/dfno.py:310-311,333-338).  rocBLAS
+// handles the T_in = 1 time lift (an outer product, M ~ 5e5, K = 1)
+// pathologically (~1.25 ms for a 250 MB write), and the unfused chain
+// costs ~4 ms/step at the flagship config.  Here the whole lift is one
+// pass: read the (tiny) input column, produce the [width, T_out] output
+// column per grid point.  Restricted to T_in == 1 (the two-phase flagship;
+// other configs use the unfused path).
+//
+// Backward (T_in == 1) recomputes the hidden activations from x and emits
+// grad-x plus all four weight/bias grads via per-wave LDS accumulators and
+// one atomic flush per block (proj_head pattern).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include "kernels.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+
+template <typename T>
+__device__ __forceinline__ T gelu_(T z) {
+  return T(0.5) * z * (T(1.0) + erf(z * T(0.7071067811865476)));
+}
+
+template <typename T>
+__device__ __forceinline__ T gelu_g_(T z) {
+  return T(0.5) * (T(1.0) + erf(z * T(0.7071067811865476))) +
+         z * exp(T(-0.5) * z * z) * T(0.3989422804014327);
+}
+
+template <typename T>
+__device__ __forceinline__ T lh_wave_sum(T v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// x: [B, C, S] (T_in == 1 folded away), out: [B, W, S, T]
+// W1: [T, 1], b1: [T], W2: [W, C], b2: [W]
+template <typename T, int CCAP, int TCAP, int WCAP>
+__global__ __launch_bounds__(kBlock) void lift_head_fwd_kernel(
+    const T* __restrict__ x, const T* __restrict__ W1, const T* __restrict__ b1,
+    const T* __restrict__ W2, const T* __restrict__ b2, T* __restrict__ out,
+    int B, int C, int W, int Tn, long S) {
+  __shared__ T w1[TCAP], bb1[TCAP], w2[WCAP * CCAP], bb2[WCAP];
+  for (int k = threadIdx.x; k < Tn; k += kBlock) { w1[k] = W1[k]; bb1[k] = b1[k]; }
+  for (int k = threadIdx.x; k < W * C; k += kBlock) w2[k] = W2[k];
+  for (int k = threadIdx.x; k < W; k += kBlock) bb2[k] = b2[k];
+  __syncthreads();
+
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = t0; t < (long)B * S; t += stride) {
+    long b = t / S;
+    long s = t % S;
+
+    T xv[CCAP];
+#pragma unroll
+    for (int c = 0; c < CCAP; ++c)
+      if (c < C) xv[c] = x[(b * C + c) * S + s];
+
+    // h[c][t] = gelu(W1[t] * x[c] + b1[t])
+    T h[CCAP][TCAP];
+#pragma unroll
+    for (int c = 0; c < CCAP; ++c) {
+      if (c < C) {
+#pragma unroll
+        for (int k = 0; k < TCAP; ++k)
+          if (k < Tn) h[c][k] = gelu_(w1[k] * xv[c] + bb1[k]);
+      }
+    }
+
+    for (int w = 0; w < W; ++w) {
+      T acc[TCAP];
+#pragma unroll
+      for (int k = 0; k < TCAP; ++k)
+        if (k < Tn) acc[k] = bb2[w];
+#pragma unroll
+      for (int c = 0; c < CCAP; ++c) {
+        if (c < C) {
+          T wv = w2[(size_t)w * C + c];
+#pragma unroll
+          for (int k = 0; k < TCAP; ++k)
+            if (k < Tn) acc[k] += wv * h[c][k];
+        }
+      }
+      T* dst = out + ((b * W + w) * S + s) * Tn;
+#pragma unroll
+      for (int k = 0; k < TCAP; ++k)
+        if (k < Tn) dst[k] = gelu_(acc[k]);
+    }
+  }
+}
+
+// gy: [B, W, S, T]; outputs gx [B, C, S] and gW1/gb1/gW2/gb2 via atomics.
+template <typename T, int CCAP, int TCAP, int WCAP>
+__global__ __launch_bounds__(kBlock) void lift_head_bwd_kernel(
+    const T* __restrict__ gy, const T* __restrict__ x,
+    const T* __restrict__ W1, const T* __restrict__ b1,
+    const T* __restrict__ W2, const T* __restrict__ b2,
+    T* __restrict__ gx, T* __restrict__ gW1, T* __restrict__ gb1,
+    T* __restrict__ gW2, T* __restrict__ gb2,
+    int B, int C, int W, int Tn, long S) {
+  __shared__ T w1[TCAP], bb1[TCAP], w2[WCAP * CCAP], bb2[WCAP];
+  // per-wave partial accumulators
+  __shared__ T a_gW1[4][TCAP], a_gb1[4][TCAP], a_gW2[4][WCAP * CCAP], a_gb2[4][WCAP];
+  for (int k = threadIdx.x; k < Tn; k += kBlock) { w1[k] = W1[k]; bb1[k] = b1[k]; }
+  for (int k = threadIdx.x; k < W * C; k += kBlock) w2[k] = W2[k];
+  for (int k = threadIdx.x; k < W; k += kBlock) bb2[k] = b2[k];
+  for (int k = threadIdx.x; k < 4 * TCAP; k += kBlock)
+    a_gW1[k / TCAP][k % TCAP] = a_gb1[k / TCAP][k % TCAP] = T(0);
+  for (int k = threadIdx.x; k < 4 * WCAP * CCAP; k += kBlock)
+    a_gW2[k / (WCAP * CCAP)][k % (WCAP * CCAP)] = T(0);
+  for (int k = threadIdx.x; k < 4 * WCAP; k += kBlock)
+    a_gb2[k / WCAP][k % WCAP] = T(0);
+  __syncthreads();
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x % 64;
+
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = t0; t < (long)B * S; t += stride) {
+    long b = t / S;
+    long s = t % S;
+
+    T xv[CCAP];
+#pragma unroll
+    for (int c = 0; c < CCAP; ++c)
+      if (c < C) xv[c] = x[(b * C + c) * S + s];
+
+    T z1[CCAP][TCAP], h[CCAP][TCAP], gh[CCAP][TCAP];
+#pragma unroll
+    for (int c = 0; c < CCAP; ++c) {
+      if (c < C) {
+#pragma unroll
+        for (int k = 0; k < TCAP; ++k) {
+          if (k < Tn) {
+            z1[c][k] = w1[k] * xv[c] + bb1[k];
+            h[c][k] = gelu_(z1[c][k]);
+            gh[c][k] = T(0);
+          }
+        }
+      }
+    }
+
+    for (int w = 0; w < W; ++w) {
+      const T* gyp = gy + ((b * W + w) * S + s) * Tn;
+      T gz2[TCAP];
+      T gb2p = T(0);
+#pragma unroll
+      for (int k = 0; k < TCAP; ++k) {
+        if (k < Tn) {
+          // recompute z2[w][k]
+          T z2 = bb2[w];
+#pragma unroll
+          for (int c = 0; c < CCAP; ++c)
+            if (c < C) z2 += w2[(size_t)w * C + c] * h[c][k];
+          gz2[k] = gyp[k] * gelu_g_(z2);
+          gb2p += gz2[k];
+        }
+      }
+      // accumulate gh and wave-reduced gW2 partials
+#pragma unroll
+      for (int c = 0; c < CCAP; ++c) {
+        if (c < C) {
+          T wv = w2[(size_t)w * C + c];
+          T gw2p = T(0);
+#pragma unroll
+          for (int k = 0; k < TCAP; ++k) {
+            if (k < Tn) {
+              gh[c][k] += wv * gz2[k];
+              gw2p += gz2[k] * h[c][k];
+            }
+          }
+          gw2p = lh_wave_sum(gw2p);
+          if (lane == 0) a_gW2[wave][w * C + c] += gw2p;
+        }
+      }
+      gb2p = lh_wave_sum(gb2p);
+      if (lane == 0) a_gb2[wave][w] += gb2p;
+    }
+
+    // gz1 = gh * gelu'(z1); gx[c] = sum_k W1[k] gz1[c][k];
+    // gW1[k] += sum_c gz1[c][k] * x[c]; gb1[k] += sum_c gz1[c][k]
+#pragma unroll
+    for (int c = 0; c < CCAP; ++c) {
+      if (c < C) {
+        T gxa = T(0);
+#pragma unroll
+        for (int k = 0; k < TCAP; ++k) {
+          if (k < Tn) {
+            T gz1 = gh[c][k] * gelu_g_(z1[c][k]);
+            gh[c][k] = gz1;  // reuse as gz1 for the reductions below
+            gxa += w1[k] * gz1;
+          }
+        }
+        gx[(b * C + c) * S + s] = gxa;
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < TCAP; ++k) {
+      if (k < Tn) {
+        T pw = T(0), pb = T(0);
+#pragma unroll
+        for (int c = 0; c < CCAP; ++c) {
+          if (c < C) {
+            pw += gh[c][k] * xv[c];
+            pb += gh[c][k];
+          }
+        }
+        pw = lh_wave_sum(pw);
+        pb = lh_wave_sum(pb);
+        if (lane == 0) { a_gW1[wave][k] += pw; a_gb1[wave][k] += pb; }
+      }
+    }
+  }
+
+  __syncthreads();
+  for (int k = threadIdx.x; k < Tn; k += kBlock) {
+    T v = a_gW1[0][k] + a_gW1[1][k] + a_gW1[2][k] + a_gW1[3][k];
+    if (v != T(0)) atomicAdd(&gW1[k], v);
+    v = a_gb1[0][k] + a_gb1[1][k] + a_gb1[2][k] + a_gb1[3][k];
+    if (v != T(0)) atomicAdd(&gb1[k], v);
+  }
+  for (int k = threadIdx.x; k < W * C; k += kBlock) {
+    T v = a_gW2[0][k] + a_gW2[1][k] + a_gW2[2][k] + a_gW2[3][k];
+    if (v != T(0)) atomicAdd(&gW2[k], v);
+  }
+  for (int k = threadIdx.x; k < W; k += kBlock) {
+    T v = a_gb2[0][k] + a_gb2[1][k] + a_gb2[2][k] + a_gb2[3][k];
+    if (v != T(0)) atomicAdd(&gb2[k], v);
+  }
+}
+
+int grid_for_l(long work) {
+  long g = (work + kBlock - 1) / kBlock;
+  long cap = 256L * 8;
+  if (g > cap) g = cap;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+void check_lf(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == at::kFloat || t.scalar_type() == at::kDouble,
+              name, " must be float32/float64");
+}
+
+}  // namespace
+
+at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
+                         const at::Tensor& b1, const at::Tensor& W2,
+                         const at::Tensor& b2) {
+  check_lf(x, "x"); check_lf(W1, "W1"); check_lf(b1, "b1");
+  check_lf(W2, "W2"); check_lf(b2, "b2");
+  TORCH_CHECK(x.dim() == 3, "x must be [B,C,S]");
+  int B = (int)x.size(0), C = (int)x.size(1);
+  long S = x.size(2);
+  int Tn = (int)W1.size(0), W = (int)W2.size(0);
+  TORCH_CHECK(W1.size(1) == 1 && (int)W2.size(1) == C, "lift_head shapes");
+  TORCH_CHECK(C <= 4 && Tn <= 32 && W <= 24, "lift_head: unsupported dims");
+
+  auto out = at::empty({B, W, S, (long)Tn}, x.options());
+  if (x.numel() == 0) return out;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int grid = grid_for_l((long)B * S);
+  AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "lift_head_fwd", [&] {
+    hipLaunchKernelGGL((lift_head_fwd_kernel<scalar_t, 4, 32, 24>), dim3(grid),
+                       dim3(kBlock), 0, stream, x.data_ptr<scalar_t>(),
+                       W1.data_ptr<scalar_t>(), b1.data_ptr<scalar_t>(),
+                       W2.data_ptr<scalar_t>(), b2.data_ptr<scalar_t>(),
+                       out.data_ptr<scalar_t>(), B, C, W, Tn, S);
+  });
+  return out;
+}
+
+std::vector<at::Tensor> lift_head_bwd(const at::Tensor& gy, const at::Tensor& x,
+                                      const at::Tensor& W1, const at::Tensor& b1,
+                                      const at::Tensor& W2, const at::Tensor& b2) {
+  check_lf(gy, "gy"); check_lf(x, "x");
+  int B = (int)x.size(0), C = (int)x.size(1);
+  long S = x.size(2);
+  int Tn = (int)W1.size(0), W = (int)W2.size(0);
+  TORCH_CHECK(C <= 4 && Tn <= 32 && W <= 24, "lift_head: unsupported dims");
+
+  auto gx = at::empty_like(x);
+  auto gW1 = at::zeros({Tn, 1}, x.options());
+  auto gb1 = at::zeros({Tn}, x.options());
+  auto gW2 = at::zeros({W, C}, x.options());
+  auto gb2 = at::zeros({W}, x.options());
+  if (x.numel() == 0) return {gx, gW1, gb1, gW2, gb2};
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  int grid = grid_for_l((long)B * S);
+  AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "lift_head_bwd", [&] {
+    hipLaunchKernelGGL((lift_head_bwd_kernel<scalar_t, 4, 32, 24>), dim3(grid),
+                       dim3(kBlock), 0, stream, gy.data_ptr<scalar_t>(),
+                       x.data_ptr<scalar_t>(), W1.data_ptr<scalar_t>(),
+                       b1.data_ptr<scalar_t>(), W2.data_ptr<scalar_t>(),
+                       b2.data_ptr<scalar_t>(), gx.data_ptr<scalar_t>(),
+                       gW1.data_ptr<scalar_t>(), gb1.data_ptr<scalar_t>(),
+                       gW2.data_ptr<scalar_t>(), gb2.data_ptr<scalar_t>(),
+                       B, C, W, Tn, S);
+  });
+  return {gx, gW1, gb1, gW2, gb2};
+}

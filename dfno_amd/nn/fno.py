@@ -73,16 +73,37 @@ class DistributedFNONd(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         self.dt_comm = 0.0
 
-        x = self.linear1(x, activation="gelu")
-        self.dt_comm += self.linear1.dt_comm
-        x = self.linear2(x, activation="gelu")
-        self.dt_comm += self.linear2.dt_comm
+        x = self._lift(x)
 
         for block in self.blocks:
             x = block(x)
             self.dt_comm += block.dt_comm
 
         x = self._projection(x)
+        return x
+
+    def _lift(self, x: torch.Tensor) -> torch.Tensor:
+        """time lift T_in -> T_out then channel lift C_in -> width, each
+        with GELU.  One fused kernel on GPU when T_in == 1 (ops.lift_head);
+        the composed path otherwise."""
+        from ..ops import lift_head, lift_head_supported
+        import time as _time
+
+        l1, l2 = self.linear1, self.linear2
+        if lift_head_supported(x, l1.in_features, l1.out_features,
+                               l2.in_features, l2.out_features):
+            t0 = _time.time()
+            W1 = l1.W_bcast(l1.W)
+            b1 = l1.b_bcast(l1.b)
+            W2 = l2.W_bcast(l2.W)
+            b2 = l2.b_bcast(l2.b)
+            self.dt_comm += _time.time() - t0
+            return lift_head(x, W1, b1, W2, b2)
+
+        x = l1(x, activation="gelu")
+        self.dt_comm += l1.dt_comm
+        x = l2(x, activation="gelu")
+        self.dt_comm += l2.dt_comm
         return x
 
     def _projection(self, x: torch.Tensor) -> torch.Tensor:

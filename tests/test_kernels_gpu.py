@@ -340,3 +340,38 @@ def test_fused_adam_matches_torch():
     for a, b in zip(p1, p2):
         assert torch.allclose(a.detach(), b.detach(), rtol=1e-5, atol=1e-6), \
             f"max {(a.detach()-b.detach()).abs().max()}"
+
+
+# ---------------------------------------------------------------------------
+# fused lift head
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("dtype,tt", [(torch.float32, 3e-4), (torch.float64, 1e-10)])
+def test_lift_head(dtype, tt):
+    from dfno_amd.ops import lift_head
+    torch.manual_seed(21)
+    B, C, S, Tn, W = 1, 2, 5000, 30, 20
+    x = torch.randn(B, C, S, 1, device="cuda", dtype=dtype, requires_grad=True)
+    W1 = torch.randn(Tn, 1, device="cuda", dtype=dtype).requires_grad_(True)
+    b1 = torch.randn(Tn, device="cuda", dtype=dtype).requires_grad_(True)
+    W2 = (torch.randn(W, C, device="cuda", dtype=dtype) / C).requires_grad_(True)
+    b2 = torch.randn(W, device="cuda", dtype=dtype).requires_grad_(True)
+
+    y = lift_head(x, W1, b1, W2, b2)
+    assert y.shape == (B, W, S, Tn)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    xr = x.detach().clone().requires_grad_(True)
+    W1r = W1.detach().clone().requires_grad_(True)
+    b1r = b1.detach().clone().requires_grad_(True)
+    W2r = W2.detach().clone().requires_grad_(True)
+    b2r = b2.detach().clone().requires_grad_(True)
+    h = F.gelu(torch.einsum("ti,bcsi->bcst", W1r, xr) + b1r.view(1, 1, 1, -1))
+    yr = F.gelu(torch.einsum("wc,bcst->bwst", W2r, h) + b2r.view(1, -1, 1, 1))
+    yr.backward(gy)
+
+    assert torch.allclose(y, yr, rtol=tt, atol=tt), f"fwd {(y-yr).abs().max()}"
+    for a, b in [(x, xr), (W1, W1r), (b1, b1r), (W2, W2r), (b2, b2r)]:
+        assert torch.allclose(a.grad, b.grad, rtol=tt * 30, atol=tt * 30), \
+            f"grad max {(a.grad-b.grad).abs().max()}"
