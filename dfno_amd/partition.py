@@ -180,7 +180,11 @@ class Partition:
         ``_comm.allreduce`` at /root/reference/training/two_phase/sleipner_dataset.py:93-96)."""
         if not (is_distributed() and self.active and self.size > 1):
             return float(value)
-        t = torch.tensor([float(value)], dtype=torch.float64)
+        # RCCL ("nccl") only reduces device tensors
+        backend = dist.get_backend(self.group)
+        dev = torch.device("cuda", torch.cuda.current_device()) \
+            if backend == "nccl" else torch.device("cpu")
+        t = torch.tensor([float(value)], dtype=torch.float64, device=dev)
         red = {"sum": dist.ReduceOp.SUM, "min": dist.ReduceOp.MIN, "max": dist.ReduceOp.MAX}[op]
         dist.all_reduce(t, op=red, group=self.group)
         return float(t.item())

@@ -57,8 +57,8 @@ np.random.seed(P_x.rank + 123)
 
 # root-generated run timestamp broadcast to all ranks (reference :50-52)
 B = dfno.Broadcast(P_0, P_x)
-ts = torch.tensor([float(int(time.time()))]) if P_0.active else \
-    zero_volume_tensor()
+ts = torch.tensor([float(int(time.time()))], device=device) if P_0.active else \
+    zero_volume_tensor(device=device)
 timestamp = int(B(ts).item())
 
 stem = args.input.stem if args.input is not None else "synthetic"
