@@ -19,21 +19,17 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include "kernels.h"
+#include "gelu_math.h"
 
 namespace {
 
 constexpr int kBlock = 256;
 
 template <typename T>
-__device__ __forceinline__ T gelu_(T z) {
-  return T(0.5) * z * (T(1.0) + erf(z * T(0.7071067811865476)));
-}
+__device__ __forceinline__ T gelu_(T z) { return dfno_gelu::gelu(z); }
 
 template <typename T>
-__device__ __forceinline__ T gelu_g_(T z) {
-  return T(0.5) * (T(1.0) + erf(z * T(0.7071067811865476))) +
-         z * exp(T(-0.5) * z * z) * T(0.3989422804014327);
-}
+__device__ __forceinline__ T gelu_g_(T z) { return dfno_gelu::gelu_grad(z); }
 
 template <typename T>
 __device__ __forceinline__ T lh_wave_sum(T v) {

@@ -21,6 +21,7 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include "kernels.h"
+#include "gelu_math.h"
 
 #define DFNO_CHECK(x, msg) TORCH_CHECK(x, msg)
 
@@ -29,17 +30,10 @@ namespace {
 constexpr int kBlock = 256;
 
 template <typename T>
-__device__ __forceinline__ T gelu_erf(T z) {
-  // exact gelu: 0.5 z (1 + erf(z / sqrt(2)))
-  return T(0.5) * z * (T(1.0) + erf(z * T(0.7071067811865476)));
-}
+__device__ __forceinline__ T gelu_erf(T z) { return dfno_gelu::gelu(z); }
 
 template <typename T>
-__device__ __forceinline__ T gelu_grad_erf(T z) {
-  // d/dz: 0.5(1+erf(z/sqrt2)) + z * exp(-z^2/2) / sqrt(2 pi)
-  return T(0.5) * (T(1.0) + erf(z * T(0.7071067811865476))) +
-         z * exp(T(-0.5) * z * z) * T(0.3989422804014327);
-}
+__device__ __forceinline__ T gelu_grad_erf(T z) { return dfno_gelu::gelu_grad(z); }
 
 // ---------------------------------------------------------------------------
 // elementwise gelu / add+gelu (vec4, grid-stride)

@@ -20,21 +20,17 @@
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include "kernels.h"
+#include "gelu_math.h"
 
 namespace {
 
 constexpr int kBlock = 256;
 
 template <typename T>
-__device__ __forceinline__ T gelu_erf_(T z) {
-  return T(0.5) * z * (T(1.0) + erf(z * T(0.7071067811865476)));
-}
+__device__ __forceinline__ T gelu_erf_(T z) { return dfno_gelu::gelu(z); }
 
 template <typename T>
-__device__ __forceinline__ T gelu_grad_erf_(T z) {
-  return T(0.5) * (T(1.0) + erf(z * T(0.7071067811865476))) +
-         z * exp(T(-0.5) * z * z) * T(0.3989422804014327);
-}
+__device__ __forceinline__ T gelu_grad_erf_(T z) { return dfno_gelu::gelu_grad(z); }
 
 // butterfly sum over the 64-lane wave
 template <typename T>
@@ -249,10 +245,8 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
       }
       T gk[VEC], dgk[VEC];
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) {
-        gk[k] = gelu_erf_(zk[k]);
-        dgk[k] = gelu_grad_erf_(zk[k]);
-      }
+      for (int k = 0; k < VEC; ++k)
+        dfno_gelu::gelu_and_grad(zk[k], gk[k], dgk[k]);
       // gz3_j = (sum_o W4[o,j] gy[o]) * gelu'(z3_j)
       T gzk[VEC];
 #pragma unroll
