@@ -27,7 +27,7 @@ __device__ __forceinline__ float erf_from_E(float ax, float E) {
 // gelu(z) = 0.5 z (1 + erf(z/sqrt2))
 __device__ __forceinline__ float gelu(float z) {
   float x = z * (float)kInvSqrt2;
-  float E = expf(-x * x);
+  float E = __expf(-x * x);
   float e = erf_from_E(fabsf(x), E);
   e = copysignf(e, x);
   return 0.5f * z * (1.0f + e);
@@ -35,7 +35,7 @@ __device__ __forceinline__ float gelu(float z) {
 
 __device__ __forceinline__ float gelu_grad(float z) {
   float x = z * (float)kInvSqrt2;
-  float E = expf(-x * x);          // == exp(-z^2/2)
+  float E = __expf(-x * x);        // == exp(-z^2/2)
   float e = copysignf(erf_from_E(fabsf(x), E), x);
   return 0.5f * (1.0f + e) + z * E * (float)kInvSqrt2Pi;
 }
@@ -43,7 +43,7 @@ __device__ __forceinline__ float gelu_grad(float z) {
 // both at once, sharing E and the erf tail
 __device__ __forceinline__ void gelu_and_grad(float z, float& g, float& dg) {
   float x = z * (float)kInvSqrt2;
-  float E = expf(-x * x);
+  float E = __expf(-x * x);
   float e = copysignf(erf_from_E(fabsf(x), E), x);
   float phi = 0.5f * (1.0f + e);
   g = z * phi;

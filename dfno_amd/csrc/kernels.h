@@ -44,6 +44,10 @@ std::vector<at::Tensor> lift_head_bwd(const at::Tensor& gy, const at::Tensor& x,
 void adam_step_(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
                 double lr, double beta1, double beta2, double eps,
                 double weight_decay, int64_t step);
+void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+                      std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                      double lr, double beta1, double beta2, double eps,
+                      double weight_decay, std::vector<int64_t> steps);
 
 // fused truncated-spectrum DFTs (see ops/fft.py):
 at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,

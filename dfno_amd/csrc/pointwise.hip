@@ -891,6 +891,18 @@ __global__ void adam_step_kernel(T* __restrict__ p, const T* __restrict__ g,
 
 }  // namespace
 
+void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+                      std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                      double lr, double beta1, double beta2, double eps,
+                      double weight_decay, std::vector<int64_t> steps) {
+  TORCH_CHECK(ps.size() == gs.size() && ps.size() == ms.size() &&
+              ps.size() == vs.size() && ps.size() == steps.size(),
+              "adam batch: length mismatch");
+  for (size_t i = 0; i < ps.size(); ++i)
+    adam_step_(ps[i], gs[i], ms[i], vs[i], lr, beta1, beta2, eps,
+               weight_decay, steps[i]);
+}
+
 void adam_step_(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
                 double lr, double beta1, double beta2, double eps,
                 double weight_decay, int64_t step) {

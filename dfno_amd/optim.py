@@ -36,6 +36,7 @@ class Adam(torch.optim.Adam):
                 fallback_groups.append(group)
                 continue
             slow = []
+            batch = ([], [], [], [], [])  # p, g, m, v, step
             for p in group["params"]:
                 if p.grad is None or p.numel() == 0:
                     continue
@@ -49,17 +50,21 @@ class Adam(torch.optim.Adam):
                     state["step"] = torch.tensor(0.0)
                     state["exp_avg"] = torch.zeros_like(p)
                     state["exp_avg_sq"] = torch.zeros_like(p)
+                if "_flat_pmv" not in state:
+                    flat = lambda t: (torch.view_as_real(t) if t.is_complex() else t).reshape(-1)
+                    state["_flat_pmv"] = (flat(p), flat(state["exp_avg"]),
+                                          flat(state["exp_avg_sq"]))
                 state["step"] += 1
-                pv = torch.view_as_real(p) if p.is_complex() else p
+                pv, mv, vv = state["_flat_pmv"]
                 gv = torch.view_as_real(p.grad) if p.grad.is_complex() else p.grad
-                mv = state["exp_avg"]
-                mv = torch.view_as_real(mv) if mv.is_complex() else mv
-                vv = state["exp_avg_sq"]
-                vv = torch.view_as_real(vv) if vv.is_complex() else vv
-                ext.adam_step_(pv.reshape(-1), gv.reshape(-1).contiguous(),
-                               mv.reshape(-1), vv.reshape(-1),
-                               lr, beta1, beta2, eps, wd,
-                               int(state["step"].item()))
+                batch[0].append(pv)
+                batch[1].append(gv.reshape(-1).contiguous())
+                batch[2].append(mv)
+                batch[3].append(vv)
+                batch[4].append(int(state["step"].item()))
+            if batch[0]:
+                ext.adam_step_batch_(batch[0], batch[1], batch[2], batch[3],
+                                     lr, beta1, beta2, eps, wd, batch[4])
             if slow:
                 fallback_groups.append({**group, "params": slow})
 
