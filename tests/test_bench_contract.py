@@ -1,0 +1,35 @@
+"""The driver depends on bench.py's CLI and JSON contract; validate it."""
+
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def test_bench_json_contract():
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, str(REPO / "bench.py"), "--gpus", "1", "--steps", "1",
+         "--warmup", "0", "--width", "4", "--num-blocks", "1"],
+        capture_output=True, text=True, timeout=900, env=env, cwd=str(REPO))
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = r.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in d, f"missing {key}"
+    assert d["metric"] == "sec/batch"
+    assert d["higher_is_better"] is False
+    assert d["scaling"] == "strong"
+    assert d["n_gpus"] == 1
+    assert d["value"] > 0
+    assert abs(d["ms_per_step"] - d["value"] * 1e3) < 1e-6
+    assert "synthetic" in d["data"]
+    for k in ("model", "global_batch", "parallelism"):
+        assert k in d["config"]

@@ -126,6 +126,9 @@ def test_fno_2d_time_equivalence(pshape, world):
 @pytest.mark.parametrize("pshape,world", [
     ((1, 1, 1, 2, 1, 1), 2),
     ((1, 1, 2, 2, 1, 1), 4),
+    # z-axis partitioned: P_m != P_x, so R1/R4 run REAL all-to-alls of the
+    # activation (the identity fast path does not apply)
+    ((1, 1, 2, 1, 2, 1), 4),
 ])
 def test_fno_3d_time_equivalence(pshape, world):
     run_dist(_fno_equiv_body, world, pshape, [1, 2, 8, 8, 6, 1], 8, 6, (3, 3, 2, 2), 2)
