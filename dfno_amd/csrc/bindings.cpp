@@ -11,6 +11,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "transposed channel contraction (grad-x)");
   m.def("channel_mix_bwd_w", &channel_mix_bwd_w,
         "split-s grad-W (+grad-bias) reduction");
+  m.def("linear_res_gelu_fwd", &linear_res_gelu_fwd,
+        "fused y = gelu(W @ x + res): returns (y, z)");
   m.def("gelu_fwd", &gelu_fwd, "exact gelu");
   m.def("gelu_bwd", &gelu_bwd, "gelu backward (gy, z) -> gz");
   m.def("add_gelu_fwd", &add_gelu_fwd, "fused residual add + gelu: returns (y, z)");

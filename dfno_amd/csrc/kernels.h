@@ -13,6 +13,10 @@ std::vector<at::Tensor> channel_mix_fwd(const at::Tensor& x, const at::Tensor& W
 // transposed contraction for grad-x: gx[b,i,s] = sum_o W[o,i] gz[b,o,s]
 at::Tensor channel_mix_fwd_t(const at::Tensor& gz, const at::Tensor& W);
 
+// fused residual linear epilogue: y = gelu(W @ x + res); returns {y, z}
+std::vector<at::Tensor> linear_res_gelu_fwd(const at::Tensor& x, const at::Tensor& W,
+                                            const at::Tensor& res);
+
 at::Tensor gelu_fwd(const at::Tensor& x);
 at::Tensor gelu_bwd(const at::Tensor& gy, const at::Tensor& z);
 std::vector<at::Tensor> add_gelu_fwd(const at::Tensor& a, const at::Tensor& b);

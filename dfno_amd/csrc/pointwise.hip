@@ -112,7 +112,8 @@ template <typename T, int IMAX, int VEC, bool ACT, bool VECTOR>
 __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
     const T* __restrict__ x, const T* __restrict__ W, const T* __restrict__ bias,
     T* __restrict__ y, T* __restrict__ z,
-    int B, int I, int O, long S, bool wt, bool has_bias, bool write_z) {
+    int B, int I, int O, long S, bool wt, bool has_bias, bool write_z,
+    const T* __restrict__ res = nullptr) {
   extern __shared__ __align__(16) char smem_raw[];
   T* Wl = reinterpret_cast<T*>(smem_raw);        // [O*I]
   T* bl = Wl + (size_t)O * I;                     // [O]
@@ -168,6 +169,11 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
               }
       }
       if constexpr (VECTOR && std::is_same<T, float>::value) {
+        if (res != nullptr) {
+          const float4 rv = *reinterpret_cast<const float4*>(
+              res + ((long)b * O) * S + (long)o * S + s);
+          acc[0] += rv.x; acc[1] += rv.y; acc[2] += rv.z; acc[3] += rv.w;
+        }
         if (write_z)
           *reinterpret_cast<float4*>(zb + (long)o * S) =
               make_float4(acc[0], acc[1], acc[2], acc[3]);
@@ -183,6 +189,7 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
 #pragma unroll
         for (int k = 0; k < VEC; ++k) {
           if (!full && k >= nv) break;
+          if (res != nullptr) acc[k] += res[((long)b * O + o) * S + s + k];
           if (write_z) zb[(long)o * S + k] = acc[k];
           yb[(long)o * S + k] = ACT ? gelu_erf(acc[k]) : acc[k];
         }
@@ -349,7 +356,8 @@ bool can_vectorize(const T* x, const T* y, const T* z, long S, bool write_z) {
 template <typename T>
 void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
                         int B, int I, int O, long S, bool wt, bool has_bias,
-                        bool act, bool write_z, hipStream_t stream) {
+                        bool act, bool write_z, hipStream_t stream,
+                        const T* res = nullptr) {
   // effective input count for residency decisions
   long nchunk_work = (long)B * ((S + 3) / 4);
   int grid = grid_for(nchunk_work, kBlock);
@@ -360,6 +368,10 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
   hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V>), dim3(grid), dim3(kBlock),       \
                      smem, stream, x, W, bias, y, z, B, I, O, S, wt,            \
                      has_bias, write_z);
+#define CMIX_LAUNCH_RES(KERNEL, CAP, A, V)                                      \
+  hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V>), dim3(grid), dim3(kBlock),       \
+                     smem, stream, x, W, bias, y, z, B, I, O, S, wt,            \
+                     has_bias, write_z, res);
 #define CMIX_DISPATCH(KERNEL, CAP)                                              \
   if (act) {                                                                    \
     if (vec) { CMIX_LAUNCH(KERNEL, CAP, true, true) }                           \
@@ -369,6 +381,22 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
     else { CMIX_LAUNCH(KERNEL, CAP, false, false) }                             \
   }
 
+#define CMIX_DISPATCH_RES(KERNEL, CAP)                                          \
+  if (act) {                                                                    \
+    if (vec) { CMIX_LAUNCH_RES(KERNEL, CAP, true, true) }                       \
+    else { CMIX_LAUNCH_RES(KERNEL, CAP, true, false) }                          \
+  } else {                                                                      \
+    if (vec) { CMIX_LAUNCH_RES(KERNEL, CAP, false, true) }                      \
+    else { CMIX_LAUNCH_RES(KERNEL, CAP, false, false) }                         \
+  }
+  if (res != nullptr) {
+    TORCH_CHECK(I <= 32, "channel_mix with residual needs I <= 32");
+    if (I <= 8) { CMIX_DISPATCH_RES(channel_mix_xres_kernel, 8) }
+    else if (I <= 16) { CMIX_DISPATCH_RES(channel_mix_xres_kernel, 16) }
+    else if (I <= 24) { CMIX_DISPATCH_RES(channel_mix_xres_kernel, 24) }
+    else { CMIX_DISPATCH_RES(channel_mix_xres_kernel, 32) }
+    return;
+  }
   if (I <= 8) { CMIX_DISPATCH(channel_mix_xres_kernel, 8) }
   else if (I <= 16) { CMIX_DISPATCH(channel_mix_xres_kernel, 16) }
   else if (I <= 24) { CMIX_DISPATCH(channel_mix_xres_kernel, 24) }
@@ -407,7 +435,10 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
     }
 #undef CMIX_LDS
   }
+#undef CMIX_DISPATCH_RES
 #undef CMIX_DISPATCH
+#undef CMIX_LAUNCH_RES
+#undef CMIX_LAUNCH
 }
 
 }  // namespace
@@ -470,6 +501,31 @@ at::Tensor channel_mix_fwd_t(const at::Tensor& gz, const at::Tensor& W) {
         /*write_z=*/false, stream);
   });
   return gx;
+}
+
+std::vector<at::Tensor> linear_res_gelu_fwd(const at::Tensor& x, const at::Tensor& W,
+                                            const at::Tensor& res) {
+  check_f(x, "x"); check_f(W, "W"); check_f(res, "res");
+  TORCH_CHECK(x.dim() == 3 && res.dim() == 3, "x/res must be [B,*,S]");
+  int B = (int)x.size(0), I = (int)x.size(1);
+  long S = x.size(2);
+  int O = (int)W.size(0);
+  TORCH_CHECK((int)W.size(1) == I, "W/I mismatch");
+  TORCH_CHECK(res.size(0) == B && (int)res.size(1) == O && res.size(2) == S,
+              "res shape mismatch");
+
+  auto y = at::empty({B, O, S}, x.options());
+  auto z = at::empty({B, O, S}, x.options());
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  if (x.numel() == 0) return {y, z};
+  AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "linear_res_gelu_fwd", [&] {
+    launch_channel_mix<scalar_t>(
+        x.data_ptr<scalar_t>(), W.data_ptr<scalar_t>(), nullptr,
+        y.data_ptr<scalar_t>(), z.data_ptr<scalar_t>(),
+        B, I, O, S, /*wt=*/false, /*has_bias=*/false, /*act=*/true,
+        /*write_z=*/true, stream, res.data_ptr<scalar_t>());
+  });
+  return {y, z};
 }
 
 at::Tensor gelu_fwd(const at::Tensor& x) {

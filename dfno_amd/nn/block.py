@@ -30,7 +30,7 @@ import torch.nn as nn
 
 from ..comm import Repartition
 from ..partition import Partition, compute_distribution_info
-from ..ops import spectral_conv, add_gelu, rfft_trunc, fft_trunc, pad_ifft, pad_irfft
+from ..ops import spectral_conv, add_gelu, linear_res_gelu, rfft_trunc, fft_trunc, pad_ifft, pad_irfft
 from .linear import BroadcastedLinear
 
 __all__ = ["DistributedFNOBlock"]
@@ -198,7 +198,16 @@ class DistributedFNOBlock(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         self.dt_comm = 0.0
 
-        y0 = self.linear(x)
+        # The residual pass-through linear (reference computes y0 = linear(x)
+        # up front, dfno.py:244) is deferred and fused into the epilogue as
+        # gelu(W x_in + y) — no y0 tensor is materialized.  Its weight is
+        # broadcast here so the collective order matches the reference.
+        x_in = x
+        t0 = time.time()
+        W_res = self.linear.W_bcast(self.linear.W)
+        b_res = self.linear.b_bcast(self.linear.b)  # unused (bias=False); keeps
+        self.linear.dt_comm = time.time() - t0      # broadcast parity
+        del b_res
 
         t0 = time.time()
         x = self.R1(x)
@@ -254,4 +263,4 @@ class DistributedFNOBlock(nn.Module):
         y = self.R4(y)
         self.dt_comm += time.time() - t0
 
-        return add_gelu(y0, y)
+        return linear_res_gelu(x_in, W_res, y)

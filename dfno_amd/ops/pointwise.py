@@ -27,7 +27,7 @@ import torch.nn.functional as F
 
 from .. import _ext
 
-__all__ = ["linear_nd", "add_gelu", "gelu"]
+__all__ = ["linear_nd", "add_gelu", "gelu", "linear_res_gelu"]
 
 _SQRT_2 = math.sqrt(2.0)
 _INV_SQRT_2PI = 1.0 / math.sqrt(2.0 * math.pi)
@@ -135,6 +135,58 @@ def linear_nd(x: torch.Tensor, W: torch.Tensor, b: Optional[torch.Tensor],
     out_shape = list(xm.shape)
     out_shape[1] = W.shape[0]
     return ym.reshape(out_shape).movedim(1, d).contiguous()
+
+
+# ---------------------------------------------------------------------------
+# fused residual-linear epilogue: gelu(W @ x + res)  (block epilogue with the
+# pass-through linear folded in — the reference computes y0 = linear(x) at
+# block start and gelu(y0 + y) at the end, dfno.py:244,291; fusing removes
+# the y0 materialization entirely)
+# ---------------------------------------------------------------------------
+
+class _LinearResGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, W, res):
+        B, I = x.shape[0], x.shape[1]
+        S = x.numel() // max(B * I, 1)
+        x3 = x.reshape(B, I, S)
+        r3 = res.reshape(B, W.shape[0], S)
+        if x.is_cuda:
+            ext = _ext.get(required=True)
+            y3, z3 = ext.linear_res_gelu_fwd(x3.contiguous(), W, r3.contiguous())
+        else:
+            z3 = torch.einsum("oi,bis->bos", W, x3) + r3
+            y3 = F.gelu(z3)
+        ctx.save_for_backward(x3, W, z3)
+        ctx.x_shape = tuple(x.shape)
+        ctx.res_shape = tuple(res.shape)
+        return y3
+
+    @staticmethod
+    def backward(ctx, gy):
+        x3, W, z3 = ctx.saved_tensors
+        gy = gy.contiguous()
+        if gy.is_cuda:
+            ext = _ext.get(required=True)
+            gz = ext.gelu_bwd(gy, z3)
+            gx = ext.channel_mix_fwd_t(gz, W)
+            if x3.shape[1] <= 32:
+                gW, _ = ext.channel_mix_bwd_w(gz, x3.contiguous(), False)
+            else:
+                gW = torch.einsum("bos,bis->oi", gz, x3)
+        else:
+            gz = gy * _gelu_grad(z3)
+            gx = torch.einsum("oi,bos->bis", W, gz)
+            gW = torch.einsum("bos,bis->oi", gz, x3)
+        return gx.reshape(ctx.x_shape), gW, gz.reshape(ctx.res_shape)
+
+
+def linear_res_gelu(x: torch.Tensor, W: torch.Tensor, res: torch.Tensor) -> torch.Tensor:
+    """gelu(W @_channel x + res): the block's residual path in one pass."""
+    out = _LinearResGeluFn.apply(x, W, res)
+    shape = list(x.shape)
+    shape[1] = W.shape[0]
+    return out.reshape(shape)
 
 
 # ---------------------------------------------------------------------------
