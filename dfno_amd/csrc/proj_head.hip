@@ -363,7 +363,7 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
                        b4.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),     \
                        B, I, M, O2, S);
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "proj_head_fwd", [&] {
-    size_t smem = sizeof(scalar_t) * ((size_t)M * I + M + (size_t)O2 * M + O2);
+    size_t smem = 0;
     bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), out.data_ptr()});
     if (I <= 24 && O2 <= 2) {
       constexpr int IM = 24, OM = 2;
