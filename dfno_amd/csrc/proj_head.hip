@@ -42,19 +42,12 @@ __device__ __forceinline__ T wave_sum(T v) {
 
 template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR>
 __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
-    const T* __restrict__ x, const T* __restrict__ W3, const T* __restrict__ b3,
-    const T* __restrict__ W4, const T* __restrict__ b4, T* __restrict__ out,
+    const T* __restrict__ x, const T* __restrict__ W3l, const T* __restrict__ b3l,
+    const T* __restrict__ W4l, const T* __restrict__ b4l, T* __restrict__ out,
     int B, int I, int M, int O2, long S) {
-  extern __shared__ __align__(16) char smem_raw[];
-  T* W3l = reinterpret_cast<T*>(smem_raw);   // [M*I]
-  T* b3l = W3l + (size_t)M * I;              // [M]
-  T* W4l = b3l + M;                          // [O2*M]
-  T* b4l = W4l + (size_t)O2 * M;             // [O2]
-  for (int k = threadIdx.x; k < M * I; k += blockDim.x) W3l[k] = W3[k];
-  for (int k = threadIdx.x; k < M; k += blockDim.x) b3l[k] = b3[k];
-  for (int k = threadIdx.x; k < O2 * M; k += blockDim.x) W4l[k] = W4[k];
-  for (int k = threadIdx.x; k < O2; k += blockDim.x) b4l[k] = b4[k];
-  __syncthreads();
+  // weights are read straight through the kernel-arg pointers: every access
+  // index is wave-uniform, so they lower to scalar loads (s-cache) instead
+  // of per-MAC LDS reads that double the issue count.
 
   long nchunks = (S + VEC - 1) / VEC;
   long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
