@@ -161,3 +161,20 @@ def test_gen_scripts(tmp_path):
         assert (tmp_path / n).exists()
         body = (tmp_path / n).read_text()
         assert "torch.distributed.run" in body
+
+
+def test_bench_harness_2rank(tmp_path):
+    # the reference-protocol harness under torchrun (gloo on CPU), the same
+    # launch pattern the driver uses for the multi-GPU scaling runs
+    r = run_script([str(REPO / "benchmarks/bench.py"),
+                    "--input-shape", "1", "1", "8", "8", "3",
+                    "--partition_shape", "1", "1", "2", "1", "1",
+                    "--modes", "2", "2", "2", "--num-timesteps", "4",
+                    "--width", "4", "--device", "cpu",
+                    "--benchmark-type", "grad",
+                    "--output-dir", str(tmp_path)], cwd=str(tmp_path), nproc=2)
+    outs = sorted(tmp_path.glob("*-grad-*-2.json"))
+    assert len(outs) == 2, [p.name for p in outs]
+    for o in outs:
+        d = json.loads(o.read_text())
+        assert "dt" in d and "dt_grad" in d
