@@ -702,22 +702,30 @@ __global__ __launch_bounds__(kBlock) void gw_outer_kernel(
 }
 
 
-// LDS-staged variant: the B tile (the I<=32 narrow operand) is staged into
-// a double-buffered LDS ring by global_load_lds DMA (no VGPR round trip, no
-// spill) and shared by the block's waves/OW rows; the next tile's DMA is
-// issued before computing the current one, so HBM latency hides under the
-// fma phase.  __syncthreads() drains the outstanding DMA (hipcc emits
-// vmcnt(0) in the barrier when glds is in flight) and doubles as the ring
-// hand-off.  Without the staging, co-resident blocks evict the shared rows
-// from L1/L2 (~4x algorithmic traffic); without the pipelining the
-// load->write pairs serialize (SQ_WAIT_ANY 76%).
+// All-glds pipelined variant.  Both operand tiles (the I<=32 narrow B rows
+// AND this block's 4*OW A rows) are staged into a 3-deep LDS ring by
+// global_load_lds DMA; the loop keeps exactly one tile's DMA in flight and
+// waits for the current tile with a COUNTED s_waitcnt vmcnt(NG) + raw
+// s_barrier.  The previous version kept A in registers via ordinary loads:
+// hipcc then inserted s_waitcnt vmcnt(0) before every A use (76 of them in
+// the unrolled body), draining the in-flight next-tile DMA each step and
+// serializing the ring at ~1.8 TB/s.  Counted vmcnt needs a compile-time
+// exact per-wave instruction count, so the issue loops are padded to fixed
+// trip counts (NB = ICAP/4 for B, OW for A) with out-of-range rows clamped
+// to a valid source row; compute masks those lanes out as before.  3
+// buffers (not 2) so a tile's DMA lands in a buffer whose last readers are
+// two barriers back.  LDS = 3*(ICAP+4*OW)*1KB -> 1-2 blocks/CU; the low
+// occupancy is by design, latency hides in the pipeline depth.
 template <typename T, int ICAP, int OW>
-__global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
+__global__ __launch_bounds__(kBlock) void gw_outer_glds3_kernel(
     const T* __restrict__ A, const T* __restrict__ Bm,
     T* __restrict__ gW, T* __restrict__ gb,
     int B, int O, int I, long S, int n_schunk, bool want_bias) {
-  constexpr int TS = 256;  // floats per s-tile
-  __shared__ float btile[2][ICAP * TS];
+  constexpr int TS = 256;           // floats per s-tile row
+  constexpr int NB = ICAP / 4;      // B-tile glds instrs per wave (padded)
+  constexpr int NG = NB + OW;       // total glds instrs per wave per tile
+  constexpr int AROWS = 4 * OW;     // A rows staged per block
+  __shared__ float ring[3][(ICAP + AROWS) * TS];
 
   const int o_tiles = (O + 4 * OW - 1) / (4 * OW);
   const int schunk = blockIdx.x / o_tiles;
@@ -725,6 +733,7 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
   const int o0 = (o_tile * 4 + wave) * OW;
+  const int ob0 = o_tile * 4 * OW;  // first A row of the whole block
 
   long chunk_sz = ((S + n_schunk - 1) / n_schunk + (TS - 1)) & ~((long)TS - 1);
   long s0 = (long)schunk * chunk_sz;
@@ -740,38 +749,61 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
 #pragma unroll
   for (int w = 0; w < OW; ++w) bacc[w] = 0.f;
 
-  const int tot_slots = I * (TS / 4);   // float4 slots per tile
-
   for (int b = 0; b < B; ++b) {
     const float* Ab = reinterpret_cast<const float*>(A) + ((long)b * O + o0) * S;
+    const float* Abl = reinterpret_cast<const float*>(A) + ((long)b * O) * S;
     const float* Bb = reinterpret_cast<const float*>(Bm) + ((long)b * I) * S;
 
-    // one tile's DMA: wave-cooperative, lane-linear LDS image
+    // one tile's DMA: exactly NG glds instructions per wave, every wave.
     auto issue_tile = [&](int buf, long st) {
-      for (int sb = wave * 64; sb < tot_slots; sb += kBlock) {
-        int slot = sb + lane;
+      float* dst = ring[buf];
+#pragma unroll
+      for (int k = 0; k < NB; ++k) {            // B rows, lane-linear image
+        int slot = wave * 64 + k * kBlock + lane;
         int fo = slot * 4;
         int row = fo / TS, col = fo % TS;
-        const float* gsrc = Bb + (long)row * S + st + col;
+        int srow = row < I ? row : I - 1;       // pad: clamp to a valid row
         __builtin_amdgcn_global_load_lds(
-            (const __attribute__((address_space(1))) void*)gsrc,
-            (__attribute__((address_space(3))) void*)&btile[buf][sb * 4],
+            (const __attribute__((address_space(1))) void*)
+                (Bb + (long)srow * S + st + col),
+            (__attribute__((address_space(3))) void*)&dst[slot * 4],
+            16, 0, 0);
+      }
+#pragma unroll
+      for (int w = 0; w < OW; ++w) {            // this wave's own A rows
+        int row = ob0 + wave * OW + w;
+        int srow = row < O ? row : O - 1;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)
+                (Abl + (long)srow * S + st + lane * 4),
+            (__attribute__((address_space(3))) void*)
+                &dst[(ICAP + wave * OW + w) * TS + lane * 4],
             16, 0, 0);
       }
     };
 
     if (s0 < full_end) {
+      const long nt = (full_end - s0) / TS;
       issue_tile(0, s0);
-      __syncthreads();   // drain tile-0 DMA
-      int cur = 0;
-      for (long st = s0; st < full_end; st += TS, cur ^= 1) {
-        if (st + TS < full_end) issue_tile(cur ^ 1, st + TS);
+      for (long t = 0; t < nt; ++t) {
+        if (t + 1 < nt) {
+          issue_tile((int)((t + 1) % 3), s0 + (t + 1) * TS);
+          asm volatile("s_waitcnt vmcnt(%0)" ::"n"(NG) : "memory");
+        } else {
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        }
+        // raw barrier (no implicit vmcnt drain), fenced so the compiler
+        // cannot hoist the LDS reads above it
+        __builtin_amdgcn_s_barrier();
+        asm volatile("" ::: "memory");
+        const float* cur = ring[t % 3];
         if (o0 < O) {
           float4 av[OW];
 #pragma unroll
           for (int w = 0; w < OW; ++w) {
             if (o0 + w < O) {
-              av[w] = *reinterpret_cast<const float4*>(Ab + (long)w * S + st + lane * 4);
+              av[w] = *reinterpret_cast<const float4*>(
+                  &cur[(ICAP + wave * OW + w) * TS + lane * 4]);
               if (want_bias) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
             }
           }
@@ -779,7 +811,7 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
           for (int i = 0; i < ICAP; ++i) {
             if (i < I) {
               const float4 bv =
-                  *reinterpret_cast<const float4*>(&btile[cur][i * TS + lane * 4]);
+                  *reinterpret_cast<const float4*>(&cur[i * TS + lane * 4]);
 #pragma unroll
               for (int w = 0; w < OW; ++w) {
                 if (o0 + w < O)
@@ -789,8 +821,8 @@ __global__ __launch_bounds__(kBlock) void gw_outer_lds_kernel(
             }
           }
         }
-        __syncthreads();  // drain next-tile DMA + ring hand-off
       }
+      __syncthreads();  // next b (or tail) may overwrite ring buffers
     }
 
     // ragged tail of the last (partial) tile, direct from global
@@ -851,10 +883,31 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
   bool vec = is_f32 && (S % 4 == 0) &&
              ((reinterpret_cast<uintptr_t>(gz.data_ptr()) & 15) == 0) &&
              ((reinterpret_cast<uintptr_t>(x.data_ptr()) & 15) == 0);
-  // rows per wave: wider cuts B re-reads but costs registers/occupancy
-  const int OW = (O >= 8) ? ((I <= 8) ? 4 : 2) : 1;
+  static const bool no_glds = []() {
+    const char* e = getenv("DFNO_GW_NO_GLDS");  // A/B knob: force generic path
+    return e && e[0] == '1';
+  }();
+  // rows per wave: wider cuts B re-reads but costs registers/occupancy.
+  // The glds3 OW=5 variant covers O<=20 in ONE o-tile so B is read once.
+  const bool ow5 = vec && !no_glds && I > 8 && I <= 20 && O <= 20 && O >= 8;
+  const int OW = ow5 ? 5 : ((O >= 8) ? ((I <= 8) ? 4 : 2) : 1);
   int o_tiles = (O + 4 * OW - 1) / (4 * OW);
-  int n_schunk = (int)std::max(1L, std::min((long)(4096 / o_tiles), S / (64 * 16)));
+  static const long ns_cap = []() {
+    const char* e = getenv("DFNO_GW_NSCHUNK");  // sweep knob
+    return e ? atol(e) : 0L;
+  }();
+  // Grid sizing: every block atomically flushes its whole O*I accumulator,
+  // so the flush cost scales with grid and serializes per cache line
+  // (measured: 4096 s-chunks -> 1.84 ms, 256 -> 0.52 ms on the 20x20
+  // S=7.9M flagship shape).  The glds3 kernels are 1-2 blocks/CU by LDS,
+  // so one grid-wave exactly fills the 256 CUs; the generic kernel gets a
+  // few more for latency-hiding.
+  const bool use_glds = vec && !no_glds && O >= 8;
+  int bpc = use_glds ? ((!ow5 && I <= 8) ? 2 : 1) : 4;
+  int n_schunk = (int)std::max(
+      1L, std::min((long)(256 * bpc / o_tiles), S / (64 * 16)));
+  if (ns_cap > 0)
+    n_schunk = (int)std::max(1L, std::min(ns_cap, S / (64 * 16)));
   int grid = n_schunk * o_tiles;
 
 #define GW_LAUNCH(ICAP_, OW_, BIAS_, V)                                         \
@@ -873,15 +926,16 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
       else { GW_LAUNCH(ICAP_, OW_, false, false) }                              \
     }
 #define GW_LDS(ICAP_, OW_)                                                      \
-      hipLaunchKernelGGL((gw_outer_lds_kernel<scalar_t, ICAP_, OW_>),           \
+      hipLaunchKernelGGL((gw_outer_glds3_kernel<scalar_t, ICAP_, OW_>),         \
                          dim3(grid), dim3(kBlock), 0, stream,                   \
                          gz.data_ptr<scalar_t>(),                               \
                          x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),       \
                          want_bias ? gb.data_ptr<scalar_t>() : nullptr,         \
                          B, O, I, S, n_schunk, want_bias);
   AT_DISPATCH_FLOATING_TYPES(gz.scalar_type(), "channel_mix_bwd_w", [&] {
-    if (vec && OW == 4) { GW_LDS(8, 4) }
-    else if (vec && OW == 2) {
+    if (ow5) { GW_LDS(20, 5) }
+    else if (vec && !no_glds && OW == 4) { GW_LDS(8, 4) }
+    else if (vec && !no_glds && OW == 2) {
       if (I <= 24) { GW_LDS(24, 2) } else { GW_LDS(32, 2) }
     } else if (OW == 4) { GW_DISPATCH2(8, 4) }
     else if (OW == 2) {
@@ -891,6 +945,11 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
 #undef GW_LDS
 #undef GW_DISPATCH2
 #undef GW_LAUNCH
+  // the glds3 kernels carry 72-120 KB static LDS: surface a refused launch
+  // loudly instead of silently returning the zero-initialized gW
+  hipError_t lerr = hipGetLastError();
+  TORCH_CHECK(lerr == hipSuccess, "channel_mix_bwd_w launch failed: ",
+              hipGetErrorString(lerr));
   return {gW, gb};
 }
 
