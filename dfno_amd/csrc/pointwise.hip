@@ -716,20 +716,33 @@ __global__ __launch_bounds__(kBlock) void gw_outer_kernel(
 // buffers (not 2) so a tile's DMA lands in a buffer whose last readers are
 // two barriers back.  LDS = 3*(ICAP+4*OW)*1KB -> 1-2 blocks/CU; the low
 // occupancy is by design, latency hides in the pipeline depth.
-template <typename T, int ICAP, int OW>
+template <typename T, int ICAP, int OW, int TS = 256, int NBUF = 3>
 __global__ __launch_bounds__(kBlock) void gw_outer_glds3_kernel(
     const T* __restrict__ A, const T* __restrict__ Bm,
     T* __restrict__ gW, T* __restrict__ gb,
     int B, int O, int I, long S, int n_schunk, bool want_bias) {
-  constexpr int TS = 256;           // floats per s-tile row
-  constexpr int NB = ICAP / 4;      // B-tile glds instrs per wave (padded)
-  constexpr int NG = NB + OW;       // total glds instrs per wave per tile
+  // TS = floats per s-tile row: bigger rows read longer contiguous HBM
+  // bursts per stream (TS=512 -> 2 KB sequential per row per tile).
+  static_assert((ICAP * TS) % (kBlock * 4) == 0, "B image must tile evenly");
+  static_assert((OW * TS) % 256 == 0, "A image must tile evenly");
+  constexpr int NB = ICAP * TS / (kBlock * 4);  // B glds instrs per wave
+  constexpr int NA = OW * TS / 256;             // A glds instrs per wave
+  constexpr int NG = NB + NA;                   // per wave per tile
   constexpr int AROWS = 4 * OW;     // A rows staged per block
-  __shared__ float ring[3][(ICAP + AROWS) * TS];
+  constexpr int PASS = TS >= 256 ? TS / 256 : 1;  // compute passes per tile
+  __shared__ float ring[NBUF][(ICAP + AROWS) * TS];
 
   const int o_tiles = (O + 4 * OW - 1) / (4 * OW);
-  const int schunk = blockIdx.x / o_tiles;
-  const int o_tile = blockIdx.x % o_tiles;
+  // XCD-aware decode: workgroups are dealt round-robin to the 8 XCDs, each
+  // with its own L2.  All o-tiles of one s-chunk share the B rows, so place
+  // them on ONE XCD (ids congruent mod 8): the 2nd..Nth o-tile then reads B
+  // from that XCD's L2 instead of re-fetching HBM (gW3's O=128 shape was 16x
+  // amplified).  Identity when o_tiles == 1; host rounds n_schunk up to a
+  // multiple of 8 when o_tiles > 1 so the decode stays a bijection.
+  const int xr = blockIdx.x % 8;
+  const int xq = blockIdx.x / 8;
+  const int o_tile = xq % o_tiles;
+  const int schunk = (xq / o_tiles) * 8 + xr;
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x % 64;
   const int o0 = (o_tile * 4 + wave) * OW;
@@ -770,14 +783,16 @@ __global__ __launch_bounds__(kBlock) void gw_outer_glds3_kernel(
             16, 0, 0);
       }
 #pragma unroll
-      for (int w = 0; w < OW; ++w) {            // this wave's own A rows
+      for (int k = 0; k < NA; ++k) {            // this wave's own A rows
+        int fa = k * 256 + lane * 4;            // flat offset in OW*TS image
+        int w = fa / TS, col = fa % TS;
         int row = ob0 + wave * OW + w;
         int srow = row < O ? row : O - 1;
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) void*)
-                (Abl + (long)srow * S + st + lane * 4),
+                (Abl + (long)srow * S + st + col),
             (__attribute__((address_space(3))) void*)
-                &dst[(ICAP + wave * OW + w) * TS + lane * 4],
+                &dst[(ICAP + wave * OW + w) * TS + col],
             16, 0, 0);
       }
     };
@@ -787,7 +802,13 @@ __global__ __launch_bounds__(kBlock) void gw_outer_glds3_kernel(
       issue_tile(0, s0);
       for (long t = 0; t < nt; ++t) {
         if (t + 1 < nt) {
-          issue_tile((int)((t + 1) % 3), s0 + (t + 1) * TS);
+          if constexpr (NBUF == 2) {
+            // 2-buffer ring: the issue target is the buffer read LAST
+            // iteration; fence all waves out of it first
+            __builtin_amdgcn_s_barrier();
+            asm volatile("" ::: "memory");
+          }
+          issue_tile((int)((t + 1) % NBUF), s0 + (t + 1) * TS);
           asm volatile("s_waitcnt vmcnt(%0)" ::"n"(NG) : "memory");
         } else {
           asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -796,27 +817,55 @@ __global__ __launch_bounds__(kBlock) void gw_outer_glds3_kernel(
         // cannot hoist the LDS reads above it
         __builtin_amdgcn_s_barrier();
         asm volatile("" ::: "memory");
-        const float* cur = ring[t % 3];
+        const float* cur = ring[t % NBUF];
         if (o0 < O) {
-          float4 av[OW];
+          if constexpr (TS >= 256) {
 #pragma unroll
-          for (int w = 0; w < OW; ++w) {
-            if (o0 + w < O) {
-              av[w] = *reinterpret_cast<const float4*>(
-                  &cur[(ICAP + wave * OW + w) * TS + lane * 4]);
-              if (want_bias) bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
-            }
-          }
-#pragma unroll
-          for (int i = 0; i < ICAP; ++i) {
-            if (i < I) {
-              const float4 bv =
-                  *reinterpret_cast<const float4*>(&cur[i * TS + lane * 4]);
+            for (int p = 0; p < PASS; ++p) {
+              float4 av[OW];
 #pragma unroll
               for (int w = 0; w < OW; ++w) {
-                if (o0 + w < O)
-                  acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
-                               av[w].z * bv.z + av[w].w * bv.w;
+                if (o0 + w < O) {
+                  av[w] = *reinterpret_cast<const float4*>(
+                      &cur[(ICAP + wave * OW + w) * TS + p * 256 + lane * 4]);
+                  if (want_bias)
+                    bacc[w] += av[w].x + av[w].y + av[w].z + av[w].w;
+                }
+              }
+#pragma unroll
+              for (int i = 0; i < ICAP; ++i) {
+                if (i < I) {
+                  const float4 bv = *reinterpret_cast<const float4*>(
+                      &cur[i * TS + p * 256 + lane * 4]);
+#pragma unroll
+                  for (int w = 0; w < OW; ++w) {
+                    if (o0 + w < O)
+                      acc[w][i] += av[w].x * bv.x + av[w].y * bv.y +
+                                   av[w].z * bv.z + av[w].w * bv.w;
+                  }
+                }
+              }
+            }
+          } else {                              // TS == 128: float2 per lane
+            float2 av[OW];
+#pragma unroll
+            for (int w = 0; w < OW; ++w) {
+              if (o0 + w < O) {
+                av[w] = *reinterpret_cast<const float2*>(
+                    &cur[(ICAP + wave * OW + w) * TS + lane * 2]);
+                if (want_bias) bacc[w] += av[w].x + av[w].y;
+              }
+            }
+#pragma unroll
+            for (int i = 0; i < ICAP; ++i) {
+              if (i < I) {
+                const float2 bv = *reinterpret_cast<const float2*>(
+                    &cur[i * TS + lane * 2]);
+#pragma unroll
+                for (int w = 0; w < OW; ++w) {
+                  if (o0 + w < O)
+                    acc[w][i] += av[w].x * bv.x + av[w].y * bv.y;
+                }
               }
             }
           }
@@ -903,11 +952,13 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
   // so one grid-wave exactly fills the 256 CUs; the generic kernel gets a
   // few more for latency-hiding.
   const bool use_glds = vec && !no_glds && O >= 8;
-  int bpc = use_glds ? ((!ow5 && I <= 8) ? 2 : 1) : 4;
+  int bpc = use_glds ? (ow5 ? 1 : (I <= 8 ? 2 : (o_tiles >= 8 ? 3 : 1))) : 4;
   int n_schunk = (int)std::max(
       1L, std::min((long)(256 * bpc / o_tiles), S / (64 * 16)));
   if (ns_cap > 0)
     n_schunk = (int)std::max(1L, std::min(ns_cap, S / (64 * 16)));
+  // the glds3 XCD swizzle decodes schunk mod 8: keep it a bijection
+  if (use_glds && o_tiles > 1) n_schunk = (n_schunk + 7) & ~7;
   int grid = n_schunk * o_tiles;
 
 #define GW_LAUNCH(ICAP_, OW_, BIAS_, V)                                         \
@@ -925,18 +976,28 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
       if (vec) { GW_LAUNCH(ICAP_, OW_, false, true) }                           \
       else { GW_LAUNCH(ICAP_, OW_, false, false) }                              \
     }
-#define GW_LDS(ICAP_, OW_)                                                      \
-      hipLaunchKernelGGL((gw_outer_glds3_kernel<scalar_t, ICAP_, OW_>),         \
-                         dim3(grid), dim3(kBlock), 0, stream,                   \
-                         gz.data_ptr<scalar_t>(),                               \
-                         x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),       \
-                         want_bias ? gb.data_ptr<scalar_t>() : nullptr,         \
-                         B, O, I, S, n_schunk, want_bias);
+#define GW_LDS(ICAP_, OW_, TS_, NBUF_)                                          \
+      hipLaunchKernelGGL(                                                       \
+          (gw_outer_glds3_kernel<scalar_t, ICAP_, OW_, TS_, NBUF_>),            \
+          dim3(grid), dim3(kBlock), 0, stream,                                  \
+          gz.data_ptr<scalar_t>(),                                              \
+          x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),                      \
+          want_bias ? gb.data_ptr<scalar_t>() : nullptr,                        \
+          B, O, I, S, n_schunk, want_bias);
   AT_DISPATCH_FLOATING_TYPES(gz.scalar_type(), "channel_mix_bwd_w", [&] {
-    if (ow5) { GW_LDS(20, 5) }
-    else if (vec && !no_glds && OW == 4) { GW_LDS(8, 4) }
+    if (ow5) { GW_LDS(20, 5, 256, 3) }
+    else if (vec && !no_glds && OW == 4) { GW_LDS(8, 4, 256, 3) }
     else if (vec && !no_glds && OW == 2) {
-      if (I <= 24) { GW_LDS(24, 2) } else { GW_LDS(32, 2) }
+      // big-O shapes (gW3: O=128) re-read B o_tiles times; occupancy (3
+      // blocks/CU at a 48 KB TS=128 ring) beats per-stream burst length
+      // there (measured 2.7 ms vs 4.4 ms at 1 block/CU, matmul 3.8 ms)
+      if (I <= 24) {
+        if (o_tiles >= 8) { GW_LDS(24, 2, 128, 3) }
+        else { GW_LDS(24, 2, 256, 3) }
+      } else {
+        if (o_tiles >= 8) { GW_LDS(32, 2, 128, 3) }
+        else { GW_LDS(32, 2, 256, 3) }
+      }
     } else if (OW == 4) { GW_DISPATCH2(8, 4) }
     else if (OW == 2) {
       if (I <= 24) { GW_DISPATCH2(24, 2) } else { GW_DISPATCH2(32, 2) }
