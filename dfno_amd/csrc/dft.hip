@@ -173,24 +173,42 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
 // twiddle work amortized): ~1.6x fewer VALU ops per element.
 // ---------------------------------------------------------------------------
 
-template <typename T, int LCAP>
+// G-way j-split: G threads (same wave) share one pair's j-loop, each
+// walking j = g, g+G, ... with chain step w^{-kG} and start w^{-kg}; the
+// partial sums reduce with log2(G) shfl_xor steps.  Needs G | N so the
+// chains wrap to their start after each line (w^{-kN} = 1).  Raises wave
+// occupancy G-fold for the middle-dim transforms whose outer*pairs alone
+// underfills 1024 SIMDs (the x/y-dim c2c calls of the flagship run at 0.2-
+// 0.5 waves/SIMD otherwise).
+template <typename T, int LCAP, int G = 1>
 __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long outer, int N, long inner, int m_lo, T scale) {
   // m_hi == m_lo; chains for k = 0..m_lo (the last used by the lowest suffix)
   const int nch = m_lo + 1;
-  T str[LCAP], sti[LCAP];
+  constexpr int PL = 64 / G;           // pairs handled per wave per g-group
+  const int lane = (int)(threadIdx.x % 64);
+  const int g = lane / PL;
+  T str[LCAP], sti[LCAP], cr[LCAP], ci[LCAP];
 #pragma unroll
   for (int k = 0; k < LCAP; ++k) {
-    if (k < nch)
-      sincos_t<T>(T(-2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
+    if (k < nch) {
+      sincos_t<T>(T(-2.0) * T(M_PI) * T(k * G) / T(N), &sti[k], &str[k]);
+      if constexpr (G > 1) {
+        sincos_t<T>(T(-2.0) * T(M_PI) * T(k * g) / T(N), &ci[k], &cr[k]);
+      } else {
+        cr[k] = T(1); ci[k] = T(0);
+      }
+    }
   }
 
   long pairs = inner / 2;
   long total = outer * pairs;
-  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (long t = t0; t < total; t += stride) {
+  long p0 = (((long)blockIdx.x * blockDim.x + threadIdx.x) / 64) * PL +
+            (lane % PL);
+  long pstride = ((long)gridDim.x * blockDim.x / 64) * PL;
+  const int iters = N / G;             // host guarantees G | N
+  for (long t = p0; t < total; t += pstride) {
     long o = t / pairs;
     long i = (t % pairs) * 2;
     const T* src = in + 2 * (o * N * inner + i);
@@ -198,16 +216,15 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
     // acc[ki][elem]{r,i}; ki < 2*m_lo
     T a0r[LCAP], a0i[LCAP], a1r[LCAP], a1i[LCAP];   // prefix accs (elem0/1)
     T b0r[LCAP], b0i[LCAP], b1r[LCAP], b1i[LCAP];   // suffix accs
-    T cr[LCAP], ci[LCAP];
 #pragma unroll
     for (int k = 0; k < LCAP; ++k) {
-      if (k < nch) { cr[k] = T(1); ci[k] = T(0); }
       if (k < m_lo) {
         a0r[k] = a0i[k] = a1r[k] = a1i[k] = T(0);
         b0r[k] = b0i[k] = b1r[k] = b1i[k] = T(0);
       }
     }
-    for (int j = 0; j < N; ++j) {
+    for (int sct = 0; sct < iters; ++sct) {
+      const int j = g + sct * G;
       T x0r, x0i, x1r, x1i;
       if constexpr (std::is_same<T, float>::value) {
         const float4 v = *reinterpret_cast<const float4*>(src + 2 * j * inner);
@@ -239,6 +256,25 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
       for (int k = 0; k < LCAP; ++k)
         if (k < nch) cmul_acc(cr[k], ci[k], str[k], sti[k]);
     }
+    if constexpr (G > 1) {
+      // combine the G partial sums (partners share (o, pair), differ in g)
+#pragma unroll
+      for (int k = 0; k < LCAP; ++k) {
+        if (k < m_lo) {
+          for (int off = PL; off < 64; off <<= 1) {
+            a0r[k] += __shfl_xor(a0r[k], off, 64);
+            a0i[k] += __shfl_xor(a0i[k], off, 64);
+            a1r[k] += __shfl_xor(a1r[k], off, 64);
+            a1i[k] += __shfl_xor(a1i[k], off, 64);
+            b0r[k] += __shfl_xor(b0r[k], off, 64);
+            b0i[k] += __shfl_xor(b0i[k], off, 64);
+            b1r[k] += __shfl_xor(b1r[k], off, 64);
+            b1i[k] += __shfl_xor(b1i[k], off, 64);
+          }
+        }
+      }
+      if (g != 0) continue;            // one g-group writes
+    }
     const int m = 2 * m_lo;
     T* dst = out + 2 * (o * m * inner + i);
 #pragma unroll
@@ -260,34 +296,46 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
   }
 }
 
-template <typename T, int LCAP>
+// G-way j-split synthesis: outputs j = g, g+G, ... are independent, so no
+// reduction is needed; each g-group loads the m kept modes (same lines, L1
+// broadcast) and writes its own j's.  Needs G | N for the chain wrap.
+template <typename T, int LCAP, int G = 1>
 __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis2_kernel(
     const T* __restrict__ in, T* __restrict__ out,
     long outer, int N, long inner, int m_lo, T scale) {
   const int nch = m_lo + 1;
-  T str[LCAP], sti[LCAP];
+  constexpr int PL = 64 / G;
+  const int lane = (int)(threadIdx.x % 64);
+  const int g = lane / PL;
+  T str[LCAP], sti[LCAP], cr[LCAP], ci[LCAP];
 #pragma unroll
   for (int k = 0; k < LCAP; ++k) {
-    if (k < nch)
-      sincos_t<T>(T(2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
+    if (k < nch) {
+      sincos_t<T>(T(2.0) * T(M_PI) * T(k * G) / T(N), &sti[k], &str[k]);
+      if constexpr (G > 1) {
+        sincos_t<T>(T(2.0) * T(M_PI) * T(k * g) / T(N), &ci[k], &cr[k]);
+      } else {
+        cr[k] = T(1); ci[k] = T(0);
+      }
+    }
   }
 
   const int m = 2 * m_lo;
   long pairs = inner / 2;
   long total = outer * pairs;
-  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (long t = t0; t < total; t += stride) {
+  long p0 = (((long)blockIdx.x * blockDim.x + threadIdx.x) / 64) * PL +
+            (lane % PL);
+  long pstride = ((long)gridDim.x * blockDim.x / 64) * PL;
+  const int iters = N / G;             // host guarantees G | N
+  for (long t = p0; t < total; t += pstride) {
     long o = t / pairs;
     long i = (t % pairs) * 2;
     const T* src = in + 2 * (o * m * inner + i);
 
     T p0r[LCAP], p0i[LCAP], p1r[LCAP], p1i[LCAP];   // prefix inputs
     T q0r[LCAP], q0i[LCAP], q1r[LCAP], q1i[LCAP];   // suffix inputs
-    T cr[LCAP], ci[LCAP];
 #pragma unroll
     for (int k = 0; k < LCAP; ++k) {
-      if (k < nch) { cr[k] = T(1); ci[k] = T(0); }
       if (k < m_lo) {
         p0r[k] = scale * src[2 * k * inner];
         p0i[k] = scale * src[2 * k * inner + 1];
@@ -301,7 +349,8 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis2_kernel(
       }
     }
     T* dst = out + 2 * (o * N * inner + i);
-    for (int j = 0; j < N; ++j) {
+    for (int sct = 0; sct < iters; ++sct) {
+      const int j = g + sct * G;
       T s0r = T(0), s0i = T(0), s1r = T(0), s1i = T(0);
 #pragma unroll
       for (int k = 0; k < LCAP; ++k) {
@@ -527,11 +576,35 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
   int grid = grid_for_d(outer * inner);
 
   bool paired = (m_hi == m_lo) && (inner % 2 == 0) && m_lo <= 16;
-  int grid2 = paired ? grid_for_d(outer * (inner / 2)) : grid;
+  // j-split degree: raise wave occupancy when outer*pairs alone underfills
+  // the 1024 SIMDs (x/y-dim transforms); G must divide n for the chain wrap
+  int Gsel = 1;
+  if (paired) {
+    static const long gtarget = []() {
+      const char* e = getenv("DFNO_DFT_GTARGET");  // threads wanted (tunable)
+      // default 1 = j-split off: measured on MI355X, G>1 LOSES at every
+      // flagship shape (y-dim c2c 0.119 -> 0.137 ms at G=4) -- the per-wave
+      // load stream already saturates at 0.5 waves/SIMD and narrowing the
+      // contiguous chunk from 1 KB to 256 B costs more than the extra
+      // occupancy buys.  Kept as a knob for other shapes.
+      return e ? atol(e) : 1L;
+    }();
+    long pwork = outer * (inner / 2);
+    while (Gsel < 8 && (n % (Gsel * 2) == 0) && pwork * Gsel < gtarget)
+      Gsel *= 2;
+  }
+  int grid2 = paired ? grid_for_d(outer * (inner / 2) * Gsel) : grid;
+#define DFT_LG(KERNEL, LC, ...)                                                \
+  switch (Gsel) {                                                              \
+    case 8: hipLaunchKernelGGL((KERNEL<scalar_t, LC, 8>), __VA_ARGS__); break; \
+    case 4: hipLaunchKernelGGL((KERNEL<scalar_t, LC, 4>), __VA_ARGS__); break; \
+    case 2: hipLaunchKernelGGL((KERNEL<scalar_t, LC, 2>), __VA_ARGS__); break; \
+    default: hipLaunchKernelGGL((KERNEL<scalar_t, LC, 1>), __VA_ARGS__);       \
+  }
 #define DFT_LDISPATCH(KERNEL, ...)                                             \
-  if (m_lo <= 8) { hipLaunchKernelGGL((KERNEL<scalar_t, 9>), __VA_ARGS__); }   \
-  else if (m_lo <= 12) { hipLaunchKernelGGL((KERNEL<scalar_t, 13>), __VA_ARGS__); } \
-  else { hipLaunchKernelGGL((KERNEL<scalar_t, 17>), __VA_ARGS__); }
+  if (m_lo <= 8) { DFT_LG(KERNEL, 9, __VA_ARGS__) }                            \
+  else if (m_lo <= 12) { DFT_LG(KERNEL, 13, __VA_ARGS__) }                     \
+  else { DFT_LG(KERNEL, 17, __VA_ARGS__) }
   AT_DISPATCH_FLOATING_TYPES(c10::toRealValueType(x.scalar_type()), "dft_c2c", [&] {
     auto inp = reinterpret_cast<const scalar_t*>(x.data_ptr());
     auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
@@ -554,6 +627,7 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
     }
   });
 #undef DFT_LDISPATCH
+#undef DFT_LG
   return out;
 }
 
