@@ -21,6 +21,8 @@
 
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
+#include <mutex>
+#include <unordered_map>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include "kernels.h"
@@ -182,25 +184,15 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis_kernel(
 // 0.5 waves/SIMD otherwise).
 template <typename T, int LCAP, int G = 1>
 __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
-    const T* __restrict__ in, T* __restrict__ out,
+    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long outer, int N, long inner, int m_lo, T scale) {
-  // m_hi == m_lo; chains for k = 0..m_lo (the last used by the lowest suffix)
+  // m_hi == m_lo; tw is the [N, m_lo+1, 2] table of w^{-jk} (see r2c note:
+  // register twiddle chains were 4 of every ~5 non-MAC VALU ops and cost 4
+  // LCAP register arrays; (j,k) are lane-uniform so table reads scalarize)
   const int nch = m_lo + 1;
   constexpr int PL = 64 / G;           // pairs handled per wave per g-group
   const int lane = (int)(threadIdx.x % 64);
   const int g = lane / PL;
-  T str[LCAP], sti[LCAP], cr[LCAP], ci[LCAP];
-#pragma unroll
-  for (int k = 0; k < LCAP; ++k) {
-    if (k < nch) {
-      sincos_t<T>(T(-2.0) * T(M_PI) * T(k * G) / T(N), &sti[k], &str[k]);
-      if constexpr (G > 1) {
-        sincos_t<T>(T(-2.0) * T(M_PI) * T(k * g) / T(N), &ci[k], &cr[k]);
-      } else {
-        cr[k] = T(1); ci[k] = T(0);
-      }
-    }
-  }
 
   long pairs = inner / 2;
   long total = outer * pairs;
@@ -235,26 +227,26 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
         x1r = src[2 * j * inner + 2];
         x1i = src[2 * j * inner + 3];
       }
+      auto twj = (const __attribute__((address_space(4))) T*)
+          (tw + (long)j * 2 * nch);
 #pragma unroll
       for (int k = 0; k < LCAP; ++k) {
         if (k < m_lo) {
-          // prefix mode k: w = (cr, ci)
-          a0r[k] += x0r * cr[k] - x0i * ci[k];
-          a0i[k] += x0r * ci[k] + x0i * cr[k];
-          a1r[k] += x1r * cr[k] - x1i * ci[k];
-          a1i[k] += x1r * ci[k] + x1i * cr[k];
-          // suffix mode N - k' where k' = m_lo - k: uses conj(chain[k'])
+          // prefix mode k: w^{-jk} = (cr, ci)
+          const T cr = twj[2 * k], ci = twj[2 * k + 1];
+          a0r[k] += x0r * cr - x0i * ci;
+          a0i[k] += x0r * ci + x0i * cr;
+          a1r[k] += x1r * cr - x1i * ci;
+          a1i[k] += x1r * ci + x1i * cr;
+          // suffix mode N - k' where k' = m_lo - k: uses conj(w^{-jk'})
           const int kp = m_lo - k;
-          const T dr = cr[kp], di = -ci[kp];
+          const T dr = twj[2 * kp], di = -twj[2 * kp + 1];
           b0r[k] += x0r * dr - x0i * di;
           b0i[k] += x0r * di + x0i * dr;
           b1r[k] += x1r * dr - x1i * di;
           b1i[k] += x1r * di + x1i * dr;
         }
       }
-#pragma unroll
-      for (int k = 0; k < LCAP; ++k)
-        if (k < nch) cmul_acc(cr[k], ci[k], str[k], sti[k]);
     }
     if constexpr (G > 1) {
       // combine the G partial sums (partners share (o, pair), differ in g)
@@ -301,24 +293,13 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
 // broadcast) and writes its own j's.  Needs G | N for the chain wrap.
 template <typename T, int LCAP, int G = 1>
 __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis2_kernel(
-    const T* __restrict__ in, T* __restrict__ out,
+    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long outer, int N, long inner, int m_lo, T scale) {
+  // tw is the [N, m_lo+1, 2] table of w^{+jk}; see analysis2 note
   const int nch = m_lo + 1;
   constexpr int PL = 64 / G;
   const int lane = (int)(threadIdx.x % 64);
   const int g = lane / PL;
-  T str[LCAP], sti[LCAP], cr[LCAP], ci[LCAP];
-#pragma unroll
-  for (int k = 0; k < LCAP; ++k) {
-    if (k < nch) {
-      sincos_t<T>(T(2.0) * T(M_PI) * T(k * G) / T(N), &sti[k], &str[k]);
-      if constexpr (G > 1) {
-        sincos_t<T>(T(2.0) * T(M_PI) * T(k * g) / T(N), &ci[k], &cr[k]);
-      } else {
-        cr[k] = T(1); ci[k] = T(0);
-      }
-    }
-  }
 
   const int m = 2 * m_lo;
   long pairs = inner / 2;
@@ -352,26 +333,26 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis2_kernel(
     for (int sct = 0; sct < iters; ++sct) {
       const int j = g + sct * G;
       T s0r = T(0), s0i = T(0), s1r = T(0), s1i = T(0);
+      auto twj = (const __attribute__((address_space(4))) T*)
+          (tw + (long)j * 2 * nch);
 #pragma unroll
       for (int k = 0; k < LCAP; ++k) {
         if (k < m_lo) {
           // prefix mode k with w^{+jk} = (cr, ci)
-          s0r += p0r[k] * cr[k] - p0i[k] * ci[k];
-          s0i += p0r[k] * ci[k] + p0i[k] * cr[k];
-          s1r += p1r[k] * cr[k] - p1i[k] * ci[k];
-          s1i += p1r[k] * ci[k] + p1i[k] * cr[k];
-          // suffix kglobal = N - (m_lo - k): w^{+j(N-kp)} = conj(chain[kp])
+          const T cr = twj[2 * k], ci = twj[2 * k + 1];
+          s0r += p0r[k] * cr - p0i[k] * ci;
+          s0i += p0r[k] * ci + p0i[k] * cr;
+          s1r += p1r[k] * cr - p1i[k] * ci;
+          s1i += p1r[k] * ci + p1i[k] * cr;
+          // suffix kglobal = N - (m_lo - k): w^{+j(N-kp)} = conj(w^{+jkp})
           const int kp = m_lo - k;
-          const T dr = cr[kp], di = -ci[kp];
+          const T dr = twj[2 * kp], di = -twj[2 * kp + 1];
           s0r += q0r[k] * dr - q0i[k] * di;
           s0i += q0r[k] * di + q0i[k] * dr;
           s1r += q1r[k] * dr - q1i[k] * di;
           s1i += q1r[k] * di + q1i[k] * dr;
         }
       }
-#pragma unroll
-      for (int k = 0; k < LCAP; ++k)
-        if (k < nch) cmul_acc(cr[k], ci[k], str[k], sti[k]);
       if constexpr (std::is_same<T, float>::value) {
         *reinterpret_cast<float4*>(dst + 2 * j * inner) =
             make_float4(s0r, s0i, s1r, s1i);
@@ -392,18 +373,16 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_synthesis2_kernel(
 // One 256-line tile is cooperatively staged into LDS with coalesced loads
 // (per-thread direct line reads thrash L1: 16 waves x line-span > 32 KiB,
 // measured SQ_WAIT_ANY = 72%); threads then stream their line from LDS.
+// Twiddles come from a host-precomputed [N, m, 2] table instead of per-mode
+// register recurrences: the chain update was 4 of every 6 VALU ops (PMC:
+// VALU-bound at ~85% issue), and (j, k) are lane-uniform so the table reads
+// compile to scalar loads that the k-unrolled fma stream hides entirely.
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
-    const T* __restrict__ in, T* __restrict__ out,
+    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long lines, int N, int m, T scale, bool factors) {
   extern __shared__ __align__(16) char smem_raw[];
   T* tile = reinterpret_cast<T*>(smem_raw);   // [kBlock * N]
-  T str[MCAP], sti[MCAP];
-#pragma unroll
-  for (int k = 0; k < MCAP; ++k) {
-    if (k < m)
-      sincos_t<T>(T(-2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
-  }
 
   long ntiles = (lines + kBlock - 1) / kBlock;
   for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x) {
@@ -426,33 +405,40 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
         tile[idx] = in[l0 * N + idx];
     }
     __syncthreads();
-    if ((int)threadIdx.x < nl) {
+    {
+      // compute runs on ALL threads (tail threads chew stale LDS and skip
+      // the store): a divergent guard here would block the scalarization of
+      // the lane-uniform twiddle loads below
       const T* src = tile + threadIdx.x * N;
-      T ar[MCAP], ai[MCAP], cr[MCAP], ci[MCAP];
+      T ar[MCAP], ai[MCAP];
 #pragma unroll
       for (int k = 0; k < MCAP; ++k) {
-        if (k < m) { ar[k] = T(0); ai[k] = T(0); cr[k] = T(1); ci[k] = T(0); }
+        if (k < m) { ar[k] = T(0); ai[k] = T(0); }
       }
       for (int j = 0; j < N; ++j) {
         const T x = src[j];
+        // constant-addrspace cast: lane-uniform twiddles become s_load
+        auto twj = (const __attribute__((address_space(4))) T*)
+            (tw + (long)j * 2 * m);
 #pragma unroll
         for (int k = 0; k < MCAP; ++k) {
           if (k < m) {
-            ar[k] += x * cr[k];
-            ai[k] += x * ci[k];
-            cmul_acc(cr[k], ci[k], str[k], sti[k]);
+            ar[k] += x * twj[2 * k];
+            ai[k] += x * twj[2 * k + 1];
           }
         }
       }
-      T* dst = out + 2 * (l0 + threadIdx.x) * m;
+      if ((int)threadIdx.x < nl) {   // no continue: barrier above must stay
+        T* dst = out + 2 * (l0 + threadIdx.x) * m;   // wave-convergent
 #pragma unroll
-      for (int k = 0; k < MCAP; ++k) {
-        if (k < m) {
-          T f = T(1);
-          bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
-          if (factors && !edge) f = T(2);
-          dst[2 * k] = f * scale * ar[k];
-          dst[2 * k + 1] = (factors && edge) ? T(0) : f * scale * ai[k];
+        for (int k = 0; k < MCAP; ++k) {
+          if (k < m) {
+            T f = T(1);
+            bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
+            if (factors && !edge) f = T(2);
+            dst[2 * k] = f * scale * ar[k];
+            dst[2 * k + 1] = (factors && edge) ? T(0) : f * scale * ai[k];
+          }
         }
       }
     }
@@ -467,25 +453,23 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
 // (coalesced); inputs are 64-128B contiguous per line and read directly.
 template <typename T, int MCAP>
 __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
-    const T* __restrict__ in, T* __restrict__ out,
+    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long lines, int N, int m, T scale, bool factors) {
   extern __shared__ __align__(16) char smem_raw[];
   T* tile = reinterpret_cast<T*>(smem_raw);   // [kBlock * N]
-  T str[MCAP], sti[MCAP];
-#pragma unroll
-  for (int k = 0; k < MCAP; ++k) {
-    if (k < m)
-      sincos_t<T>(T(2.0) * T(M_PI) * T(k) / T(N), &sti[k], &str[k]);
-  }
 
   long ntiles = (lines + kBlock - 1) / kBlock;
   for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x) {
     long l0 = tb * kBlock;
     int nl = (int)min((long)kBlock, lines - l0);
     __syncthreads();
-    if ((int)threadIdx.x < nl) {
-      const T* src = in + 2 * (l0 + threadIdx.x) * m;
-      T yr[MCAP], yi[MCAP], cr[MCAP], ci[MCAP];
+    {
+      // all threads compute (tail threads re-read line nl-1) so the twiddle
+      // loads stay wave-uniform and scalarize; only the LDS/store side is
+      // masked via the cooperative writeback bound below
+      long lidx = l0 + min((int)threadIdx.x, nl - 1);
+      const T* src = in + 2 * lidx * m;
+      T yr[MCAP], yi[MCAP];
 #pragma unroll
       for (int k = 0; k < MCAP; ++k) {
         if (k < m) {
@@ -495,18 +479,18 @@ __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
           yr[k] = f * src[2 * k];
           yi[k] = f * src[2 * k + 1];
           if (factors && edge) yi[k] = T(0);
-          cr[k] = T(1); ci[k] = T(0);
         }
       }
       T* dst = tile + threadIdx.x * N;
       for (int j = 0; j < N; ++j) {
         T sacc = T(0);
+        auto twj = (const __attribute__((address_space(4))) T*)
+            (tw + (long)j * 2 * m);            // lane-uniform -> s_load
 #pragma unroll
         for (int k = 0; k < MCAP; ++k) {
           if (k < m) {
-            // Re(y * w^{+jk}) with w^{+jk} = (cr, ci)
-            sacc += yr[k] * cr[k] - yi[k] * ci[k];
-            cmul_acc(cr[k], ci[k], str[k], sti[k]);
+            // Re(y * w^{+jk}) with w^{+jk} = (twj[2k], twj[2k+1])
+            sacc += yr[k] * twj[2 * k] - yi[k] * twj[2 * k + 1];
           }
         }
         dst[j] = sacc;
@@ -552,6 +536,27 @@ void split_dims(const at::Tensor& x, int dim, long& outer, long& inner) {
   else if (m <= 24) { hipLaunchKernelGGL((KERNEL<scalar_t, 24>), __VA_ARGS__); } \
   else { TORCH_CHECK(m <= 32, "dft: m > 32 unsupported natively");             \
          hipLaunchKernelGGL((KERNEL<scalar_t, 32>), __VA_ARGS__); }
+
+// Cached [N, m, 2] twiddle table (fp64 sincos on host, cast to T): one per
+// (N, m, direction, dtype) per process; ranks own one device each.
+static at::Tensor twiddle_table(int N, int m, bool analysis,
+                                const at::TensorOptions& opt) {
+  static std::mutex mu;
+  static std::unordered_map<long, at::Tensor> cache;
+  long key = ((long)N << 20) | ((long)m << 8) | ((long)analysis << 1) |
+             (opt.dtype() == at::kFloat ? 1 : 0);
+  std::lock_guard<std::mutex> lk(mu);
+  auto it = cache.find(key);
+  if (it != cache.end()) return it->second;
+  double sgn = analysis ? -1.0 : 1.0;
+  auto j = at::arange(N, at::kDouble).reshape({N, 1});
+  auto k = at::arange(m, at::kDouble).reshape({1, m});
+  auto ang = (sgn * 2.0 * M_PI / N) * j * k;
+  auto tw = at::stack({ang.cos(), ang.sin()}, 2).to(opt).contiguous();
+  cache[key] = tw;
+  return tw;
+}
+
 
 }  // namespace
 
@@ -605,17 +610,21 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
   if (m_lo <= 8) { DFT_LG(KERNEL, 9, __VA_ARGS__) }                            \
   else if (m_lo <= 12) { DFT_LG(KERNEL, 13, __VA_ARGS__) }                     \
   else { DFT_LG(KERNEL, 17, __VA_ARGS__) }
+  at::Tensor tw;
+  if (paired)
+    tw = twiddle_table((int)n, (int)m_lo + 1, analysis,
+                       x.options().dtype(c10::toRealValueType(x.scalar_type())));
   AT_DISPATCH_FLOATING_TYPES(c10::toRealValueType(x.scalar_type()), "dft_c2c", [&] {
     auto inp = reinterpret_cast<const scalar_t*>(x.data_ptr());
     auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
     if (paired && analysis) {
       DFT_LDISPATCH(dft_c2c_analysis2_kernel, dim3(grid2), dim3(kBlock), 0,
-                    stream, inp, op, outer, (int)n, inner, (int)m_lo,
-                    (scalar_t)scale)
+                    stream, inp, op, tw.data_ptr<scalar_t>(), outer, (int)n,
+                    inner, (int)m_lo, (scalar_t)scale)
     } else if (paired) {
       DFT_LDISPATCH(dft_c2c_synthesis2_kernel, dim3(grid2), dim3(kBlock), 0,
-                    stream, inp, op, outer, (int)n, inner, (int)m_lo,
-                    (scalar_t)scale)
+                    stream, inp, op, tw.data_ptr<scalar_t>(), outer, (int)n,
+                    inner, (int)m_lo, (scalar_t)scale)
     } else if (analysis) {
       DFT_MDISPATCH(dft_c2c_analysis_kernel, dim3(grid), dim3(kBlock), 0, stream,
                     inp, op, outer, (int)n, inner, (int)m_lo, (int)m_hi,
@@ -648,12 +657,15 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   long ntiles = (lines + kBlock - 1) / kBlock;
   int grid = (int)std::min(ntiles, 4096L);
+  auto tw = twiddle_table(N, (int)m, /*analysis=*/true,
+                          x.options().dtype(x.scalar_type()));
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "dft_r2c", [&] {
     size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = x.data_ptr<scalar_t>();
     auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
     DFT_MDISPATCH(dft_r2c_last_kernel, dim3(grid), dim3(kBlock), smem, stream,
-                  inp, op, lines, N, (int)m, (scalar_t)scale, factors)
+                  inp, op, tw.data_ptr<scalar_t>(), lines, N, (int)m,
+                  (scalar_t)scale, factors)
   });
   return out;
 }
@@ -678,12 +690,15 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   long ntiles = (lines + kBlock - 1) / kBlock;
   int grid = (int)std::min(ntiles, 4096L);
+  auto tw = twiddle_table(N, m, /*analysis=*/false,
+                          out.options().dtype(out.scalar_type()));
   AT_DISPATCH_FLOATING_TYPES(out.scalar_type(), "dft_c2r", [&] {
     size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = reinterpret_cast<const scalar_t*>(y.data_ptr());
     auto op = out.data_ptr<scalar_t>();
     DFT_MDISPATCH(dft_c2r_last_kernel, dim3(grid), dim3(kBlock), smem, stream,
-                  inp, op, lines, N, (int)m, (scalar_t)scale, factors)
+                  inp, op, tw.data_ptr<scalar_t>(), lines, N, (int)m,
+                  (scalar_t)scale, factors)
   });
   return out;
 }
