@@ -215,12 +215,19 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_analysis2_kernel(
         b0r[k] = b0i[k] = b1r[k] = b1i[k] = T(0);
       }
     }
+    // software pipeline: the next j's load issues BEFORE this j's ~200-fma
+    // block (the scheduler will not hoist it past that body on its own;
+    // measured 6.3 TB/s for a bare read loop vs 1.9 here without this)
+    float4 v;
+    if constexpr (std::is_same<T, float>::value)
+      v = *reinterpret_cast<const float4*>(src + 2 * g * inner);
     for (int sct = 0; sct < iters; ++sct) {
       const int j = g + sct * G;
       T x0r, x0i, x1r, x1i;
       if constexpr (std::is_same<T, float>::value) {
-        const float4 v = *reinterpret_cast<const float4*>(src + 2 * j * inner);
         x0r = v.x; x0i = v.y; x1r = v.z; x1i = v.w;
+        if (sct + 1 < iters)
+          v = *reinterpret_cast<const float4*>(src + 2 * (j + G) * inner);
       } else {
         x0r = src[2 * j * inner];
         x0i = src[2 * j * inner + 1];
