@@ -1065,6 +1065,53 @@ __global__ void adam_step_kernel(T* __restrict__ p, const T* __restrict__ g,
   }
 }
 
+// Multi-tensor Adam: the whole parameter set updates in ONE launch.  The
+// pointer/size table travels by value in kernargs (fits: 40 tensors x 48 B
+// < the 4 KB kernarg segment); blocks grid-stride over the concatenated
+// vec4 index space and locate their tensor with a short uniform search.
+constexpr int kAdamMaxT = 40;
+struct AdamTab {
+  float* p[kAdamMaxT];
+  const float* g[kAdamMaxT];
+  float* m[kAdamMaxT];
+  float* v[kAdamMaxT];
+  long end4[kAdamMaxT];   // exclusive prefix sum of n/4 per tensor
+  float c1[kAdamMaxT], c2[kAdamMaxT];
+  int nt;
+};
+
+__global__ __launch_bounds__(kBlock) void adam_multi_kernel(
+    AdamTab tab, long total4, float lr, float b1, float b2, float eps,
+    float wd) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  int t = 0;
+  for (long i = i0; i < total4; i += stride) {
+    while (i >= tab.end4[t]) ++t;          // monotone: i increases
+    long base = (t == 0) ? 0 : tab.end4[t - 1];
+    long j = (i - base) * 4;
+    float4 pv = *reinterpret_cast<float4*>(tab.p[t] + j);
+    const float4 gv = *reinterpret_cast<const float4*>(tab.g[t] + j);
+    float4 mv = *reinterpret_cast<float4*>(tab.m[t] + j);
+    float4 vv = *reinterpret_cast<float4*>(tab.v[t] + j);
+    float pr[4] = {pv.x, pv.y, pv.z, pv.w};
+    float gr[4] = {gv.x, gv.y, gv.z, gv.w};
+    float mr[4] = {mv.x, mv.y, mv.z, mv.w};
+    float vr[4] = {vv.x, vv.y, vv.z, vv.w};
+    const float c1 = tab.c1[t], c2 = tab.c2[t];
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float gg = gr[k] + wd * pr[k];
+      mr[k] = b1 * mr[k] + (1.f - b1) * gg;
+      vr[k] = b2 * vr[k] + (1.f - b2) * gg * gg;
+      pr[k] -= lr * (mr[k] / c1) / (sqrtf(vr[k] / c2) + eps);
+    }
+    *reinterpret_cast<float4*>(tab.p[t] + j) = make_float4(pr[0], pr[1], pr[2], pr[3]);
+    *reinterpret_cast<float4*>(tab.m[t] + j) = make_float4(mr[0], mr[1], mr[2], mr[3]);
+    *reinterpret_cast<float4*>(tab.v[t] + j) = make_float4(vr[0], vr[1], vr[2], vr[3]);
+  }
+}
+
 }  // namespace
 
 void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
@@ -1074,9 +1121,46 @@ void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
   TORCH_CHECK(ps.size() == gs.size() && ps.size() == ms.size() &&
               ps.size() == vs.size() && ps.size() == steps.size(),
               "adam batch: length mismatch");
-  for (size_t i = 0; i < ps.size(); ++i)
-    adam_step_(ps[i], gs[i], ms[i], vs[i], lr, beta1, beta2, eps,
-               weight_decay, steps[i]);
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  AdamTab tab;
+  tab.nt = 0;
+  long total4 = 0;
+  auto flush = [&]() {
+    if (tab.nt == 0) return;
+    int grid = grid_for(total4, kBlock);
+    hipLaunchKernelGGL(adam_multi_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                       tab, total4, (float)lr, (float)beta1, (float)beta2,
+                       (float)eps, (float)weight_decay);
+    tab.nt = 0;
+    total4 = 0;
+  };
+  for (size_t i = 0; i < ps.size(); ++i) {
+    long n = ps[i].numel();
+    bool vec = ps[i].scalar_type() == at::kFloat && (n % 4 == 0) && n > 0 &&
+               ((reinterpret_cast<uintptr_t>(ps[i].data_ptr()) & 15) == 0) &&
+               ((reinterpret_cast<uintptr_t>(gs[i].data_ptr()) & 15) == 0) &&
+               ((reinterpret_cast<uintptr_t>(ms[i].data_ptr()) & 15) == 0) &&
+               ((reinterpret_cast<uintptr_t>(vs[i].data_ptr()) & 15) == 0);
+    if (!vec) {  // rare: odd-sized or fp64 tensor keeps the single-tensor path
+      adam_step_(ps[i], gs[i], ms[i], vs[i], lr, beta1, beta2, eps,
+                 weight_decay, steps[i]);
+      continue;
+    }
+    int t = tab.nt++;
+    tab.p[t] = ps[i].data_ptr<float>();
+    tab.g[t] = gs[i].data_ptr<float>();
+    tab.m[t] = ms[i].data_ptr<float>();
+    tab.v[t] = vs[i].data_ptr<float>();
+    total4 += n / 4;
+    tab.end4[t] = total4;
+    tab.c1[t] = (float)(1.0 - std::pow(beta1, (double)steps[i]));
+    tab.c2[t] = (float)(1.0 - std::pow(beta2, (double)steps[i]));
+    if (tab.nt == kAdamMaxT) flush();
+  }
+  flush();
+  hipError_t lerr = hipGetLastError();
+  TORCH_CHECK(lerr == hipSuccess, "adam_multi launch failed: ",
+              hipGetErrorString(lerr));
 }
 
 void adam_step_(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,
