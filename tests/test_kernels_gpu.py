@@ -254,9 +254,12 @@ def test_dft_rfft_trunc(dtype, tt, shape, dim, N, m):
 
 @pytest.mark.parametrize("dtype,tt", [(torch.complex64, 3e-4), (torch.complex128, 1e-11)])
 @pytest.mark.parametrize("shape,dim,mlo,mhi", [
-    ((1, 20, 64, 24, 8), 2, 12, 12),     # z-dim with inner>1
+    ((1, 20, 64, 24, 8), 2, 12, 12),     # z-dim with inner>1 (radix-8x8)
     ((1, 6, 10, 5), 1, 3, 2),
     ((2, 5, 12), 2, 4, 3),               # last dim c2c
+    ((2, 20, 64, 4), 2, 12, 12),         # radix-8x8 with tiny inner (pairs=2)
+    ((1, 4, 64, 3), 2, 10, 10),          # N=64 odd inner -> non-paired fallback
+    ((1, 3, 64, 8), 2, 16, 16),          # radix-8x8 at the LCAP=17 boundary
 ])
 def test_dft_fft_trunc_and_pad_ifft(dtype, tt, shape, dim, mlo, mhi):
     from dfno_amd.ops.fft import fft_trunc, pad_ifft, _t_fft_trunc, _t_pad_ifft
