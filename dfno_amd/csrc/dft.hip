@@ -900,6 +900,7 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
   });
 #undef DFT_LDISPATCH
 #undef DFT_LG
+  DFNO_CHECK_LAUNCH("dft_c2c");
   return out;
 }
 
@@ -930,6 +931,7 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
                   inp, op, tw.data_ptr<scalar_t>(), lines, N, (int)m,
                   (scalar_t)scale, factors)
   });
+  DFNO_CHECK_LAUNCH("dft_r2c");
   return out;
 }
 
@@ -963,6 +965,7 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
                   inp, op, tw.data_ptr<scalar_t>(), lines, N, (int)m,
                   (scalar_t)scale, factors)
   });
+  DFNO_CHECK_LAUNCH("dft_c2r");
   return out;
 }
 

@@ -3,6 +3,16 @@
 #pragma once
 
 #include <torch/extension.h>
+// Surface a refused/failed kernel launch loudly instead of silently
+// returning stale/zero output (several kernels carry large static LDS or
+// max-VGPR configurations where a bad launch is conceivable).
+#define DFNO_CHECK_LAUNCH(what)                                            \
+  do {                                                                     \
+    hipError_t _e = hipGetLastError();                                     \
+    TORCH_CHECK(_e == hipSuccess, what, " launch failed: ",                \
+                hipGetErrorString(_e));                                    \
+  } while (0)
+
 #include <vector>
 
 // fused channel-contraction linear: y[b,o,s] = act(sum_i W[o,i] x[b,i,s] + b[o])

@@ -276,6 +276,7 @@ at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
                        W2.data_ptr<scalar_t>(), b2.data_ptr<scalar_t>(),
                        out.data_ptr<scalar_t>(), B, C, W, Tn, S);
   });
+  DFNO_CHECK_LAUNCH("lift_head");
   return out;
 }
 
@@ -307,5 +308,6 @@ std::vector<at::Tensor> lift_head_bwd(const at::Tensor& gy, const at::Tensor& x,
                        gW2.data_ptr<scalar_t>(), gb2.data_ptr<scalar_t>(),
                        B, C, W, Tn, S);
   });
+  DFNO_CHECK_LAUNCH("lift_head");
   return {gx, gW1, gb1, gW2, gb2};
 }

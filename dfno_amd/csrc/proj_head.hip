@@ -366,6 +366,7 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
       if (vec) { PH_LAUNCH_F(true) } else { PH_LAUNCH_F(false) }
     }
   });
+  DFNO_CHECK_LAUNCH("proj_head");
 #undef PH_LAUNCH_F
   return out;
 }
@@ -411,6 +412,7 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
       if (vec) { PH_LAUNCH_B(true) } else { PH_LAUNCH_B(false) }
     }
   });
+  DFNO_CHECK_LAUNCH("proj_head");
 #undef PH_LAUNCH_B
   return {gz3, gb3, gW4, gb4};
 }

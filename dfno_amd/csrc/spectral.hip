@@ -371,6 +371,7 @@ void spectral_corners_fwd(const at::Tensor& x, std::vector<at::Tensor> ws,
   } else {
     launch_corners<double, false>(x, ws, y, starts, B, I, O, Ftot);
   }
+  DFNO_CHECK_LAUNCH("spectral");
 }
 
 void spectral_corners_bwd_x(const at::Tensor& gy, std::vector<at::Tensor> ws,
@@ -390,4 +391,5 @@ void spectral_corners_bwd_x(const at::Tensor& gy, std::vector<at::Tensor> ws,
   } else {
     launch_corners<double, true>(gy, ws, gx, starts, B, I, O, Ftot);
   }
+  DFNO_CHECK_LAUNCH("spectral");
 }
