@@ -938,6 +938,10 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
   }();
   // rows per wave: wider cuts B re-reads but costs registers/occupancy.
   // The glds3 OW=5 variant covers O<=20 in ONE o-tile so B is read once.
+  // For big-O shapes (gW3: O=128), wider OW (4: 9 GB, 6: 7.8 GB vs OW=2's
+  // 14 GB of o-tile-amplified traffic) measured NO faster: per-block
+  // throughput here scales with resident blocks (3/CU at the 48 KB OW=2
+  // ring vs 2/CU at 60-72 KB), which exactly offsets the traffic cut.
   const bool ow5 = vec && !no_glds && I > 8 && I <= 20 && O <= 20 && O >= 8;
   const int OW = ow5 ? 5 : ((O >= 8) ? ((I <= 8) ? 4 : 2) : 1);
   int o_tiles = (O + 4 * OW - 1) / (4 * OW);
