@@ -149,3 +149,4 @@ if __name__ == "__main__":
     bench(args.input_shape, args.partition_shape, args.width, args.modes,
           args.num_timesteps, args.device, args.num_gpus, args.benchmark_type,
           args.output_dir)
+    dfno.finalize_distributed()

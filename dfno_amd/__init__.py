@@ -14,6 +14,7 @@ normalisation helpers — plus the native comm primitives.
 from .partition import (
     Partition,
     init_distributed,
+    finalize_distributed,
     is_distributed,
     zero_volume_tensor,
     compute_distribution_info,

@@ -74,6 +74,20 @@ def init_distributed(backend: Optional[str] = None, timeout_s: int = 900) -> Non
     dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
 
 
+def finalize_distributed() -> None:
+    """Tear down the process group (and the cached sub-groups) at exit.
+
+    Safe to call when serial / already finalized.  Mirrors the reference's
+    implicit MPI_Finalize-at-exit; without it recent torch warns about a
+    leaked process group on interpreter shutdown.
+    """
+    if not is_distributed():
+        return
+    _GROUP_REGISTRY.clear()
+    dist.barrier()
+    dist.destroy_process_group()
+
+
 # Registry: member-rank tuple -> ProcessGroup (deduplicated across Partition
 # objects so each distinct rank set costs one communicator).
 _GROUP_REGISTRY: Dict[Tuple[int, ...], object] = {}

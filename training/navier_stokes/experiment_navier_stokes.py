@@ -224,3 +224,5 @@ for i in range(args.num_epochs):
                     plt.xlabel("Epoch")
                     plt.ylabel("Loss")
                     plt.savefig(out_dir / f"curves_{j:04d}.png")
+
+dfno.finalize_distributed()

@@ -109,3 +109,4 @@ def main():
 
 if __name__ == "__main__":
     main()
+    dfno.finalize_distributed()

@@ -167,3 +167,4 @@ def main():
 
 if __name__ == "__main__":
     main()
+    dfno.finalize_distributed()
