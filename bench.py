@@ -80,6 +80,10 @@ def main():
                                   dtype=torch.float32)
     criterion = dfno.DistributedRelativeLpLoss(P_x)
     from dfno_amd.optim import Adam as FusedAdam
+    if torch.cuda.is_available():
+        # fail loudly rather than silently timing a torch-eager fallback
+        from dfno_amd import _ext
+        _ext.get(required=True)
     optimizer = FusedAdam(model.parameters(), lr=1e-3)
 
     # synthetic local shards of the global tensors
