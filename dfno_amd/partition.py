@@ -84,7 +84,9 @@ def finalize_distributed() -> None:
     if not is_distributed():
         return
     _GROUP_REGISTRY.clear()
-    dist.barrier()
+    # no barrier here: on a world that never ran a collective it would
+    # initialize RCCL at exit and print its banner AFTER the bench's JSON
+    # line; destroy_process_group alone is sufficient and silent
     dist.destroy_process_group()
 
 
