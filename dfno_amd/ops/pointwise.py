@@ -27,6 +27,12 @@ import torch.nn.functional as F
 
 from .. import _ext
 
+
+def _native_dt(t: torch.Tensor) -> bool:
+    """Native kernels are fp32/fp64; other dtypes (bf16/fp16) take the
+    composed-torch path (still on GPU via rocBLAS/eager)."""
+    return t.dtype in (torch.float32, torch.float64)
+
 __all__ = ["linear_nd", "add_gelu", "gelu", "linear_res_gelu"]
 
 _SQRT_2 = math.sqrt(2.0)
@@ -50,7 +56,7 @@ class _ChannelMixFn(torch.autograd.Function):
         B, I = x.shape[0], x.shape[1]
         S = x.numel() // max(B * I, 1)
         x3 = x.reshape(B, I, S)
-        if x.is_cuda:
+        if x.is_cuda and _native_dt(x):
             ext = _ext.get(required=True)
             y3, z3 = ext.channel_mix_fwd(x3, W, b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device), act)
         else:
@@ -70,7 +76,7 @@ class _ChannelMixFn(torch.autograd.Function):
         act = ctx.act
         gy = gy.contiguous()
         if act:
-            if gy.is_cuda:
+            if gy.is_cuda and _native_dt(gy):
                 ext = _ext.get(required=True)
                 gz = ext.gelu_bwd(gy, z3)
             else:
@@ -78,13 +84,13 @@ class _ChannelMixFn(torch.autograd.Function):
         else:
             gz = gy
         # grad x: contraction with W^T
-        if gy.is_cuda:
+        if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             gx = ext.channel_mix_fwd_t(gz, W)  # sum_o W[o,i] gz[b,o,s]
         else:
             gx = torch.einsum("oi,bos->bis", W, gz)
         # grad W / b
-        if gy.is_cuda and x3.shape[1] <= 32:
+        if gy.is_cuda and _native_dt(gy) and x3.shape[1] <= 32:
             ext = _ext.get(required=True)
             gW, gb = ext.channel_mix_bwd_w(gz.contiguous(), x3, ctx.has_bias)
             if not ctx.has_bias:
@@ -151,7 +157,7 @@ class _LinearResGeluFn(torch.autograd.Function):
         S = x.numel() // max(B * I, 1)
         x3 = x.reshape(B, I, S)
         r3 = res.reshape(B, W.shape[0], S)
-        if x.is_cuda:
+        if x.is_cuda and _native_dt(x):
             ext = _ext.get(required=True)
             y3, z3 = ext.linear_res_gelu_fwd(x3.contiguous(), W, r3.contiguous())
         else:
@@ -166,7 +172,7 @@ class _LinearResGeluFn(torch.autograd.Function):
     def backward(ctx, gy):
         x3, W, z3 = ctx.saved_tensors
         gy = gy.contiguous()
-        if gy.is_cuda:
+        if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             gz = ext.gelu_bwd(gy, z3)
             gx = ext.channel_mix_fwd_t(gz, W)
@@ -196,7 +202,7 @@ def linear_res_gelu(x: torch.Tensor, W: torch.Tensor, res: torch.Tensor) -> torc
 class _AddGeluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, a, bt):
-        if a.is_cuda:
+        if a.is_cuda and _native_dt(a):
             ext = _ext.get(required=True)
             y, z = ext.add_gelu_fwd(a.contiguous(), bt.contiguous())
         else:
@@ -209,7 +215,7 @@ class _AddGeluFn(torch.autograd.Function):
     def backward(ctx, gy):
         (z,) = ctx.saved_tensors
         gy = gy.contiguous()
-        if gy.is_cuda:
+        if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             gz = ext.gelu_bwd(gy, z)
         else:
@@ -225,7 +231,7 @@ def add_gelu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
 class _GeluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
-        if x.is_cuda:
+        if x.is_cuda and _native_dt(x):
             ext = _ext.get(required=True)
             y = ext.gelu_fwd(x.contiguous())
         else:
@@ -237,7 +243,7 @@ class _GeluFn(torch.autograd.Function):
     def backward(ctx, gy):
         (x,) = ctx.saved_tensors
         gy = gy.contiguous()
-        if gy.is_cuda:
+        if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             return ext.gelu_bwd(gy, x)
         return gy * _gelu_grad(x)

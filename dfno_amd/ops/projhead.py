@@ -51,7 +51,7 @@ def proj_head(x: torch.Tensor, W3, b3, W4, b4) -> torch.Tensor:
     """
     b3f = b3.reshape(-1)
     b4f = b4.reshape(-1)
-    if x.is_cuda:
+    if x.is_cuda and x.dtype in (torch.float32, torch.float64):
         out3 = _ProjHeadFn.apply(x, W3, b3f, W4, b4f)
         out_shape = list(x.shape)
         out_shape[1] = W4.shape[0]

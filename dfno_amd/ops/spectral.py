@@ -41,7 +41,7 @@ class _SpectralConvFn(torch.autograd.Function):
         B, I = x.shape[0], x.shape[1]
         fdims = list(x.shape[2:])
         y = torch.zeros((B, out_channels, *fdims), dtype=x.dtype, device=x.device)
-        if x.is_cuda and x.numel() > 0:
+        if x.is_cuda and x.numel() > 0 and x.dtype in (torch.complex64, torch.complex128):
             ext = _ext.get(required=True)
             xc = x.contiguous()
             ext.spectral_corners_fwd(
@@ -65,7 +65,7 @@ class _SpectralConvFn(torch.autograd.Function):
         gy = gy.contiguous()
         gx = torch.zeros_like(x)
         gws = []
-        if gy.is_cuda and gy.numel() > 0:
+        if gy.is_cuda and gy.numel() > 0 and gy.dtype in (torch.complex64, torch.complex128):
             ext = _ext.get(required=True)
             ext.spectral_corners_bwd_x(
                 gy, [w.contiguous() for w in weights], gx,
