@@ -55,6 +55,10 @@ use_cuda, _, _, device, _ = dfno.get_env(P_x)
 torch.manual_seed(P_x.rank + 123)
 np.random.seed(P_x.rank + 123)
 
+# the reference runs NS training under autograd anomaly detection
+# (experiment_navier_stokes.py:54); kept for behavioral parity
+torch.set_anomaly_enabled(True)
+
 # root-generated run timestamp broadcast to all ranks (reference :50-52)
 B = dfno.Broadcast(P_0, P_x)
 ts = torch.tensor([float(int(time.time()))], device=device) if P_0.active else \
