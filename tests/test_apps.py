@@ -107,6 +107,23 @@ def test_train_two_phase_serial(tmp_path):
     assert "training finished." in r.stdout
 
 
+def test_train_two_phase_resume(tmp_path):
+    common = [str(REPO / "training/two_phase/train_two_phase.py"),
+              "--data", "synthetic", "--shape", "8", "8", "4", "6",
+              "--width", "4", "--modes", "2", "2", "2", "2",
+              "--num-train", "2", "--num-valid", "1",
+              "--checkpoint-interval", "1", "--out-dir", str(tmp_path)]
+    run_script(common + ["--num-epochs", "1"], cwd=str(REPO))
+    assert (tmp_path / "train_state_0001_0000.pt").exists()
+    r = run_script(common + ["--num-epochs", "2", "--resume", str(tmp_path)],
+                   cwd=str(REPO))
+    # resumed run starts at epoch 1 (0-indexed) and finishes epoch 2
+    assert "resumed epoch 1" in r.stdout
+    assert "epoch = 1, batch = 0" in r.stdout
+    assert "epoch = 0, batch" not in r.stdout
+    assert (tmp_path / "model_0002_0000.pt").exists()
+
+
 def test_train_and_test_two_phase_2rank(tmp_path):
     run_script([str(REPO / "training/two_phase/train_two_phase.py"),
                 "--data", "synthetic", "--shape", "8", "8", "4", "6",

@@ -43,6 +43,12 @@ def main():
     p.add_argument("--num-valid", type=int, default=200)
     p.add_argument("--num-epochs", type=int, default=100)
     p.add_argument("--checkpoint-interval", type=int, default=10)
+    p.add_argument("--resume", type=str, default=None, metavar="DIR",
+                   help="resume from DIR's latest train_state_*.pt (model + "
+                        "optimizer + epoch; per-rank sharded, same partition "
+                        "shape required). The reference cannot resume "
+                        "training (optimizer state never saved, SURVEY.md "
+                        "section 5); this extends it.")
     p.add_argument("--batch-size", type=int, default=1)
     p.add_argument("--width", type=int, default=20)
     p.add_argument("--modes", type=int, nargs=4, default=(12, 12, 12, 8))
@@ -103,7 +109,24 @@ def main():
         out_dir.mkdir(parents=True, exist_ok=True)
     train_accs, valid_accs = [], []
 
-    for i in range(args.num_epochs):
+    start_epoch = 0
+    if args.resume:
+        rdir = Path(args.resume)
+        rk = max(P_x.rank, 0)
+        cands = sorted(rdir.glob(f"train_state_*_{rk:04d}.pt"))
+        if not cands:
+            raise FileNotFoundError(
+                f"rank {rk}: no train_state_*_{rk:04d}.pt under {rdir}")
+        state = torch.load(cands[-1], map_location=device, weights_only=False)
+        model.load_state_dict(state["model"])
+        optimizer.load_state_dict(state["optimizer"])
+        start_epoch = int(state["epoch"])
+        train_accs = list(state.get("train_accs", []))
+        valid_accs = list(state.get("valid_accs", []))
+        print(f"rank = {P_x.rank}, resumed epoch {start_epoch} from {cands[-1]}")
+    P_x.barrier()
+
+    for i in range(start_epoch, args.num_epochs):
         model.train()
         train_loss, n_train_batch = 0.0, 0
         for j, (x, y) in enumerate(train_loader):
@@ -157,6 +180,13 @@ def main():
             path = out_dir / f"model_{i + 1:04d}_{max(P_x.rank, 0):04d}.pt"
             torch.save(model.state_dict(), path)
             print(f"rank = {P_x.rank}, saved model: {path}")
+            # full training state for --resume (model file above keeps the
+            # reference's exact checkpoint layout; this adds optimizer+epoch)
+            torch.save({"model": model.state_dict(),
+                        "optimizer": optimizer.state_dict(),
+                        "epoch": i + 1,
+                        "train_accs": train_accs, "valid_accs": valid_accs},
+                       out_dir / f"train_state_{i + 1:04d}_{max(P_x.rank, 0):04d}.pt")
 
     path = out_dir / f"model_{max(P_x.rank, 0):04d}.pt"
     torch.save(model.state_dict(), path)
