@@ -104,6 +104,12 @@ class _BroadcastFn(torch.autograd.Function):
             meta = obj[0]
             module._meta = meta
         shape, dtype = meta
+        if is_root and (tuple(x.shape) != shape or x.dtype != dtype):
+            raise RuntimeError(
+                f"Broadcast: payload changed from {shape}/{dtype} to "
+                f"{tuple(x.shape)}/{x.dtype}; a Broadcast module is bound to "
+                "one payload signature (non-root ranks cache it) - build a "
+                "new module for a different tensor")
         if is_root:
             buf = _comm_clone(x)
         else:
