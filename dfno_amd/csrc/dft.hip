@@ -591,6 +591,107 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_syn_kernel(
 // R2C (last dim): out[l, k] = fac_k * scale * sum_j in[l, j] * w^{-jk}
 // ---------------------------------------------------------------------------
 
+// glds double-buffered variant (fp32): the cooperative load->ds_write stage
+// serializes against the compute phase (probe: full 0.40 ms vs 0.19 ms for
+// stage+write alone at the flagship shape; phase-skewing co-resident blocks
+// changes nothing).  Staging tile t+1 by global_load_lds DMA while t is
+// being consumed, with a counted s_waitcnt vmcnt(NG) + raw barriers, runs
+// the same shape at 5.4 TB/s -- 2.3x the staged version (r2cprobe.hip).
+template <typename T, int MCAP, int NG, int NT = 0>
+__global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
+    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
+    long lines, int N_, int m, T scale, bool factors) {
+  // NT > 0 pins the transform length AND the mode count (m == MCAP, host-
+  // checked) at compile time: the j-loop fully unrolls with all twiddle
+  // offsets folded to immediates, batching the s_loads and exposing the fma
+  // ILP.  With runtime N/m the loop serializes on per-iteration scalar-load
+  // waits (0.42 ms) or, unrolled with runtime-m addressing, spills SGPRs;
+  // the folded version runs 0.18 ms at the flagship shape (r2cprobe.hip).
+  const int N = NT > 0 ? NT : N_;
+  const int mm = NT > 0 ? MCAP : m;
+  if constexpr (!std::is_same<T, float>::value) return;  // float-only path
+  extern __shared__ __align__(16) char smem_raw[];
+  float* ring = reinterpret_cast<float*>(smem_raw);  // [2][kBlock * N]
+  const int tilef = kBlock * N;                      // floats per tile
+  // the counted s_waitcnt needs a compile-time-exact per-wave instruction
+  // count: host picks NG = ceil(N/4) rounded up to a supported value and
+  // issues are padded to exactly NG glds (pad slots re-stage the tile's
+  // last 16 bytes; a whole-wave all-lanes-same-address pad instruction is
+  // NOT free, so NG tracks N instead of a single worst case)
+  const int wave = (int)(threadIdx.x / 64), lane = (int)(threadIdx.x % 64);
+  const long nfl = lines * N;
+  long ntiles = (lines + kBlock - 1) / kBlock;
+
+  auto issue = [&](int buf, long tb) {
+    const float* src = reinterpret_cast<const float*>(in);
+    const long base = tb * (long)tilef;
+    float* dst = ring + (long)buf * tilef;
+#pragma unroll
+    for (int k = 0; k < NG; ++k) {
+      long fo = (long)(wave * 64 + k * kBlock + lane) * 4;
+      if (fo + 4 > tilef) fo = tilef - 4;            // pad: clamp in-tile
+      long gfo = base + fo;
+      if (gfo + 4 > nfl) gfo = nfl - 4;              // clamp last tile
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)(src + gfo),
+          (__attribute__((address_space(3))) void*)&dst[fo], 16, 0, 0);
+    }
+  };
+
+  if (0 < ntiles) issue(0, blockIdx.x < ntiles ? blockIdx.x : 0);
+  long c = 0;
+  for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x, ++c) {
+    long nxt = tb + gridDim.x;
+    if (nxt < ntiles) {
+      issue((int)((c + 1) & 1), nxt);
+      asm volatile("s_waitcnt vmcnt(%0)" ::"n"(NG) : "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+
+    const float* tile = ring + (long)(c & 1) * tilef;
+    const float* src = tile + threadIdx.x * N;
+    T ar[MCAP], ai[MCAP];
+#pragma unroll
+    for (int k = 0; k < MCAP; ++k) {
+      if (k < mm) { ar[k] = T(0); ai[k] = T(0); }
+    }
+#pragma unroll
+    for (int j = 0; j < (NT > 0 ? NT : 64); ++j) {
+      if (NT == 0 && j >= N) break;
+      const T x = (T)src[j];
+      auto twj = (const __attribute__((address_space(4))) T*)
+          (tw + (long)(j * 2) * mm);
+#pragma unroll
+      for (int k = 0; k < MCAP; ++k) {
+        if (k < mm) {
+          ar[k] += x * twj[2 * k];
+          ai[k] += x * twj[2 * k + 1];
+        }
+      }
+    }
+    long l0 = tb * kBlock;
+    if (l0 + (long)threadIdx.x < lines) {
+      T* dst = out + 2 * (l0 + threadIdx.x) * mm;
+#pragma unroll
+      for (int k = 0; k < MCAP; ++k) {
+        if (k < mm) {
+          T f = T(1);
+          bool edge = (k == 0) || (N % 2 == 0 && 2 * k == N);
+          if (factors && !edge) f = T(2);
+          dst[2 * k] = f * scale * ar[k];
+          dst[2 * k + 1] = (factors && edge) ? T(0) : f * scale * ai[k];
+        }
+      }
+    }
+    // all waves out of ring[c&1] before its re-issue next iteration
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+  }
+}
+
 // One 256-line tile is cooperatively staged into LDS with coalesced loads
 // (per-thread direct line reads thrash L1: 16 waves x line-span > 32 KiB,
 // measured SQ_WAIT_ANY = 72%); threads then stream their line from LDS.
@@ -923,14 +1024,39 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
   int grid = (int)std::min(ntiles, 4096L);
   auto tw = twiddle_table(N, (int)m, /*analysis=*/true,
                           x.options().dtype(x.scalar_type()));
+#define R2CG(MC, NGV, NTV)                                                     \
+      hipLaunchKernelGGL((dft_r2c_glds_kernel<scalar_t, MC, NGV, NTV>),        \
+                         dim3(grid), dim3(kBlock), smem2, stream, inp, op,     \
+                         tw.data_ptr<scalar_t>(), lines, N, (int)m,            \
+                         (scalar_t)scale, factors);
+#define R2CG_M(NGV, NTV)                                                       \
+      if (m <= 8) { R2CG(8, NGV, NTV) } else if (m <= 16) { R2CG(16, NGV, NTV) } \
+      else if (m <= 24) { R2CG(24, NGV, NTV) } else { R2CG(32, NGV, NTV) }
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "dft_r2c", [&] {
     size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = x.data_ptr<scalar_t>();
     auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
-    DFT_MDISPATCH(dft_r2c_last_kernel, dim3(grid), dim3(kBlock), smem, stream,
-                  inp, op, tw.data_ptr<scalar_t>(), lines, N, (int)m,
-                  (scalar_t)scale, factors)
+    bool glds_ok = std::is_same<scalar_t, float>::value &&
+                   ((reinterpret_cast<uintptr_t>(inp) & 15) == 0) &&
+                   ((long)kBlock * N) % 4 == 0 && lines * N >= 4;
+    if (glds_ok) {
+      size_t smem2 = 2 * smem;   // double-buffered ring
+      // common N with m on an MCAP boundary get the fully-folded fast path
+      bool mexact = (m == 8 || m == 16 || m == 24 || m == 32);
+      if (mexact && N == 30) { R2CG_M(8, 30) }
+      else if (mexact && N == 64) { R2CG_M(16, 64) }
+      else if (mexact && N == 40) { R2CG_M(16, 40) }
+      else if (mexact && N == 32) { R2CG_M(8, 32) }
+      else if (N <= 32) { R2CG_M(8, 0) }   // ceil(N/4) <= 8 glds per wave
+      else { R2CG_M(16, 0) }
+    } else {
+      DFT_MDISPATCH(dft_r2c_last_kernel, dim3(grid), dim3(kBlock), smem,
+                    stream, inp, op, tw.data_ptr<scalar_t>(), lines, N,
+                    (int)m, (scalar_t)scale, factors)
+    }
   });
+#undef R2CG_M
+#undef R2CG
   DFNO_CHECK_LAUNCH("dft_r2c");
   return out;
 }
