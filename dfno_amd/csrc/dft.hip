@@ -773,10 +773,14 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
 
 // Output lines are written through an LDS tile and stored cooperatively
 // (coalesced); inputs are 64-128B contiguous per line and read directly.
-template <typename T, int MCAP>
+template <typename T, int MCAP, int NT = 0>
 __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
-    long lines, int N, int m, T scale, bool factors) {
+    long lines, int N_, int m_, T scale, bool factors) {
+  // NT > 0 pins N and m (== MCAP) at compile time: full unroll + folded
+  // twiddle offsets, as in dft_r2c_glds_kernel
+  const int N = NT > 0 ? NT : N_;
+  const int m = NT > 0 ? MCAP : m_;
   extern __shared__ __align__(16) char smem_raw[];
   T* tile = reinterpret_cast<T*>(smem_raw);   // [kBlock * N]
 
@@ -804,10 +808,12 @@ __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
         }
       }
       T* dst = tile + threadIdx.x * N;
-      for (int j = 0; j < N; ++j) {
+#pragma unroll
+      for (int j = 0; j < (NT > 0 ? NT : 64); ++j) {
+        if (NT == 0 && j >= N) break;
         T sacc = T(0);
         auto twj = (const __attribute__((address_space(4))) T*)
-            (tw + (long)j * 2 * m);            // lane-uniform -> s_load
+            (tw + (long)(j * 2) * m);          // lane-uniform -> s_load
 #pragma unroll
         for (int k = 0; k < MCAP; ++k) {
           if (k < m) {
@@ -1083,14 +1089,33 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
   int grid = (int)std::min(ntiles, 4096L);
   auto tw = twiddle_table(N, m, /*analysis=*/false,
                           out.options().dtype(out.scalar_type()));
+#define C2RG(MC, NTV)                                                          \
+      hipLaunchKernelGGL((dft_c2r_last_kernel<scalar_t, MC, NTV>),             \
+                         dim3(grid), dim3(kBlock), smem, stream, inp, op,      \
+                         tw.data_ptr<scalar_t>(), lines, N, (int)m,            \
+                         (scalar_t)scale, factors);
   AT_DISPATCH_FLOATING_TYPES(out.scalar_type(), "dft_c2r", [&] {
     size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = reinterpret_cast<const scalar_t*>(y.data_ptr());
     auto op = out.data_ptr<scalar_t>();
-    DFT_MDISPATCH(dft_c2r_last_kernel, dim3(grid), dim3(kBlock), smem, stream,
-                  inp, op, tw.data_ptr<scalar_t>(), lines, N, (int)m,
-                  (scalar_t)scale, factors)
+    bool mexact = std::is_same<scalar_t, float>::value &&
+                  (m == 8 || m == 16 || m == 24 || m == 32);
+    if (mexact && N == 30) {
+      if (m == 8) { C2RG(8, 30) } else if (m == 16) { C2RG(16, 30) }
+      else if (m == 24) { C2RG(24, 30) } else { C2RG(32, 30) }
+    } else if (mexact && N == 64) {
+      if (m == 8) { C2RG(8, 64) } else if (m == 16) { C2RG(16, 64) }
+      else if (m == 24) { C2RG(24, 64) } else { C2RG(32, 64) }
+    } else if (mexact && N == 40) {
+      if (m == 8) { C2RG(8, 40) } else if (m == 16) { C2RG(16, 40) }
+      else if (m == 24) { C2RG(24, 40) } else { C2RG(32, 40) }
+    } else {
+      DFT_MDISPATCH(dft_c2r_last_kernel, dim3(grid), dim3(kBlock), smem,
+                    stream, inp, op, tw.data_ptr<scalar_t>(), lines, N,
+                    (int)m, (scalar_t)scale, factors)
+    }
   });
+#undef C2RG
   DFNO_CHECK_LAUNCH("dft_c2r");
   return out;
 }
