@@ -40,11 +40,18 @@ __device__ __forceinline__ T wave_sum(T v) {
   return v;
 }
 
-template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR>
+template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR,
+          int MT = 0, int IT = 0>
 __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ W3l, const T* __restrict__ b3l,
     const T* __restrict__ W4l, const T* __restrict__ b4l, T* __restrict__ out,
-    int B, int I, int M, int O2, long S) {
+    int B, int I_, int M_, int O2, long S) {
+  // MT/IT > 0 pin the hidden width and input channels at compile time: the
+  // M-loop fully unrolls with weight offsets folded into the scalar loads
+  // (runtime-M serializes a scalar-load wait per hidden unit; the same fix
+  // bought 1.9x on the r2c kernels)
+  const int I = IT > 0 ? IT : I_;
+  const int M = MT > 0 ? MT : M_;
   // weights are read straight through the kernel-arg pointers: every access
   // index is wave-uniform, so they lower to scalar loads (s-cache) instead
   // of per-MAC LDS reads that double the issue count.
@@ -89,7 +96,9 @@ __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
           }
     }
 
-    for (int j = 0; j < M; ++j) {
+#pragma unroll 8
+    for (int j = 0; j < (MT > 0 ? MT : 512); ++j) {
+      if (MT == 0 && j >= M) break;
       T zk[VEC];
       T bj = b3l[j];
 #pragma unroll
@@ -135,14 +144,17 @@ __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
 
 // Backward: one pass producing gx, gb3, gW4, gb4 and materializing gz3
 // (grad wrt z3 = pre-gelu hidden) for the grad-W3 library GEMM.
-template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR>
+template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR,
+          int MT = 0, int IT = 0>
 __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ W3, const T* __restrict__ b3,
     const T* __restrict__ W4,
     T* __restrict__ gz3,
     T* __restrict__ gb3, T* __restrict__ gW4, T* __restrict__ gb4,
-    int B, int I, int M, int O2, long S) {
+    int B, int I_, int M_, int O2, long S) {
+  const int I = IT > 0 ? IT : I_;     // see fwd kernel note
+  const int M = MT > 0 ? MT : M_;
   extern __shared__ __align__(16) char smem_raw[];
   T* W3l = reinterpret_cast<T*>(smem_raw);   // [M*I]
   T* b3l = W3l + (size_t)M * I;              // [M]
@@ -222,7 +234,9 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
 
 
     T* gz3b = gz3 + ((long)b * M) * S + s;
-    for (int j = 0; j < M; ++j) {
+#pragma unroll 8
+    for (int j = 0; j < (MT > 0 ? MT : 512); ++j) {
+      if (MT == 0 && j >= M) break;
       // recompute z3 and gelu pieces
       T zk[VEC];
       T bj = b3l[j];
@@ -348,6 +362,13 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_p((long)B * ((S + 3) / 4));
 
+#define PH_LAUNCH_FT(V)                                                       \
+    hipLaunchKernelGGL((proj_head_fwd_kernel<scalar_t, IM, OM, 4, V, 128, 20>), \
+                       dim3(grid), dim3(kBlock), smem, stream,                \
+                       x.data_ptr<scalar_t>(), W3.data_ptr<scalar_t>(),       \
+                       b3.data_ptr<scalar_t>(), W4.data_ptr<scalar_t>(),      \
+                       b4.data_ptr<scalar_t>(), out.data_ptr<scalar_t>(),     \
+                       B, I, M, O2, S);
 #define PH_LAUNCH_F(V)                                                        \
     hipLaunchKernelGGL((proj_head_fwd_kernel<scalar_t, IM, OM, 4, V>),        \
                        dim3(grid), dim3(kBlock), smem, stream,                \
@@ -358,7 +379,10 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "proj_head_fwd", [&] {
     size_t smem = 0;
     bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), out.data_ptr()});
-    if (I <= 24 && O2 <= 2) {
+    if (I == 20 && M == 128 && O2 <= 2) {
+      constexpr int IM = 24, OM = 2;   // flagship: fold M/I at compile time
+      if (vec) { PH_LAUNCH_FT(true) } else { PH_LAUNCH_FT(false) }
+    } else if (I <= 24 && O2 <= 2) {
       constexpr int IM = 24, OM = 2;
       if (vec) { PH_LAUNCH_F(true) } else { PH_LAUNCH_F(false) }
     } else {
@@ -390,6 +414,15 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_p((long)B * ((S + 3) / 4));
 
+#define PH_LAUNCH_BT(V)                                                       \
+    hipLaunchKernelGGL((proj_head_bwd_kernel<scalar_t, IM, OM, 4, V, 128, 20>), \
+                       dim3(grid), dim3(kBlock), smem, stream,                \
+                       gy.data_ptr<scalar_t>(), x.data_ptr<scalar_t>(),       \
+                       W3.data_ptr<scalar_t>(), b3.data_ptr<scalar_t>(),      \
+                       W4.data_ptr<scalar_t>(), gz3.data_ptr<scalar_t>(),     \
+                       gb3.data_ptr<scalar_t>(),                              \
+                       gW4.data_ptr<scalar_t>(), gb4.data_ptr<scalar_t>(),    \
+                       B, I, M, O2, S);
 #define PH_LAUNCH_B(V)                                                        \
     hipLaunchKernelGGL((proj_head_bwd_kernel<scalar_t, IM, OM, 4, V>),        \
                        dim3(grid), dim3(kBlock), smem, stream,                \
@@ -404,7 +437,10 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
         ((size_t)M * I + M + (size_t)O2 * M + 4 * (size_t)M + 4 * (size_t)O2 * M);
     TORCH_CHECK(smem <= 160 * 1024, "proj_head_bwd: LDS overflow");
     bool vec = vec_ok<scalar_t>(S, {x.data_ptr(), gy.data_ptr(), gz3.data_ptr()});
-    if (I <= 24 && O2 <= 2) {
+    if (I == 20 && M == 128 && O2 <= 2) {
+      constexpr int IM = 24, OM = 2;   // flagship fold
+      if (vec) { PH_LAUNCH_BT(true) } else { PH_LAUNCH_BT(false) }
+    } else if (I <= 24 && O2 <= 2) {
       constexpr int IM = 24, OM = 2;
       if (vec) { PH_LAUNCH_B(true) } else { PH_LAUNCH_B(false) }
     } else {
