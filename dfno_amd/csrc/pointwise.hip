@@ -108,12 +108,15 @@ __global__ void add_gelu_kernel(const T* __restrict__ a, const T* __restrict__ b
 // W (O x I or transposed) staged in LDS; reads are wave-uniform broadcasts.
 // ---------------------------------------------------------------------------
 
-template <typename T, int IMAX, int VEC, bool ACT, bool VECTOR>
+template <typename T, int IMAX, int VEC, bool ACT, bool VECTOR,
+          int IT = 0, int OT = 0>
 __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
     const T* __restrict__ x, const T* __restrict__ W, const T* __restrict__ bias,
     T* __restrict__ y, T* __restrict__ z,
-    int B, int I, int O, long S, bool wt, bool has_bias, bool write_z,
+    int B, int I_, int O_, long S, bool wt, bool has_bias, bool write_z,
     const T* __restrict__ res = nullptr) {
+  const int I = IT > 0 ? IT : I_;     // see ores note
+  const int O = OT > 0 ? OT : O_;
   extern __shared__ __align__(16) char smem_raw[];
   T* Wl = reinterpret_cast<T*>(smem_raw);        // [O*I]
   T* bl = Wl + (size_t)O * I;                     // [O]
@@ -155,7 +158,9 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
 
     T* yb = y + ((long)b * O) * S + s;
     T* zb = write_z ? z + ((long)b * O) * S + s : nullptr;
-    for (int o = 0; o < O; ++o) {
+#pragma unroll 4
+    for (int o = 0; o < (OT > 0 ? OT : 512); ++o) {
+      if (OT == 0 && o >= O) break;
       T acc[VEC];
       T bv = has_bias ? bl[o] : T(0);
 #pragma unroll
@@ -203,11 +208,17 @@ __global__ __launch_bounds__(kBlock) void channel_mix_xres_kernel(
 // Used for the projection head 128 -> 1 and for grad-x of the lift 1 -> C.
 // ---------------------------------------------------------------------------
 
-template <typename T, int OMAX, int VEC, bool ACT, bool VECTOR>
+template <typename T, int OMAX, int VEC, bool ACT, bool VECTOR,
+          int IT = 0, int OT = 0>
 __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
     const T* __restrict__ x, const T* __restrict__ W, const T* __restrict__ bias,
     T* __restrict__ y, T* __restrict__ z,
-    int B, int I, int O, long S, bool wt, bool has_bias, bool write_z) {
+    int B, int I_, int O_, long S, bool wt, bool has_bias, bool write_z) {
+  // IT/OT > 0 pin the channel counts at compile time: the I-stream loop
+  // fully unrolls with folded LDS offsets (see proj_head / dft note: the
+  // runtime-bound loop serializes on a per-iteration uniform-load wait)
+  const int I = IT > 0 ? IT : I_;
+  const int O = OT > 0 ? OT : O_;
   extern __shared__ __align__(16) char smem_raw[];
   T* Wl = reinterpret_cast<T*>(smem_raw);
   T* bl = Wl + (size_t)O * I;
@@ -237,7 +248,9 @@ __global__ __launch_bounds__(kBlock) void channel_mix_ores_kernel(
     }
 
     const T* xb = x + ((long)b * I) * S + s;
-    for (int i = 0; i < I; ++i) {
+#pragma unroll 8
+    for (int i = 0; i < (IT > 0 ? IT : 512); ++i) {
+      if (IT == 0 && i >= I) break;
       T xv[VEC];
       if constexpr (VECTOR && std::is_same<T, float>::value) {
         const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
@@ -364,14 +377,33 @@ void launch_channel_mix(const T* x, const T* W, const T* bias, T* y, T* z,
   size_t smem = sizeof(T) * ((size_t)O * I + O);
   const bool vec = can_vectorize(x, y, z, S, write_z);
 
+  // fold the flagship channel counts at compile time (see kernel note)
+  const bool f2020 = (I == 20 && O == 20);
+  const bool f12820 = (I == 128 && O == 20);
 #define CMIX_LAUNCH(KERNEL, CAP, A, V)                                          \
-  hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V>), dim3(grid), dim3(kBlock),       \
-                     smem, stream, x, W, bias, y, z, B, I, O, S, wt,            \
-                     has_bias, write_z);
+  if (f2020) {                                                                  \
+    hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V, 20, 20>), dim3(grid),           \
+                       dim3(kBlock), smem, stream, x, W, bias, y, z, B, I, O,   \
+                       S, wt, has_bias, write_z);                               \
+  } else if (f12820) {                                                          \
+    hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V, 128, 20>), dim3(grid),          \
+                       dim3(kBlock), smem, stream, x, W, bias, y, z, B, I, O,   \
+                       S, wt, has_bias, write_z);                               \
+  } else {                                                                      \
+    hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V>), dim3(grid), dim3(kBlock),     \
+                       smem, stream, x, W, bias, y, z, B, I, O, S, wt,          \
+                       has_bias, write_z);                                      \
+  }
 #define CMIX_LAUNCH_RES(KERNEL, CAP, A, V)                                      \
-  hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V>), dim3(grid), dim3(kBlock),       \
-                     smem, stream, x, W, bias, y, z, B, I, O, S, wt,            \
-                     has_bias, write_z, res);
+  if (f2020) {                                                                  \
+    hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V, 20, 20>), dim3(grid),           \
+                       dim3(kBlock), smem, stream, x, W, bias, y, z, B, I, O,   \
+                       S, wt, has_bias, write_z, res);                          \
+  } else {                                                                      \
+    hipLaunchKernelGGL((KERNEL<T, CAP, 4, A, V>), dim3(grid), dim3(kBlock),     \
+                       smem, stream, x, W, bias, y, z, B, I, O, S, wt,          \
+                       has_bias, write_z, res);                                 \
+  }
 #define CMIX_DISPATCH(KERNEL, CAP)                                              \
   if (act) {                                                                    \
     if (vec) { CMIX_LAUNCH(KERNEL, CAP, true, true) }                           \
