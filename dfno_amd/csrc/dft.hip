@@ -439,10 +439,13 @@ __device__ __forceinline__ void dft8(const T* xr, const T* xi, T* Ar, T* Ai) {
   }
 }
 
-template <typename T, int LCAP>
+template <typename T, int LCAP, int MLT = 0>
 __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_ana_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
-    long outer, long inner, int m_lo, T scale) {
+    long outer, long inner, int m_lo_, T scale) {
+  // MLT > 0 pins m_lo at compile time (folded table strides / no
+  // predicates; see the r2c fast-path note)
+  const int m_lo = MLT > 0 ? MLT : m_lo_;
   constexpr int N = 64;
   const int nch = m_lo + 1;
   long pairs = inner / 2;
@@ -520,10 +523,13 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_ana_kernel(
 // Radix-8x8 synthesis: X[8a+b] = sum_r w8^{+ar} C_b[r] with C_b[r] the
 // per-residue accumulation of the kept modes times w64^{+bk} (prefix) /
 // conj(w64^{+b kp}) (suffix).  tw here is the synthesis-signed table.
-template <typename T, int LCAP>
+template <typename T, int LCAP, int MLT = 0>
 __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_syn_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
-    long outer, long inner, int m_lo, T scale) {
+    long outer, long inner, int m_lo_, T scale) {
+  // MLT > 0 pins m_lo at compile time (folded table strides / no
+  // predicates; see the r2c fast-path note)
+  const int m_lo = MLT > 0 ? MLT : m_lo_;
   constexpr int N = 64;
   const int nch = m_lo + 1;
   const int m = 2 * m_lo;
@@ -952,7 +958,12 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
     if (paired && analysis && n == 64 &&
         c10::toRealValueType(x.scalar_type()) == at::kFloat && !no_radix) {
       // radix-8x8 path (see kernel comment); grid2 already sized for pairs
-      if (m_lo <= 8) {
+      if (m_lo == 12) {       // flagship: fully folded
+        hipLaunchKernelGGL((dft_c2c_radix8_ana_kernel<scalar_t, 13, 12>),
+                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
+                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
+                           (scalar_t)scale);
+      } else if (m_lo <= 8) {
         hipLaunchKernelGGL((dft_c2c_radix8_ana_kernel<scalar_t, 9>),
                            dim3(grid2), dim3(kBlock), 0, stream, inp, op,
                            tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
@@ -975,7 +986,12 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
     } else if (paired && !analysis && n == 64 &&
                c10::toRealValueType(x.scalar_type()) == at::kFloat &&
                !no_radix) {
-      if (m_lo <= 8) {
+      if (m_lo == 12) {       // flagship: fully folded
+        hipLaunchKernelGGL((dft_c2c_radix8_syn_kernel<scalar_t, 13, 12>),
+                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
+                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
+                           (scalar_t)scale);
+      } else if (m_lo <= 8) {
         hipLaunchKernelGGL((dft_c2c_radix8_syn_kernel<scalar_t, 9>),
                            dim3(grid2), dim3(kBlock), 0, stream, inp, op,
                            tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,

@@ -40,7 +40,7 @@ __device__ __forceinline__ T lh_wave_sum(T v) {
 
 // x: [B, C, S] (T_in == 1 folded away), out: [B, W, S, T]
 // W1: [T, 1], b1: [T], W2: [W, C], b2: [W]
-template <typename T, int CCAP, int TCAP, int WCAP>
+template <typename T, int CCAP, int TCAP, int WCAP, int WT = 0>
 __global__ __launch_bounds__(kBlock) void lift_head_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ W1, const T* __restrict__ b1,
     const T* __restrict__ W2, const T* __restrict__ b2, T* __restrict__ out,
@@ -73,7 +73,9 @@ __global__ __launch_bounds__(kBlock) void lift_head_fwd_kernel(
       }
     }
 
-    for (int w = 0; w < W; ++w) {
+    #pragma unroll 4
+    for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
+      if (WT == 0 && w >= W) break;
       T acc[TCAP];
 #pragma unroll
       for (int k = 0; k < TCAP; ++k)
@@ -96,7 +98,7 @@ __global__ __launch_bounds__(kBlock) void lift_head_fwd_kernel(
 }
 
 // gy: [B, W, S, T]; outputs gx [B, C, S] and gW1/gb1/gW2/gb2 via atomics.
-template <typename T, int CCAP, int TCAP, int WCAP>
+template <typename T, int CCAP, int TCAP, int WCAP, int WT = 0>
 __global__ __launch_bounds__(kBlock) void lift_head_bwd_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ W1, const T* __restrict__ b1,
@@ -147,7 +149,9 @@ __global__ __launch_bounds__(kBlock) void lift_head_bwd_kernel(
       }
     }
 
-    for (int w = 0; w < W; ++w) {
+    #pragma unroll 4
+    for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
+      if (WT == 0 && w >= W) break;
       const T* gyp = gy + ((b * W + w) * S + s) * Tn;
       T gz2[TCAP];
       T gb2p = T(0);
@@ -270,11 +274,19 @@ at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_l((long)B * S);
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "lift_head_fwd", [&] {
-    hipLaunchKernelGGL((lift_head_fwd_kernel<scalar_t, 4, 32, 24>), dim3(grid),
-                       dim3(kBlock), 0, stream, x.data_ptr<scalar_t>(),
-                       W1.data_ptr<scalar_t>(), b1.data_ptr<scalar_t>(),
-                       W2.data_ptr<scalar_t>(), b2.data_ptr<scalar_t>(),
-                       out.data_ptr<scalar_t>(), B, C, W, Tn, S);
+    if (W == 20) {
+      hipLaunchKernelGGL((lift_head_fwd_kernel<scalar_t, 4, 32, 24, 20>), dim3(grid),
+                         dim3(kBlock), 0, stream, x.data_ptr<scalar_t>(),
+                         W1.data_ptr<scalar_t>(), b1.data_ptr<scalar_t>(),
+                         W2.data_ptr<scalar_t>(), b2.data_ptr<scalar_t>(),
+                         out.data_ptr<scalar_t>(), B, C, W, Tn, S);
+    } else {
+      hipLaunchKernelGGL((lift_head_fwd_kernel<scalar_t, 4, 32, 24>), dim3(grid),
+                         dim3(kBlock), 0, stream, x.data_ptr<scalar_t>(),
+                         W1.data_ptr<scalar_t>(), b1.data_ptr<scalar_t>(),
+                         W2.data_ptr<scalar_t>(), b2.data_ptr<scalar_t>(),
+                         out.data_ptr<scalar_t>(), B, C, W, Tn, S);
+    }
   });
   DFNO_CHECK_LAUNCH("lift_head");
   return out;
@@ -299,14 +311,25 @@ std::vector<at::Tensor> lift_head_bwd(const at::Tensor& gy, const at::Tensor& x,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_l((long)B * S);
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "lift_head_bwd", [&] {
-    hipLaunchKernelGGL((lift_head_bwd_kernel<scalar_t, 4, 32, 24>), dim3(grid),
-                       dim3(kBlock), 0, stream, gy.data_ptr<scalar_t>(),
-                       x.data_ptr<scalar_t>(), W1.data_ptr<scalar_t>(),
-                       b1.data_ptr<scalar_t>(), W2.data_ptr<scalar_t>(),
-                       b2.data_ptr<scalar_t>(), gx.data_ptr<scalar_t>(),
-                       gW1.data_ptr<scalar_t>(), gb1.data_ptr<scalar_t>(),
-                       gW2.data_ptr<scalar_t>(), gb2.data_ptr<scalar_t>(),
-                       B, C, W, Tn, S);
+    if (W == 20) {
+      hipLaunchKernelGGL((lift_head_bwd_kernel<scalar_t, 4, 32, 24, 20>), dim3(grid),
+                         dim3(kBlock), 0, stream, gy.data_ptr<scalar_t>(),
+                         x.data_ptr<scalar_t>(), W1.data_ptr<scalar_t>(),
+                         b1.data_ptr<scalar_t>(), W2.data_ptr<scalar_t>(),
+                         b2.data_ptr<scalar_t>(), gx.data_ptr<scalar_t>(),
+                         gW1.data_ptr<scalar_t>(), gb1.data_ptr<scalar_t>(),
+                         gW2.data_ptr<scalar_t>(), gb2.data_ptr<scalar_t>(),
+                         B, C, W, Tn, S);
+    } else {
+      hipLaunchKernelGGL((lift_head_bwd_kernel<scalar_t, 4, 32, 24>), dim3(grid),
+                         dim3(kBlock), 0, stream, gy.data_ptr<scalar_t>(),
+                         x.data_ptr<scalar_t>(), W1.data_ptr<scalar_t>(),
+                         b1.data_ptr<scalar_t>(), W2.data_ptr<scalar_t>(),
+                         b2.data_ptr<scalar_t>(), gx.data_ptr<scalar_t>(),
+                         gW1.data_ptr<scalar_t>(), gb1.data_ptr<scalar_t>(),
+                         gW2.data_ptr<scalar_t>(), gb2.data_ptr<scalar_t>(),
+                         B, C, W, Tn, S);
+    }
   });
   DFNO_CHECK_LAUNCH("lift_head");
   return {gx, gW1, gb1, gW2, gb2};
