@@ -37,4 +37,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "all corner boxes in one launch");
   m.def("spectral_corners_bwd_x", &spectral_corners_bwd_x,
         "all corner boxes adjoint wrt x in one launch");
+  m.def("spectral_corners_bwd_w", &spectral_corners_bwd_w,
+        "multi-corner spectral grad-W (gw[i,o,e] = sum_b x conj(gy))");
 }

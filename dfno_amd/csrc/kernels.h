@@ -84,3 +84,6 @@ void spectral_corners_fwd(const at::Tensor& x, std::vector<at::Tensor> ws,
                           at::Tensor& y, std::vector<std::vector<int64_t>> starts);
 void spectral_corners_bwd_x(const at::Tensor& gy, std::vector<at::Tensor> ws,
                             at::Tensor& gx, std::vector<std::vector<int64_t>> starts);
+void spectral_corners_bwd_w(const at::Tensor& x, const at::Tensor& gy,
+                            std::vector<at::Tensor> gws,
+                            std::vector<std::vector<int64_t>> starts);

@@ -215,6 +215,59 @@ __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
   }
 }
 
+// BWD-W: gw[i, o, e] = sum_b gy[b, o, F(e)] * conj(x[b, i, F(e)]) per
+// corner -- at batch ~1 this is a pure per-frequency outer product; fusing
+// the corner gather into the addressing replaces the python-side per-corner
+// slice + conj materialization + complex-mul einsum chain (~0.6 ms/step of
+// eager kernels at the flagship config).  One thread owns (corner, e, o)
+// and streams the I outputs; writes for fixed (i, o) are e-consecutive
+// across lanes (coalesced), as are the x/gy reads.
+template <typename T, int ICAP>
+__global__ __launch_bounds__(kBlock) void spectral_corners_bwd_w_kernel(
+    const T* __restrict__ x, const T* __restrict__ gy, MultiGeom<T> mg,
+    int B, int I, int O, long Ftot) {
+  const long total = mg.cum[mg.ncorners] * O;
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < total; t += stride) {
+    long work = t % mg.cum[mg.ncorners];
+    int o = (int)(t / mg.cum[mg.ncorners]);
+    int c = 0;
+    while (work >= mg.cum[c + 1]) ++c;
+    long e = work - mg.cum[c];
+    const BoxGeom& g = mg.g[c];
+    T* gw = const_cast<T*>(mg.w[c]);
+
+    long f = box_to_global(e, g);
+    T accr[ICAP], acci[ICAP];
+#pragma unroll
+    for (int k = 0; k < ICAP; ++k) { accr[k] = T(0); acci[k] = T(0); }
+    for (int b = 0; b < B; ++b) {
+      const T* gyb = gy + 2 * (((long)b * O + o) * Ftot + f);
+      const T gr = gyb[0], gi = gyb[1];
+      const T* xb = x + 2 * (((long)b * I) * Ftot + f);
+#pragma unroll
+      for (int k = 0; k < ICAP; ++k) {
+        if (k < I) {
+          // acc += gy * conj(x)  (the reference's einsum conjugates x)
+          const T xr = xb[2 * (long)k * Ftot], xi = xb[2 * (long)k * Ftot + 1];
+          accr[k] += xr * gr + xi * gi;
+          acci[k] += xr * gi - xi * gr;
+        }
+      }
+    }
+#pragma unroll
+    for (int k = 0; k < ICAP; ++k) {
+      if (k < I) {
+        long widx = 2 * (((long)k * O + o) * g.nelem + e);
+        gw[widx] = accr[k];
+        gw[widx + 1] = acci[k];
+      }
+    }
+  }
+}
+
 int grid_for_s(long work) {
   long g = (work + kBlock - 1) / kBlock;
   long cap = 256L * 16;
@@ -371,6 +424,57 @@ void spectral_corners_fwd(const at::Tensor& x, std::vector<at::Tensor> ws,
   } else {
     launch_corners<double, false>(x, ws, y, starts, B, I, O, Ftot);
   }
+  DFNO_CHECK_LAUNCH("spectral");
+}
+
+void spectral_corners_bwd_w(const at::Tensor& x, const at::Tensor& gy,
+                            std::vector<at::Tensor> gws,
+                            std::vector<std::vector<int64_t>> starts) {
+  check_c(x, "x"); check_c(gy, "gy");
+  TORCH_CHECK(gws.size() == starts.size(), "gws/starts size mismatch");
+  int B = (int)x.size(0), I = (int)x.size(1), O = (int)gy.size(1);
+  TORCH_CHECK(I <= 32, "spectral_bwd_w: I must be <= 32");
+  long Ftot = 1;
+  for (int d = 2; d < x.dim(); ++d) Ftot *= x.size(d);
+  for (auto& w : gws) {
+    check_c(w, "gw");
+    TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "spectral: gw shape");
+  }
+  if (B == 0 || gws.empty()) return;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+#define SBW(T, ICAP_)                                                          \
+  {                                                                            \
+    size_t idx = 0;                                                            \
+    while (idx < gws.size()) {                                                 \
+      MultiGeom<T> mg{};                                                       \
+      mg.ncorners = 0;                                                         \
+      mg.cum[0] = 0;                                                           \
+      while (idx < gws.size() && mg.ncorners < kMaxCorners) {                  \
+        BoxGeom g = make_geom(x, gws[idx], starts[idx]);                       \
+        if (g.nelem == 0) { ++idx; continue; }                                 \
+        int c = mg.ncorners++;                                                 \
+        mg.g[c] = g;                                                           \
+        mg.w[c] = reinterpret_cast<const T*>(gws[idx].data_ptr());             \
+        mg.cum[c + 1] = mg.cum[c] + g.nelem;                                   \
+        ++idx;                                                                 \
+      }                                                                        \
+      if (mg.ncorners == 0) continue;                                          \
+      int grid = grid_for_s(mg.cum[mg.ncorners] * O);                          \
+      hipLaunchKernelGGL((spectral_corners_bwd_w_kernel<T, ICAP_>),            \
+                         dim3(grid), dim3(kBlock), 0, stream,                  \
+                         reinterpret_cast<const T*>(x.data_ptr()),             \
+                         reinterpret_cast<const T*>(gy.data_ptr()), mg,        \
+                         B, I, O, Ftot);                                       \
+    }                                                                          \
+  }
+  if (x.scalar_type() == at::kComplexFloat) {
+    if (I <= 8) { SBW(float, 8) } else if (I <= 24) { SBW(float, 24) }
+    else { SBW(float, 32) }
+  } else {
+    if (I <= 8) { SBW(double, 8) } else if (I <= 24) { SBW(double, 24) }
+    else { SBW(double, 32) }
+  }
+#undef SBW
   DFNO_CHECK_LAUNCH("spectral");
 }
 

@@ -67,12 +67,17 @@ class _SpectralConvFn(torch.autograd.Function):
         gws = []
         if gy.is_cuda and gy.numel() > 0 and gy.dtype in (torch.complex64, torch.complex128):
             ext = _ext.get(required=True)
+            starts = [[a for a, _ in bounds] for bounds in bounds_list]
             ext.spectral_corners_bwd_x(
-                gy, [w.contiguous() for w in weights], gx,
-                [[a for a, _ in bounds] for bounds in bounds_list])
-            for w, bounds in zip(weights, bounds_list):
-                sl = _corner_slices(bounds)
-                gws.append(torch.einsum("bo...,bi...->io...", gy[sl], x[sl].conj()))
+                gy, [w.contiguous() for w in weights], gx, starts)
+            if x.shape[1] <= 32:
+                gws = [torch.empty_like(w) for w in weights]
+                ext.spectral_corners_bwd_w(x.contiguous(), gy, gws, starts)
+            else:
+                for w, bounds in zip(weights, bounds_list):
+                    sl = _corner_slices(bounds)
+                    gws.append(torch.einsum("bo...,bi...->io...",
+                                            gy[sl], x[sl].conj()))
         else:
             for w, bounds in zip(weights, bounds_list):
                 sl = _corner_slices(bounds)
