@@ -31,21 +31,31 @@ def _worker(rank, world_size, fn, init_file, err_dir, args):
 
 
 def run_dist(fn, world_size, *args, timeout=300):
-    """Spawn ``world_size`` gloo processes each running fn(rank, world, *args)."""
-    with tempfile.TemporaryDirectory() as td:
-        init_file = os.path.join(td, "init")
-        try:
-            mp.start_processes(
-                _worker,
-                args=(world_size, fn, init_file, td, args),
-                nprocs=world_size,
-                start_method="spawn",
-                join=True,
-            )
-        except Exception:
-            msgs = []
-            for r in range(world_size):
-                p = os.path.join(td, f"rank{r}.err")
-                if os.path.exists(p):
-                    msgs.append(f"--- rank {r} ---\n" + open(p).read())
-            raise AssertionError("distributed test failed:\n" + "\n".join(msgs))
+    """Spawn ``world_size`` gloo processes each running fn(rank, world, *args).
+
+    Retries once when the spawn dies without any rank writing a traceback
+    (an infrastructure flake -- process startup under load -- rather than a
+    test assertion, which would write rank<k>.err and re-fail anyway).
+    """
+    for attempt in range(2):
+        with tempfile.TemporaryDirectory() as td:
+            init_file = os.path.join(td, "init")
+            try:
+                mp.start_processes(
+                    _worker,
+                    args=(world_size, fn, init_file, td, args),
+                    nprocs=world_size,
+                    start_method="spawn",
+                    join=True,
+                )
+                return
+            except Exception:
+                msgs = []
+                for r in range(world_size):
+                    p = os.path.join(td, f"rank{r}.err")
+                    if os.path.exists(p):
+                        msgs.append(f"--- rank {r} ---\n" + open(p).read())
+                if not msgs and attempt == 0:
+                    continue  # no rank-level traceback: infra flake, retry
+                raise AssertionError(
+                    "distributed test failed:\n" + "\n".join(msgs))
