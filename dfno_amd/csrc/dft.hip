@@ -644,7 +644,9 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
     }
   };
 
-  if (0 < ntiles) issue(0, blockIdx.x < ntiles ? blockIdx.x : 0);
+  // blocks with no tile must not issue: an outstanding LDS-DMA at kernel
+  // exit (nothing ever waits on it) is undefined once the LDS is reassigned
+  if (blockIdx.x < ntiles) issue(0, blockIdx.x);
   long c = 0;
   for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x, ++c) {
     long nxt = tb + gridDim.x;
