@@ -124,6 +124,30 @@ def test_train_two_phase_resume(tmp_path):
     assert (tmp_path / "model_0002_0000.pt").exists()
 
 
+def test_train_two_phase_resume_missing_state(tmp_path):
+    # --resume against a dir with no train_state files must fail loudly
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, str(REPO / "training/two_phase/train_two_phase.py"),
+         "--data", "synthetic", "--shape", "8", "8", "4", "6",
+         "--width", "4", "--modes", "2", "2", "2", "2",
+         "--num-train", "2", "--num-valid", "1", "--num-epochs", "1",
+         "--resume", str(tmp_path), "--out-dir", str(tmp_path)],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=300)
+    assert r.returncode != 0
+    assert "train_state" in (r.stderr + r.stdout)
+
+
+def test_finalize_distributed_idempotent():
+    import dfno_amd
+
+    # serial: both calls are no-ops and must not raise
+    dfno_amd.finalize_distributed()
+    dfno_amd.finalize_distributed()
+
+
 def test_train_and_test_two_phase_2rank(tmp_path):
     run_script([str(REPO / "training/two_phase/train_two_phase.py"),
                 "--data", "synthetic", "--shape", "8", "8", "4", "6",
