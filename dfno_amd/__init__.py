@@ -50,3 +50,5 @@ from .utils import (
 )
 
 __version__ = "0.1.0"
+
+from .checkpoint import reshard_checkpoint  # noqa: E402

@@ -48,7 +48,36 @@ def is_distributed() -> bool:
     return dist.is_available() and dist.is_initialized()
 
 
+# Offline impersonation: checkpoint tooling constructs models "as" an
+# arbitrary rank of an arbitrary partition without a process group (all
+# collectives are identity in serial mode; only the shard geometry depends
+# on the rank).  See dfno_amd/checkpoint.py.
+_FORCED_RANK: Optional[int] = None
+
+
+class as_rank:
+    """Context manager: make ``world_rank()`` report ``r`` (serial only)."""
+
+    def __init__(self, r: int):
+        self.r = int(r)
+
+    def __enter__(self):
+        global _FORCED_RANK
+        if is_distributed():
+            raise RuntimeError("as_rank is for offline (serial) use only")
+        self._prev = _FORCED_RANK
+        _FORCED_RANK = self.r
+        return self
+
+    def __exit__(self, *exc):
+        global _FORCED_RANK
+        _FORCED_RANK = self._prev
+        return False
+
+
 def world_rank() -> int:
+    if _FORCED_RANK is not None and not is_distributed():
+        return _FORCED_RANK
     return dist.get_rank() if is_distributed() else 0
 
 
