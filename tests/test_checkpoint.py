@@ -86,3 +86,28 @@ def test_resharded_loads_into_real_2rank_model(tmp_path):
     d2 = tmp_path / "p2"
     reshard_checkpoint(tmp_path, PSERIAL, P2, out_dir=d2, **CFG)
     run_dist(_load_dist_body, 2, str(d2), CFG, P2)
+
+
+# 2D+time NS-style config on a (1,1,2,2,1) world: the pencil P_y folds onto
+# a PREFIX of ranks (odd transform count), so some ranks hold no spectral
+# shards at all -- the harvest/emit passes must handle them.
+CFG2D = dict(in_shape=[1, 1, 8, 8, 4], out_t=6, width=4,
+             modes=(2, 2, 2), num_blocks=2)
+P2D_SERIAL = (1, 1, 1, 1, 1)
+P2D_4 = (1, 1, 2, 2, 1)
+
+
+def test_reshard_2d_inactive_py_ranks(tmp_path):
+    torch.manual_seed(3)
+    m = _build_model(P2D_SERIAL, 0, **CFG2D)
+    torch.save(m.state_dict(), tmp_path / "model_0000.pt")
+    d4 = tmp_path / "p4"
+    paths = reshard_checkpoint(tmp_path, P2D_SERIAL, P2D_4, out_dir=d4, **CFG2D)
+    assert len(paths) == 4
+    back = tmp_path / "back"
+    reshard_checkpoint(d4, P2D_4, P2D_SERIAL, out_dir=back, **CFG2D)
+    src = m.state_dict()
+    out = torch.load(back / "model_0000.pt", weights_only=False)
+    assert set(src) == set(out)
+    for k in src:
+        assert torch.equal(src[k], out[k]), k
