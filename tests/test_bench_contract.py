@@ -33,3 +33,23 @@ def test_bench_json_contract():
     assert "synthetic" in d["data"]
     for k in ("model", "global_batch", "parallelism"):
         assert k in d["config"]
+
+
+def test_bench_json_contract_4rank():
+    # the driver launches N>1 exactly like this (torchrun, one rank per GPU);
+    # on CPU the world runs over gloo with the same (1,1,2,2,1,1) partition
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=4", "--master-addr", "127.0.0.1",
+         "--master-port", "29537", str(REPO / "bench.py"), "--gpus", "4",
+         "--steps", "1", "--warmup", "0", "--width", "4", "--num-blocks", "1"],
+        capture_output=True, text=True, timeout=900, env=env, cwd=str(REPO))
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = r.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    assert d["n_gpus"] == 4
+    assert "2x2" in d["config"]["parallelism"]
+    assert d["value"] > 0
