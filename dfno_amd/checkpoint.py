@@ -109,15 +109,22 @@ def reshard_checkpoint(src_dir, src_pshape: Sequence[int],
         state = model.state_dict()
         for k in state:
             if ".weights." not in k:
+                if k not in root_state:
+                    continue
                 if q == 0:
                     # destination root takes the harvested full parameter
-                    if k in root_state:
-                        if tuple(root_state[k].shape) != tuple(state[k].shape):
-                            raise RuntimeError(
-                                f"{k}: shape {tuple(root_state[k].shape)} in "
-                                f"source vs {tuple(state[k].shape)} expected")
-                        state[k] = root_state[k]
-                # non-root ranks keep their zero-volume placeholders
+                    if tuple(root_state[k].shape) != tuple(state[k].shape):
+                        raise RuntimeError(
+                            f"{k}: shape {tuple(root_state[k].shape)} in "
+                            f"source vs {tuple(state[k].shape)} expected")
+                    state[k] = root_state[k]
+                elif (state[k].numel() > 0
+                      and tuple(state[k].shape) == tuple(root_state[k].shape)):
+                    # replicated full-size state (batchnorm weight/bias/
+                    # running stats, step counters) lives on EVERY rank, not
+                    # just root — copy it so ranks stay synchronized; only
+                    # zero-volume root-stored placeholders are kept local.
+                    state[k] = root_state[k].clone()
         for bi, block in enumerate(model.blocks):
             for k, cid in enumerate(block.corner_ids):
                 full = corners.get((bi, cid))

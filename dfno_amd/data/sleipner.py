@@ -153,6 +153,21 @@ class DistributedSleipnerDataset3D(Dataset):
         i = self.samples[index]
 
         fname = self._cache_name(i)
+        multi = self.P_feat.active and int(np.prod(self.P_feat.shape)) > 1
+        if multi and torch.utils.data.get_worker_info() is not None:
+            # forked DataLoader workers must not issue collectives: they
+            # would race/desync the parent's process group
+            raise RuntimeError(
+                "DistributedSleipnerDataset3D requires num_workers=0 in a "
+                "multi-rank partition (collectives run in __getitem__)")
+        if self.cache is not None and multi:
+            # make the cache-hit decision COLLECTIVE so a per-rank cache
+            # divergence (partial prior run) cannot desync the normalization
+            # allreduces below: all ranks take the same branch.
+            hit = self.P_feat.allreduce_scalar(
+                1.0 if fname in self.cache else 0.0, "min")
+            if hit < 1.0:
+                self.cache.discard(fname)
         if self.cache is not None and fname in self.cache:
             import h5py
 

@@ -43,12 +43,20 @@ class DistributedBatchNorm(nn.Module):
         self._ar = AllReduceSum(P_x)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        C = x.shape[1]
+        # Zero-volume ranks (1-D empty placeholder) still participate in the
+        # packed allreduce with zeroed sums so the partition never desyncs.
+        zero_volume = x.numel() == 0
+        C = self.num_features if zero_volume else x.shape[1]
         red_dims = [d for d in range(x.dim()) if d != 1]
         if self.training:
-            cnt = torch.tensor([x.numel() / max(C, 1)], device=x.device, dtype=x.dtype)
-            s = torch.sum(x, dim=red_dims)
-            ss = torch.sum(x * x, dim=red_dims)
+            if zero_volume:
+                cnt = torch.zeros(1, device=x.device, dtype=x.dtype)
+                s = torch.zeros(C, device=x.device, dtype=x.dtype)
+                ss = torch.zeros(C, device=x.device, dtype=x.dtype)
+            else:
+                cnt = torch.tensor([x.numel() / max(C, 1)], device=x.device, dtype=x.dtype)
+                s = torch.sum(x, dim=red_dims)
+                ss = torch.sum(x * x, dim=red_dims)
             packed = torch.cat([s, ss, cnt])
             packed = self._ar(packed)
             n = packed[-1].clamp_min(1.0)
@@ -63,6 +71,9 @@ class DistributedBatchNorm(nn.Module):
         else:
             mean = self.running_mean.to(x.dtype).to(x.device)
             var = self.running_var.to(x.dtype).to(x.device)
+
+        if zero_volume:
+            return x
 
         shape = [1, C] + [1] * (x.dim() - 2)
         xh = (x - mean.reshape(shape)) / torch.sqrt(var.reshape(shape) + self.eps)

@@ -150,6 +150,9 @@ class _PadIrfftFn(torch.autograd.Function):
 
 def rfft_trunc(x, dim, m):
     d = dim % x.dim()
+    # Clamp to the half-spectrum size: modes > n//2+1 keep the whole spectrum
+    # (matches the reference's graceful [:m] slice, /root/reference/dfno/dfno.py:195).
+    m = min(m, x.shape[d] // 2 + 1)
     if _native_ok(x, x.shape[d], m) and d == x.dim() - 1:
         return _RfftTruncFn.apply(x, d, m)
     return _t_rfft_trunc(x, dim, m)
@@ -157,13 +160,18 @@ def rfft_trunc(x, dim, m):
 
 def fft_trunc(x, dim, m_lo, m_hi):
     d = dim % x.dim()
-    if _native_ok(x, x.shape[d], m_lo + m_hi):
+    n = x.shape[d]
+    m_lo = min(m_lo, n)
+    m_hi = min(m_hi, n - m_lo)
+    if _native_ok(x, n, m_lo + m_hi):
         return _FftTruncFn.apply(x, d, m_lo, m_hi)
     return _t_fft_trunc(x, dim, m_lo, m_hi)
 
 
 def pad_ifft(y, dim, n, m_lo, m_hi):
     d = dim % y.dim()
+    m_lo = min(m_lo, n)
+    m_hi = min(m_hi, n - m_lo)
     if _native_ok(y, n, m_lo + m_hi):
         return _PadIfftFn.apply(y, d, n, m_lo, m_hi)
     return _t_pad_ifft(y, dim, n, m_lo, m_hi)
@@ -171,6 +179,7 @@ def pad_ifft(y, dim, n, m_lo, m_hi):
 
 def pad_irfft(y, dim, n_half, n_out, m):
     d = dim % y.dim()
+    m = min(m, n_half)
     if _native_ok(y, n_out, m) and d == y.dim() - 1:
         return _PadIrfftFn.apply(y, d, n_half, n_out, m)
     return _t_pad_irfft(y, dim, n_half, n_out, m)

@@ -18,6 +18,29 @@ __all__ = ["Adam"]
 
 
 class Adam(torch.optim.Adam):
+    """Flat views of (p, exp_avg, exp_avg_sq) are cached OUTSIDE self.state
+    (keyed by id(p)) so state_dict()/load_state_dict() never serialize them;
+    a cached view is rebuilt whenever any underlying storage pointer changed
+    (e.g. after load_state_dict replaced the moment buffers)."""
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self._flat_cache = {}  # id(p) -> (ptrs, (pv, mv, vv))
+
+    def _flat_views(self, p, state):
+        def flat(t):
+            r = torch.view_as_real(t) if t.is_complex() else t
+            assert r.is_contiguous(), "fused Adam requires contiguous params"
+            return r.view(-1)
+        ptrs = (p.data_ptr(), state["exp_avg"].data_ptr(),
+                state["exp_avg_sq"].data_ptr())
+        cached = self._flat_cache.get(id(p))
+        if cached is None or cached[0] != ptrs:
+            cached = (ptrs, (flat(p), flat(state["exp_avg"]),
+                             flat(state["exp_avg_sq"])))
+            self._flat_cache[id(p)] = cached
+        return cached[1]
+
     @torch.no_grad()
     def step(self, closure=None):
         loss = None
@@ -42,7 +65,7 @@ class Adam(torch.optim.Adam):
                     continue
                 if not (p.is_cuda and p.dtype in (torch.float32, torch.complex64,
                                                   torch.float64, torch.complex128)
-                        and not p.grad.is_sparse):
+                        and not p.grad.is_sparse and p.is_contiguous()):
                     slow.append(p)
                     continue
                 state = self.state[p]
@@ -50,12 +73,9 @@ class Adam(torch.optim.Adam):
                     state["step"] = torch.tensor(0.0)
                     state["exp_avg"] = torch.zeros_like(p)
                     state["exp_avg_sq"] = torch.zeros_like(p)
-                if "_flat_pmv" not in state:
-                    flat = lambda t: (torch.view_as_real(t) if t.is_complex() else t).reshape(-1)
-                    state["_flat_pmv"] = (flat(p), flat(state["exp_avg"]),
-                                          flat(state["exp_avg_sq"]))
+                state.pop("_flat_pmv", None)  # stale entry from old checkpoints
                 state["step"] += 1
-                pv, mv, vv = state["_flat_pmv"]
+                pv, mv, vv = self._flat_views(p, state)
                 gv = torch.view_as_real(p.grad) if p.grad.is_complex() else p.grad
                 batch[0].append(pv)
                 batch[1].append(gv.reshape(-1).contiguous())
