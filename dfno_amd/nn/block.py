@@ -21,7 +21,6 @@ MI355X-native differences:
 
 from __future__ import annotations
 
-import time
 from typing import Dict, List, Tuple
 
 import numpy as np
@@ -31,6 +30,7 @@ import torch.nn as nn
 from ..comm import Repartition
 from ..partition import Partition, compute_distribution_info
 from ..ops import spectral_conv, add_gelu, linear_res_gelu, rfft_trunc, fft_trunc, pad_ifft, pad_irfft
+from ..timing import comm_region
 from .linear import BroadcastedLinear
 
 __all__ = ["DistributedFNOBlock"]
@@ -203,15 +203,15 @@ class DistributedFNOBlock(nn.Module):
         # gelu(W x_in + y) — no y0 tensor is materialized.  Its weight is
         # broadcast here so the collective order matches the reference.
         x_in = x
-        t0 = time.time()
-        W_res = self.linear.W_bcast(self.linear.W)
-        b_res = self.linear.b_bcast(self.linear.b)  # unused (bias=False); keeps
-        self.linear.dt_comm = time.time() - t0      # broadcast parity
+        with comm_region() as r:
+            W_res = self.linear.W_bcast(self.linear.W)
+            b_res = self.linear.b_bcast(self.linear.b)  # unused (bias=False);
+        self.linear.dt_comm = r.host_dt                 # broadcast parity
         del b_res
 
-        t0 = time.time()
-        x = self.R1(x)
-        self.dt_comm += time.time() - t0
+        with comm_region() as r:
+            x = self.R1(x)
+        self.dt_comm += r.host_dt
 
         saved_last: Dict[int, int] = {}   # pre-truncation extent per dim
         outermost = self.dim_m[-1]
@@ -226,9 +226,9 @@ class DistributedFNOBlock(nn.Module):
         else:
             x = torch.empty(0, dtype=self.dtype_complex, device=x.device)
 
-        t0 = time.time()
-        x = self.R2(x)
-        self.dt_comm += time.time() - t0
+        with comm_region() as r:
+            x = self.R2(x)
+        self.dt_comm += r.host_dt
 
         if x.numel() > 0:
             for dim in reversed(self.dim_y):
@@ -245,9 +245,9 @@ class DistributedFNOBlock(nn.Module):
         else:
             y = x
 
-        t0 = time.time()
-        y = self.R3(y)
-        self.dt_comm += time.time() - t0
+        with comm_region() as r:
+            y = self.R3(y)
+        self.dt_comm += r.host_dt
 
         if y.numel() > 0:
             for dim in self.dim_m[:-1]:
@@ -259,8 +259,8 @@ class DistributedFNOBlock(nn.Module):
         else:
             y = torch.empty(0, dtype=self.dtype, device=y.device)
 
-        t0 = time.time()
-        y = self.R4(y)
-        self.dt_comm += time.time() - t0
+        with comm_region() as r:
+            y = self.R4(y)
+        self.dt_comm += r.host_dt
 
         return linear_res_gelu(x_in, W_res, y)

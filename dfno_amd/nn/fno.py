@@ -87,19 +87,23 @@ class DistributedFNONd(nn.Module):
         with GELU.  One fused kernel on GPU when T_in == 1 (ops.lift_head);
         the composed path otherwise."""
         from ..ops import lift_head, lift_head_supported
-        import time as _time
+        from ..timing import comm_region
+        from ..dispatch import note_fallback
 
         l1, l2 = self.linear1, self.linear2
         if lift_head_supported(x, l1.in_features, l1.out_features,
                                l2.in_features, l2.out_features):
-            t0 = _time.time()
-            W1 = l1.W_bcast(l1.W)
-            b1 = l1.b_bcast(l1.b)
-            W2 = l2.W_bcast(l2.W)
-            b2 = l2.b_bcast(l2.b)
-            self.dt_comm += _time.time() - t0
+            with comm_region() as r:
+                W1 = l1.W_bcast(l1.W)
+                b1 = l1.b_bcast(l1.b)
+                W2 = l2.W_bcast(l2.W)
+                b2 = l2.b_bcast(l2.b)
+            self.dt_comm += r.host_dt
             return lift_head(x, W1, b1, W2, b2)
 
+        if x.is_cuda:
+            note_fallback("lift_head", "unsupported lift shape/dtype for the "
+                          "fused kernel (composed linears instead)")
         x = l1(x, activation="gelu")
         self.dt_comm += l1.dt_comm
         x = l2(x, activation="gelu")
@@ -110,21 +114,26 @@ class DistributedFNONd(nn.Module):
         """width -> 128 -> gelu -> 1 head.  On GPU this is a single fused
         kernel (ops.proj_head) instead of two linears with a GELU pass."""
         from ..ops import proj_head, proj_head_supported
-        import time as _time
+        from ..timing import comm_region
+        from ..dispatch import note_fallback
 
         l3, l4 = self.linear3, self.linear4
         supported = (x.is_cuda and x.dtype in (torch.float32, torch.float64)
                      and l3.in_features <= 32 and l3.out_features <= 512
                      and l4.out_features <= 8)
         if supported:
-            t0 = _time.time()
-            W3 = l3.W_bcast(l3.W)
-            b3 = l3.b_bcast(l3.b)
-            W4 = l4.W_bcast(l4.W)
-            b4 = l4.b_bcast(l4.b)
-            self.dt_comm += _time.time() - t0
+            with comm_region() as r:
+                W3 = l3.W_bcast(l3.W)
+                b3 = l3.b_bcast(l3.b)
+                W4 = l4.W_bcast(l4.W)
+                b4 = l4.b_bcast(l4.b)
+            self.dt_comm += r.host_dt
             return proj_head(x, W3, b3, W4, b4)
 
+        if x.is_cuda:
+            note_fallback("proj_head", f"shape out of fused-kernel range "
+                          f"(in={l3.in_features}, mid={l3.out_features}, "
+                          f"out={l4.out_features}) or dtype {x.dtype}")
         x = l3(x, activation="gelu")
         self.dt_comm += l3.dt_comm
         x = l4(x)

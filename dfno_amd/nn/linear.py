@@ -11,8 +11,6 @@ einsum, with optional fused GELU.
 
 from __future__ import annotations
 
-import time
-
 import numpy as np
 import torch
 import torch.nn as nn
@@ -20,6 +18,7 @@ import torch.nn as nn
 from ..comm import Broadcast
 from ..partition import Partition, create_root_partition, zero_volume_tensor
 from ..ops import linear_nd
+from ..timing import comm_region
 
 __all__ = ["BroadcastedLinear"]
 
@@ -59,9 +58,9 @@ class BroadcastedLinear(nn.Module):
     def forward(self, x: torch.Tensor, activation: str = None) -> torch.Tensor:
         self.dt_comm = 0.0
 
-        t0 = time.time()
-        W = self.W_bcast(self.W)
-        b = self.b_bcast(self.b)
-        self.dt_comm += time.time() - t0
+        with comm_region() as r:
+            W = self.W_bcast(self.W)
+            b = self.b_bcast(self.b)
+        self.dt_comm += r.host_dt
 
         return linear_nd(x, W, b if self.bias else None, self.dim, activation)

@@ -25,16 +25,23 @@ from __future__ import annotations
 import torch
 
 from .. import _ext
+from ..dispatch import note_fallback
 
 __all__ = ["rfft_trunc", "fft_trunc", "pad_ifft", "pad_irfft"]
 
 _MAX_N = 64
 
 
-def _native_ok(x: torch.Tensor, n: int, m: int) -> bool:
-    return (x.is_cuda and n <= _MAX_N and m <= 32
-            and x.dtype in (torch.float32, torch.complex64,
-                            torch.float64, torch.complex128))
+def _native_ok(x: torch.Tensor, n: int, m: int, op: str) -> bool:
+    if not x.is_cuda:
+        return False
+    ok = (n <= _MAX_N and m <= 32
+          and x.dtype in (torch.float32, torch.complex64,
+                          torch.float64, torch.complex128))
+    if not ok:
+        note_fallback(op, f"N={n} (cap {_MAX_N}) / m={m} (cap 32) / "
+                      f"dtype {x.dtype} outside the native-DFT range")
+    return ok
 
 
 # ---------------------------------------------------------------------------
@@ -153,8 +160,10 @@ def rfft_trunc(x, dim, m):
     # Clamp to the half-spectrum size: modes > n//2+1 keep the whole spectrum
     # (matches the reference's graceful [:m] slice, /root/reference/dfno/dfno.py:195).
     m = min(m, x.shape[d] // 2 + 1)
-    if _native_ok(x, x.shape[d], m) and d == x.dim() - 1:
+    if d == x.dim() - 1 and _native_ok(x, x.shape[d], m, "rfft_trunc"):
         return _RfftTruncFn.apply(x, d, m)
+    if x.is_cuda and d != x.dim() - 1:
+        note_fallback("rfft_trunc", f"non-last transform dim {d}")
     return _t_rfft_trunc(x, dim, m)
 
 
@@ -163,7 +172,7 @@ def fft_trunc(x, dim, m_lo, m_hi):
     n = x.shape[d]
     m_lo = min(m_lo, n)
     m_hi = min(m_hi, n - m_lo)
-    if _native_ok(x, n, m_lo + m_hi):
+    if _native_ok(x, n, m_lo + m_hi, "fft_trunc"):
         return _FftTruncFn.apply(x, d, m_lo, m_hi)
     return _t_fft_trunc(x, dim, m_lo, m_hi)
 
@@ -172,7 +181,7 @@ def pad_ifft(y, dim, n, m_lo, m_hi):
     d = dim % y.dim()
     m_lo = min(m_lo, n)
     m_hi = min(m_hi, n - m_lo)
-    if _native_ok(y, n, m_lo + m_hi):
+    if _native_ok(y, n, m_lo + m_hi, "pad_ifft"):
         return _PadIfftFn.apply(y, d, n, m_lo, m_hi)
     return _t_pad_ifft(y, dim, n, m_lo, m_hi)
 
@@ -180,6 +189,8 @@ def pad_ifft(y, dim, n, m_lo, m_hi):
 def pad_irfft(y, dim, n_half, n_out, m):
     d = dim % y.dim()
     m = min(m, n_half)
-    if _native_ok(y, n_out, m) and d == y.dim() - 1:
+    if d == y.dim() - 1 and _native_ok(y, n_out, m, "pad_irfft"):
         return _PadIrfftFn.apply(y, d, n_half, n_out, m)
+    if y.is_cuda and d != y.dim() - 1:
+        note_fallback("pad_irfft", f"non-last transform dim {d}")
     return _t_pad_irfft(y, dim, n_half, n_out, m)

@@ -26,12 +26,17 @@ import torch
 import torch.nn.functional as F
 
 from .. import _ext
+from ..dispatch import note_fallback
 
 
-def _native_dt(t: torch.Tensor) -> bool:
-    """Native kernels are fp32/fp64; other dtypes (bf16/fp16) take the
-    composed-torch path (still on GPU via rocBLAS/eager)."""
-    return t.dtype in (torch.float32, torch.float64)
+def _native_dt(t: torch.Tensor, op: str = "channel_mix") -> bool:
+    """Native kernels are fp32/fp64 (bf16 goes through the dedicated bf16
+    MFMA path where wired); anything else takes the composed-torch path
+    (still on GPU via rocBLAS/eager) with a warn-once."""
+    ok = t.dtype in (torch.float32, torch.float64)
+    if not ok and t.is_cuda:
+        note_fallback(op, f"dtype {t.dtype} has no native kernel")
+    return ok
 
 __all__ = ["linear_nd", "add_gelu", "gelu", "linear_res_gelu"]
 

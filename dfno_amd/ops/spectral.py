@@ -49,6 +49,9 @@ class _SpectralConvFn(torch.autograd.Function):
                 [[a for a, _ in bounds] for bounds in bounds_list])
             x_saved = xc
         else:
+            if x.is_cuda and x.numel() > 0:
+                from ..dispatch import note_fallback
+                note_fallback("spectral_conv", f"dtype {x.dtype} not complex64/128")
             x_saved = x
             for w, bounds in zip(weights, bounds_list):
                 sl = _corner_slices(bounds)
@@ -74,6 +77,9 @@ class _SpectralConvFn(torch.autograd.Function):
                 gws = [torch.empty_like(w) for w in weights]
                 ext.spectral_corners_bwd_w(x.contiguous(), gy, gws, starts)
             else:
+                from ..dispatch import note_fallback
+                note_fallback("spectral_corners_bwd_w",
+                              f"in_channels {x.shape[1]} > 32 (einsum grad-W)")
                 for w, bounds in zip(weights, bounds_list):
                     sl = _corner_slices(bounds)
                     gws.append(torch.einsum("bo...,bi...->io...",
