@@ -30,7 +30,8 @@
 namespace {
 
 constexpr int kBlock = 256;
-constexpr int kMaxN = 64;
+constexpr int kMaxN = 64;    // cap of the fully-tuned N<=64 paths
+constexpr int kMaxNBig = 256;  // cap of the generalized radix-8xNB / LB paths
 
 template <typename T>
 __device__ __forceinline__ void sincos_t(T a, T* s, T* c);
@@ -439,14 +440,20 @@ __device__ __forceinline__ void dft8(const T* xr, const T* xi, T* Ar, T* Ai) {
   }
 }
 
-template <typename T, int LCAP, int MLT = 0>
+// Generalized to N = 8 * NBT via decimation j = NBT*a + b (a in [0,8),
+// b in [0,NBT)): w_N^{(NBT a + b)k} = w_8^{a (k mod 8)} * w_N^{bk}, so
+// stage 1 is an 8-point DFT over the stride-NBT comb of each b and stage 2
+// accumulates kept modes with the first NBT rows of the [N, m_lo+1] table.
+// NBT=8 is the round-1 N=64 kernel; 16/32 cover the 128/256 grids of the
+// weak-scaling configs (reference gen_scripts.py:140-153) natively.
+template <typename T, int LCAP, int MLT = 0, int NBT = 8>
 __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_ana_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long outer, long inner, int m_lo_, T scale) {
   // MLT > 0 pins m_lo at compile time (folded table strides / no
   // predicates; see the r2c fast-path note)
   const int m_lo = MLT > 0 ? MLT : m_lo_;
-  constexpr int N = 64;
+  constexpr int N = 8 * NBT;
   const int nch = m_lo + 1;
   long pairs = inner / 2;
   long total = outer * pairs;
@@ -466,12 +473,12 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_ana_kernel(
         b0r[k] = b0i[k] = b1r[k] = b1i[k] = T(0);
       }
     }
-    for (int b = 0; b < 8; ++b) {
+    for (int b = 0; b < NBT; ++b) {
       T x0r[8], x0i[8], x1r[8], x1i[8];
 #pragma unroll
       for (int a = 0; a < 8; ++a) {
         const float4 v = *reinterpret_cast<const float4*>(
-            src + 2 * (8 * a + b) * inner);
+            src + 2 * (NBT * a + b) * inner);
         x0r[a] = v.x; x0i[a] = v.y; x1r[a] = v.z; x1i[a] = v.w;
       }
       T A0r[8], A0i[8], A1r[8], A1i[8];
@@ -520,17 +527,17 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_ana_kernel(
   }
 }
 
-// Radix-8x8 synthesis: X[8a+b] = sum_r w8^{+ar} C_b[r] with C_b[r] the
-// per-residue accumulation of the kept modes times w64^{+bk} (prefix) /
-// conj(w64^{+b kp}) (suffix).  tw here is the synthesis-signed table.
-template <typename T, int LCAP, int MLT = 0>
+// Radix-8xNB synthesis: X[NBT a + b] = sum_r w8^{+ar} C_b[r] with C_b[r] the
+// per-residue accumulation of the kept modes times w_N^{+bk} (prefix) /
+// conj(w_N^{+b kp}) (suffix).  tw here is the synthesis-signed table.
+template <typename T, int LCAP, int MLT = 0, int NBT = 8>
 __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_syn_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long outer, long inner, int m_lo_, T scale) {
   // MLT > 0 pins m_lo at compile time (folded table strides / no
   // predicates; see the r2c fast-path note)
   const int m_lo = MLT > 0 ? MLT : m_lo_;
-  constexpr int N = 64;
+  constexpr int N = 8 * NBT;
   const int nch = m_lo + 1;
   const int m = 2 * m_lo;
   long pairs = inner / 2;
@@ -557,7 +564,7 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_syn_kernel(
       }
     }
     T* dst = out + 2 * (o * N * inner + i);
-    for (int b = 0; b < 8; ++b) {
+    auto syn_b = [&](int b) {
       T C0r[8], C0i[8], C1r[8], C1i[8];
 #pragma unroll
       for (int r = 0; r < 8; ++r) { C0r[r] = C0i[r] = C1r[r] = C1i[r] = T(0); }
@@ -586,9 +593,17 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_syn_kernel(
       dft8<T, 1>(C1r, C1i, X1r, X1i);
 #pragma unroll
       for (int a = 0; a < 8; ++a) {
-        *reinterpret_cast<float4*>(dst + 2 * (8 * a + b) * inner) =
+        *reinterpret_cast<float4*>(dst + 2 * (NBT * a + b) * inner) =
             make_float4(X0r[a], X0i[a], X1r[a], X1i[a]);
       }
+    };
+    if constexpr (NBT == 8) {
+      for (int b = 0; b < 8; ++b) syn_b(b);
+    } else {
+      // cap the b-unroll: full unroll at NBT=16/32 runs the register file
+      // to 256 VGPR (1 wave/SIMD) with no twiddle-fold benefit
+#pragma clang loop unroll_count(2)
+      for (int b = 0; b < NBT; ++b) syn_b(b);
     }
   }
 }
@@ -707,31 +722,33 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
 // register recurrences: the chain update was 4 of every 6 VALU ops (PMC:
 // VALU-bound at ~85% issue), and (j, k) are lane-uniform so the table reads
 // compile to scalar loads that the k-unrolled fma stream hides entirely.
-template <typename T, int MCAP>
-__global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
+template <typename T, int MCAP, int LB = kBlock>
+__global__ __launch_bounds__(LB) void dft_r2c_last_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long lines, int N, int m, T scale, bool factors) {
+  // LB = block size = lines per tile; 128/64 for N = 128/256 keep the
+  // [LB * N] LDS tile within the 160 KiB budget (2 blocks/CU at 64 KiB).
   extern __shared__ __align__(16) char smem_raw[];
-  T* tile = reinterpret_cast<T*>(smem_raw);   // [kBlock * N]
+  T* tile = reinterpret_cast<T*>(smem_raw);   // [LB * N]
 
-  long ntiles = (lines + kBlock - 1) / kBlock;
+  long ntiles = (lines + LB - 1) / LB;
   for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x) {
-    long l0 = tb * kBlock;
-    int nl = (int)min((long)kBlock, lines - l0);
+    long l0 = tb * LB;
+    int nl = (int)min((long)LB, lines - l0);
     __syncthreads();
     if constexpr (std::is_same<T, float>::value) {
       const long base = l0 * N;
       if ((nl * N) % 4 == 0 && (base % 4 == 0) &&
           ((reinterpret_cast<uintptr_t>(in) & 15) == 0)) {
-        for (int idx = threadIdx.x * 4; idx < nl * N; idx += kBlock * 4)
+        for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4)
           *reinterpret_cast<float4*>(&tile[idx]) =
               *reinterpret_cast<const float4*>(in + base + idx);
       } else {
-        for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+        for (int idx = threadIdx.x; idx < nl * N; idx += LB)
           tile[idx] = in[base + idx];
       }
     } else {
-      for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+      for (int idx = threadIdx.x; idx < nl * N; idx += LB)
         tile[idx] = in[l0 * N + idx];
     }
     __syncthreads();
@@ -781,21 +798,21 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_last_kernel(
 
 // Output lines are written through an LDS tile and stored cooperatively
 // (coalesced); inputs are 64-128B contiguous per line and read directly.
-template <typename T, int MCAP, int NT = 0>
-__global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
+template <typename T, int MCAP, int NT = 0, int LB = kBlock>
+__global__ __launch_bounds__(LB) void dft_c2r_last_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long lines, int N_, int m_, T scale, bool factors) {
   // NT > 0 pins N and m (== MCAP) at compile time: full unroll + folded
-  // twiddle offsets, as in dft_r2c_glds_kernel
+  // twiddle offsets, as in dft_r2c_glds_kernel.  LB: see dft_r2c_last_kernel.
   const int N = NT > 0 ? NT : N_;
   const int m = NT > 0 ? MCAP : m_;
   extern __shared__ __align__(16) char smem_raw[];
-  T* tile = reinterpret_cast<T*>(smem_raw);   // [kBlock * N]
+  T* tile = reinterpret_cast<T*>(smem_raw);   // [LB * N]
 
-  long ntiles = (lines + kBlock - 1) / kBlock;
+  long ntiles = (lines + LB - 1) / LB;
   for (long tb = blockIdx.x; tb < ntiles; tb += gridDim.x) {
-    long l0 = tb * kBlock;
-    int nl = (int)min((long)kBlock, lines - l0);
+    long l0 = tb * LB;
+    int nl = (int)min((long)LB, lines - l0);
     __syncthreads();
     {
       // all threads compute (tail threads re-read line nl-1) so the twiddle
@@ -816,9 +833,7 @@ __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
         }
       }
       T* dst = tile + threadIdx.x * N;
-#pragma unroll
-      for (int j = 0; j < (NT > 0 ? NT : 64); ++j) {
-        if (NT == 0 && j >= N) break;
+      auto syn_j = [&](int j) {
         T sacc = T(0);
         auto twj = (const __attribute__((address_space(4))) T*)
             (tw + (long)(j * 2) * m);          // lane-uniform -> s_load
@@ -830,6 +845,12 @@ __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
           }
         }
         dst[j] = sacc;
+      };
+      if constexpr (NT > 0) {
+#pragma unroll
+        for (int j = 0; j < NT; ++j) syn_j(j);
+      } else {
+        for (int j = 0; j < N; ++j) syn_j(j);
       }
     }
     __syncthreads();
@@ -837,15 +858,15 @@ __global__ __launch_bounds__(kBlock) void dft_c2r_last_kernel(
       const long base = l0 * N;
       if ((nl * N) % 4 == 0 && (base % 4 == 0) &&
           ((reinterpret_cast<uintptr_t>(out) & 15) == 0)) {
-        for (int idx = threadIdx.x * 4; idx < nl * N; idx += kBlock * 4)
+        for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4)
           *reinterpret_cast<float4*>(out + base + idx) =
               *reinterpret_cast<const float4*>(&tile[idx]);
       } else {
-        for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+        for (int idx = threadIdx.x; idx < nl * N; idx += LB)
           out[base + idx] = tile[idx];
       }
     } else {
-      for (int idx = threadIdx.x; idx < nl * N; idx += kBlock)
+      for (int idx = threadIdx.x; idx < nl * N; idx += LB)
         out[l0 * N + idx] = tile[idx];
     }
   }
@@ -901,7 +922,7 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "dft_c2c: contiguous GPU input");
   TORCH_CHECK(x.scalar_type() == at::kComplexFloat || x.scalar_type() == at::kComplexDouble,
               "dft_c2c: complex input");
-  TORCH_CHECK(n <= kMaxN, "dft_c2c: N too large");
+  TORCH_CHECK(n <= kMaxNBig, "dft_c2c: N too large");
   const int m = (int)(m_lo + m_hi);
   TORCH_CHECK(m <= 64 && m <= n, "dft_c2c: bad mode count");
   TORCH_CHECK(x.size(dim) == (analysis ? n : m), "dft_c2c: dim extent mismatch");
@@ -954,61 +975,47 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
   if (paired)
     tw = twiddle_table((int)n, (int)m_lo + 1, analysis,
                        x.options().dtype(c10::toRealValueType(x.scalar_type())));
+  // radix-8xNB eligibility: N in {64, 128, 256} (NBT 8/16/32), fp32, paired
+  bool radix_ok = paired && (n == 64 || n == 128 || n == 256) &&
+                  c10::toRealValueType(x.scalar_type()) == at::kFloat &&
+                  !no_radix;
+#define DFT_RADIX_LC(KERNEL, NBT)                                              \
+  if (n == 64 && m_lo == 12) { /* flagship: fully folded */                    \
+    hipLaunchKernelGGL((KERNEL<scalar_t, 13, 12, 8>), dim3(grid2),             \
+                       dim3(kBlock), 0, stream, inp, op,                       \
+                       tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,       \
+                       (scalar_t)scale);                                       \
+  } else if (m_lo <= 8) {                                                      \
+    hipLaunchKernelGGL((KERNEL<scalar_t, 9, 0, NBT>), dim3(grid2),             \
+                       dim3(kBlock), 0, stream, inp, op,                       \
+                       tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,       \
+                       (scalar_t)scale);                                       \
+  } else if (m_lo <= 12) {                                                     \
+    hipLaunchKernelGGL((KERNEL<scalar_t, 13, 0, NBT>), dim3(grid2),            \
+                       dim3(kBlock), 0, stream, inp, op,                       \
+                       tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,       \
+                       (scalar_t)scale);                                       \
+  } else {                                                                     \
+    hipLaunchKernelGGL((KERNEL<scalar_t, 17, 0, NBT>), dim3(grid2),            \
+                       dim3(kBlock), 0, stream, inp, op,                       \
+                       tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,       \
+                       (scalar_t)scale);                                       \
+  }
+#define DFT_RADIX(KERNEL)                                                      \
+  if (n == 256) { DFT_RADIX_LC(KERNEL, 32) }                                   \
+  else if (n == 128) { DFT_RADIX_LC(KERNEL, 16) }                              \
+  else { DFT_RADIX_LC(KERNEL, 8) }
   AT_DISPATCH_FLOATING_TYPES(c10::toRealValueType(x.scalar_type()), "dft_c2c", [&] {
     auto inp = reinterpret_cast<const scalar_t*>(x.data_ptr());
     auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
-    if (paired && analysis && n == 64 &&
-        c10::toRealValueType(x.scalar_type()) == at::kFloat && !no_radix) {
-      // radix-8x8 path (see kernel comment); grid2 already sized for pairs
-      if (m_lo == 12) {       // flagship: fully folded
-        hipLaunchKernelGGL((dft_c2c_radix8_ana_kernel<scalar_t, 13, 12>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      } else if (m_lo <= 8) {
-        hipLaunchKernelGGL((dft_c2c_radix8_ana_kernel<scalar_t, 9>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      } else if (m_lo <= 12) {
-        hipLaunchKernelGGL((dft_c2c_radix8_ana_kernel<scalar_t, 13>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      } else {
-        hipLaunchKernelGGL((dft_c2c_radix8_ana_kernel<scalar_t, 17>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      }
+    if (radix_ok && analysis) {
+      DFT_RADIX(dft_c2c_radix8_ana_kernel)
     } else if (paired && analysis) {
       DFT_LDISPATCH(dft_c2c_analysis2_kernel, dim3(grid2), dim3(kBlock), 0,
                     stream, inp, op, tw.data_ptr<scalar_t>(), outer, (int)n,
                     inner, (int)m_lo, (scalar_t)scale)
-    } else if (paired && !analysis && n == 64 &&
-               c10::toRealValueType(x.scalar_type()) == at::kFloat &&
-               !no_radix) {
-      if (m_lo == 12) {       // flagship: fully folded
-        hipLaunchKernelGGL((dft_c2c_radix8_syn_kernel<scalar_t, 13, 12>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      } else if (m_lo <= 8) {
-        hipLaunchKernelGGL((dft_c2c_radix8_syn_kernel<scalar_t, 9>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      } else if (m_lo <= 12) {
-        hipLaunchKernelGGL((dft_c2c_radix8_syn_kernel<scalar_t, 13>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      } else {
-        hipLaunchKernelGGL((dft_c2c_radix8_syn_kernel<scalar_t, 17>),
-                           dim3(grid2), dim3(kBlock), 0, stream, inp, op,
-                           tw.data_ptr<scalar_t>(), outer, inner, (int)m_lo,
-                           (scalar_t)scale);
-      }
+    } else if (radix_ok && !analysis) {
+      DFT_RADIX(dft_c2c_radix8_syn_kernel)
     } else if (paired) {
       DFT_LDISPATCH(dft_c2c_synthesis2_kernel, dim3(grid2), dim3(kBlock), 0,
                     stream, inp, op, tw.data_ptr<scalar_t>(), outer, (int)n,
@@ -1023,6 +1030,8 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
                     (scalar_t)scale)
     }
   });
+#undef DFT_RADIX
+#undef DFT_RADIX_LC
 #undef DFT_LDISPATCH
 #undef DFT_LG
   DFNO_CHECK_LAUNCH("dft_c2c");
@@ -1034,7 +1043,8 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "dft_r2c: contiguous GPU input");
   TORCH_CHECK(dim == x.dim() - 1, "dft_r2c: last-dim only");
   const int N = (int)x.size(dim);
-  TORCH_CHECK(N <= kMaxN && m <= N, "dft_r2c: bad sizes");
+  TORCH_CHECK(N <= kMaxNBig && m <= N, "dft_r2c: bad sizes");
+  TORCH_CHECK(m <= 32, "dft_r2c: m > 32 unsupported natively");
   long lines = x.numel() / std::max(N, 1);
 
   auto sizes = x.sizes().vec();
@@ -1056,10 +1066,27 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
 #define R2CG_M(NGV, NTV)                                                       \
       if (m <= 8) { R2CG(8, NGV, NTV) } else if (m <= 16) { R2CG(16, NGV, NTV) } \
       else if (m <= 24) { R2CG(24, NGV, NTV) } else { R2CG(32, NGV, NTV) }
+#define R2C_BIG(MC, LBV)                                                       \
+      { long nt2 = (lines + LBV - 1) / LBV;                                    \
+        int grid2 = (int)std::min(nt2, 2048L);                                 \
+        size_t smem2 = sizeof(scalar_t) * (size_t)LBV * N;                     \
+        hipLaunchKernelGGL((dft_r2c_last_kernel<scalar_t, MC, LBV>),           \
+                           dim3(grid2), dim3(LBV), smem2, stream, inp, op,     \
+                           tw.data_ptr<scalar_t>(), lines, N, (int)m,          \
+                           (scalar_t)scale, factors); }
+#define R2C_BIG_M(LBV)                                                         \
+      if (m <= 8) { R2C_BIG(8, LBV) } else if (m <= 16) { R2C_BIG(16, LBV) }   \
+      else if (m <= 24) { R2C_BIG(24, LBV) } else { R2C_BIG(32, LBV) }
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "dft_r2c", [&] {
     size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = x.data_ptr<scalar_t>();
     auto op = reinterpret_cast<scalar_t*>(out.data_ptr());
+    if (N > kMaxN) {
+      // big-N path: smaller lines-per-tile keeps the LDS tile <= 64 KiB
+      // (2 blocks/CU); one thread per line, table-driven naive DFT
+      if (N <= 128) { R2C_BIG_M(128) } else { R2C_BIG_M(64) }
+      return;
+    }
     bool glds_ok = std::is_same<scalar_t, float>::value &&
                    ((reinterpret_cast<uintptr_t>(inp) & 15) == 0) &&
                    ((long)kBlock * N) % 4 == 0 && lines * N >= 4;
@@ -1079,6 +1106,8 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
                     (int)m, (scalar_t)scale, factors)
     }
   });
+#undef R2C_BIG_M
+#undef R2C_BIG
 #undef R2CG_M
 #undef R2CG
   DFNO_CHECK_LAUNCH("dft_r2c");
@@ -1093,7 +1122,8 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
               "dft_c2r: complex input");
   const int m = (int)y.size(dim);
   const int N = (int)n_out;
-  TORCH_CHECK(N <= kMaxN && m <= N, "dft_c2r: bad sizes");
+  TORCH_CHECK(N <= kMaxNBig && m <= N, "dft_c2r: bad sizes");
+  TORCH_CHECK(m <= 32, "dft_c2r: m > 32 unsupported natively");
   long lines = y.numel() / std::max(m, 1);
 
   auto sizes = y.sizes().vec();
@@ -1112,10 +1142,25 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
                          dim3(grid), dim3(kBlock), smem, stream, inp, op,      \
                          tw.data_ptr<scalar_t>(), lines, N, (int)m,            \
                          (scalar_t)scale, factors);
+#define C2R_BIG(MC, LBV)                                                       \
+      { long nt2 = (lines + LBV - 1) / LBV;                                    \
+        int grid2 = (int)std::min(nt2, 2048L);                                 \
+        size_t smem2 = sizeof(scalar_t) * (size_t)LBV * N;                     \
+        hipLaunchKernelGGL((dft_c2r_last_kernel<scalar_t, MC, 0, LBV>),        \
+                           dim3(grid2), dim3(LBV), smem2, stream, inp, op,     \
+                           tw.data_ptr<scalar_t>(), lines, N, (int)m,          \
+                           (scalar_t)scale, factors); }
+#define C2R_BIG_M(LBV)                                                         \
+      if (m <= 8) { C2R_BIG(8, LBV) } else if (m <= 16) { C2R_BIG(16, LBV) }   \
+      else if (m <= 24) { C2R_BIG(24, LBV) } else { C2R_BIG(32, LBV) }
   AT_DISPATCH_FLOATING_TYPES(out.scalar_type(), "dft_c2r", [&] {
     size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = reinterpret_cast<const scalar_t*>(y.data_ptr());
     auto op = out.data_ptr<scalar_t>();
+    if (N > kMaxN) {
+      if (N <= 128) { C2R_BIG_M(128) } else { C2R_BIG_M(64) }
+      return;
+    }
     bool mexact = std::is_same<scalar_t, float>::value &&
                   (m == 8 || m == 16 || m == 24 || m == 32);
     if (mexact && N == 30) {
@@ -1133,6 +1178,8 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
                     (int)m, (scalar_t)scale, factors)
     }
   });
+#undef C2R_BIG_M
+#undef C2R_BIG
 #undef C2RG
   DFNO_CHECK_LAUNCH("dft_c2r");
   return out;

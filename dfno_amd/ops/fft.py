@@ -29,7 +29,11 @@ from ..dispatch import note_fallback
 
 __all__ = ["rfft_trunc", "fft_trunc", "pad_ifft", "pad_irfft"]
 
-_MAX_N = 64
+# N <= 64: fully-tuned glds/radix-8x8 kernels.  64 < N <= 256: generalized
+# radix-8xNB (c2c) and lines-per-tile (r2c/c2r) kernels — covers the 128^3 /
+# 256^3 weak-scaling grids and nt <= 256 temporal scaling natively
+# (VERDICT.md round-1 item 2).
+_MAX_N = 256
 
 
 def _native_ok(x: torch.Tensor, n: int, m: int, op: str) -> bool:
