@@ -30,6 +30,7 @@ SOURCES = [
     _CSRC / "proj_head.hip",
     _CSRC / "dft.hip",
     _CSRC / "lift_head.hip",
+    _CSRC / "pack.hip",
 ]
 
 

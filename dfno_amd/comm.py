@@ -44,6 +44,9 @@ __all__ = [
     "AllReduceSum",
     "ZeroVolumeCorrectorFunction",
     "all_reduce_",
+    "repartition_issue",
+    "repartition_complete",
+    "reset_chain",
 ]
 
 
@@ -78,18 +81,66 @@ def all_reduce_(t: torch.Tensor, P: Partition, op: str = "sum") -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
+# comm ordering chain
+# ---------------------------------------------------------------------------
+#
+# Matched collectives must leave the autograd engine in the SAME order on
+# every rank.  With a purely sequential graph that holds by construction,
+# but parallel branches (the channel-chunked pencil pipeline, the loss's two
+# SumReduces) let the engine interleave comm backward nodes differently on
+# ranks whose local graphs differ (zero-volume ranks have fewer compute
+# nodes) — observed deadlock: two ranks parked in a Broadcast-adjoint reduce
+# while two waited on a repartition-adjoint p2p.  Fix: every comm Function
+# takes and emits a zero-size ordering token; the token chain adds a graph
+# edge between consecutive collectives, so backward visits them in exactly
+# reverse forward order on every rank.  The chain is reset at each model
+# forward (nn/fno.py) and spans model + loss.
+
+_CHAIN = {"tok": None}
+
+
+def chain_token(ref: torch.Tensor) -> torch.Tensor:
+    """Chain link for a comm op on ``ref``.
+
+    Only ops whose backward will RUN (grad mode + ref requires grad) join
+    the chain; for the rest a detached dummy is returned so comm on
+    detached tensors (collectors, eval) neither gains requires_grad nor
+    links grad-comm ordering through it.  ``ref.requires_grad`` is uniform
+    across ranks for every model comm site (activations from the lift /
+    parameter placeholders), which the ordering argument requires.
+    """
+    if not (torch.is_grad_enabled() and ref.requires_grad):
+        return torch.zeros(0, device=ref.device)
+    t = _CHAIN["tok"]
+    if t is None or t.device != ref.device or not t.requires_grad:
+        t = torch.zeros(0, device=ref.device, requires_grad=True)
+    return t
+
+
+def set_chain(t: torch.Tensor) -> None:
+    # advance the chain only when the new link carries grad history;
+    # non-grad comm must not sever the ordering between grad comm ops
+    if t is not None and t.grad_fn is not None:
+        _CHAIN["tok"] = t
+
+
+def reset_chain() -> None:
+    _CHAIN["tok"] = None
+
+
+# ---------------------------------------------------------------------------
 # Broadcast (root-stored weights -> all ranks each forward; adjoint = reduce)
 # ---------------------------------------------------------------------------
 
 class _BroadcastFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, module: "Broadcast") -> torch.Tensor:
+    def forward(ctx, x: torch.Tensor, tok: torch.Tensor, module: "Broadcast"):
         ctx.module = module
         P = module.P_dst
         if not (is_distributed() and P.active and P.size > 1):
             # serial / single-member partition: identity
             ctx.was_root = True
-            return x.clone()
+            return x.clone(), tok.new_empty(0)
         root_world = module.P_src.ranks[0]
         is_root = world_rank() == root_world
         ctx.was_root = is_root
@@ -120,23 +171,25 @@ class _BroadcastFn(torch.autograd.Function):
             out = torch.view_as_complex(buf)
         else:
             out = buf
-        return out
+        return out, tok.new_empty(0)
 
     @staticmethod
-    def backward(ctx, grad_out: torch.Tensor):
+    def backward(ctx, grad_out: torch.Tensor, grad_tok):
         module: Broadcast = ctx.module
         P = module.P_dst
+        tg = torch.empty(0, device=grad_out.device)
         if not (is_distributed() and P.active and P.size > 1):
-            return grad_out, None
+            return grad_out, tg, None
         root_world = module.P_src.ranks[0]
         buf = _comm_clone(grad_out)
         dist.reduce(buf, dst=root_world, op=dist.ReduceOp.SUM, group=P.group)
         if ctx.was_root:
             if grad_out.is_complex():
                 buf = torch.view_as_complex(buf)
-            return buf, None
+            return buf, tg, None
         # non-root: parameter is a zero-volume placeholder
-        return zero_volume_tensor(device=grad_out.device, dtype=grad_out.dtype), None
+        return (zero_volume_tensor(device=grad_out.device, dtype=grad_out.dtype),
+                tg, None)
 
 
 class Broadcast(torch.nn.Module):
@@ -165,7 +218,9 @@ class Broadcast(torch.nn.Module):
         return x.device
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return _BroadcastFn.apply(x, self)
+        out, tok = _BroadcastFn.apply(x, chain_token(x), self)
+        set_chain(tok)
+        return out
 
 
 # ---------------------------------------------------------------------------
@@ -174,29 +229,30 @@ class Broadcast(torch.nn.Module):
 
 class _SumReduceFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, module: "SumReduce") -> torch.Tensor:
+    def forward(ctx, x: torch.Tensor, tok: torch.Tensor, module: "SumReduce"):
         ctx.module = module
         ctx.in_shape = tuple(x.shape)
         ctx.in_dtype = x.dtype
         ctx.in_device = x.device
         P = module.P_src
         if not (is_distributed() and P.active and P.size > 1):
-            return x.clone()
+            return x.clone(), tok.new_empty(0)
         root_world = module.P_dst.ranks[0]
         buf = _comm_clone(x)
         dist.reduce(buf, dst=root_world, op=dist.ReduceOp.SUM, group=P.group)
         if world_rank() == root_world:
             if x.is_complex():
                 buf = torch.view_as_complex(buf)
-            return buf
-        return zero_volume_tensor(device=x.device, dtype=x.dtype)
+            return buf, tok.new_empty(0)
+        return zero_volume_tensor(device=x.device, dtype=x.dtype), tok.new_empty(0)
 
     @staticmethod
-    def backward(ctx, grad_out: torch.Tensor):
+    def backward(ctx, grad_out: torch.Tensor, grad_tok):
         module: SumReduce = ctx.module
         P = module.P_src
+        tg = torch.empty(0, device=ctx.in_device)
         if not (is_distributed() and P.active and P.size > 1):
-            return grad_out, None
+            return grad_out, tg, None
         root_world = module.P_dst.ranks[0]
         if world_rank() == root_world:
             buf = _comm_clone(grad_out)
@@ -206,7 +262,7 @@ class _SumReduceFn(torch.autograd.Function):
         dist.broadcast(buf, src=root_world, group=P.group)
         if ctx.in_dtype.is_complex:
             buf = torch.view_as_complex(buf)
-        return buf, None
+        return buf, tg, None
 
 
 class SumReduce(torch.nn.Module):
@@ -222,7 +278,9 @@ class SumReduce(torch.nn.Module):
         self.P_dst = P_dst
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return _SumReduceFn.apply(x, self)
+        out, tok = _SumReduceFn.apply(x, chain_token(x), self)
+        set_chain(tok)
+        return out
 
 
 # ---------------------------------------------------------------------------
@@ -231,26 +289,27 @@ class SumReduce(torch.nn.Module):
 
 class _AllReduceSumFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, P: Partition) -> torch.Tensor:
+    def forward(ctx, x: torch.Tensor, tok: torch.Tensor, P: Partition):
         ctx.P = P
         if not (is_distributed() and P.active and P.size > 1):
-            return x.clone()
+            return x.clone(), tok.new_empty(0)
         buf = _comm_clone(x)
         dist.all_reduce(buf, op=dist.ReduceOp.SUM, group=P.group)
         if x.is_complex():
             buf = torch.view_as_complex(buf)
-        return buf
+        return buf, tok.new_empty(0)
 
     @staticmethod
-    def backward(ctx, grad_out: torch.Tensor):
+    def backward(ctx, grad_out: torch.Tensor, grad_tok):
         P = ctx.P
+        tg = torch.empty(0, device=grad_out.device)
         if not (is_distributed() and P.active and P.size > 1):
-            return grad_out, None
+            return grad_out, tg, None
         buf = _comm_clone(grad_out)
         dist.all_reduce(buf, op=dist.ReduceOp.SUM, group=P.group)
         if grad_out.is_complex():
             buf = torch.view_as_complex(buf)
-        return buf, None
+        return buf, tg, None
 
 
 class AllReduceSum(torch.nn.Module):
@@ -261,7 +320,9 @@ class AllReduceSum(torch.nn.Module):
         self.P = P
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return _AllReduceSumFn.apply(x, self.P)
+        out, tok = _AllReduceSumFn.apply(x, chain_token(x), self.P)
+        set_chain(tok)
+        return out
 
 
 # ---------------------------------------------------------------------------
@@ -337,6 +398,9 @@ class _RepartitionPlan:
         self.sends.sort(key=lambda t: t[0])
         self.recvs.sort(key=lambda t: t[0])
 
+        # per-(dtype, device) packed-exchange descriptors (built lazily)
+        self._packed: Dict[Tuple, "_PackedMeta"] = {}
+
         # Identity detection: this rank keeps its whole block and exchanges
         # nothing -> the repartition is a no-op here (e.g. P_x == P_m for the
         # two-phase partition (1,1,1,N,1,1)).  Skipping it avoids a full
@@ -366,9 +430,188 @@ def _get_plan(P_src: Partition, P_dst: Partition, gshape: Tuple[int, ...]) -> _R
     return plan
 
 
-def _execute_plan(plan: _RepartitionPlan, x: torch.Tensor,
-                  dtype: torch.dtype, device) -> torch.Tensor:
-    """Run a repartition plan on tensor ``x`` (zero-volume on inactive src)."""
+_PACK_REC = 20   # longs per piece record (csrc/pack.hip)
+
+
+def _box_record(shape: Tuple[int, ...], box, wpe: int, flat_off: int):
+    """One pack descriptor record for a box of a contiguous tensor.
+
+    ``shape``: element-space local tensor shape; ``box``: per-dim (start,
+    stop); ``wpe``: words per element (2 for complex viewed as real words).
+    Contiguous inner runs are merged host-side so the kernel's divmod chain
+    is 2-3 deep in practice.
+    """
+    nd = len(shape)
+    estr = [1] * nd
+    for d in range(nd - 2, -1, -1):
+        estr[d] = estr[d + 1] * shape[d + 1]
+    lens = [hi - lo for lo, hi in box]
+    numel = 1
+    for l in lens:
+        numel *= l
+    numel *= wpe
+    dims = [int(l) for l in lens] + ([wpe] if wpe > 1 else [])
+    strs = [estr[d] * wpe for d in range(nd)] + ([1] if wpe > 1 else [])
+    off = sum(lo * estr[d] for d, (lo, _) in enumerate(box)) * wpe
+    # merge contiguous runs, inner -> outer
+    blocks = []  # (len, stride), innermost first
+    for l, s in zip(reversed(dims), reversed(strs)):
+        if l == 1:
+            continue
+        if blocks and s == blocks[-1][0] * blocks[-1][1]:
+            blocks[-1] = (blocks[-1][0] * l, blocks[-1][1])
+        else:
+            blocks.append((l, s))
+    if not blocks:
+        blocks = [(1, 1)]
+    blocks.reverse()  # outer -> inner
+    assert len(blocks) <= 8, "pack box rank > 8 after merging"
+    mdims = [l for l, _ in blocks] + [1] * (8 - len(blocks))
+    mstrs = [s for _, s in blocks] + [0] * (8 - len(blocks))
+    return [flat_off, off, numel, len(blocks), *mdims, *mstrs], numel
+
+
+class _PackedMeta:
+    """Cached device descriptors for one (plan, dtype, device)."""
+
+    __slots__ = ("send_desc", "send_total", "send_ranges", "send_max",
+                 "recv_desc", "recv_total", "recv_ranges", "recv_max",
+                 "word_dtype")
+
+    def __init__(self, plan: "_RepartitionPlan", dtype: torch.dtype, device):
+        wpe = 2 if dtype.is_complex else 1
+        self.word_dtype = (torch.float64 if dtype in (torch.complex128, torch.float64)
+                           else torch.float32)
+        src_shape = None
+        if plan.P_src.active:
+            sb = block_bounds(plan.P_src, plan.gshape, plan.P_src.rank)
+            src_shape = tuple(b - a for a, b in sb)
+
+        recs, ranges, off, mx = [], [], 0, 1
+        for peer, ssl in plan.sends:
+            box = [(s.start or 0, s.stop) for s in ssl]
+            rec, n = _box_record(src_shape, box, wpe, off)
+            recs.append(rec)
+            ranges.append((peer, off, n))
+            off += n
+            mx = max(mx, n)
+        self.send_desc = (torch.tensor(recs, dtype=torch.int64, device=device)
+                          if recs else None)
+        self.send_total, self.send_ranges, self.send_max = off, ranges, mx
+
+        recs, ranges, off, mx = [], [], 0, 1
+        for peer, dsl, shp in plan.recvs:
+            box = [(s.start or 0, s.stop) for s in dsl]
+            rec, n = _box_record(tuple(plan.out_shape), box, wpe, off)
+            recs.append(rec)
+            ranges.append((peer, off, n))
+            off += n
+            mx = max(mx, n)
+        self.recv_desc = (torch.tensor(recs, dtype=torch.int64, device=device)
+                          if recs else None)
+        self.recv_total, self.recv_ranges, self.recv_max = off, ranges, mx
+
+
+def _issue_plan_packed(plan: "_RepartitionPlan", x: torch.Tensor,
+                       out: torch.Tensor, dtype: torch.dtype, device):
+    """GPU packed exchange: one pack kernel -> grouped isend/irecv on slices
+    of a flat staging buffer -> one unpack kernel at completion.  Returns
+    None when the extension is unavailable (caller falls back to slicing)."""
+    from . import _ext
+    ext = _ext.get(required=False)
+    if ext is None:
+        return None
+    if not x.is_contiguous():
+        x = x.contiguous()
+
+    key = (dtype, x.device.index)
+    meta = plan._packed.get(key)
+    if meta is None:
+        meta = _PackedMeta(plan, dtype, device)
+        plan._packed[key] = meta
+
+    word = meta.word_dtype
+    # reinterpret the contiguous storage as flat words (complex -> real pairs)
+    x_words = _as_real(x).reshape(-1) if x.numel() else torch.empty(0, dtype=word, device=device)
+    out_words = _as_real(out).reshape(-1) if out.numel() else torch.empty(0, dtype=word, device=device)
+
+    ops = []
+    flat_send = None
+    if meta.send_desc is not None:
+        flat_send = torch.empty(meta.send_total, dtype=word, device=device)
+        ext.pack_boxes(x_words, flat_send, meta.send_desc, meta.send_max)
+        for peer, off, n in meta.send_ranges:
+            ops.append(dist.P2POp(dist.isend, flat_send.narrow(0, off, n), peer))
+    flat_recv = None
+    if meta.recv_desc is not None:
+        flat_recv = torch.empty(meta.recv_total, dtype=word, device=device)
+        for peer, off, n in meta.recv_ranges:
+            ops.append(dist.P2POp(dist.irecv, flat_recv.narrow(0, off, n), peer))
+
+    if plan.local_copy is not None:
+        ssl, dsl = plan.local_copy
+        out[dsl] = x[ssl]
+
+    reqs = dist.batch_isend_irecv(ops) if ops else ()
+
+    unpack = None
+    if flat_recv is not None:
+        def unpack(fr=flat_recv, ow=out_words, d=meta.recv_desc, m=meta.recv_max):
+            ext.unpack_boxes(fr, ow, d, m)
+    return _PendingRepart(plan, out, reqs, (), dtype,
+                          send_bufs=(flat_send,) if flat_send is not None else (),
+                          unpack=unpack)
+
+
+class _PendingRepart:
+    """In-flight repartition: sends/recvs issued, completion deferred.
+
+    The issue/complete split is the comm/compute overlap hook (SURVEY.md K9,
+    VERDICT.md round-1 item 1): issue ALL chunk exchanges first, then
+    complete+compute chunk by chunk — on RCCL the exchanges run on the NCCL
+    stream while the default stream computes earlier chunks; on gloo the
+    transfers progress on background threads during host compute.
+    """
+
+    __slots__ = ("plan", "out", "reqs", "recv_bufs", "dtype", "done",
+                 "send_bufs", "_unpack")
+
+    def __init__(self, plan, out, reqs, recv_bufs, dtype,
+                 send_bufs=(), unpack=None):
+        self.plan = plan
+        self.out = out
+        self.reqs = reqs
+        self.recv_bufs = recv_bufs
+        self.dtype = dtype
+        self.done = False
+        self.send_bufs = send_bufs
+        self._unpack = unpack
+
+    def complete(self) -> torch.Tensor:
+        if self.done:
+            return self.out
+        for r in self.reqs:
+            r.wait()
+        if self._unpack is not None:
+            self._unpack()
+        else:
+            for buf, dsl, shp in self.recv_bufs:
+                if self.dtype.is_complex:
+                    self.out[dsl] = torch.view_as_complex(buf)
+                else:
+                    self.out[dsl] = buf
+        self.done = True
+        # release send/recv staging
+        self.reqs = ()
+        self.recv_bufs = ()
+        self.send_bufs = ()
+        self._unpack = None
+        return self.out
+
+
+def _issue_plan(plan: _RepartitionPlan, x: torch.Tensor,
+                dtype: torch.dtype, device) -> _PendingRepart:
+    """Pack + post all sends/recvs of a repartition plan; defer completion."""
     P_src, P_dst = plan.P_src, plan.P_dst
 
     if plan.out_shape is None:
@@ -380,7 +623,12 @@ def _execute_plan(plan: _RepartitionPlan, x: torch.Tensor,
         if plan.local_copy is not None:
             ssl, dsl = plan.local_copy
             out[dsl] = x[ssl]
-        return out
+        return _PendingRepart(plan, out, (), (), dtype)
+
+    if x.is_cuda and (plan.sends or plan.recvs or plan.local_copy):
+        pending = _issue_plan_packed(plan, x, out, dtype, device)
+        if pending is not None:
+            return pending
 
     ops = []
     send_bufs = []
@@ -399,36 +647,120 @@ def _execute_plan(plan: _RepartitionPlan, x: torch.Tensor,
         ssl, dsl = plan.local_copy
         out[dsl] = x[ssl]
 
-    if ops:
-        reqs = dist.batch_isend_irecv(ops)
-        for r in reqs:
-            r.wait()
+    reqs = dist.batch_isend_irecv(ops) if ops else ()
+    # send buffers stay alive until completion (explicitly referenced)
+    return _PendingRepart(plan, out, reqs, recv_bufs, dtype, send_bufs=send_bufs)
 
-    for buf, dsl, shp in recv_bufs:
-        if dtype.is_complex:
-            out[dsl] = torch.view_as_complex(buf)
-        else:
-            out[dsl] = buf
+
+def _execute_plan(plan: _RepartitionPlan, x: torch.Tensor,
+                  dtype: torch.dtype, device) -> torch.Tensor:
+    """Run a repartition plan on tensor ``x`` (zero-volume on inactive src)."""
+    return _issue_plan(plan, x, dtype, device).complete()
+
+
+# ---------------------------------------------------------------------------
+# split-phase (issue/complete) repartition for comm/compute overlap
+# ---------------------------------------------------------------------------
+#
+# The pipelined pencil chain (nn/block.py) issues ALL channel-chunk exchanges
+# before completing/computing any of them: on RCCL the exchanges proceed on
+# the NCCL stream while the default stream transforms earlier chunks; on gloo
+# they progress on background threads under host compute.  The split is two
+# autograd Functions linked by a zero-volume token; the in-flight state lives
+# in a ticket registry.
+#
+# BACKWARD runs the adjoint exchange SYNCHRONOUSLY inside the complete-fn's
+# backward node (the finished gradient is handed to the issue-fn's backward
+# via the registry).  Splitting the adjoint's post/wait across two backward
+# nodes deadlocks: the autograd engine schedules per-rank-different compute
+# nodes (zero-volume ranks have fewer) between them, so collectives leave
+# graph order differently on different ranks (observed: two ranks parked in
+# a Broadcast-adjoint reduce while two waited on the adjoint p2p).  Forward
+# overlap — the measured win — is unaffected: its issue/complete interleave
+# is explicit, identical code on every rank.
+
+import itertools as _itertools
+
+_TICKETS: Dict[int, "_PendingRepart"] = {}
+_ADJ_GRADS: Dict[int, torch.Tensor] = {}
+_ticket_counter = _itertools.count(1)
+
+
+class _RepartIssueFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, tok: torch.Tensor, module: "Repartition",
+                gshape: Tuple[int, ...], ticket: int) -> torch.Tensor:
+        plan = _get_plan(module.P_src, module.P_dst, gshape)
+        _TICKETS[ticket] = _issue_plan(plan, x, x.dtype, x.device)
+        ctx.ticket = ticket
+        return torch.empty(0, dtype=x.dtype, device=x.device)
+
+    @staticmethod
+    def backward(ctx, g_token: torch.Tensor):
+        gx = _ADJ_GRADS.pop(ctx.ticket)
+        return gx, torch.empty(0, device=gx.device), None, None, None
+
+
+class _RepartCompleteFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, token: torch.Tensor, tok: torch.Tensor,
+                module: "Repartition", gshape: Tuple[int, ...],
+                ticket: int):
+        ctx.module = module
+        ctx.gshape = gshape
+        ctx.ticket = ticket
+        return _TICKETS.pop(ticket).complete(), tok.new_empty(0)
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor, grad_tok):
+        module: Repartition = ctx.module
+        plan = _get_plan(module.P_dst, module.P_src, ctx.gshape)
+        _ADJ_GRADS[ctx.ticket] = _issue_plan(
+            plan, grad_out.contiguous(), grad_out.dtype,
+            grad_out.device).complete()
+        token_grad = torch.empty(0, dtype=grad_out.dtype, device=grad_out.device)
+        return token_grad, torch.empty(0, device=grad_out.device), None, None, None
+
+
+def repartition_issue(R: "Repartition", x: torch.Tensor, global_shape):
+    """Post the exchange for ``x``; returns a handle for repartition_complete."""
+    g = tuple(int(s) for s in global_shape)
+    plan = _get_plan(R.P_src, R.P_dst, g)
+    if plan.is_identity:
+        return ("id", x)
+    t = next(_ticket_counter)
+    token = _RepartIssueFn.apply(x, chain_token(x), R, g, t)
+    set_chain(token)  # the issue's own output token doubles as the chain link
+    return ("tok", token, R, g, t)
+
+
+def repartition_complete(handle) -> torch.Tensor:
+    if handle[0] == "id":
+        return handle[1]
+    _, token, R, g, t = handle
+    out, tok = _RepartCompleteFn.apply(token, chain_token(token), R, g, t)
+    set_chain(tok)
     return out
 
 
 class _RepartitionFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, module: "Repartition", gshape: Tuple[int, ...]) -> torch.Tensor:
+    def forward(ctx, x: torch.Tensor, tok: torch.Tensor, module: "Repartition",
+                gshape: Tuple[int, ...]):
         ctx.module = module
         ctx.gshape = gshape
         ctx.dtype = x.dtype
         ctx.device = x.device
         plan = _get_plan(module.P_src, module.P_dst, gshape)
-        return _execute_plan(plan, x, x.dtype, x.device)
+        return _execute_plan(plan, x, x.dtype, x.device), tok.new_empty(0)
 
     @staticmethod
-    def backward(ctx, grad_out: torch.Tensor):
+    def backward(ctx, grad_out: torch.Tensor, grad_tok):
         module: Repartition = ctx.module
         # adjoint = reversed repartition on the gradient
         plan = _get_plan(module.P_dst, module.P_src, ctx.gshape)
         gx = _execute_plan(plan, grad_out.contiguous(), ctx.dtype, ctx.device)
-        return gx, None, None
+        return gx, torch.empty(0, device=ctx.device), None, None
 
 
 class Repartition(torch.nn.Module):
@@ -489,7 +821,9 @@ class Repartition(torch.nn.Module):
         plan = _get_plan(self.P_src, self.P_dst, g)
         if plan.is_identity:
             return x
-        return _RepartitionFn.apply(x, self, g)
+        out, tok = _RepartitionFn.apply(x, chain_token(x), self, g)
+        set_chain(tok)
+        return out
 
 
 # DistDL exposes the same op under a second name; the reference's NS trainer

@@ -39,4 +39,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "all corner boxes adjoint wrt x in one launch");
   m.def("spectral_corners_bwd_w", &spectral_corners_bwd_w,
         "multi-corner spectral grad-W (gw[i,o,e] = sum_b x conj(gy))");
+  m.def("pack_boxes", &pack_boxes,
+        "gather repartition boxes into one flat staging buffer");
+  m.def("unpack_boxes", &unpack_boxes,
+        "scatter one flat staging buffer into repartition boxes");
 }

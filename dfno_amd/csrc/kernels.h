@@ -71,6 +71,13 @@ at::Tensor dft_rfft_trunc_adj(const at::Tensor& gy, int64_t dim, int64_t n);
 at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m);
 at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m);
 
+// repartition pack/unpack (pack.hip): gather/scatter block-intersection
+// boxes between a contiguous tensor (word view) and one flat staging buffer
+void pack_boxes(const at::Tensor& src, at::Tensor& flat,
+                const at::Tensor& desc, int64_t max_numel);
+void unpack_boxes(const at::Tensor& flat, at::Tensor& dst,
+                  const at::Tensor& desc, int64_t max_numel);
+
 // corner-block spectral contraction on the truncated complex spectrum:
 //   y[b,o,f] += sum_i x[b,i,f] * w[i,o,f_box]  for f in the corner box
 void spectral_corner_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,

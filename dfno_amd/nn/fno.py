@@ -71,7 +71,9 @@ class DistributedFNONd(nn.Module):
         self.dt_comm = 0.0
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        self.dt_comm = 0.0
+        from ..comm import reset_chain
+        reset_chain()  # fresh comm ordering chain per forward (spans the
+        self.dt_comm = 0.0  # model AND the loss that follows)
 
         x = self._lift(x)
 

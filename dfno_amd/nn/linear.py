@@ -60,7 +60,10 @@ class BroadcastedLinear(nn.Module):
 
         with comm_region() as r:
             W = self.W_bcast(self.W)
-            b = self.b_bcast(self.b)
+            # bias-less linears keep the parameter (state-dict parity) but
+            # never ship it: a dead broadcast would still run its adjoint
+            # reduce through the comm ordering chain
+            b = self.b_bcast(self.b) if self.bias else None
         self.dt_comm += r.host_dt
 
-        return linear_nd(x, W, b if self.bias else None, self.dim, activation)
+        return linear_nd(x, W, b, self.dim, activation)
