@@ -27,6 +27,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dft_c2c", &dft_c2c, "truncated/padded complex DFT along a dim");
   m.def("dft_rfft_trunc", &dft_rfft_trunc, "real->kept-low-modes DFT (last dim)");
   m.def("dft_rfft_trunc_adj", &dft_rfft_trunc_adj, "adjoint of dft_rfft_trunc");
+  m.def("dft_rfft_trunc_adj_acc", &dft_rfft_trunc_adj_acc,
+        "adjoint of dft_rfft_trunc with fused accumulate addend");
   m.def("dft_pad_irfft", &dft_pad_irfft, "kept modes -> real inverse (last dim)");
   m.def("dft_pad_irfft_adj", &dft_pad_irfft_adj, "adjoint of dft_pad_irfft");
   m.def("spectral_corner_fwd", &spectral_corner_fwd,

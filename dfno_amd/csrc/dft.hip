@@ -798,10 +798,14 @@ __global__ __launch_bounds__(LB) void dft_r2c_last_kernel(
 
 // Output lines are written through an LDS tile and stored cooperatively
 // (coalesced); inputs are 64-128B contiguous per line and read directly.
+// acc != nullptr fuses a same-shape addend into the writeback (the block
+// residual's input-gradient accumulate: out = idft + acc in one pass instead
+// of a separate aten add over the full activation — docs/ROADMAP.md item 4).
 template <typename T, int MCAP, int NT = 0, int LB = kBlock>
 __global__ __launch_bounds__(LB) void dft_c2r_last_kernel(
     const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
-    long lines, int N_, int m_, T scale, bool factors) {
+    long lines, int N_, int m_, T scale, bool factors,
+    const T* __restrict__ acc = nullptr) {
   // NT > 0 pins N and m (== MCAP) at compile time: full unroll + folded
   // twiddle offsets, as in dft_r2c_glds_kernel.  LB: see dft_r2c_last_kernel.
   const int N = NT > 0 ? NT : N_;
@@ -857,17 +861,27 @@ __global__ __launch_bounds__(LB) void dft_c2r_last_kernel(
     if constexpr (std::is_same<T, float>::value) {
       const long base = l0 * N;
       if ((nl * N) % 4 == 0 && (base % 4 == 0) &&
-          ((reinterpret_cast<uintptr_t>(out) & 15) == 0)) {
-        for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4)
-          *reinterpret_cast<float4*>(out + base + idx) =
-              *reinterpret_cast<const float4*>(&tile[idx]);
+          ((reinterpret_cast<uintptr_t>(out) & 15) == 0) &&
+          (acc == nullptr || (reinterpret_cast<uintptr_t>(acc) & 15) == 0)) {
+        if (acc != nullptr) {
+          for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4) {
+            const float4 t4 = *reinterpret_cast<const float4*>(&tile[idx]);
+            const float4 a4 = *reinterpret_cast<const float4*>(acc + base + idx);
+            *reinterpret_cast<float4*>(out + base + idx) =
+                make_float4(t4.x + a4.x, t4.y + a4.y, t4.z + a4.z, t4.w + a4.w);
+          }
+        } else {
+          for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4)
+            *reinterpret_cast<float4*>(out + base + idx) =
+                *reinterpret_cast<const float4*>(&tile[idx]);
+        }
       } else {
         for (int idx = threadIdx.x; idx < nl * N; idx += LB)
-          out[base + idx] = tile[idx];
+          out[base + idx] = tile[idx] + (acc ? acc[base + idx] : 0.f);
       }
     } else {
       for (int idx = threadIdx.x; idx < nl * N; idx += LB)
-        out[l0 * N + idx] = tile[idx];
+        out[l0 * N + idx] = tile[idx] + (acc ? acc[l0 * N + idx] : T(0));
     }
   }
 }
@@ -1115,7 +1129,8 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
 }
 
 static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
-                               double scale, bool factors) {
+                               double scale, bool factors,
+                               const at::Tensor& accum = at::Tensor()) {
   TORCH_CHECK(y.is_cuda() && y.is_contiguous(), "dft_c2r: contiguous GPU input");
   TORCH_CHECK(dim == y.dim() - 1, "dft_c2r: last-dim only");
   TORCH_CHECK(y.scalar_type() == at::kComplexFloat || y.scalar_type() == at::kComplexDouble,
@@ -1141,7 +1156,7 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
       hipLaunchKernelGGL((dft_c2r_last_kernel<scalar_t, MC, NTV>),             \
                          dim3(grid), dim3(kBlock), smem, stream, inp, op,      \
                          tw.data_ptr<scalar_t>(), lines, N, (int)m,            \
-                         (scalar_t)scale, factors);
+                         (scalar_t)scale, factors, accp);
 #define C2R_BIG(MC, LBV)                                                       \
       { long nt2 = (lines + LBV - 1) / LBV;                                    \
         int grid2 = (int)std::min(nt2, 2048L);                                 \
@@ -1149,14 +1164,22 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
         hipLaunchKernelGGL((dft_c2r_last_kernel<scalar_t, MC, 0, LBV>),        \
                            dim3(grid2), dim3(LBV), smem2, stream, inp, op,     \
                            tw.data_ptr<scalar_t>(), lines, N, (int)m,          \
-                           (scalar_t)scale, factors); }
+                           (scalar_t)scale, factors, accp); }
 #define C2R_BIG_M(LBV)                                                         \
       if (m <= 8) { C2R_BIG(8, LBV) } else if (m <= 16) { C2R_BIG(16, LBV) }   \
       else if (m <= 24) { C2R_BIG(24, LBV) } else { C2R_BIG(32, LBV) }
+  if (accum.defined() && accum.numel() > 0) {
+    TORCH_CHECK(accum.is_cuda() && accum.is_contiguous() &&
+                accum.numel() == out.numel() &&
+                accum.scalar_type() == out.scalar_type(),
+                "dft_c2r: bad accumulate tensor");
+  }
   AT_DISPATCH_FLOATING_TYPES(out.scalar_type(), "dft_c2r", [&] {
     size_t smem = sizeof(scalar_t) * (size_t)kBlock * N;
     auto inp = reinterpret_cast<const scalar_t*>(y.data_ptr());
     auto op = out.data_ptr<scalar_t>();
+    const scalar_t* accp = (accum.defined() && accum.numel() > 0)
+                               ? accum.data_ptr<scalar_t>() : nullptr;
     if (N > kMaxN) {
       if (N <= 128) { C2R_BIG_M(128) } else { C2R_BIG_M(64) }
       return;
@@ -1175,7 +1198,7 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
     } else {
       DFT_MDISPATCH(dft_c2r_last_kernel, dim3(grid), dim3(kBlock), smem,
                     stream, inp, op, tw.data_ptr<scalar_t>(), lines, N,
-                    (int)m, (scalar_t)scale, factors)
+                    (int)m, (scalar_t)scale, factors, accp)
     }
   });
 #undef C2R_BIG_M
@@ -1192,6 +1215,13 @@ at::Tensor dft_rfft_trunc(const at::Tensor& x, int64_t dim, int64_t m) {
 at::Tensor dft_rfft_trunc_adj(const at::Tensor& gy, int64_t dim, int64_t n) {
   // gx_j = Re(sum_k gY_k w^{+jk}) — c2r with unit factors / unit scale
   return dft_c2r_impl(gy, dim, n, 1.0, /*factors=*/false);
+}
+
+at::Tensor dft_rfft_trunc_adj_acc(const at::Tensor& gy, int64_t dim, int64_t n,
+                                  const at::Tensor& accum) {
+  // adjoint with a fused addend: gx = idft(gy) + accum in one writeback
+  // (the block residual's input-grad accumulate)
+  return dft_c2r_impl(gy, dim, n, 1.0, /*factors=*/false, accum);
 }
 
 at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m) {

@@ -68,6 +68,8 @@ at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,
                    int64_t m_lo, int64_t m_hi, bool analysis, double scale);
 at::Tensor dft_rfft_trunc(const at::Tensor& x, int64_t dim, int64_t m);
 at::Tensor dft_rfft_trunc_adj(const at::Tensor& gy, int64_t dim, int64_t n);
+at::Tensor dft_rfft_trunc_adj_acc(const at::Tensor& gy, int64_t dim, int64_t n,
+                                  const at::Tensor& accum);
 at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m);
 at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m);
 

@@ -20,6 +20,9 @@
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
+#include <map>
+#include <vector>
+
 #include "kernels.h"
 #include "gelu_math.h"
 
@@ -1148,6 +1151,43 @@ __global__ __launch_bounds__(kBlock) void adam_multi_kernel(
   }
 }
 
+// Uniform-size variant: a group of SAME-length tensors (the 32 spectral
+// corner weights dominate the parameter set) updates with blockIdx.y as the
+// tensor index — no per-iteration tensor search / end4 kernarg loads at all
+// (docs/ROADMAP.md item 4: runtime-trip-count folding for the Adam loop).
+__global__ __launch_bounds__(kBlock) void adam_multi_uniform_kernel(
+    AdamTab tab, long n4, float lr, float b1, float b2, float eps, float wd) {
+  const int t = blockIdx.y;
+  float* __restrict__ p = tab.p[t];
+  const float* __restrict__ g = tab.g[t];
+  float* __restrict__ m = tab.m[t];
+  float* __restrict__ v = tab.v[t];
+  const float c1 = tab.c1[t], c2 = tab.c2[t];
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n4; i += stride) {
+    long j = i * 4;
+    float4 pv = *reinterpret_cast<float4*>(p + j);
+    const float4 gv = *reinterpret_cast<const float4*>(g + j);
+    float4 mv = *reinterpret_cast<float4*>(m + j);
+    float4 vv = *reinterpret_cast<float4*>(v + j);
+    float pr[4] = {pv.x, pv.y, pv.z, pv.w};
+    float gr[4] = {gv.x, gv.y, gv.z, gv.w};
+    float mr[4] = {mv.x, mv.y, mv.z, mv.w};
+    float vr[4] = {vv.x, vv.y, vv.z, vv.w};
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float gg = gr[k] + wd * pr[k];
+      mr[k] = b1 * mr[k] + (1.f - b1) * gg;
+      vr[k] = b2 * vr[k] + (1.f - b2) * gg * gg;
+      pr[k] -= lr * (mr[k] / c1) / (sqrtf(vr[k] / c2) + eps);
+    }
+    *reinterpret_cast<float4*>(p + j) = make_float4(pr[0], pr[1], pr[2], pr[3]);
+    *reinterpret_cast<float4*>(m + j) = make_float4(mr[0], mr[1], mr[2], mr[3]);
+    *reinterpret_cast<float4*>(v + j) = make_float4(vr[0], vr[1], vr[2], vr[3]);
+  }
+}
+
 }  // namespace
 
 void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
@@ -1158,18 +1198,11 @@ void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
               ps.size() == vs.size() && ps.size() == steps.size(),
               "adam batch: length mismatch");
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  AdamTab tab;
-  tab.nt = 0;
-  long total4 = 0;
-  auto flush = [&]() {
-    if (tab.nt == 0) return;
-    int grid = grid_for(total4, kBlock);
-    hipLaunchKernelGGL(adam_multi_kernel, dim3(grid), dim3(kBlock), 0, stream,
-                       tab, total4, (float)lr, (float)beta1, (float)beta2,
-                       (float)eps, (float)weight_decay);
-    tab.nt = 0;
-    total4 = 0;
-  };
+
+  // bucket vec4-eligible tensors by length: same-size groups (>= 2) take
+  // the uniform kernel, the ragged rest the searched multi-tensor kernel
+  std::vector<size_t> vec_idx;
+  std::map<long, std::vector<size_t>> by_len;
   for (size_t i = 0; i < ps.size(); ++i) {
     long n = ps[i].numel();
     bool vec = ps[i].scalar_type() == at::kFloat && (n % 4 == 0) && n > 0 &&
@@ -1182,18 +1215,59 @@ void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                  weight_decay, steps[i]);
       continue;
     }
-    int t = tab.nt++;
+    by_len[n].push_back(i);
+  }
+
+  auto fill = [&](AdamTab& tab, size_t i, int t) {
     tab.p[t] = ps[i].data_ptr<float>();
     tab.g[t] = gs[i].data_ptr<float>();
     tab.m[t] = ms[i].data_ptr<float>();
     tab.v[t] = vs[i].data_ptr<float>();
-    total4 += n / 4;
-    tab.end4[t] = total4;
     tab.c1[t] = (float)(1.0 - std::pow(beta1, (double)steps[i]));
     tab.c2[t] = (float)(1.0 - std::pow(beta2, (double)steps[i]));
-    if (tab.nt == kAdamMaxT) flush();
+  };
+
+  AdamTab rag;
+  rag.nt = 0;
+  long total4 = 0;
+  auto flush_ragged = [&]() {
+    if (rag.nt == 0) return;
+    int grid = grid_for(total4, kBlock);
+    hipLaunchKernelGGL(adam_multi_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                       rag, total4, (float)lr, (float)beta1, (float)beta2,
+                       (float)eps, (float)weight_decay);
+    rag.nt = 0;
+    total4 = 0;
+  };
+
+  for (auto& [n, idxs] : by_len) {
+    if (idxs.size() < 2) {
+      for (size_t i : idxs) {
+        int t = rag.nt++;
+        fill(rag, i, t);
+        total4 += n / 4;
+        rag.end4[t] = total4;
+        if (rag.nt == kAdamMaxT) flush_ragged();
+      }
+      continue;
+    }
+    long n4 = n / 4;
+    for (size_t off = 0; off < idxs.size(); off += kAdamMaxT) {
+      AdamTab tab;
+      int cnt = (int)std::min<size_t>(kAdamMaxT, idxs.size() - off);
+      for (int t = 0; t < cnt; ++t) fill(tab, idxs[off + t], t);
+      tab.nt = cnt;
+      // size grid.x so that grid.x * cnt covers the chip at ~8 blocks/CU
+      long gx = (n4 + kBlock - 1) / kBlock;
+      long cap = std::max(1L, (256L * 8) / cnt);
+      if (gx > cap) gx = cap;
+      hipLaunchKernelGGL(adam_multi_uniform_kernel, dim3((int)gx, cnt),
+                         dim3(kBlock), 0, stream, tab, n4, (float)lr,
+                         (float)beta1, (float)beta2, (float)eps,
+                         (float)weight_decay);
+    }
   }
-  flush();
+  flush_ragged();
   hipError_t lerr = hipGetLastError();
   TORCH_CHECK(lerr == hipSuccess, "adam_multi launch failed: ",
               hipGetErrorString(lerr));

@@ -149,12 +149,16 @@ struct MultiGeom {
   int ncorners;
 };
 
-template <typename T, int OTILE, bool CONJT>
+// NIN > 0 pins the contraction depth (n_in) at compile time: the channel
+// loop fully unrolls with folded weight offsets (the round-1 probe lesson:
+// runtime trip counts serialize on a uniform-load wait per iteration,
+// docs/ROADMAP.md item 4; folding recovered 1.5-2x on the dft/head kernels).
+template <typename T, int OTILE, bool CONJT, int NIN = 0>
 __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
     const T* __restrict__ x, MultiGeom<T> mg, T* __restrict__ y,
     int B, int I, int O, long Ftot) {
   const int n_out = CONJT ? I : O;
-  const int n_in = CONJT ? O : I;
+  const int n_in = NIN > 0 ? NIN : (CONJT ? O : I);
   const long ntiles = (n_out + OTILE - 1) / OTILE;
   const long total = mg.cum[mg.ncorners] * B;
 
@@ -181,7 +185,7 @@ __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
     for (int k = 0; k < OTILE; ++k) { accr[k] = T(0); acci[k] = T(0); }
 
     const T* xb = x + 2 * (((long)b * n_in) * Ftot + f);
-    for (int i = 0; i < n_in; ++i) {
+    auto chan = [&](int i) {
       T xr = xb[2 * (long)i * Ftot];
       T xi = xb[2 * (long)i * Ftot + 1];
       if (CONJT) {
@@ -202,6 +206,12 @@ __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
           }
         }
       }
+    };
+    if constexpr (NIN > 0) {
+#pragma unroll
+      for (int i = 0; i < NIN; ++i) chan(i);
+    } else {
+      for (int i = 0; i < n_in; ++i) chan(i);
     }
 
     T* yb = y + 2 * (((long)b * n_out + o0) * Ftot + f);
@@ -400,10 +410,29 @@ static void launch_corners(const at::Tensor& x,
     }
     if (mg.ncorners == 0) continue;
     int grid = grid_for_s(mg.cum[mg.ncorners] * B);
-    hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT>), dim3(grid),
-                       dim3(kBlock), 0, stream,
-                       reinterpret_cast<const T*>(x.data_ptr()), mg,
-                       reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
+    const int n_in = CONJT ? O : I;
+    // compile-time channel depth for the common widths (flagship 20)
+    if (n_in == 20) {
+      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT, 20>),
+                         dim3(grid), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const T*>(x.data_ptr()), mg,
+                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
+    } else if (n_in == 32) {
+      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT, 32>),
+                         dim3(grid), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const T*>(x.data_ptr()), mg,
+                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
+    } else if (n_in == 8) {
+      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT, 8>),
+                         dim3(grid), dim3(kBlock), 0, stream,
+                         reinterpret_cast<const T*>(x.data_ptr()), mg,
+                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
+    } else {
+      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT>), dim3(grid),
+                         dim3(kBlock), 0, stream,
+                         reinterpret_cast<const T*>(x.data_ptr()), mg,
+                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
+    }
   }
 }
 

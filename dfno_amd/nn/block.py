@@ -298,17 +298,41 @@ class DistributedFNOBlock(nn.Module):
         if nch > 1:
             y = self._forward_pipelined(x, nch)
         else:
-            y = self._forward_seq(x)
+            y, x_in = self._forward_seq(x)
 
         return linear_res_gelu(x_in, W_res, y)
 
-    def _forward_seq(self, x: torch.Tensor) -> torch.Tensor:
+    def _forward_seq(self, x: torch.Tensor):
+        """Returns (chain output, epilogue input).  The epilogue input is
+        either ``x`` itself or its StashGradFn alias when the residual
+        input-gradient fuses into the rfft adjoint (ops/fft.py stash)."""
+        from ..ops.fft import (StashGradFn, new_stash_key, rfft_trunc_stash,
+                               stash_fusable)
+
         saved: Dict[int, int] = {}   # pre-truncation extent per dim
+        x_orig = x
         with comm_region() as r:
             x = self.R1(x)
         self.dt_comm += r.host_dt
 
-        x = self._fwd_m(x, saved)
+        x_epi = x_orig
+        outermost = self.dim_m[-1]
+        if (x is x_orig
+                and stash_fusable(x, outermost, self.restrict_prefixes[outermost])):
+            # R1 is an identity here: the chain's input-grad producer is the
+            # rfft adjoint — fold the epilogue's gradient into its writeback
+            key = new_stash_key()
+            saved[outermost] = x.shape[outermost] // 2 + 1
+            xm, tok = rfft_trunc_stash(x, outermost,
+                                       self.restrict_prefixes[outermost], key)
+            for dim in reversed(self.dim_m[:-1]):
+                saved[dim] = xm.shape[dim]
+                xm = fft_trunc(xm, dim, self.restrict_prefixes[dim],
+                               self.restrict_suffixes.get(dim, 0))
+            x_epi = StashGradFn.apply(x_orig, tok, key)
+            x = xm
+        else:
+            x = self._fwd_m(x, saved)
 
         with comm_region() as r:
             x = self.R2(x)
@@ -330,7 +354,7 @@ class DistributedFNOBlock(nn.Module):
         with comm_region() as r:
             y = self.R4(y)
         self.dt_comm += r.host_dt
-        return y
+        return y, x_epi
 
     def _forward_pipelined(self, x: torch.Tensor, nch: int) -> torch.Tensor:
         """Channel-chunked software pipeline over the pencil chain.

@@ -156,6 +156,86 @@ class _PadIrfftFn(torch.autograd.Function):
 
 
 # ---------------------------------------------------------------------------
+# fused residual-gradient accumulate (docs/ROADMAP.md item 4 tail)
+# ---------------------------------------------------------------------------
+#
+# The block input feeds BOTH the spectral chain and the residual epilogue;
+# the autograd engine would sum the two input gradients with a full-tensor
+# aten add (~0.27 ms per block at the flagship).  Instead the epilogue-side
+# gradient is STASHED (its fanout branch contributes None, so the engine
+# never adds) and folded into the rfft adjoint's writeback, which is the
+# LAST producer of the chain-side gradient.  A zero-size token from the
+# rfft forward into the stash node makes the backward ordering a hard graph
+# dependency: the stash backward must run before the rfft adjoint.
+
+import itertools as _itertools
+
+_GRAD_STASH = {}
+_stash_keys = _itertools.count(1)
+
+
+def new_stash_key() -> int:
+    return next(_stash_keys)
+
+
+class StashGradFn(torch.autograd.Function):
+    """Alias ``x`` for a second consumer whose input-gradient is stashed
+    (fused into the rfft adjoint) instead of engine-added."""
+
+    @staticmethod
+    def forward(ctx, x, tok, key):
+        ctx.key = key
+        return x.view_as(x)
+
+    @staticmethod
+    def backward(ctx, g):
+        _GRAD_STASH[ctx.key] = g.contiguous()
+        # None for x: the engine must NOT add this branch (the stashed value
+        # is injected by _RfftTruncStashFn.backward); empty grad for tok
+        # keeps the ordering edge alive.
+        return None, torch.empty(0, device=g.device), None
+
+
+class _RfftTruncStashFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, dim, m, key):
+        ext = _ext.get(required=True)
+        ctx.dim, ctx.m, ctx.n, ctx.key = dim, m, x.shape[dim], key
+        y = ext.dft_rfft_trunc(x.contiguous(), dim, m)
+        tok = torch.empty(0, device=x.device)
+        return y, tok
+
+    @staticmethod
+    def backward(ctx, gy, gtok):
+        ext = _ext.get(required=True)
+        acc = _GRAD_STASH.pop(ctx.key, None)
+        if acc is None:
+            raise RuntimeError(
+                "rfft stash: epilogue gradient missing (stash backward did "
+                "not run before the rfft adjoint)")
+        gx = ext.dft_rfft_trunc_adj_acc(gy.contiguous(), ctx.dim, ctx.n, acc)
+        return gx, None, None, None
+
+
+def rfft_trunc_stash(x, dim, m, key):
+    """Native rfft_trunc variant returning (y, ordering-token) whose adjoint
+    adds the gradient stashed under ``key``.  Caller must route the block
+    input's second use through ``StashGradFn(x, tok, key)``."""
+    d = dim % x.dim()
+    m = min(m, x.shape[d] // 2 + 1)
+    return _RfftTruncStashFn.apply(x, d, m, key)
+
+
+def stash_fusable(x, dim, m) -> bool:
+    d = dim % x.dim()
+    m = min(m, x.shape[d] // 2 + 1)
+    return (d == x.dim() - 1 and x.is_cuda and x.numel() > 0
+            and torch.is_grad_enabled() and x.requires_grad
+            and x.shape[d] <= _MAX_N and m <= 32
+            and x.dtype in (torch.float32, torch.float64))
+
+
+# ---------------------------------------------------------------------------
 # public API
 # ---------------------------------------------------------------------------
 
