@@ -101,3 +101,53 @@ for name, runtype, mode in [
 ]:
     create_runscript(name, runs, list(args.local_shape), list(args.modes),
                      runtype, mode=mode)
+
+
+# ---------------------------------------------------------------------------
+# 256^3 8-GPU model-parallel capability point (BASELINE.json config #4):
+# fixed 256^3 global grid, width 20, modes (16,16,16,8), nt 32, partition
+# (1,1,2,2,2,1) — the 288 GB-HBM memory-capability demonstration.
+# ---------------------------------------------------------------------------
+
+def mem_budget_gib(gshape, pshape, width, modes, out_t, num_blocks):
+    """Rough fp32 peak-memory estimate per rank.  Forward saves per block
+    ~2 full-width activations (block input + pre-gelu z; the native DFT
+    autograd Functions save only dims), the projection head saves its input
+    plus the 128-channel mid activation, and backward adds ~60% transient
+    headroom; parameters + grads + Adam moments ride on top."""
+    import numpy as _np
+    vol_local = _np.prod([g // p for g, p in zip(gshape[2:5], pshape[2:5])]) * out_t
+    act = width * vol_local * 4               # one full-width activation, bytes
+    proj = 128 * vol_local * 4                # projection mid activation
+    weights = num_blocks * (2 ** 3) * width * width * _np.prod(
+        [m for m in modes[:-1]]) * modes[-1] * 8 / max(_np.prod(pshape[2:5]), 1)
+    total = (1.6 * (num_blocks * 2.2 * act + act + proj)  # fwd saved + bwd live
+             + 4 * weights)                               # w + grad + adam m,v
+    return total / 2**30
+
+
+def create_capability_script():
+    g = 256
+    width, modes, out_t, blocks = 20, (16, 16, 16, 8), 32, 4
+    pshape = (1, 1, 2, 2, 2, 1)
+    est = mem_budget_gib([1, 2, g, g, g, 1], pshape, width, modes, out_t, blocks)
+    assert est < 288 * 0.85, f"256^3 config estimated {est:.0f} GiB/rank > HBM budget"
+    fname = Path("capability_256cubed_gpu.sh")
+    with open(fname, "w") as f:
+        f.write(
+            "#!/bin/bash\n"
+            f"# 256^3 8-GPU model-parallel capability point (BASELINE config #4)\n"
+            f"# estimated peak memory: ~{est:.0f} GiB/rank of 288 GiB HBM3E\n"
+            "set -x\n"
+            "python -m torch.distributed.run --nnodes=1 --nproc-per-node 8 "
+            "--master-addr 127.0.0.1 --master-port 29519 "
+            f"../benchmarks/bench.py --input-shape 1 2 {g} {g} {g} 1 "
+            f"--modes {' '.join(map(str, modes))} "
+            f"--partition_shape {' '.join(map(str, pshape))} "
+            f"--width {width} --num-timesteps {out_t} --device cuda "
+            "--benchmark-type grad --output-dir capability_256cubed\n")
+    os.chmod(fname, 0o755)
+    print(f"created script: {fname.name} (est. {est:.0f} GiB/rank)")
+
+
+create_capability_script()
