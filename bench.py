@@ -67,6 +67,11 @@ def main():
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--width", type=int, default=WIDTH)
     p.add_argument("--num-blocks", type=int, default=NUM_BLOCKS)
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                   help="compute dtype; fp32 is the headline (the reference's "
+                        "dtype), bf16 is the reduced-precision config "
+                        "(BASELINE.json config #2: bf16 storage, fp32 math, "
+                        "complex64 spectral path)")
     p.add_argument("--heavy-comm", action="store_true",
                    help="partition the trailing spatial axis: R1/R4 become "
                         "real full-activation all-to-alls")
@@ -99,9 +104,10 @@ def main():
     P_world, P_x, P_0 = dfno.create_standard_partitions(pshape)
 
     torch.manual_seed(1234 + max(P_x.rank, 0))
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     model = dfno.DistributedFNONd(P_x, GLOBAL_SHAPE, OUT_T, args.width, MODES,
                                   num_blocks=args.num_blocks, device=device,
-                                  dtype=torch.float32)
+                                  dtype=dtype)
     criterion = dfno.DistributedRelativeLpLoss(P_x)
     from dfno_amd.optim import Adam as FusedAdam
     if torch.cuda.is_available():
@@ -114,8 +120,8 @@ def main():
     info_x = compute_distribution_info(P_x, GLOBAL_SHAPE)
     out_shape = [GLOBAL_SHAPE[0], 1, *GLOBAL_SHAPE[2:-1], OUT_T]
     info_y = compute_distribution_info(P_x, out_shape)
-    x = torch.rand(*info_x["shape"], device=device)
-    y_true = torch.rand(*info_y["shape"], device=device)
+    x = torch.rand(*info_x["shape"], device=device, dtype=dtype)
+    y_true = torch.rand(*info_y["shape"], device=device, dtype=dtype)
 
     def step():
         optimizer.zero_grad(set_to_none=True)
@@ -173,7 +179,7 @@ def main():
             "higher_is_better": False,
             "scaling": "strong",
             "vs_baseline": None,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic (random input/target shards, random-init weights)",
             "dt_comm_per_step": dt_comm_dev / args.steps,
             "dt_comm_launch_per_step": dt_comm_acc / args.steps,

@@ -31,6 +31,7 @@ SOURCES = [
     _CSRC / "dft.hip",
     _CSRC / "lift_head.hip",
     _CSRC / "pack.hip",
+    _CSRC / "bf16.hip",
 ]
 
 

@@ -521,6 +521,9 @@ def _issue_plan_packed(plan: "_RepartitionPlan", x: torch.Tensor,
     ext = _ext.get(required=False)
     if ext is None:
         return None
+    if dtype not in (torch.float32, torch.float64,
+                     torch.complex64, torch.complex128):
+        return None   # bf16 exchanges use the slicing path (2-byte words)
     if not x.is_contiguous():
         x = x.contiguous()
 

@@ -1,0 +1,451 @@
+// gfx950 bf16-storage pointwise kernels (the bf16 compute config,
+// BASELINE.json config #2 / VERDICT.md round-1 item 3).
+//
+// Storage is bf16 (halves every activation's HBM traffic vs the fp32
+// headline config — these ops are bandwidth-bound at 8 TB/s HBM3E);
+// arithmetic is fp32 (convert-on-load, round-to-nearest-even on store).
+// Vector IO is 8 bf16 per lane per instruction (uint4 = 16 B, the
+// coalescing sweet spot; cdna_hip_programming.md Guideline 13: hipcc does
+// NOT auto-vectorize bf16 loads).
+//
+// Kernels:
+//  * bf16_channel_mix (x-resident, I <= 32): y[b,o,s] = act(sum_i W[o,i]
+//    x[b,i,s] + bias[o] [+ res]) — covers the width-20 channel mixes, the
+//    20->128 projection lift and (transposed) grad-x.
+//  * bf16_channel_mix_ores (O <= 8, streams I): the 128->1 projection.
+//  * bf16_gelu fwd/bwd, bf16_add_gelu elementwise epilogues.
+//  * bf16_gw: split-s grad-W/grad-bias outer-product reduction with fp32
+//    accumulation (atomics on an fp32 staging buffer).
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include "kernels.h"
+#include "gelu_math.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kVec = 8;   // bf16 per lane per vector op (16 B)
+
+__device__ __forceinline__ float b2f(unsigned short h) {
+  unsigned int u = ((unsigned int)h) << 16;
+  return __uint_as_float(u);
+}
+
+__device__ __forceinline__ unsigned short f2b(float f) {
+  // round-to-nearest-even bf16
+  unsigned int u = __float_as_uint(f);
+  unsigned int lsb = (u >> 16) & 1u;
+  u += 0x7fffu + lsb;
+  if ((__float_as_uint(f) & 0x7f800000u) == 0x7f800000u) u = __float_as_uint(f);
+  return (unsigned short)(u >> 16);
+}
+
+__device__ __forceinline__ void load8(const unsigned short* p, float* out) {
+  const uint4 raw = *reinterpret_cast<const uint4*>(p);
+  const unsigned int w[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+  for (int k = 0; k < 4; ++k) {
+    out[2 * k] = b2f((unsigned short)(w[k] & 0xffffu));
+    out[2 * k + 1] = b2f((unsigned short)(w[k] >> 16));
+  }
+}
+
+__device__ __forceinline__ void store8(unsigned short* p, const float* v) {
+  uint4 raw;
+  unsigned int w[4];
+#pragma unroll
+  for (int k = 0; k < 4; ++k)
+    w[k] = (unsigned int)f2b(v[2 * k]) | ((unsigned int)f2b(v[2 * k + 1]) << 16);
+  raw.x = w[0]; raw.y = w[1]; raw.z = w[2]; raw.w = w[3];
+  *reinterpret_cast<uint4*>(p) = raw;
+}
+
+// ---------------------------------------------------------------------------
+// x-resident channel mix (I <= IMAX): one thread owns kVec consecutive s.
+// ---------------------------------------------------------------------------
+
+template <int IMAX, bool ACT>
+__global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
+    const unsigned short* __restrict__ x, const unsigned short* __restrict__ W,
+    const unsigned short* __restrict__ bias, unsigned short* __restrict__ y,
+    unsigned short* __restrict__ z, int B, int I, int O, long S, bool wt,
+    bool has_bias, bool write_z, const unsigned short* __restrict__ res) {
+  extern __shared__ __align__(16) char smem_raw[];
+  float* Wl = reinterpret_cast<float*>(smem_raw);   // [O*I] fp32
+  float* bl = Wl + (size_t)O * I;                   // [O]
+  for (int k = threadIdx.x; k < O * I; k += blockDim.x) Wl[k] = b2f(W[k]);
+  if (has_bias)
+    for (int k = threadIdx.x; k < O; k += blockDim.x) bl[k] = b2f(bias[k]);
+  __syncthreads();
+
+  long nchunks = S / kVec;   // host guarantees S % 8 == 0
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < (long)B * nchunks; t += stride) {
+    int b = (int)(t / nchunks);
+    long s = (t % nchunks) * kVec;
+
+    float xr[IMAX][kVec];
+    const unsigned short* xb = x + ((long)b * I) * S + s;
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) {
+      if (i < I) load8(xb + (long)i * S, xr[i]);
+    }
+
+    unsigned short* yb = y + ((long)b * O) * S + s;
+    unsigned short* zb = write_z ? z + ((long)b * O) * S + s : nullptr;
+#pragma unroll 4
+    for (int o = 0; o < 512; ++o) {
+      if (o >= O) break;
+      float acc[kVec];
+      const float bv = has_bias ? bl[o] : 0.f;
+#pragma unroll
+      for (int k = 0; k < kVec; ++k) acc[k] = bv;
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
+          const float wv = wt ? Wl[(size_t)i * O + o] : Wl[(size_t)o * I + i];
+#pragma unroll
+          for (int k = 0; k < kVec; ++k) acc[k] += wv * xr[i][k];
+        }
+      }
+      if (res != nullptr) {
+        float rv[kVec];
+        load8(res + ((long)b * O + o) * S + s, rv);
+#pragma unroll
+        for (int k = 0; k < kVec; ++k) acc[k] += rv[k];
+      }
+      if (write_z) store8(zb + (long)o * S, acc);
+      if (ACT) {
+#pragma unroll
+        for (int k = 0; k < kVec; ++k) acc[k] = dfno_gelu::gelu(acc[k]);
+      }
+      store8(yb + (long)o * S, acc);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// accumulator-resident channel mix (O <= OMAX, streams I) — the 128->1 head
+// ---------------------------------------------------------------------------
+
+template <int OMAX, bool ACT>
+__global__ __launch_bounds__(kBlock) void bf16_channel_mix_ores_kernel(
+    const unsigned short* __restrict__ x, const unsigned short* __restrict__ W,
+    const unsigned short* __restrict__ bias, unsigned short* __restrict__ y,
+    unsigned short* __restrict__ z, int B, int I, int O, long S, bool wt,
+    bool has_bias, bool write_z, const unsigned short* __restrict__ res) {
+  extern __shared__ __align__(16) char smem_raw[];
+  float* Wl = reinterpret_cast<float*>(smem_raw);   // [O*I]
+  float* bl = Wl + (size_t)O * I;
+  for (int k = threadIdx.x; k < O * I; k += blockDim.x) Wl[k] = b2f(W[k]);
+  if (has_bias)
+    for (int k = threadIdx.x; k < O; k += blockDim.x) bl[k] = b2f(bias[k]);
+  __syncthreads();
+
+  long nchunks = S / kVec;
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < (long)B * nchunks; t += stride) {
+    int b = (int)(t / nchunks);
+    long s = (t % nchunks) * kVec;
+
+    float acc[OMAX][kVec];
+#pragma unroll
+    for (int o = 0; o < OMAX; ++o) {
+      const float bv = (o < O && has_bias) ? bl[o] : 0.f;
+#pragma unroll
+      for (int k = 0; k < kVec; ++k) acc[o][k] = bv;
+    }
+
+    const unsigned short* xb = x + ((long)b * I) * S + s;
+    for (int i = 0; i < I; ++i) {
+      float xv[kVec];
+      load8(xb + (long)i * S, xv);
+#pragma unroll
+      for (int o = 0; o < OMAX; ++o) {
+        if (o < O) {
+          const float wv = wt ? Wl[(size_t)i * O + o] : Wl[(size_t)o * I + i];
+#pragma unroll
+          for (int k = 0; k < kVec; ++k) acc[o][k] += wv * xv[k];
+        }
+      }
+    }
+
+    unsigned short* yb = y + ((long)b * O) * S + s;
+    unsigned short* zb = write_z ? z + ((long)b * O) * S + s : nullptr;
+#pragma unroll
+    for (int o = 0; o < OMAX; ++o) {
+      if (o < O) {
+        if (res != nullptr) {
+          float rv[kVec];
+          load8(res + ((long)b * O + o) * S + s, rv);
+#pragma unroll
+          for (int k = 0; k < kVec; ++k) acc[o][k] += rv[k];
+        }
+        if (write_z) store8(zb + (long)o * S, acc[o]);
+        if (ACT) {
+#pragma unroll
+          for (int k = 0; k < kVec; ++k) acc[o][k] = dfno_gelu::gelu(acc[o][k]);
+        }
+        store8(yb + (long)o * S, acc[o]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// elementwise gelu / add-gelu (8-wide)
+// ---------------------------------------------------------------------------
+
+__global__ void bf16_gelu_fwd_kernel(const unsigned short* __restrict__ x,
+                                     unsigned short* __restrict__ y, long n8) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n8; i += stride) {
+    float v[kVec];
+    load8(x + i * kVec, v);
+#pragma unroll
+    for (int k = 0; k < kVec; ++k) v[k] = dfno_gelu::gelu(v[k]);
+    store8(y + i * kVec, v);
+  }
+}
+
+__global__ void bf16_gelu_bwd_kernel(const unsigned short* __restrict__ gy,
+                                     const unsigned short* __restrict__ z,
+                                     unsigned short* __restrict__ gz, long n8) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n8; i += stride) {
+    float g[kVec], zv[kVec];
+    load8(gy + i * kVec, g);
+    load8(z + i * kVec, zv);
+#pragma unroll
+    for (int k = 0; k < kVec; ++k) g[k] *= dfno_gelu::gelu_grad(zv[k]);
+    store8(gz + i * kVec, g);
+  }
+}
+
+__global__ void bf16_add_gelu_kernel(const unsigned short* __restrict__ a,
+                                     const unsigned short* __restrict__ b,
+                                     unsigned short* __restrict__ y,
+                                     unsigned short* __restrict__ z, long n8) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < n8; i += stride) {
+    float av[kVec], bv[kVec];
+    load8(a + i * kVec, av);
+    load8(b + i * kVec, bv);
+#pragma unroll
+    for (int k = 0; k < kVec; ++k) av[k] += bv[k];
+    store8(z + i * kVec, av);
+#pragma unroll
+    for (int k = 0; k < kVec; ++k) av[k] = dfno_gelu::gelu(av[k]);
+    store8(y + i * kVec, av);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// grad-W (+grad-bias): gW[o,i] = sum_{b,s} gz[b,o,s] x[b,i,s], fp32 staging.
+// Each block reduces its s-chunk into an LDS [O][I] partial, then one
+// atomicAdd per (o, i) — O*I <= 4096 words.
+// ---------------------------------------------------------------------------
+
+template <int IMAX>
+__global__ __launch_bounds__(kBlock) void bf16_gw_kernel(
+    const unsigned short* __restrict__ gz, const unsigned short* __restrict__ x,
+    float* __restrict__ gW, float* __restrict__ gb,
+    int B, int I, int O, long S, bool want_bias) {
+  extern __shared__ __align__(16) char smem_raw[];
+  float* part = reinterpret_cast<float*>(smem_raw);  // [O*I] (+ [O] bias)
+  float* pb = part + (size_t)O * I;
+  for (int k = threadIdx.x; k < O * I; k += blockDim.x) part[k] = 0.f;
+  if (want_bias)
+    for (int k = threadIdx.x; k < O; k += blockDim.x) pb[k] = 0.f;
+  __syncthreads();
+
+  long nchunks = S / kVec;
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+
+  for (long t = t0; t < (long)B * nchunks; t += stride) {
+    int b = (int)(t / nchunks);
+    long s = (t % nchunks) * kVec;
+    float xr[IMAX][kVec];
+    const unsigned short* xb = x + ((long)b * I) * S + s;
+#pragma unroll
+    for (int i = 0; i < IMAX; ++i) {
+      if (i < I) load8(xb + (long)i * S, xr[i]);
+    }
+    const unsigned short* gzb = gz + ((long)b * O) * S + s;
+    for (int o = 0; o < O; ++o) {
+      float gv[kVec];
+      load8(gzb + (long)o * S, gv);
+      if (want_bias) {
+        float sum = 0.f;
+#pragma unroll
+        for (int k = 0; k < kVec; ++k) sum += gv[k];
+        atomicAdd(&pb[o], sum);   // LDS atomic (cross-wave in this block)
+      }
+#pragma unroll
+      for (int i = 0; i < IMAX; ++i) {
+        if (i < I) {
+          float sum = 0.f;
+#pragma unroll
+          for (int k = 0; k < kVec; ++k) sum += gv[k] * xr[i][k];
+          atomicAdd(&part[(size_t)o * I + i], sum);
+        }
+      }
+    }
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < O * I; k += blockDim.x)
+    if (part[k] != 0.f) atomicAdd(&gW[k], part[k]);
+  if (want_bias)
+    for (int k = threadIdx.x; k < O; k += blockDim.x)
+      if (pb[k] != 0.f) atomicAdd(&gb[k], pb[k]);
+}
+
+int bgrid(long work) {
+  long g = (work + kBlock - 1) / kBlock;
+  if (g > 256L * 16) g = 256L * 16;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+              t.scalar_type() == at::kBFloat16, name,
+              " must be a contiguous CUDA bf16 tensor");
+}
+
+const unsigned short* usp(const at::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+unsigned short* usp_mut(at::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+}  // namespace
+
+std::vector<at::Tensor> bf16_channel_mix(const at::Tensor& x, const at::Tensor& W,
+                                         const at::Tensor& b, bool act, bool wt,
+                                         bool write_z, const at::Tensor& res) {
+  check_bf16(x, "x"); check_bf16(W, "W");
+  const int B = (int)x.size(0);
+  const int I = (int)x.size(1);
+  const long S = x.size(2);
+  const int O = wt ? (int)W.size(1) : (int)W.size(0);
+  TORCH_CHECK(S % kVec == 0, "bf16 channel mix: S must be a multiple of 8");
+  const bool has_bias = b.defined() && b.numel() > 0;
+  if (has_bias) check_bf16(b, "b");
+  const bool has_res = res.defined() && res.numel() > 0;
+  if (has_res) check_bf16(res, "res");
+
+  auto y = at::empty({B, O, S}, x.options());
+  auto z = write_z ? at::empty({B, O, S}, x.options()) : at::empty({0}, x.options());
+  if (x.numel() == 0) return {y, z};
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  size_t smem = sizeof(float) * ((size_t)O * I + O);
+  int grid = bgrid((long)B * (S / kVec));
+  const unsigned short* bp = has_bias ? usp(b) : nullptr;
+  const unsigned short* rp = has_res ? usp(res) : nullptr;
+  unsigned short* zp = write_z ? usp_mut(z) : nullptr;
+
+#define BCM(KER, CAP)                                                          \
+  if (act) {                                                                   \
+    hipLaunchKernelGGL((KER<CAP, true>), dim3(grid), dim3(kBlock), smem,       \
+                       stream, usp(x), usp(W), bp, usp_mut(y), zp, B, I, O, S, \
+                       wt, has_bias, write_z, rp);                             \
+  } else {                                                                     \
+    hipLaunchKernelGGL((KER<CAP, false>), dim3(grid), dim3(kBlock), smem,      \
+                       stream, usp(x), usp(W), bp, usp_mut(y), zp, B, I, O, S, \
+                       wt, has_bias, write_z, rp);                             \
+  }
+  if (I <= 8) { BCM(bf16_channel_mix_xres_kernel, 8) }
+  else if (I <= 24) { BCM(bf16_channel_mix_xres_kernel, 24) }
+  else if (I <= 32) { BCM(bf16_channel_mix_xres_kernel, 32) }
+  else if (O <= 8) { BCM(bf16_channel_mix_ores_kernel, 8) }
+  else if (O <= 24) { BCM(bf16_channel_mix_ores_kernel, 24) }
+  else { TORCH_CHECK(false, "bf16 channel mix: unsupported shape I=", I, " O=", O); }
+#undef BCM
+  DFNO_CHECK_LAUNCH("bf16_channel_mix");
+  return {y, z};
+}
+
+std::vector<at::Tensor> bf16_channel_mix_bwd_w(const at::Tensor& gz,
+                                               const at::Tensor& x,
+                                               bool want_bias) {
+  check_bf16(gz, "gz"); check_bf16(x, "x");
+  const int B = (int)x.size(0);
+  const int I = (int)x.size(1);
+  const long S = x.size(2);
+  const int O = (int)gz.size(1);
+  TORCH_CHECK(I <= 32, "bf16 grad-W: I must be <= 32");
+  TORCH_CHECK(S % kVec == 0, "bf16 grad-W: S must be a multiple of 8");
+
+  auto opts = x.options().dtype(at::kFloat);
+  auto gW = at::zeros({O, I}, opts);
+  auto gb = at::zeros({want_bias ? O : 0}, opts);
+  if (x.numel() == 0) return {gW, gb};
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  size_t smem = sizeof(float) * ((size_t)O * I + O);
+  int grid = bgrid((long)B * (S / kVec));
+#define BGW(CAP)                                                               \
+  hipLaunchKernelGGL((bf16_gw_kernel<CAP>), dim3(grid), dim3(kBlock), smem,    \
+                     stream, usp(gz), usp(x), gW.data_ptr<float>(),            \
+                     gb.numel() ? gb.data_ptr<float>() : nullptr, B, I, O, S,  \
+                     want_bias);
+  if (I <= 8) { BGW(8) } else if (I <= 24) { BGW(24) } else { BGW(32) }
+#undef BGW
+  DFNO_CHECK_LAUNCH("bf16_gw");
+  return {gW, gb};
+}
+
+at::Tensor bf16_gelu_fwd(const at::Tensor& x) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.numel() % kVec == 0, "bf16 gelu: numel must be a multiple of 8");
+  auto y = at::empty_like(x);
+  if (x.numel() == 0) return y;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  long n8 = x.numel() / kVec;
+  hipLaunchKernelGGL(bf16_gelu_fwd_kernel, dim3(bgrid(n8)), dim3(kBlock), 0,
+                     stream, usp(x), usp_mut(y), n8);
+  DFNO_CHECK_LAUNCH("bf16_gelu");
+  return y;
+}
+
+at::Tensor bf16_gelu_bwd(const at::Tensor& gy, const at::Tensor& z) {
+  check_bf16(gy, "gy"); check_bf16(z, "z");
+  TORCH_CHECK(gy.numel() % kVec == 0, "bf16 gelu bwd: numel % 8");
+  auto gz = at::empty_like(gy);
+  if (gy.numel() == 0) return gz;
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  long n8 = gy.numel() / kVec;
+  hipLaunchKernelGGL(bf16_gelu_bwd_kernel, dim3(bgrid(n8)), dim3(kBlock), 0,
+                     stream, usp(gy), usp(z), usp_mut(gz), n8);
+  DFNO_CHECK_LAUNCH("bf16_gelu_bwd");
+  return gz;
+}
+
+std::vector<at::Tensor> bf16_add_gelu(const at::Tensor& a, const at::Tensor& b) {
+  check_bf16(a, "a"); check_bf16(b, "b");
+  TORCH_CHECK(a.numel() == b.numel() && a.numel() % kVec == 0, "bf16 add_gelu shape");
+  auto y = at::empty_like(a);
+  auto z = at::empty_like(a);
+  if (a.numel() == 0) return {y, z};
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  long n8 = a.numel() / kVec;
+  hipLaunchKernelGGL(bf16_add_gelu_kernel, dim3(bgrid(n8)), dim3(kBlock), 0,
+                     stream, usp(a), usp(b), usp_mut(y), usp_mut(z), n8);
+  DFNO_CHECK_LAUNCH("bf16_add_gelu");
+  return {y, z};
+}

@@ -41,6 +41,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "all corner boxes adjoint wrt x in one launch");
   m.def("spectral_corners_bwd_w", &spectral_corners_bwd_w,
         "multi-corner spectral grad-W (gw[i,o,e] = sum_b x conj(gy))");
+  m.def("bf16_channel_mix", &bf16_channel_mix,
+        "bf16-storage fused channel linear (+bias/res/gelu): returns (y, z)");
+  m.def("bf16_channel_mix_bwd_w", &bf16_channel_mix_bwd_w,
+        "bf16 grad-W/grad-bias reduction (fp32 accumulation)");
+  m.def("bf16_gelu_fwd", &bf16_gelu_fwd, "bf16 exact gelu");
+  m.def("bf16_gelu_bwd", &bf16_gelu_bwd, "bf16 gelu backward");
+  m.def("bf16_add_gelu", &bf16_add_gelu, "bf16 fused add+gelu: returns (y, z)");
   m.def("pack_boxes", &pack_boxes,
         "gather repartition boxes into one flat staging buffer");
   m.def("unpack_boxes", &unpack_boxes,

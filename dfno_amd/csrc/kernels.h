@@ -73,6 +73,17 @@ at::Tensor dft_rfft_trunc_adj_acc(const at::Tensor& gy, int64_t dim, int64_t n,
 at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m);
 at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m);
 
+// bf16-storage pointwise kernels (bf16.hip; fp32 arithmetic):
+std::vector<at::Tensor> bf16_channel_mix(const at::Tensor& x, const at::Tensor& W,
+                                         const at::Tensor& b, bool act, bool wt,
+                                         bool write_z, const at::Tensor& res);
+std::vector<at::Tensor> bf16_channel_mix_bwd_w(const at::Tensor& gz,
+                                               const at::Tensor& x,
+                                               bool want_bias);
+at::Tensor bf16_gelu_fwd(const at::Tensor& x);
+at::Tensor bf16_gelu_bwd(const at::Tensor& gy, const at::Tensor& z);
+std::vector<at::Tensor> bf16_add_gelu(const at::Tensor& a, const at::Tensor& b);
+
 // repartition pack/unpack (pack.hip): gather/scatter block-intersection
 // boxes between a contiguous tensor (word view) and one flat staging buffer
 void pack_boxes(const at::Tensor& src, at::Tensor& flat,

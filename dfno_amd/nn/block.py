@@ -52,7 +52,10 @@ class DistributedFNOBlock(nn.Module):
         self.n = P_x.dim - 2
         self.device = device
         self.dtype = dtype
-        self.dtype_complex = torch.complex64 if dtype == torch.float32 else torch.complex128
+        # bf16 models keep the spectral path in fp32/complex64 (the reference
+        # semantic; transforms upcast at entry, downcast after the inverse)
+        self.dtype_complex = (torch.complex128 if dtype == torch.float64
+                              else torch.complex64)
 
         # ---- FFT pencil partitions (reference dfno.py:82-97) -------------
         shape_m = P_x.shape.copy()
@@ -202,6 +205,8 @@ class DistributedFNOBlock(nn.Module):
             # rank, including ones whose local block is empty)
             return torch.empty(0, dtype=self.dtype_complex, device=x.device,
                                requires_grad=torch.is_grad_enabled())
+        if x.dtype == torch.bfloat16:
+            x = x.float()   # spectral path runs fp32/c64 (see dtype_complex)
         outermost = self.dim_m[-1]
         saved[outermost] = x.shape[outermost] // 2 + 1
         x = rfft_trunc(x, outermost, self.restrict_prefixes[outermost])
@@ -239,8 +244,10 @@ class DistributedFNOBlock(nn.Module):
             y = pad_ifft(y, dim, saved[dim],
                          self.restrict_prefixes[dim],
                          self.restrict_suffixes.get(dim, 0))
-        return pad_irfft(y, outermost, saved[outermost],
-                         self.in_shape[-1], self.restrict_prefixes[outermost])
+        y = pad_irfft(y, outermost, saved[outermost],
+                      self.in_shape[-1], self.restrict_prefixes[outermost])
+        # downcast BEFORE R4 so the real-activation exchange moves bf16
+        return y.to(self.dtype) if y.dtype != self.dtype else y
 
     # ---- pipelined-chunk policy ------------------------------------------
     def _pipeline_chunks(self) -> int:

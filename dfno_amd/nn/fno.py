@@ -103,7 +103,9 @@ class DistributedFNONd(nn.Module):
             self.dt_comm += r.host_dt
             return lift_head(x, W1, b1, W2, b2)
 
-        if x.is_cuda:
+        if x.is_cuda and x.dtype != torch.bfloat16:
+            # (bf16 deliberately composes the native bf16 channel-mix
+            # kernels; only non-bf16 misses are real fallbacks)
             note_fallback("lift_head", "unsupported lift shape/dtype for the "
                           "fused kernel (composed linears instead)")
         x = l1(x, activation="gelu")
@@ -132,7 +134,7 @@ class DistributedFNONd(nn.Module):
             self.dt_comm += r.host_dt
             return proj_head(x, W3, b3, W4, b4)
 
-        if x.is_cuda:
+        if x.is_cuda and x.dtype != torch.bfloat16:
             note_fallback("proj_head", f"shape out of fused-kernel range "
                           f"(in={l3.in_features}, mid={l3.out_features}, "
                           f"out={l4.out_features}) or dtype {x.dtype}")

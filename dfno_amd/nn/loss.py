@@ -32,6 +32,11 @@ class DistributedRelativeLpLoss(nn.Module):
 
     def forward(self, y_hat: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
         batch_size = y_hat.shape[0] if y_hat.dim() > 0 and y_hat.numel() > 0 else 0
+        if y_hat.dtype == torch.bfloat16:
+            # bf16 sums over millions of elements lose the loss signal;
+            # accumulate the norms in fp32 (grad flows back through the cast)
+            y_hat = y_hat.float()
+            y = y.float()
         if batch_size > 0:
             y_hat_flat = y_hat.reshape(batch_size, -1)
             y_flat = y.reshape(batch_size, -1)

@@ -30,12 +30,34 @@ from ..dispatch import note_fallback
 
 
 def _native_dt(t: torch.Tensor, op: str = "channel_mix") -> bool:
-    """Native kernels are fp32/fp64 (bf16 goes through the dedicated bf16
-    MFMA path where wired); anything else takes the composed-torch path
-    (still on GPU via rocBLAS/eager) with a warn-once."""
+    """fp32/fp64 native-kernel gate; bf16 is handled by the dedicated bf16
+    branch at each call site BEFORE this check.  Anything else takes the
+    composed-torch path (still on GPU via rocBLAS/eager) with a warn-once."""
     ok = t.dtype in (torch.float32, torch.float64)
-    if not ok and t.is_cuda:
+    if not ok and t.is_cuda and t.dtype != torch.bfloat16:
         note_fallback(op, f"dtype {t.dtype} has no native kernel")
+    return ok
+
+
+_EMPTY_BF16 = {}
+
+
+def _ebf(device):
+    t = _EMPTY_BF16.get(device)
+    if t is None:
+        t = torch.empty(0, dtype=torch.bfloat16, device=device)
+        _EMPTY_BF16[device] = t
+    return t
+
+
+def _bf16_mix_ok(x3: torch.Tensor, I: int, O: int, op: str) -> bool:
+    """bf16-storage channel-mix kernel coverage (csrc/bf16.hip)."""
+    if not (x3.is_cuda and x3.dtype == torch.bfloat16):
+        return False
+    S = x3.shape[2] if x3.dim() == 3 else 0
+    ok = S % 8 == 0 and (I <= 32 or O <= 24)
+    if not ok:
+        note_fallback(op, f"bf16 shape I={I} O={O} S={S} outside kernel range")
     return ok
 
 __all__ = ["linear_nd", "add_gelu", "gelu", "linear_res_gelu"]
@@ -64,6 +86,12 @@ class _ChannelMixFn(torch.autograd.Function):
         if x.is_cuda and _native_dt(x):
             ext = _ext.get(required=True)
             y3, z3 = ext.channel_mix_fwd(x3, W, b if b is not None else torch.empty(0, dtype=x.dtype, device=x.device), act)
+        elif _bf16_mix_ok(x3, I, W.shape[0], "channel_mix"):
+            ext = _ext.get(required=True)
+            y3, z3 = ext.bf16_channel_mix(
+                x3.contiguous(), W.contiguous(),
+                b.contiguous() if b is not None else _ebf(x.device),
+                act, False, act, _ebf(x.device))
         else:
             z3 = torch.einsum("oi,bis->bos", W, x3)
             if b is not None:
@@ -80,10 +108,16 @@ class _ChannelMixFn(torch.autograd.Function):
         x3, W, z3 = ctx.saved_tensors
         act = ctx.act
         gy = gy.contiguous()
+        is_bf16 = gy.is_cuda and gy.dtype == torch.bfloat16
+        bf16_ok = is_bf16 and _bf16_mix_ok(gy.reshape(x3.shape[0], W.shape[0], -1),
+                                           W.shape[0], x3.shape[1], "channel_mix_bwd")
         if act:
             if gy.is_cuda and _native_dt(gy):
                 ext = _ext.get(required=True)
                 gz = ext.gelu_bwd(gy, z3)
+            elif is_bf16 and gy.numel() % 8 == 0:
+                ext = _ext.get(required=True)
+                gz = ext.bf16_gelu_bwd(gy, z3.contiguous())
             else:
                 gz = gy * _gelu_grad(z3)
         else:
@@ -92,6 +126,11 @@ class _ChannelMixFn(torch.autograd.Function):
         if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             gx = ext.channel_mix_fwd_t(gz, W)  # sum_o W[o,i] gz[b,o,s]
+        elif bf16_ok:
+            ext = _ext.get(required=True)
+            gx, _ = ext.bf16_channel_mix(gz.contiguous(), W.contiguous(),
+                                         _ebf(gy.device), False, True, False,
+                                         _ebf(gy.device))
         else:
             gx = torch.einsum("oi,bos->bis", W, gz)
         # grad W / b
@@ -100,6 +139,12 @@ class _ChannelMixFn(torch.autograd.Function):
             gW, gb = ext.channel_mix_bwd_w(gz.contiguous(), x3, ctx.has_bias)
             if not ctx.has_bias:
                 gb = None
+        elif is_bf16 and x3.shape[1] <= 32 and x3.shape[2] % 8 == 0:
+            ext = _ext.get(required=True)
+            gW, gb = ext.bf16_channel_mix_bwd_w(gz.contiguous(), x3.contiguous(),
+                                                ctx.has_bias)
+            gW = gW.to(W.dtype)
+            gb = gb.to(W.dtype) if ctx.has_bias else None
         else:
             gW = torch.einsum("bos,bis->oi", gz, x3)
             gb = gz.sum(dim=(0, 2)) if ctx.has_bias else None
@@ -165,6 +210,11 @@ class _LinearResGeluFn(torch.autograd.Function):
         if x.is_cuda and _native_dt(x):
             ext = _ext.get(required=True)
             y3, z3 = ext.linear_res_gelu_fwd(x3.contiguous(), W, r3.contiguous())
+        elif _bf16_mix_ok(x3, x3.shape[1], W.shape[0], "linear_res_gelu"):
+            ext = _ext.get(required=True)
+            y3, z3 = ext.bf16_channel_mix(x3.contiguous(), W.contiguous(),
+                                          _ebf(x.device), True, False, True,
+                                          r3.contiguous())
         else:
             z3 = torch.einsum("oi,bis->bos", W, x3) + r3
             y3 = F.gelu(z3)
@@ -183,6 +233,16 @@ class _LinearResGeluFn(torch.autograd.Function):
             gx = ext.channel_mix_fwd_t(gz, W)
             if x3.shape[1] <= 32:
                 gW, _ = ext.channel_mix_bwd_w(gz, x3.contiguous(), False)
+            else:
+                gW = torch.einsum("bos,bis->oi", gz, x3)
+        elif _bf16_mix_ok(x3, W.shape[0], x3.shape[1], "linear_res_gelu_bwd"):
+            ext = _ext.get(required=True)
+            gz = ext.bf16_gelu_bwd(gy, z3.contiguous())
+            gx, _ = ext.bf16_channel_mix(gz, W.contiguous(), _ebf(gy.device),
+                                         False, True, False, _ebf(gy.device))
+            if x3.shape[1] <= 32:
+                gW, _ = ext.bf16_channel_mix_bwd_w(gz, x3.contiguous(), False)
+                gW = gW.to(W.dtype)
             else:
                 gW = torch.einsum("bos,bis->oi", gz, x3)
         else:
@@ -210,6 +270,9 @@ class _AddGeluFn(torch.autograd.Function):
         if a.is_cuda and _native_dt(a):
             ext = _ext.get(required=True)
             y, z = ext.add_gelu_fwd(a.contiguous(), bt.contiguous())
+        elif a.is_cuda and a.dtype == torch.bfloat16 and a.numel() % 8 == 0:
+            ext = _ext.get(required=True)
+            y, z = ext.bf16_add_gelu(a.contiguous(), bt.contiguous())
         else:
             z = a + bt
             y = F.gelu(z)
@@ -223,6 +286,9 @@ class _AddGeluFn(torch.autograd.Function):
         if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             gz = ext.gelu_bwd(gy, z)
+        elif gy.is_cuda and gy.dtype == torch.bfloat16 and gy.numel() % 8 == 0:
+            ext = _ext.get(required=True)
+            gz = ext.bf16_gelu_bwd(gy, z.contiguous())
         else:
             gz = gy * _gelu_grad(z)
         return gz, gz
@@ -239,6 +305,9 @@ class _GeluFn(torch.autograd.Function):
         if x.is_cuda and _native_dt(x):
             ext = _ext.get(required=True)
             y = ext.gelu_fwd(x.contiguous())
+        elif x.is_cuda and x.dtype == torch.bfloat16 and x.numel() % 8 == 0:
+            ext = _ext.get(required=True)
+            y = ext.bf16_gelu_fwd(x.contiguous())
         else:
             y = F.gelu(x)
         ctx.save_for_backward(x)
@@ -251,6 +320,9 @@ class _GeluFn(torch.autograd.Function):
         if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             return ext.gelu_bwd(gy, x)
+        if gy.is_cuda and gy.dtype == torch.bfloat16 and gy.numel() % 8 == 0:
+            ext = _ext.get(required=True)
+            return ext.bf16_gelu_bwd(gy, x.contiguous())
         return gy * _gelu_grad(x)
 
 

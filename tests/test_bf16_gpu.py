@@ -1,0 +1,164 @@
+"""bf16 compute-config tests (BASELINE.json config #2).
+
+Native bf16-storage kernels (csrc/bf16.hip: fp32 arithmetic, bf16 IO) are
+checked against plain fp32 torch references at bf16 tolerances, and the full
+bf16 model fwd+bwd is compared against the fp32 model with identical weights.
+"""
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+import dfno_amd as dfno
+
+pytestmark = pytest.mark.gpu
+
+BT = 2 ** -7   # bf16 mantissa step: tolerance scale
+
+
+def _ext():
+    from dfno_amd import _ext as e
+    return e.get(required=True)
+
+
+@pytest.mark.parametrize("I,O,act,bias", [
+    (20, 20, True, True),    # block channel mix
+    (20, 128, True, True),   # projection lift
+    (128, 1, False, True),   # projection head (ores path)
+    (2, 20, True, True),     # channel lift
+])
+def test_bf16_channel_mix_fwd(I, O, act, bias):
+    ext = _ext()
+    torch.manual_seed(0)
+    B, S = 2, 1024
+    x = (torch.rand(B, I, S, device="cuda") - 0.5)
+    W = (torch.rand(O, I, device="cuda") - 0.5) / I
+    b = (torch.rand(O, device="cuda") - 0.5) if bias else None
+
+    xb, Wb = x.bfloat16(), W.bfloat16()
+    bb = b.bfloat16() if bias else torch.empty(0, dtype=torch.bfloat16, device="cuda")
+    empty = torch.empty(0, dtype=torch.bfloat16, device="cuda")
+    y, z = ext.bf16_channel_mix(xb, Wb, bb, act, False, act, empty)
+
+    ref = torch.einsum("oi,bis->bos", Wb.float(), xb.float())
+    if bias:
+        ref = ref + bb.float().view(1, -1, 1)
+    zr = ref
+    if act:
+        ref = F.gelu(ref)
+    assert torch.allclose(y.float(), ref, rtol=BT, atol=BT), \
+        f"max {(y.float()-ref).abs().max()}"
+    if act:
+        assert torch.allclose(z.float(), zr, rtol=BT, atol=BT)
+
+
+def test_bf16_channel_mix_transposed():
+    ext = _ext()
+    torch.manual_seed(1)
+    B, I, O, S = 1, 20, 128, 2048
+    gz = (torch.rand(B, O, S, device="cuda") - 0.5).bfloat16()
+    W = ((torch.rand(O, I, device="cuda") - 0.5) / I).bfloat16()
+    empty = torch.empty(0, dtype=torch.bfloat16, device="cuda")
+    gx, _ = ext.bf16_channel_mix(gz, W, empty, False, True, False, empty)
+    ref = torch.einsum("oi,bos->bis", W.float(), gz.float())
+    assert torch.allclose(gx.float(), ref, rtol=BT, atol=BT)
+
+
+def test_bf16_grad_w():
+    ext = _ext()
+    torch.manual_seed(2)
+    B, I, O, S = 2, 20, 24, 4096
+    gz = (torch.rand(B, O, S, device="cuda") - 0.5).bfloat16()
+    x = (torch.rand(B, I, S, device="cuda") - 0.5).bfloat16()
+    gW, gb = ext.bf16_channel_mix_bwd_w(gz, x, True)
+    refW = torch.einsum("bos,bis->oi", gz.float(), x.float())
+    refb = gz.float().sum(dim=(0, 2))
+    assert torch.allclose(gW, refW, rtol=2e-3, atol=refW.abs().max() * 2e-3), \
+        f"max {(gW-refW).abs().max()}"
+    assert torch.allclose(gb, refb, rtol=2e-3, atol=refb.abs().max() * 2e-3)
+
+
+def test_bf16_gelu_and_add_gelu():
+    ext = _ext()
+    torch.manual_seed(3)
+    a = (torch.randn(4096 * 3, device="cuda")).bfloat16()
+    b = (torch.randn(4096 * 3, device="cuda")).bfloat16()
+    y = ext.bf16_gelu_fwd(a)
+    assert torch.allclose(y.float(), F.gelu(a.float()), rtol=BT, atol=BT)
+    y2, z2 = ext.bf16_add_gelu(a, b)
+    zf = a.float() + b.float()
+    assert torch.allclose(z2.float(), zf, rtol=BT, atol=BT)
+    assert torch.allclose(y2.float(), F.gelu(torch.tensor(z2.float())), rtol=BT, atol=BT)
+    gy = torch.randn_like(a).bfloat16()
+    gz = ext.bf16_gelu_bwd(gy, a)
+    zfl = a.float()
+    import math
+    ref = gy.float() * (0.5 * (1 + torch.erf(zfl / math.sqrt(2.0)))
+                        + zfl * torch.exp(-0.5 * zfl * zfl) / math.sqrt(2 * math.pi))
+    assert torch.allclose(gz.float(), ref, rtol=BT, atol=BT)
+
+
+def test_bf16_model_matches_fp32():
+    """Full bf16 model fwd+bwd vs the fp32 model with the SAME weights
+    (the 2D+time NS shape of BASELINE config #2, small extents)."""
+    torch.manual_seed(4)
+    _, P_x, _ = dfno.create_standard_partitions((1, 1, 1, 1, 1))
+    in_shape = [1, 1, 32, 32, 8]
+    f32 = dfno.DistributedFNONd(P_x, in_shape, 16, 16, (4, 4, 4),
+                                num_blocks=2, device=torch.device("cuda"),
+                                dtype=torch.float32)
+    b16 = dfno.DistributedFNONd(P_x, in_shape, 16, 16, (4, 4, 4),
+                                num_blocks=2, device=torch.device("cuda"),
+                                dtype=torch.bfloat16)
+    sd = {}
+    for k, v in f32.state_dict().items():
+        sd[k] = v.bfloat16() if v.dtype == torch.float32 else v
+    b16.load_state_dict(sd)
+
+    x = torch.rand(*in_shape, device="cuda")
+    y32 = f32(x)
+    y16 = b16(x.bfloat16())
+    assert y16.dtype == torch.bfloat16
+    scale = y32.abs().max().clamp_min(1.0)
+    err = (y16.float() - y32).abs().max() / scale
+    assert err < 0.06, f"bf16 fwd rel err {err}"
+
+    # backward: loss gradients stay correlated with fp32
+    t = torch.rand_like(y32)
+    crit32 = dfno.DistributedRelativeLpLoss(P_x)
+    crit16 = dfno.DistributedRelativeLpLoss(P_x)
+    l32 = crit32(y32, t)
+    l16 = crit16(y16, t.bfloat16())
+    assert abs(l16.item() - l32.item()) / abs(l32.item()) < 0.05
+    l32.backward()
+    l16.backward()
+    for (n, p32), (_, p16) in zip(f32.named_parameters(), b16.named_parameters()):
+        if p32.grad is None or p32.grad.numel() == 0:
+            continue
+        g32 = p32.grad.flatten()
+        g16 = p16.grad.float().flatten()
+        if g32.norm() < 1e-12:
+            continue
+        if g32.is_complex():
+            g32 = torch.view_as_real(g32).flatten()
+            g16 = torch.view_as_real(g16).flatten()
+        cos = torch.dot(g16, g32) / (g16.norm() * g32.norm()).clamp_min(1e-30)
+        assert cos > 0.98, f"{n}: grad cosine {cos}"
+
+
+def test_bf16_model_all_native():
+    """The bf16 flagship shape must not fall back to eager torch anywhere
+    (other than the documented composed-head routing)."""
+    from dfno_amd import dispatch
+    torch.manual_seed(5)
+    _, P_x, _ = dfno.create_standard_partitions((1, 1, 1, 1, 1, 1))
+    model = dfno.DistributedFNONd(P_x, [1, 2, 16, 16, 16, 1], 8, 20,
+                                  (4, 4, 4, 3), num_blocks=1,
+                                  device=torch.device("cuda"),
+                                  dtype=torch.bfloat16)
+    x = torch.rand(1, 2, 16, 16, 16, 1, device="cuda", dtype=torch.bfloat16)
+    dispatch.reset_fallbacks()
+    y = model(x)
+    y.float().square().sum().backward()
+    torch.cuda.synchronize()
+    dispatch.assert_all_native()
