@@ -46,6 +46,8 @@ __all__ = [
     "all_reduce_",
     "repartition_issue",
     "repartition_complete",
+    "begin_chain",
+    "end_chain",
     "reset_chain",
 ]
 
@@ -96,20 +98,23 @@ def all_reduce_(t: torch.Tensor, P: Partition, op: str = "sum") -> torch.Tensor:
 # reverse forward order on every rank.  The chain is reset at each model
 # forward (nn/fno.py) and spans model + loss.
 
-_CHAIN = {"tok": None}
+_CHAIN = {"tok": None, "active": False}
 
 
 def chain_token(ref: torch.Tensor) -> torch.Tensor:
     """Chain link for a comm op on ``ref``.
 
-    Only ops whose backward will RUN (grad mode + ref requires grad) join
-    the chain; for the rest a detached dummy is returned so comm on
-    detached tensors (collectors, eval) neither gains requires_grad nor
-    links grad-comm ordering through it.  ``ref.requires_grad`` is uniform
-    across ranks for every model comm site (activations from the lift /
-    parameter placeholders), which the ordering argument requires.
+    The chain is SCOPED to one model forward (begin_chain/end_chain in
+    DistributedFNONd.forward): only there do parallel graph branches (the
+    channel-chunked pipeline) exist, and only there may tokens link comm
+    ops — an unscoped chain would thread tokens across independently
+    backwarded graphs (two models, collector gathers), making the second
+    backward traverse an already-freed first graph.  Within the scope,
+    only ops whose backward will RUN (grad mode + ref requires grad) join;
+    ``ref.requires_grad`` is uniform across ranks at every model comm site,
+    which the ordering argument requires.
     """
-    if not (torch.is_grad_enabled() and ref.requires_grad):
+    if not (_CHAIN["active"] and torch.is_grad_enabled() and ref.requires_grad):
         return torch.zeros(0, device=ref.device)
     t = _CHAIN["tok"]
     if t is None or t.device != ref.device or not t.requires_grad:
@@ -120,12 +125,22 @@ def chain_token(ref: torch.Tensor) -> torch.Tensor:
 def set_chain(t: torch.Tensor) -> None:
     # advance the chain only when the new link carries grad history;
     # non-grad comm must not sever the ordering between grad comm ops
-    if t is not None and t.grad_fn is not None:
+    if _CHAIN["active"] and t is not None and t.grad_fn is not None:
         _CHAIN["tok"] = t
 
 
-def reset_chain() -> None:
+def begin_chain() -> None:
     _CHAIN["tok"] = None
+    _CHAIN["active"] = True
+
+
+def end_chain() -> None:
+    _CHAIN["tok"] = None
+    _CHAIN["active"] = False
+
+
+def reset_chain() -> None:
+    end_chain()
 
 
 # ---------------------------------------------------------------------------

@@ -71,18 +71,21 @@ class DistributedFNONd(nn.Module):
         self.dt_comm = 0.0
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        from ..comm import reset_chain
-        reset_chain()  # fresh comm ordering chain per forward (spans the
-        self.dt_comm = 0.0  # model AND the loss that follows)
+        from ..comm import begin_chain, end_chain
+        begin_chain()   # comm ordering chain scoped to this forward
+        try:
+            self.dt_comm = 0.0
 
-        x = self._lift(x)
+            x = self._lift(x)
 
-        for block in self.blocks:
-            x = block(x)
-            self.dt_comm += block.dt_comm
+            for block in self.blocks:
+                x = block(x)
+                self.dt_comm += block.dt_comm
 
-        x = self._projection(x)
-        return x
+            x = self._projection(x)
+            return x
+        finally:
+            end_chain()
 
     def _lift(self, x: torch.Tensor) -> torch.Tensor:
         """time lift T_in -> T_out then channel lift C_in -> width, each
