@@ -165,6 +165,7 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_ores_kernel(
     }
 
     const unsigned short* xb = x + ((long)b * I) * S + s;
+#pragma unroll 4
     for (int i = 0; i < I; ++i) {
       float xv[kVec];
       load8(xb + (long)i * S, xv);
@@ -252,64 +253,79 @@ __global__ void bf16_add_gelu_kernel(const unsigned short* __restrict__ a,
 }
 
 // ---------------------------------------------------------------------------
-// grad-W (+grad-bias): gW[o,i] = sum_{b,s} gz[b,o,s] x[b,i,s], fp32 staging.
-// Each block reduces its s-chunk into an LDS [O][I] partial, then one
-// atomicAdd per (o, i) — O*I <= 4096 words.
+// grad-W (+grad-bias): gW[o,i] = sum_{b,s} gz[b,o,s] x[b,i,s], fp32 accum.
+// LDS-tiled outer product: each s-tile of TS elements is staged (converted
+// to fp32, row stride TS+1 so lanes with consecutive i hit distinct banks),
+// then every thread owns ceil(O*I/256) fixed (o,i) pairs and accumulates in
+// REGISTERS over the tile — no atomics in the hot loop (the first version's
+// per-element LDS atomics serialized to 14 ms/launch); one global atomicAdd
+// per pair at thread end.
 // ---------------------------------------------------------------------------
 
-template <int IMAX>
+template <int NP>
 __global__ __launch_bounds__(kBlock) void bf16_gw_kernel(
     const unsigned short* __restrict__ gz, const unsigned short* __restrict__ x,
     float* __restrict__ gW, float* __restrict__ gb,
     int B, int I, int O, long S, bool want_bias) {
+  constexpr int TS = 128;          // s-elements per tile
+  constexpr int LD = TS + 1;       // row stride (bank-conflict pad)
   extern __shared__ __align__(16) char smem_raw[];
-  float* part = reinterpret_cast<float*>(smem_raw);  // [O*I] (+ [O] bias)
-  float* pb = part + (size_t)O * I;
-  for (int k = threadIdx.x; k < O * I; k += blockDim.x) part[k] = 0.f;
-  if (want_bias)
-    for (int k = threadIdx.x; k < O; k += blockDim.x) pb[k] = 0.f;
-  __syncthreads();
+  float* xs = reinterpret_cast<float*>(smem_raw);    // [I][LD]
+  float* gs = xs + (size_t)32 * LD;                  // [O][LD] (I <= 32)
 
-  long nchunks = S / kVec;
-  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-
-  for (long t = t0; t < (long)B * nchunks; t += stride) {
-    int b = (int)(t / nchunks);
-    long s = (t % nchunks) * kVec;
-    float xr[IMAX][kVec];
-    const unsigned short* xb = x + ((long)b * I) * S + s;
+  // fixed (o, i) pairs of this thread
+  int po[NP], pi[NP];
+  float acc[NP];
 #pragma unroll
-    for (int i = 0; i < IMAX; ++i) {
-      if (i < I) load8(xb + (long)i * S, xr[i]);
-    }
-    const unsigned short* gzb = gz + ((long)b * O) * S + s;
-    for (int o = 0; o < O; ++o) {
-      float gv[kVec];
-      load8(gzb + (long)o * S, gv);
-      if (want_bias) {
-        float sum = 0.f;
-#pragma unroll
-        for (int k = 0; k < kVec; ++k) sum += gv[k];
-        atomicAdd(&pb[o], sum);   // LDS atomic (cross-wave in this block)
-      }
-#pragma unroll
-      for (int i = 0; i < IMAX; ++i) {
-        if (i < I) {
-          float sum = 0.f;
-#pragma unroll
-          for (int k = 0; k < kVec; ++k) sum += gv[k] * xr[i][k];
-          atomicAdd(&part[(size_t)o * I + i], sum);
-        }
-      }
-    }
+  for (int p = 0; p < NP; ++p) {
+    const int idx = (int)threadIdx.x + p * kBlock;
+    po[p] = idx / I;
+    pi[p] = idx - po[p] * I;
+    acc[p] = 0.f;
   }
-  __syncthreads();
-  for (int k = threadIdx.x; k < O * I; k += blockDim.x)
-    if (part[k] != 0.f) atomicAdd(&gW[k], part[k]);
-  if (want_bias)
-    for (int k = threadIdx.x; k < O; k += blockDim.x)
-      if (pb[k] != 0.f) atomicAdd(&gb[k], pb[k]);
+  float accb = 0.f;                // bias partial (thread o = threadIdx.x)
+
+  const long ntiles = (S / TS) * (long)B;   // host guarantees S % TS == 0
+  for (long t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const int b = (int)(t / (S / TS));
+    const long s0 = (t % (S / TS)) * TS;
+    // stage: 8 bf16 per thread per instruction, convert to fp32
+    for (int r = threadIdx.x; r < (I + O) * (TS / kVec); r += kBlock) {
+      const int row = r / (TS / kVec);
+      const int col = (r - row * (TS / kVec)) * kVec;
+      float v[kVec];
+      if (row < I) {
+        load8(x + ((long)b * I + row) * S + s0 + col, v);
+#pragma unroll
+        for (int k = 0; k < kVec; ++k) xs[row * LD + col + k] = v[k];
+      } else {
+        const int ro = row - I;
+        load8(gz + ((long)b * O + ro) * S + s0 + col, v);
+#pragma unroll
+        for (int k = 0; k < kVec; ++k) gs[ro * LD + col + k] = v[k];
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int p = 0; p < NP; ++p) {
+      if (po[p] < O) {
+        const float* gr = gs + po[p] * LD;
+        const float* xr = xs + pi[p] * LD;
+#pragma unroll 8
+        for (int k = 0; k < TS; ++k) acc[p] += gr[k] * xr[k];
+      }
+    }
+    if (want_bias && (int)threadIdx.x < O) {
+      const float* gr = gs + threadIdx.x * LD;
+#pragma unroll 8
+      for (int k = 0; k < TS; ++k) accb += gr[k];
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int p = 0; p < NP; ++p)
+    if (po[p] < O) atomicAdd(&gW[(size_t)po[p] * I + pi[p]], acc[p]);
+  if (want_bias && (int)threadIdx.x < O) atomicAdd(&gb[threadIdx.x], accb);
 }
 
 int bgrid(long work) {
@@ -389,7 +405,8 @@ std::vector<at::Tensor> bf16_channel_mix_bwd_w(const at::Tensor& gz,
   const long S = x.size(2);
   const int O = (int)gz.size(1);
   TORCH_CHECK(I <= 32, "bf16 grad-W: I must be <= 32");
-  TORCH_CHECK(S % kVec == 0, "bf16 grad-W: S must be a multiple of 8");
+  TORCH_CHECK(O <= 128, "bf16 grad-W: O must be <= 128");
+  TORCH_CHECK(S % 128 == 0, "bf16 grad-W: S must be a multiple of 128");
 
   auto opts = x.options().dtype(at::kFloat);
   auto gW = at::zeros({O, I}, opts);
@@ -397,14 +414,19 @@ std::vector<at::Tensor> bf16_channel_mix_bwd_w(const at::Tensor& gz,
   if (x.numel() == 0) return {gW, gb};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  size_t smem = sizeof(float) * ((size_t)O * I + O);
-  int grid = bgrid((long)B * (S / kVec));
-#define BGW(CAP)                                                               \
-  hipLaunchKernelGGL((bf16_gw_kernel<CAP>), dim3(grid), dim3(kBlock), smem,    \
+  constexpr int LD = 129;
+  size_t smem = sizeof(float) * (size_t)(32 + O) * LD;
+  long ntiles = (S / 128) * (long)B;
+  int grid = (int)std::min(ntiles, 2048L);
+  const int pairs = O * I;
+#define BGW(NP)                                                                \
+  hipLaunchKernelGGL((bf16_gw_kernel<NP>), dim3(grid), dim3(kBlock), smem,     \
                      stream, usp(gz), usp(x), gW.data_ptr<float>(),            \
                      gb.numel() ? gb.data_ptr<float>() : nullptr, B, I, O, S,  \
                      want_bias);
-  if (I <= 8) { BGW(8) } else if (I <= 24) { BGW(24) } else { BGW(32) }
+  if (pairs <= 256) { BGW(1) } else if (pairs <= 512) { BGW(2) }
+  else if (pairs <= 1024) { BGW(4) } else if (pairs <= 2048) { BGW(8) }
+  else { BGW(16) }
 #undef BGW
   DFNO_CHECK_LAUNCH("bf16_gw");
   return {gW, gb};

@@ -139,7 +139,8 @@ class _ChannelMixFn(torch.autograd.Function):
             gW, gb = ext.channel_mix_bwd_w(gz.contiguous(), x3, ctx.has_bias)
             if not ctx.has_bias:
                 gb = None
-        elif is_bf16 and x3.shape[1] <= 32 and x3.shape[2] % 8 == 0:
+        elif (is_bf16 and x3.shape[1] <= 32 and W.shape[0] <= 128
+              and x3.shape[2] % 128 == 0):
             ext = _ext.get(required=True)
             gW, gb = ext.bf16_channel_mix_bwd_w(gz.contiguous(), x3.contiguous(),
                                                 ctx.has_bias)
@@ -240,7 +241,8 @@ class _LinearResGeluFn(torch.autograd.Function):
             gz = ext.bf16_gelu_bwd(gy, z3.contiguous())
             gx, _ = ext.bf16_channel_mix(gz, W.contiguous(), _ebf(gy.device),
                                          False, True, False, _ebf(gy.device))
-            if x3.shape[1] <= 32:
+            if (x3.shape[1] <= 32 and W.shape[0] <= 128
+                    and x3.shape[2] % 128 == 0):
                 gW, _ = ext.bf16_channel_mix_bwd_w(gz, x3.contiguous(), False)
                 gW = gW.to(W.dtype)
             else:
