@@ -136,12 +136,14 @@ def test_bf16_model_matches_fp32():
         if p32.grad is None or p32.grad.numel() == 0:
             continue
         g32 = p32.grad.flatten()
-        g16 = p16.grad.float().flatten()
-        if g32.norm() < 1e-12:
-            continue
+        g16 = p16.grad.flatten()
         if g32.is_complex():
             g32 = torch.view_as_real(g32).flatten()
             g16 = torch.view_as_real(g16).flatten()
+        g32 = g32.float()
+        g16 = g16.float()
+        if g32.norm() < 1e-12:
+            continue
         cos = torch.dot(g16, g32) / (g16.norm() * g32.norm()).clamp_min(1e-30)
         assert cos > 0.98, f"{n}: grad cosine {cos}"
 
