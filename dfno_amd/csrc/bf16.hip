@@ -35,13 +35,13 @@ __device__ __forceinline__ float b2f(unsigned short h) {
   return __uint_as_float(u);
 }
 
-__device__ __forceinline__ unsigned short f2b(float f) {
-  // round-to-nearest-even bf16
-  unsigned int u = __float_as_uint(f);
-  unsigned int lsb = (u >> 16) & 1u;
-  u += 0x7fffu + lsb;
-  if ((__float_as_uint(f) & 0x7f800000u) == 0x7f800000u) u = __float_as_uint(f);
-  return (unsigned short)(u >> 16);
+// Pack a float pair to bf16x2 with round-to-nearest-even: the compiler
+// lowers the __hip_bfloat162 conversion to v_cvt_pk_bf16_f32 (one VALU op
+// per pair; the first integer-math version cost ~5 VALU per element and
+// made every store-side kernel conversion-bound).
+__device__ __forceinline__ unsigned int f2b2(float lo, float hi) {
+  __hip_bfloat162 h2 = __float22bfloat162_rn(float2{lo, hi});
+  return *reinterpret_cast<unsigned int*>(&h2);
 }
 
 __device__ __forceinline__ void load8(const unsigned short* p, float* out) {
@@ -56,24 +56,41 @@ __device__ __forceinline__ void load8(const unsigned short* p, float* out) {
 
 __device__ __forceinline__ void store8(unsigned short* p, const float* v) {
   uint4 raw;
-  unsigned int w[4];
-#pragma unroll
-  for (int k = 0; k < 4; ++k)
-    w[k] = (unsigned int)f2b(v[2 * k]) | ((unsigned int)f2b(v[2 * k + 1]) << 16);
-  raw.x = w[0]; raw.y = w[1]; raw.z = w[2]; raw.w = w[3];
+  raw.x = f2b2(v[0], v[1]);
+  raw.y = f2b2(v[2], v[3]);
+  raw.z = f2b2(v[4], v[5]);
+  raw.w = f2b2(v[6], v[7]);
   *reinterpret_cast<uint4*>(p) = raw;
+}
+
+__device__ __forceinline__ void load4(const unsigned short* p, float* out) {
+  const uint2 raw = *reinterpret_cast<const uint2*>(p);
+  const unsigned int w[2] = {raw.x, raw.y};
+#pragma unroll
+  for (int k = 0; k < 2; ++k) {
+    out[2 * k] = b2f((unsigned short)(w[k] & 0xffffu));
+    out[2 * k + 1] = b2f((unsigned short)(w[k] >> 16));
+  }
+}
+
+__device__ __forceinline__ void store4(unsigned short* p, const float* v) {
+  uint2 raw;
+  raw.x = f2b2(v[0], v[1]);
+  raw.y = f2b2(v[2], v[3]);
+  *reinterpret_cast<uint2*>(p) = raw;
 }
 
 // ---------------------------------------------------------------------------
 // x-resident channel mix (I <= IMAX): one thread owns kVec consecutive s.
 // ---------------------------------------------------------------------------
 
-template <int IMAX, bool ACT>
+template <int IMAX, int VEC, bool ACT>
 __global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ W,
     const unsigned short* __restrict__ bias, unsigned short* __restrict__ y,
     unsigned short* __restrict__ z, int B, int I, int O, long S, bool wt,
     bool has_bias, bool write_z, const unsigned short* __restrict__ res) {
+  // VEC=4 for the wide-IMAX variants (xr[24][8] = 192 VGPRs would spill)
   extern __shared__ __align__(16) char smem_raw[];
   float* Wl = reinterpret_cast<float*>(smem_raw);   // [O*I] fp32
   float* bl = Wl + (size_t)O * I;                   // [O]
@@ -82,19 +99,26 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
     for (int k = threadIdx.x; k < O; k += blockDim.x) bl[k] = b2f(bias[k]);
   __syncthreads();
 
-  long nchunks = S / kVec;   // host guarantees S % 8 == 0
+  auto loadv = [](const unsigned short* p, float* v) {
+    if constexpr (VEC == 8) load8(p, v); else load4(p, v);
+  };
+  auto storev = [](unsigned short* p, const float* v) {
+    if constexpr (VEC == 8) store8(p, v); else store4(p, v);
+  };
+
+  long nchunks = S / VEC;   // host guarantees S % 8 == 0
   long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
 
   for (long t = t0; t < (long)B * nchunks; t += stride) {
     int b = (int)(t / nchunks);
-    long s = (t % nchunks) * kVec;
+    long s = (t % nchunks) * VEC;
 
-    float xr[IMAX][kVec];
+    float xr[IMAX][VEC];
     const unsigned short* xb = x + ((long)b * I) * S + s;
 #pragma unroll
     for (int i = 0; i < IMAX; ++i) {
-      if (i < I) load8(xb + (long)i * S, xr[i]);
+      if (i < I) loadv(xb + (long)i * S, xr[i]);
     }
 
     unsigned short* yb = y + ((long)b * O) * S + s;
@@ -102,30 +126,30 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
 #pragma unroll 4
     for (int o = 0; o < 512; ++o) {
       if (o >= O) break;
-      float acc[kVec];
+      float acc[VEC];
       const float bv = has_bias ? bl[o] : 0.f;
 #pragma unroll
-      for (int k = 0; k < kVec; ++k) acc[k] = bv;
+      for (int k = 0; k < VEC; ++k) acc[k] = bv;
 #pragma unroll
       for (int i = 0; i < IMAX; ++i) {
         if (i < I) {
           const float wv = wt ? Wl[(size_t)i * O + o] : Wl[(size_t)o * I + i];
 #pragma unroll
-          for (int k = 0; k < kVec; ++k) acc[k] += wv * xr[i][k];
+          for (int k = 0; k < VEC; ++k) acc[k] += wv * xr[i][k];
         }
       }
       if (res != nullptr) {
-        float rv[kVec];
-        load8(res + ((long)b * O + o) * S + s, rv);
+        float rv[VEC];
+        loadv(res + ((long)b * O + o) * S + s, rv);
 #pragma unroll
-        for (int k = 0; k < kVec; ++k) acc[k] += rv[k];
+        for (int k = 0; k < VEC; ++k) acc[k] += rv[k];
       }
-      if (write_z) store8(zb + (long)o * S, acc);
+      if (write_z) storev(zb + (long)o * S, acc);
       if (ACT) {
 #pragma unroll
-        for (int k = 0; k < kVec; ++k) acc[k] = dfno_gelu::gelu(acc[k]);
+        for (int k = 0; k < VEC; ++k) acc[k] = dfno_gelu::gelu(acc[k]);
       }
-      store8(yb + (long)o * S, acc);
+      storev(yb + (long)o * S, acc);
     }
   }
 }
@@ -134,12 +158,14 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
 // accumulator-resident channel mix (O <= OMAX, streams I) — the 128->1 head
 // ---------------------------------------------------------------------------
 
-template <int OMAX, bool ACT>
+template <int OMAX, int VEC, bool ACT>
 __global__ __launch_bounds__(kBlock) void bf16_channel_mix_ores_kernel(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ W,
     const unsigned short* __restrict__ bias, unsigned short* __restrict__ y,
     unsigned short* __restrict__ z, int B, int I, int O, long S, bool wt,
     bool has_bias, bool write_z, const unsigned short* __restrict__ res) {
+  // VEC=4 for the wide-OMAX variants: acc[24][8] would eat 192 VGPRs and
+  // spill; [24][4] keeps the whole accumulator set resident.
   extern __shared__ __align__(16) char smem_raw[];
   float* Wl = reinterpret_cast<float*>(smem_raw);   // [O*I]
   float* bl = Wl + (size_t)O * I;
@@ -148,33 +174,40 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_ores_kernel(
     for (int k = threadIdx.x; k < O; k += blockDim.x) bl[k] = b2f(bias[k]);
   __syncthreads();
 
-  long nchunks = S / kVec;
+  auto loadv = [](const unsigned short* p, float* v) {
+    if constexpr (VEC == 8) load8(p, v); else load4(p, v);
+  };
+  auto storev = [](unsigned short* p, const float* v) {
+    if constexpr (VEC == 8) store8(p, v); else store4(p, v);
+  };
+
+  long nchunks = S / VEC;
   long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
 
   for (long t = t0; t < (long)B * nchunks; t += stride) {
     int b = (int)(t / nchunks);
-    long s = (t % nchunks) * kVec;
+    long s = (t % nchunks) * VEC;
 
-    float acc[OMAX][kVec];
+    float acc[OMAX][VEC];
 #pragma unroll
     for (int o = 0; o < OMAX; ++o) {
       const float bv = (o < O && has_bias) ? bl[o] : 0.f;
 #pragma unroll
-      for (int k = 0; k < kVec; ++k) acc[o][k] = bv;
+      for (int k = 0; k < VEC; ++k) acc[o][k] = bv;
     }
 
     const unsigned short* xb = x + ((long)b * I) * S + s;
 #pragma unroll 4
     for (int i = 0; i < I; ++i) {
-      float xv[kVec];
-      load8(xb + (long)i * S, xv);
+      float xv[VEC];
+      loadv(xb + (long)i * S, xv);
 #pragma unroll
       for (int o = 0; o < OMAX; ++o) {
         if (o < O) {
           const float wv = wt ? Wl[(size_t)i * O + o] : Wl[(size_t)o * I + i];
 #pragma unroll
-          for (int k = 0; k < kVec; ++k) acc[o][k] += wv * xv[k];
+          for (int k = 0; k < VEC; ++k) acc[o][k] += wv * xv[k];
         }
       }
     }
@@ -185,17 +218,17 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_ores_kernel(
     for (int o = 0; o < OMAX; ++o) {
       if (o < O) {
         if (res != nullptr) {
-          float rv[kVec];
-          load8(res + ((long)b * O + o) * S + s, rv);
+          float rv[VEC];
+          loadv(res + ((long)b * O + o) * S + s, rv);
 #pragma unroll
-          for (int k = 0; k < kVec; ++k) acc[o][k] += rv[k];
+          for (int k = 0; k < VEC; ++k) acc[o][k] += rv[k];
         }
-        if (write_z) store8(zb + (long)o * S, acc[o]);
+        if (write_z) storev(zb + (long)o * S, acc[o]);
         if (ACT) {
 #pragma unroll
-          for (int k = 0; k < kVec; ++k) acc[o][k] = dfno_gelu::gelu(acc[o][k]);
+          for (int k = 0; k < VEC; ++k) acc[o][k] = dfno_gelu::gelu(acc[o][k]);
         }
-        store8(yb + (long)o * S, acc[o]);
+        storev(yb + (long)o * S, acc[o]);
       }
     }
   }
@@ -268,12 +301,20 @@ __global__ __launch_bounds__(kBlock) void bf16_gw_kernel(
     float* __restrict__ gW, float* __restrict__ gb,
     int B, int I, int O, long S, bool want_bias) {
   constexpr int TS = 128;          // s-elements per tile
-  constexpr int LD = TS + 1;       // row stride (bank-conflict pad)
+  constexpr int LD = TS + 4;       // row stride: float4 reads stay 16 B-
+                                   // aligned, banks spread (132 % 64 = 4)
+  constexpr int OSL = 32;          // O-slab per blockIdx.y: caps the LDS
+                                   // footprint at (32+32)*132*4 = 33 KiB so
+                                   // 4 blocks/CU stay resident (the O=128
+                                   // single-slab version ran 1 block/CU)
   extern __shared__ __align__(16) char smem_raw[];
   float* xs = reinterpret_cast<float*>(smem_raw);    // [I][LD]
-  float* gs = xs + (size_t)32 * LD;                  // [O][LD] (I <= 32)
+  float* gs = xs + (size_t)32 * LD;                  // [O_sl][LD]
 
-  // fixed (o, i) pairs of this thread
+  const int o0 = (int)blockIdx.y * OSL;
+  const int O_sl = min(OSL, O - o0);
+
+  // fixed (o, i) pairs of this thread within the slab
   int po[NP], pi[NP];
   float acc[NP];
 #pragma unroll
@@ -290,7 +331,7 @@ __global__ __launch_bounds__(kBlock) void bf16_gw_kernel(
     const int b = (int)(t / (S / TS));
     const long s0 = (t % (S / TS)) * TS;
     // stage: 8 bf16 per thread per instruction, convert to fp32
-    for (int r = threadIdx.x; r < (I + O) * (TS / kVec); r += kBlock) {
+    for (int r = threadIdx.x; r < (I + O_sl) * (TS / kVec); r += kBlock) {
       const int row = r / (TS / kVec);
       const int col = (r - row * (TS / kVec)) * kVec;
       float v[kVec];
@@ -300,7 +341,7 @@ __global__ __launch_bounds__(kBlock) void bf16_gw_kernel(
         for (int k = 0; k < kVec; ++k) xs[row * LD + col + k] = v[k];
       } else {
         const int ro = row - I;
-        load8(gz + ((long)b * O + ro) * S + s0 + col, v);
+        load8(gz + ((long)b * O + o0 + ro) * S + s0 + col, v);
 #pragma unroll
         for (int k = 0; k < kVec; ++k) gs[ro * LD + col + k] = v[k];
       }
@@ -308,24 +349,32 @@ __global__ __launch_bounds__(kBlock) void bf16_gw_kernel(
     __syncthreads();
 #pragma unroll
     for (int p = 0; p < NP; ++p) {
-      if (po[p] < O) {
+      if (po[p] < O_sl) {
         const float* gr = gs + po[p] * LD;
         const float* xr = xs + pi[p] * LD;
-#pragma unroll 8
-        for (int k = 0; k < TS; ++k) acc[p] += gr[k] * xr[k];
+#pragma unroll
+        for (int k = 0; k < TS; k += 4) {
+          const float4 g4 = *reinterpret_cast<const float4*>(gr + k);
+          const float4 x4 = *reinterpret_cast<const float4*>(xr + k);
+          acc[p] += g4.x * x4.x + g4.y * x4.y + g4.z * x4.z + g4.w * x4.w;
+        }
       }
     }
-    if (want_bias && (int)threadIdx.x < O) {
+    if (want_bias && (int)threadIdx.x < O_sl) {
       const float* gr = gs + threadIdx.x * LD;
-#pragma unroll 8
-      for (int k = 0; k < TS; ++k) accb += gr[k];
+#pragma unroll
+      for (int k = 0; k < TS; k += 4) {
+        const float4 g4 = *reinterpret_cast<const float4*>(gr + k);
+        accb += g4.x + g4.y + g4.z + g4.w;
+      }
     }
     __syncthreads();
   }
 #pragma unroll
   for (int p = 0; p < NP; ++p)
-    if (po[p] < O) atomicAdd(&gW[(size_t)po[p] * I + pi[p]], acc[p]);
-  if (want_bias && (int)threadIdx.x < O) atomicAdd(&gb[threadIdx.x], accb);
+    if (po[p] < O_sl) atomicAdd(&gW[(size_t)(o0 + po[p]) * I + pi[p]], acc[p]);
+  if (want_bias && (int)threadIdx.x < O_sl)
+    atomicAdd(&gb[o0 + threadIdx.x], accb);
 }
 
 int bgrid(long work) {
@@ -375,22 +424,35 @@ std::vector<at::Tensor> bf16_channel_mix(const at::Tensor& x, const at::Tensor& 
   const unsigned short* rp = has_res ? usp(res) : nullptr;
   unsigned short* zp = write_z ? usp_mut(z) : nullptr;
 
-#define BCM(KER, CAP)                                                          \
+#define BCM(KER, CAP, VEC)                                                     \
   if (act) {                                                                   \
-    hipLaunchKernelGGL((KER<CAP, true>), dim3(grid), dim3(kBlock), smem,       \
+    hipLaunchKernelGGL((KER<CAP, VEC, true>), dim3(grid), dim3(kBlock), smem,  \
                        stream, usp(x), usp(W), bp, usp_mut(y), zp, B, I, O, S, \
                        wt, has_bias, write_z, rp);                             \
   } else {                                                                     \
-    hipLaunchKernelGGL((KER<CAP, false>), dim3(grid), dim3(kBlock), smem,      \
+    hipLaunchKernelGGL((KER<CAP, VEC, false>), dim3(grid), dim3(kBlock), smem, \
                        stream, usp(x), usp(W), bp, usp_mut(y), zp, B, I, O, S, \
                        wt, has_bias, write_z, rp);                             \
   }
-  if (I <= 8) { BCM(bf16_channel_mix_xres_kernel, 8) }
-  else if (I <= 24) { BCM(bf16_channel_mix_xres_kernel, 24) }
-  else if (I <= 32) { BCM(bf16_channel_mix_xres_kernel, 32) }
-  else if (O <= 8) { BCM(bf16_channel_mix_ores_kernel, 8) }
-  else if (O <= 24) { BCM(bf16_channel_mix_ores_kernel, 24) }
+#define BCMO(CAP, VEC)                                                         \
+  if (act) {                                                                   \
+    hipLaunchKernelGGL((bf16_channel_mix_ores_kernel<CAP, VEC, true>),         \
+                       dim3(grid), dim3(kBlock), smem, stream, usp(x), usp(W), \
+                       bp, usp_mut(y), zp, B, I, O, S, wt, has_bias, write_z,  \
+                       rp);                                                    \
+  } else {                                                                     \
+    hipLaunchKernelGGL((bf16_channel_mix_ores_kernel<CAP, VEC, false>),        \
+                       dim3(grid), dim3(kBlock), smem, stream, usp(x), usp(W), \
+                       bp, usp_mut(y), zp, B, I, O, S, wt, has_bias, write_z,  \
+                       rp);                                                    \
+  }
+  if (I <= 8) { BCM(bf16_channel_mix_xres_kernel, 8, 8) }
+  else if (I <= 24) { BCM(bf16_channel_mix_xres_kernel, 24, 4) }
+  else if (I <= 32) { BCM(bf16_channel_mix_xres_kernel, 32, 4) }
+  else if (O <= 8) { BCMO(8, 8) }
+  else if (O <= 24) { BCMO(24, 4) }
   else { TORCH_CHECK(false, "bf16 channel mix: unsupported shape I=", I, " O=", O); }
+#undef BCMO
 #undef BCM
   DFNO_CHECK_LAUNCH("bf16_channel_mix");
   return {y, z};
@@ -414,19 +476,20 @@ std::vector<at::Tensor> bf16_channel_mix_bwd_w(const at::Tensor& gz,
   if (x.numel() == 0) return {gW, gb};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  constexpr int LD = 129;
-  size_t smem = sizeof(float) * (size_t)(32 + O) * LD;
+  constexpr int LD = 132;
+  const int nslab = (O + 31) / 32;
+  const int o_sl = std::min(O, 32);
+  size_t smem = sizeof(float) * (size_t)(32 + o_sl) * LD;
   long ntiles = (S / 128) * (long)B;
   int grid = (int)std::min(ntiles, 2048L);
-  const int pairs = O * I;
+  const int pairs = o_sl * I;
 #define BGW(NP)                                                                \
-  hipLaunchKernelGGL((bf16_gw_kernel<NP>), dim3(grid), dim3(kBlock), smem,     \
-                     stream, usp(gz), usp(x), gW.data_ptr<float>(),            \
+  hipLaunchKernelGGL((bf16_gw_kernel<NP>), dim3(grid, nslab), dim3(kBlock),    \
+                     smem, stream, usp(gz), usp(x), gW.data_ptr<float>(),      \
                      gb.numel() ? gb.data_ptr<float>() : nullptr, B, I, O, S,  \
                      want_bias);
   if (pairs <= 256) { BGW(1) } else if (pairs <= 512) { BGW(2) }
-  else if (pairs <= 1024) { BGW(4) } else if (pairs <= 2048) { BGW(8) }
-  else { BGW(16) }
+  else { BGW(4) }
 #undef BGW
   DFNO_CHECK_LAUNCH("bf16_gw");
   return {gW, gb};
