@@ -377,6 +377,98 @@ __global__ __launch_bounds__(kBlock) void bf16_gw_kernel(
     atomicAdd(&gb[o0 + threadIdx.x], accb);
 }
 
+// ---------------------------------------------------------------------------
+// MFMA grad-W: gW[o,i] = sum_s gz[o,s] x[i,s] on v_mfma_f32_16x16x32_bf16.
+// The one truly GEMM-shaped hot op of the model (K = S ~ 10^7): the VALU
+// version above is arithmetic-limited (~10 flop per bf16-byte at width 20,
+// right at the fp32 VALU roofline); the matrix cores run the same contraction
+// at the 2.5 PF bf16 rate, making it memory-bound.  Fragment loads need no
+// LDS: for both A (gz) and B (x) a lane's 8 k-elements are 8 CONSECUTIVE
+// s-positions of one channel row — one 16-byte load each, rows spread over
+// lanes (lane groups at k-offsets 0/8/16/24 tile 64 B per row).  Bias falls
+// out of one extra MFMA against a ones-column B fragment.
+// ---------------------------------------------------------------------------
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ bf16x8 load_frag(const unsigned short* row_base,
+                                            long k0) {
+  // 8 consecutive bf16 at s = k0 (16 B)
+  return *reinterpret_cast<const bf16x8*>(row_base + k0);
+}
+
+template <int NPAIR>
+__global__ __launch_bounds__(kBlock) void bf16_gw_mfma_kernel(
+    const unsigned short* __restrict__ gz, const unsigned short* __restrict__ x,
+    float* __restrict__ gW, float* __restrict__ gb,
+    int B, int I, int O, long S, bool want_bias) {
+  constexpr int TS = 256;          // s per grid tile (8 MFMA steps of K=32)
+  const int lane = (int)(threadIdx.x & 63);
+  const int wave = (int)(threadIdx.x >> 6);
+  const int row16 = lane & 15;     // A row / B col within the 16-tile
+  const int kgrp = lane >> 4;      // k-group (8 k each)
+
+  const int oT = (O + 15) / 16, iT = (I + 15) / 16;
+  const int npairs = oT * iT;      // host guarantees npairs <= 4 * NPAIR
+
+  f32x4 acc[NPAIR];
+  f32x4 accb[NPAIR];               // bias (only i-tile 0 pairs contribute)
+#pragma unroll
+  for (int p = 0; p < NPAIR; ++p) {
+    acc[p] = f32x4{0.f, 0.f, 0.f, 0.f};
+    accb[p] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+
+  bf16x8 ones = bf16x8{};
+  if (kgrp >= 0) {                 // B[k][0] = 1 for the bias column
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ones[j] = (row16 == 0) ? (__bf16)1.f : (__bf16)0.f;
+  }
+
+  const long ntiles = (S / TS) * (long)B;
+  for (long t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const int b = (int)(t / (S / TS));
+    const long s0 = (t % (S / TS)) * TS;
+    for (int pp = 0; pp < NPAIR; ++pp) {
+      const int p = wave + 4 * pp;
+      if (p >= npairs) break;
+      const int ot = p / iT, it = p - ot * iT;
+      const int o = ot * 16 + row16;
+      const int i = it * 16 + row16;
+      const unsigned short* gr =
+          (o < O) ? gz + ((long)b * O + o) * S + s0 : nullptr;
+      const unsigned short* xr =
+          (i < I) ? x + ((long)b * I + i) * S + s0 : nullptr;
+#pragma unroll
+      for (int ks = 0; ks < TS; ks += 32) {
+        const long k0 = ks + kgrp * 8;
+        bf16x8 a = gr ? load_frag(gr, k0) : bf16x8{};
+        bf16x8 bb = xr ? load_frag(xr, k0) : bf16x8{};
+        acc[pp] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bb, acc[pp], 0, 0, 0);
+        if (want_bias && it == 0)
+          accb[pp] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, ones, accb[pp], 0, 0, 0);
+      }
+    }
+  }
+
+  // writeback: D[row][col], row = (lane>>4)*4 + r, col = lane&15
+#pragma unroll
+  for (int pp = 0; pp < NPAIR; ++pp) {
+    const int p = wave + 4 * pp;
+    if (p >= npairs) continue;
+    const int ot = p / iT, it = p - ot * iT;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int o = ot * 16 + kgrp * 4 + r;
+      const int i = it * 16 + row16;
+      if (o < O && i < I) atomicAdd(&gW[(size_t)o * I + i], acc[pp][r]);
+      if (want_bias && it == 0 && o < O && row16 == 0)
+        atomicAdd(&gb[o], accb[pp][r]);
+    }
+  }
+}
+
 int bgrid(long work) {
   long g = (work + kBlock - 1) / kBlock;
   if (g > 256L * 16) g = 256L * 16;
@@ -476,6 +568,28 @@ std::vector<at::Tensor> bf16_channel_mix_bwd_w(const at::Tensor& gz,
   if (x.numel() == 0) return {gW, gb};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  static const bool no_mfma = []() {
+    const char* e = getenv("DFNO_BF16_GW_NO_MFMA");  // A/B knob
+    return e && e[0] == '1';
+  }();
+  const int mfma_pairs = ((O + 15) / 16) * ((I + 15) / 16);
+  if (!no_mfma && S % 256 == 0 && mfma_pairs <= 16) {
+    long ntiles = (S / 256) * (long)B;
+    int grid = (int)std::min(ntiles, 2048L);
+    if (mfma_pairs <= 4) {
+      hipLaunchKernelGGL((bf16_gw_mfma_kernel<1>), dim3(grid), dim3(kBlock), 0,
+                         stream, usp(gz), usp(x), gW.data_ptr<float>(),
+                         gb.numel() ? gb.data_ptr<float>() : nullptr, B, I, O,
+                         S, want_bias);
+    } else {
+      hipLaunchKernelGGL((bf16_gw_mfma_kernel<4>), dim3(grid), dim3(kBlock), 0,
+                         stream, usp(gz), usp(x), gW.data_ptr<float>(),
+                         gb.numel() ? gb.data_ptr<float>() : nullptr, B, I, O,
+                         S, want_bias);
+    }
+    DFNO_CHECK_LAUNCH("bf16_gw_mfma");
+    return {gW, gb};
+  }
   constexpr int LD = 132;
   const int nslab = (O + 31) / 32;
   const int o_sl = std::min(O, 32);

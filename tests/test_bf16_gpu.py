@@ -65,9 +65,10 @@ def test_bf16_channel_mix_transposed():
 
 
 def test_bf16_grad_w():
+    # S % 256 != 0 -> the LDS-tiled VALU kernel (the MFMA path's fallback)
     ext = _ext()
     torch.manual_seed(2)
-    B, I, O, S = 2, 20, 24, 4096
+    B, I, O, S = 2, 20, 24, 4224
     gz = (torch.rand(B, O, S, device="cuda") - 0.5).bfloat16()
     x = (torch.rand(B, I, S, device="cuda") - 0.5).bfloat16()
     gW, gb = ext.bf16_channel_mix_bwd_w(gz, x, True)
@@ -76,6 +77,30 @@ def test_bf16_grad_w():
     assert torch.allclose(gW, refW, rtol=2e-3, atol=refW.abs().max() * 2e-3), \
         f"max {(gW-refW).abs().max()}"
     assert torch.allclose(gb, refb, rtol=2e-3, atol=refb.abs().max() * 2e-3)
+
+
+@pytest.mark.parametrize("I,O,bias,S", [
+    (20, 20, False, 4096),    # block residual grad-W (NPAIR=1)
+    (20, 24, True, 4096),
+    (20, 128, True, 7936),    # projection lift grad-W (NPAIR=4, ragged tiles)
+    (2, 20, True, 4096),      # channel-lift grad-W (ragged i-tile)
+])
+def test_bf16_grad_w_mfma(I, O, bias, S):
+    """The MFMA grad-W path (v_mfma_f32_16x16x32_bf16) against einsum.
+    Asymmetric inputs so an operand/output transpose cannot pass."""
+    ext = _ext()
+    torch.manual_seed(7)
+    B = 1
+    gz = (torch.rand(B, O, S, device="cuda") - 0.3).bfloat16()
+    x = (torch.rand(B, I, S, device="cuda") - 0.7).bfloat16()
+    gW, gb = ext.bf16_channel_mix_bwd_w(gz, x, bias)
+    refW = torch.einsum("bos,bis->oi", gz.float(), x.float())
+    assert torch.allclose(gW, refW, rtol=2e-3, atol=refW.abs().max() * 2e-3), \
+        f"max {(gW-refW).abs().max()} of {refW.abs().max()}"
+    if bias:
+        refb = gz.float().sum(dim=(0, 2))
+        assert torch.allclose(gb, refb, rtol=2e-3, atol=refb.abs().max() * 2e-3), \
+            f"bias max {(gb-refb).abs().max()}"
 
 
 def test_bf16_gelu_and_add_gelu():
