@@ -72,6 +72,9 @@ def main():
                         "dtype), bf16 is the reduced-precision config "
                         "(BASELINE.json config #2: bf16 storage, fp32 math, "
                         "complex64 spectral path)")
+    p.add_argument("--spectral-fp8", action="store_true",
+                   help="e4m3 storage for the spectral corner weights "
+                        "(BASELINE.json config #5); masters stay complex64")
     p.add_argument("--heavy-comm", action="store_true",
                    help="partition the trailing spatial axis: R1/R4 become "
                         "real full-activation all-to-alls")
@@ -107,7 +110,7 @@ def main():
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     model = dfno.DistributedFNONd(P_x, GLOBAL_SHAPE, OUT_T, args.width, MODES,
                                   num_blocks=args.num_blocks, device=device,
-                                  dtype=dtype)
+                                  dtype=dtype, spectral_fp8=args.spectral_fp8)
     criterion = dfno.DistributedRelativeLpLoss(P_x)
     from dfno_amd.optim import Adam as FusedAdam
     if torch.cuda.is_available():
@@ -194,6 +197,7 @@ def main():
                 "num_blocks": args.num_blocks,
                 "parallelism": "spatial-model-parallel " + "x".join(map(str, pshape)),
                 "step": "fwd+relative-Lp-loss+bwd+adam",
+                "spectral_weights": "fp8-e4m3" if args.spectral_fp8 else "complex64",
             },
         }))
 

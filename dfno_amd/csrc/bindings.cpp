@@ -41,6 +41,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "all corner boxes adjoint wrt x in one launch");
   m.def("spectral_corners_bwd_w", &spectral_corners_bwd_w,
         "multi-corner spectral grad-W (gw[i,o,e] = sum_b x conj(gy))");
+  m.def("spectral_corners_fwd_fp8", &spectral_corners_fwd_fp8,
+        "fp8-weight spectral contraction (e4m3 packed pairs + per-corner scale)");
+  m.def("spectral_corners_bwd_x_fp8", &spectral_corners_bwd_x_fp8,
+        "fp8-weight spectral contraction adjoint wrt x");
   m.def("bf16_channel_mix", &bf16_channel_mix,
         "bf16-storage fused channel linear (+bias/res/gelu): returns (y, z)");
   m.def("bf16_channel_mix_bwd_w", &bf16_channel_mix_bwd_w,

@@ -107,3 +107,11 @@ void spectral_corners_bwd_x(const at::Tensor& gy, std::vector<at::Tensor> ws,
 void spectral_corners_bwd_w(const at::Tensor& x, const at::Tensor& gy,
                             std::vector<at::Tensor> gws,
                             std::vector<std::vector<int64_t>> starts);
+
+// fp8 (e4m3 packed-pair) spectral-weight variants; scales dequant per corner
+void spectral_corners_fwd_fp8(const at::Tensor& x, std::vector<at::Tensor> w16s,
+                              std::vector<double> scales, at::Tensor& y,
+                              std::vector<std::vector<int64_t>> starts);
+void spectral_corners_bwd_x_fp8(const at::Tensor& gy, std::vector<at::Tensor> w16s,
+                                std::vector<double> scales, at::Tensor& gx,
+                                std::vector<std::vector<int64_t>> starts);

@@ -58,6 +58,20 @@ __device__ __forceinline__ void cmac_conj(T& ar, T& ai, T br, T bi, T cr, T ci) 
   ai += br * ci - bi * cr;
 }
 
+// OCP e4m3fn dequant (gfx950's FP8 format — NOT the MI300X fnuz variant):
+// 1 sign, 4 exponent (bias 7), 3 mantissa; S.1111.111 is NaN, no infinities.
+__device__ __forceinline__ float fp8_e4m3_to_f32(unsigned int v) {
+  const unsigned int s = v >> 7, e = (v >> 3) & 0xf, m = v & 7;
+  float f;
+  if (e == 0) {
+    f = (float)m * (1.f / 512.f);                 // subnormal: m * 2^-9
+  } else {
+    // (1 + m/8) * 2^(e-7) via exponent-bit construction
+    f = __uint_as_float(((e + 120u) << 23) | (m << 20));
+  }
+  return s ? -f : f;
+}
+
 __device__ __forceinline__ long box_to_global(long e, const BoxGeom& g) {
   long off = g.origin;
   // innermost dim last in box[]; decompose right-to-left
@@ -144,7 +158,9 @@ constexpr int kMaxCorners = 8;
 template <typename T>
 struct MultiGeom {
   BoxGeom g[kMaxCorners];
-  const T* w[kMaxCorners];
+  const T* w[kMaxCorners];    // complex-interleaved reals, or (FP8) packed
+                              // re/im fp8 byte pairs reinterpreted
+  float scale[kMaxCorners];   // FP8 dequant scale per corner
   long cum[kMaxCorners + 1];  // cumulative work (nelem * ntiles) per corner
   int ncorners;
 };
@@ -153,7 +169,7 @@ struct MultiGeom {
 // loop fully unrolls with folded weight offsets (the round-1 probe lesson:
 // runtime trip counts serialize on a uniform-load wait per iteration,
 // docs/ROADMAP.md item 4; folding recovered 1.5-2x on the dft/head kernels).
-template <typename T, int OTILE, bool CONJT, int NIN = 0>
+template <typename T, int OTILE, bool CONJT, int NIN = 0, bool FP8 = false>
 __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
     const T* __restrict__ x, MultiGeom<T> mg, T* __restrict__ y,
     int B, int I, int O, long Ftot) {
@@ -173,6 +189,7 @@ __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
     work -= mg.cum[c];
     const BoxGeom& g = mg.g[c];
     const T* w = mg.w[c];
+    const float wscale = mg.scale[c];
     long e = work % g.nelem;
     int tile = (int)(work / g.nelem);
 
@@ -185,6 +202,20 @@ __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
     for (int k = 0; k < OTILE; ++k) { accr[k] = T(0); acci[k] = T(0); }
 
     const T* xb = x + 2 * (((long)b * n_in) * Ftot + f);
+    // complex weight load: fp32/fp64 interleaved reals, or (FP8) a packed
+    // re/im byte pair dequantized on the fly (the weight stream is what
+    // this kernel is bandwidth-bound on — fp8 storage cuts it 4x)
+    auto wload = [&](long pidx, T& wr, T& wi) {
+      if constexpr (FP8) {
+        const unsigned short pv =
+            reinterpret_cast<const unsigned short*>(w)[pidx];
+        wr = (T)(fp8_e4m3_to_f32(pv & 0xffu) * wscale);
+        wi = (T)(fp8_e4m3_to_f32(pv >> 8) * wscale);
+      } else {
+        wr = w[2 * pidx];
+        wi = w[2 * pidx + 1];
+      }
+    };
     auto chan = [&](int i) {
       T xr = xb[2 * (long)i * Ftot];
       T xi = xb[2 * (long)i * Ftot + 1];
@@ -192,17 +223,18 @@ __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
 #pragma unroll
         for (int k = 0; k < OTILE; ++k) {
           if (k < olim) {
-            long widx = 2 * ((((long)(o0 + k)) * O + i) * g.nelem + e);
-            cmac_conj(accr[k], acci[k], w[widx], w[widx + 1], xr, xi);
+            T wr, wi;
+            wload((((long)(o0 + k)) * O + i) * g.nelem + e, wr, wi);
+            cmac_conj(accr[k], acci[k], wr, wi, xr, xi);
           }
         }
       } else {
-        const T* wb = w + 2 * (((long)i * O + o0) * g.nelem + e);
 #pragma unroll
         for (int k = 0; k < OTILE; ++k) {
           if (k < olim) {
-            cmac(accr[k], acci[k], xr, xi, wb[2 * (long)k * g.nelem],
-                 wb[2 * (long)k * g.nelem + 1]);
+            T wr, wi;
+            wload(((long)i * O + o0 + k) * g.nelem + e, wr, wi);
+            cmac(accr[k], acci[k], xr, xi, wr, wi);
           }
         }
       }
@@ -383,12 +415,13 @@ void spectral_corner_bwd_x(const at::Tensor& gy, const at::Tensor& w, at::Tensor
 
 // single-launch multi-corner entries ----------------------------------------
 
-template <typename T, bool CONJT>
+template <typename T, bool CONJT, bool FP8 = false>
 static void launch_corners(const at::Tensor& x,
                            const std::vector<at::Tensor>& ws,
                            at::Tensor& y,
                            const std::vector<std::vector<int64_t>>& starts,
-                           int B, int I, int O, long Ftot) {
+                           int B, int I, int O, long Ftot,
+                           const std::vector<double>* scales = nullptr) {
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   constexpr int OT = 8;
   const int n_out = CONJT ? I : O;
@@ -405,34 +438,24 @@ static void launch_corners(const at::Tensor& x,
       int c = mg.ncorners++;
       mg.g[c] = g;
       mg.w[c] = reinterpret_cast<const T*>(ws[idx].data_ptr());
+      mg.scale[c] = scales ? (float)(*scales)[idx] : 1.f;
       mg.cum[c + 1] = mg.cum[c] + g.nelem * ntiles;
       ++idx;
     }
     if (mg.ncorners == 0) continue;
     int grid = grid_for_s(mg.cum[mg.ncorners] * B);
     const int n_in = CONJT ? O : I;
+#define SPC_LAUNCH(NINV)                                                       \
+    hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT, NINV, FP8>),     \
+                       dim3(grid), dim3(kBlock), 0, stream,                    \
+                       reinterpret_cast<const T*>(x.data_ptr()), mg,           \
+                       reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
     // compile-time channel depth for the common widths (flagship 20)
-    if (n_in == 20) {
-      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT, 20>),
-                         dim3(grid), dim3(kBlock), 0, stream,
-                         reinterpret_cast<const T*>(x.data_ptr()), mg,
-                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
-    } else if (n_in == 32) {
-      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT, 32>),
-                         dim3(grid), dim3(kBlock), 0, stream,
-                         reinterpret_cast<const T*>(x.data_ptr()), mg,
-                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
-    } else if (n_in == 8) {
-      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT, 8>),
-                         dim3(grid), dim3(kBlock), 0, stream,
-                         reinterpret_cast<const T*>(x.data_ptr()), mg,
-                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
-    } else {
-      hipLaunchKernelGGL((spectral_corners_kernel<T, OT, CONJT>), dim3(grid),
-                         dim3(kBlock), 0, stream,
-                         reinterpret_cast<const T*>(x.data_ptr()), mg,
-                         reinterpret_cast<T*>(y.data_ptr()), B, I, O, Ftot);
-    }
+    if (n_in == 20) { SPC_LAUNCH(20) }
+    else if (n_in == 32) { SPC_LAUNCH(32) }
+    else if (n_in == 8) { SPC_LAUNCH(8) }
+    else { SPC_LAUNCH(0) }
+#undef SPC_LAUNCH
   }
 }
 
@@ -454,6 +477,55 @@ void spectral_corners_fwd(const at::Tensor& x, std::vector<at::Tensor> ws,
     launch_corners<double, false>(x, ws, y, starts, B, I, O, Ftot);
   }
   DFNO_CHECK_LAUNCH("spectral");
+}
+
+// fp8 spectral-weight variants (BASELINE.json config #5): weights arrive as
+// packed e4m3 re/im byte pairs (uint16 view, same index space as the master
+// complex tensor) with one dequant scale per corner; the spectrum stays
+// complex64.  grad-W is unchanged (straight-through to the fp32 master).
+
+static void check_w16(const at::Tensor& t) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+              t.scalar_type() == at::kUInt16,
+              "spectral fp8: weights must be contiguous uint16 (packed e4m3 pairs)");
+}
+
+void spectral_corners_fwd_fp8(const at::Tensor& x, std::vector<at::Tensor> w16s,
+                              std::vector<double> scales, at::Tensor& y,
+                              std::vector<std::vector<int64_t>> starts) {
+  check_c(x, "x"); check_c(y, "y");
+  TORCH_CHECK(w16s.size() == starts.size() && scales.size() == w16s.size(),
+              "fp8 ws/starts/scales size mismatch");
+  TORCH_CHECK(x.scalar_type() == at::kComplexFloat, "fp8 spectral: c64 only");
+  int B = (int)x.size(0), I = (int)x.size(1), O = (int)y.size(1);
+  long Ftot = 1;
+  for (int d = 2; d < x.dim(); ++d) Ftot *= x.size(d);
+  for (auto& w : w16s) {
+    check_w16(w);
+    TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "fp8 w shape");
+  }
+  if (B == 0 || w16s.empty()) return;
+  launch_corners<float, false, true>(x, w16s, y, starts, B, I, O, Ftot, &scales);
+  DFNO_CHECK_LAUNCH("spectral_fp8");
+}
+
+void spectral_corners_bwd_x_fp8(const at::Tensor& gy, std::vector<at::Tensor> w16s,
+                                std::vector<double> scales, at::Tensor& gx,
+                                std::vector<std::vector<int64_t>> starts) {
+  check_c(gy, "gy"); check_c(gx, "gx");
+  TORCH_CHECK(w16s.size() == starts.size() && scales.size() == w16s.size(),
+              "fp8 ws/starts/scales size mismatch");
+  TORCH_CHECK(gy.scalar_type() == at::kComplexFloat, "fp8 spectral: c64 only");
+  int B = (int)gy.size(0), O = (int)gy.size(1), I = (int)gx.size(1);
+  long Ftot = 1;
+  for (int d = 2; d < gy.dim(); ++d) Ftot *= gy.size(d);
+  for (auto& w : w16s) {
+    check_w16(w);
+    TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "fp8 w shape");
+  }
+  if (B == 0 || w16s.empty()) return;
+  launch_corners<float, true, true>(gy, w16s, gx, starts, B, I, O, Ftot, &scales);
+  DFNO_CHECK_LAUNCH("spectral_fp8");
 }
 
 void spectral_corners_bwd_w(const at::Tensor& x, const at::Tensor& gy,

@@ -42,10 +42,14 @@ __all__ = ["DistributedFNOBlock"]
 class DistributedFNOBlock(nn.Module):
 
     def __init__(self, P_x: Partition, in_shape, modes,
-                 device=torch.device("cpu"), dtype=torch.float32):
+                 device=torch.device("cpu"), dtype=torch.float32,
+                 spectral_fp8: bool = False):
         super().__init__()
 
         self.P_x = P_x
+        # fp8 (e4m3) storage for the corner weights in the contraction
+        # (BASELINE.json config #5); masters stay complex64
+        self.spectral_fp8 = spectral_fp8
         self.in_shape = [int(s) for s in in_shape]
         self.width = self.in_shape[1]
         self.modes = [int(m) for m in modes]
@@ -347,7 +351,8 @@ class DistributedFNOBlock(nn.Module):
 
         if x.numel() > 0:
             x = self._fwd_y(x, saved)
-            y = spectral_conv(x, list(self.weights), self.corner_bounds, self.width)
+            y = spectral_conv(x, list(self.weights), self.corner_bounds,
+                              self.width, fp8=self.spectral_fp8)
             y = self._inv_y(y, saved)
         else:
             y = x
@@ -410,7 +415,8 @@ class DistributedFNOBlock(nn.Module):
 
         # spectral contraction mixes channels: needs the full spectrum
         if x.numel() > 0:
-            y = spectral_conv(x, list(self.weights), self.corner_bounds, self.width)
+            y = spectral_conv(x, list(self.weights), self.corner_bounds,
+                              self.width, fp8=self.spectral_fp8)
         else:
             y = x
 

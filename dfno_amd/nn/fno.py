@@ -28,10 +28,12 @@ class DistributedFNONd(nn.Module):
 
     def __init__(self, P_x: Partition, in_shape, out_timesteps: int, width: int,
                  modes, num_blocks: int = 4,
-                 device=torch.device("cpu"), dtype=torch.float32):
+                 device=torch.device("cpu"), dtype=torch.float32,
+                 spectral_fp8: bool = False):
         super().__init__()
 
         self.P_x = P_x
+        self.spectral_fp8 = spectral_fp8
         self.in_shape = [int(s) for s in in_shape]
         self.out_timesteps = out_timesteps
         self.width = width
@@ -61,7 +63,8 @@ class DistributedFNONd(nn.Module):
 
         self.blocks = nn.ModuleList([
             DistributedFNOBlock(self.P_x, self.block_in_shape, self.modes,
-                                device=device, dtype=dtype)
+                                device=device, dtype=dtype,
+                                spectral_fp8=spectral_fp8)
             for _ in range(num_blocks)
         ])
 
@@ -72,6 +75,10 @@ class DistributedFNONd(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         from ..comm import begin_chain, end_chain
+        if self.spectral_fp8 and torch.is_grad_enabled():
+            # refresh the cached e4m3 weight copies once per training step
+            from ..ops.spectral import bump_quant_epoch
+            bump_quant_epoch()
         begin_chain()   # comm ordering chain scoped to this forward
         try:
             self.dt_comm = 0.0

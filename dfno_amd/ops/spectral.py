@@ -35,13 +35,74 @@ def _corner_slices(bounds: Sequence[Tuple[int, int]]):
     return (slice(None), slice(None)) + tuple(slice(a, b) for a, b in bounds)
 
 
+# ---------------------------------------------------------------------------
+# fp8 (e4m3) spectral-weight quantization (BASELINE.json config #5).
+#
+# The contraction is weight-stream bandwidth-bound; storing the corner
+# weights as OCP e4m3 (one byte per real/imag component, one fp32 scale per
+# corner, amax/448 scaling) cuts that stream 4x.  The fp32/complex64 master
+# weights keep training (straight-through: grad-W is computed against the
+# master, forward/bwd-x use the quantized copy).  Quantized copies are cached
+# and refreshed once per training step (``bump_quant_epoch`` from the model
+# forward under grad mode); eval reuses the cache indefinitely.
+# ---------------------------------------------------------------------------
+
+_E4M3_MAX = 448.0
+_QUANT_EPOCH = [0]
+_FP8_CACHE = {}   # id(w) -> (epoch, w16, scale)
+
+
+def bump_quant_epoch() -> None:
+    _QUANT_EPOCH[0] += 1
+
+
+def _quantize_fp8(w: torch.Tensor):
+    """complex64 [I,O,*box] -> (uint16 packed e4m3 pairs [I,O,*box], scale)."""
+    wr = torch.view_as_real(w.detach()).contiguous()
+    amax = wr.abs().amax().clamp_min(1e-30)
+    scale = float(amax) / _E4M3_MAX
+    w8 = (wr / scale).to(torch.float8_e4m3fn)
+    w16 = w8.view(torch.uint8).view(torch.uint16).squeeze(-1).contiguous()
+    return w16, scale
+
+
+def _fp8_weights(weights):
+    w16s, scales = [], []
+    for w in weights:
+        ent = _FP8_CACHE.get(id(w))
+        if ent is None or ent[0] != _QUANT_EPOCH[0]:
+            w16, s = _quantize_fp8(w)
+            ent = (_QUANT_EPOCH[0], w16, s)
+            _FP8_CACHE[id(w)] = ent
+        w16s.append(ent[1])
+        scales.append(ent[2])
+    return w16s, scales
+
+
+def dequantize_fp8(w16: torch.Tensor, scale: float) -> torch.Tensor:
+    """Reference dequant (tests): uint16 packed pairs -> complex64."""
+    w8 = w16.unsqueeze(-1).view(torch.uint8).view(torch.float8_e4m3fn)
+    return torch.view_as_complex(w8.to(torch.float32) * scale)
+
+
 class _SpectralConvFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, bounds_list, out_channels: int, *weights: torch.Tensor):
+    def forward(ctx, x: torch.Tensor, bounds_list, out_channels: int,
+                fp8: bool, *weights: torch.Tensor):
         B, I = x.shape[0], x.shape[1]
         fdims = list(x.shape[2:])
+        ctx.fp8 = fp8 = bool(fp8 and x.is_cuda and x.dtype == torch.complex64)
         y = torch.zeros((B, out_channels, *fdims), dtype=x.dtype, device=x.device)
-        if x.is_cuda and x.numel() > 0 and x.dtype in (torch.complex64, torch.complex128):
+        if fp8 and x.numel() > 0:
+            ext = _ext.get(required=True)
+            xc = x.contiguous()
+            w16s, scales = _fp8_weights(weights)
+            ext.spectral_corners_fwd_fp8(
+                xc, w16s, scales, y,
+                [[a for a, _ in bounds] for bounds in bounds_list])
+            ctx.fp8_wq = (w16s, scales)
+            x_saved = xc
+        elif x.is_cuda and x.numel() > 0 and x.dtype in (torch.complex64, torch.complex128):
             ext = _ext.get(required=True)
             xc = x.contiguous()
             ext.spectral_corners_fwd(
@@ -71,8 +132,12 @@ class _SpectralConvFn(torch.autograd.Function):
         if gy.is_cuda and gy.numel() > 0 and gy.dtype in (torch.complex64, torch.complex128):
             ext = _ext.get(required=True)
             starts = [[a for a, _ in bounds] for bounds in bounds_list]
-            ext.spectral_corners_bwd_x(
-                gy, [w.contiguous() for w in weights], gx, starts)
+            if getattr(ctx, "fp8", False):
+                w16s, scales = ctx.fp8_wq
+                ext.spectral_corners_bwd_x_fp8(gy, w16s, scales, gx, starts)
+            else:
+                ext.spectral_corners_bwd_x(
+                    gy, [w.contiguous() for w in weights], gx, starts)
             if x.shape[1] <= 32:
                 gws = [torch.empty_like(w) for w in weights]
                 ext.spectral_corners_bwd_w(x.contiguous(), gy, gws, starts)
@@ -89,18 +154,20 @@ class _SpectralConvFn(torch.autograd.Function):
                 sl = _corner_slices(bounds)
                 gx[sl] = torch.einsum("bo...,io...->bi...", gy[sl], w.conj())
                 gws.append(torch.einsum("bo...,bi...->io...", gy[sl], x[sl].conj()))
-        return (gx, None, None, *gws)
+        return (gx, None, None, None, *gws)
 
 
 def spectral_conv(x: torch.Tensor, weights: List[torch.Tensor],
                   bounds_list: List[List[Tuple[int, int]]],
-                  out_channels: int = None) -> torch.Tensor:
+                  out_channels: int = None, fp8: bool = False) -> torch.Tensor:
     """Apply the corner-block spectral contraction.
 
     x: [B, I, *F] complex; weights[c]: [I, O, *box_c] complex;
     bounds_list[c]: per-frequency-dim (start, stop) of corner c within F.
+    fp8: read the weights through their cached e4m3 quantization (the fp32
+    masters keep the gradients — straight-through).
     Returns [B, O, *F] with zeros outside the corner boxes.
     """
     if out_channels is None:
         out_channels = weights[0].shape[1] if weights else x.shape[1]
-    return _SpectralConvFn.apply(x, bounds_list, out_channels, *weights)
+    return _SpectralConvFn.apply(x, bounds_list, out_channels, fp8, *weights)
