@@ -76,9 +76,12 @@ class DistributedFNONd(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         from ..comm import begin_chain, end_chain
         if self.spectral_fp8 and torch.is_grad_enabled():
-            # refresh the cached e4m3 weight copies once per training step
-            from ..ops.spectral import bump_quant_epoch
+            # refresh the cached e4m3 weight copies once per training step;
+            # prequantize ALL blocks' corners together (one scale sync)
+            from ..ops.spectral import bump_quant_epoch, _fp8_weights
             bump_quant_epoch()
+            if x.is_cuda:
+                _fp8_weights([w for blk in self.blocks for w in blk.weights])
         begin_chain()   # comm ordering chain scoped to this forward
         try:
             self.dt_comm = 0.0
