@@ -45,6 +45,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fp8-weight spectral contraction (e4m3 packed pairs + per-corner scale)");
   m.def("spectral_corners_bwd_x_fp8", &spectral_corners_bwd_x_fp8,
         "fp8-weight spectral contraction adjoint wrt x");
+  m.def("fp8_quant_corners", &fp8_quant_corners,
+        "device-side e4m3 requantization of the corner masters");
   m.def("bf16_channel_mix", &bf16_channel_mix,
         "bf16-storage fused channel linear (+bias/res/gelu): returns (y, z)");
   m.def("bf16_channel_mix_bwd_w", &bf16_channel_mix_bwd_w,

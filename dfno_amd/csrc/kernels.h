@@ -108,10 +108,13 @@ void spectral_corners_bwd_w(const at::Tensor& x, const at::Tensor& gy,
                             std::vector<at::Tensor> gws,
                             std::vector<std::vector<int64_t>> starts);
 
-// fp8 (e4m3 packed-pair) spectral-weight variants; scales dequant per corner
+// fp8 (e4m3 packed-pair) spectral-weight variants; device amax per corner
+// (scale = amax/448 computed on device, no host sync)
 void spectral_corners_fwd_fp8(const at::Tensor& x, std::vector<at::Tensor> w16s,
-                              std::vector<double> scales, at::Tensor& y,
+                              std::vector<at::Tensor> amaxes, at::Tensor& y,
                               std::vector<std::vector<int64_t>> starts);
 void spectral_corners_bwd_x_fp8(const at::Tensor& gy, std::vector<at::Tensor> w16s,
-                                std::vector<double> scales, at::Tensor& gx,
+                                std::vector<at::Tensor> amaxes, at::Tensor& gx,
                                 std::vector<std::vector<int64_t>> starts);
+void fp8_quant_corners(std::vector<at::Tensor> ws, std::vector<at::Tensor> w16s,
+                       std::vector<at::Tensor> amaxes);

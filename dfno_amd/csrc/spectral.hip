@@ -19,6 +19,7 @@
 // I*O-element weight column.
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_fp8.h>
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
@@ -160,10 +161,16 @@ struct MultiGeom {
   BoxGeom g[kMaxCorners];
   const T* w[kMaxCorners];    // complex-interleaved reals, or (FP8) packed
                               // re/im fp8 byte pairs reinterpreted
-  float scale[kMaxCorners];   // FP8 dequant scale per corner
+  const float* amax[kMaxCorners];  // FP8: device amax per corner (scale =
+                                   // amax/448, computed on device — no
+                                   // host sync anywhere in the fp8 path)
   long cum[kMaxCorners + 1];  // cumulative work (nelem * ntiles) per corner
   int ncorners;
 };
+
+__device__ __forceinline__ float fp8_scale_from_amax(const float* amax) {
+  return fmaxf(*amax, 1e-30f) * (1.f / 448.f);
+}
 
 // NIN > 0 pins the contraction depth (n_in) at compile time: the channel
 // loop fully unrolls with folded weight offsets (the round-1 probe lesson:
@@ -189,7 +196,7 @@ __global__ __launch_bounds__(kBlock) void spectral_corners_kernel(
     work -= mg.cum[c];
     const BoxGeom& g = mg.g[c];
     const T* w = mg.w[c];
-    const float wscale = mg.scale[c];
+    const float wscale = FP8 ? fp8_scale_from_amax(mg.amax[c]) : 1.f;
     long e = work % g.nelem;
     int tile = (int)(work / g.nelem);
 
@@ -415,13 +422,56 @@ void spectral_corner_bwd_x(const at::Tensor& gy, const at::Tensor& w, at::Tensor
 
 // single-launch multi-corner entries ----------------------------------------
 
+// device-side fp8 quantization: one amax-reduce kernel + one encode kernel
+// over all corners (the eager-torch version cost ~160 launches and a
+// device->host scale sync per step)
+
+__global__ __launch_bounds__(kBlock) void fp8_amax_kernel(
+    const float* const* __restrict__ srcs, const long* __restrict__ numels,
+    float* const* __restrict__ amaxes) {
+  const int c = blockIdx.y;
+  const float* src = srcs[c];
+  const long n = numels[c];
+  float m = 0.f;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    m = fmaxf(m, fabsf(src[i]));
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    m = fmaxf(m, __shfl_down(m, off, 64));
+  __shared__ float wred[4];
+  if ((threadIdx.x & 63) == 0) wred[threadIdx.x >> 6] = m;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    m = fmaxf(fmaxf(wred[0], wred[1]), fmaxf(wred[2], wred[3]));
+    // non-negative floats compare correctly as raw uints
+    atomicMax(reinterpret_cast<unsigned int*>(amaxes[c]),
+              __float_as_uint(m));
+  }
+}
+
+__global__ __launch_bounds__(kBlock) void fp8_encode_kernel(
+    const float* const* __restrict__ srcs, const long* __restrict__ numels,
+    float* const* __restrict__ amaxes, unsigned char* const* __restrict__ dsts) {
+  const int c = blockIdx.y;
+  const float* src = srcs[c];
+  unsigned char* dst = dsts[c];
+  const long n = numels[c];
+  const float inv = 1.f / fp8_scale_from_amax(amaxes[c]);
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    __hip_fp8_e4m3 h(src[i] * inv);
+    dst[i] = h.__x;
+  }
+}
+
 template <typename T, bool CONJT, bool FP8 = false>
 static void launch_corners(const at::Tensor& x,
                            const std::vector<at::Tensor>& ws,
                            at::Tensor& y,
                            const std::vector<std::vector<int64_t>>& starts,
                            int B, int I, int O, long Ftot,
-                           const std::vector<double>* scales = nullptr) {
+                           const std::vector<at::Tensor>* amaxes = nullptr) {
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   constexpr int OT = 8;
   const int n_out = CONJT ? I : O;
@@ -438,7 +488,7 @@ static void launch_corners(const at::Tensor& x,
       int c = mg.ncorners++;
       mg.g[c] = g;
       mg.w[c] = reinterpret_cast<const T*>(ws[idx].data_ptr());
-      mg.scale[c] = scales ? (float)(*scales)[idx] : 1.f;
+      mg.amax[c] = amaxes ? (*amaxes)[idx].data_ptr<float>() : nullptr;
       mg.cum[c + 1] = mg.cum[c] + g.nelem * ntiles;
       ++idx;
     }
@@ -491,11 +541,11 @@ static void check_w16(const at::Tensor& t) {
 }
 
 void spectral_corners_fwd_fp8(const at::Tensor& x, std::vector<at::Tensor> w16s,
-                              std::vector<double> scales, at::Tensor& y,
+                              std::vector<at::Tensor> amaxes, at::Tensor& y,
                               std::vector<std::vector<int64_t>> starts) {
   check_c(x, "x"); check_c(y, "y");
-  TORCH_CHECK(w16s.size() == starts.size() && scales.size() == w16s.size(),
-              "fp8 ws/starts/scales size mismatch");
+  TORCH_CHECK(w16s.size() == starts.size() && amaxes.size() == w16s.size(),
+              "fp8 ws/starts/amaxes size mismatch");
   TORCH_CHECK(x.scalar_type() == at::kComplexFloat, "fp8 spectral: c64 only");
   int B = (int)x.size(0), I = (int)x.size(1), O = (int)y.size(1);
   long Ftot = 1;
@@ -505,16 +555,16 @@ void spectral_corners_fwd_fp8(const at::Tensor& x, std::vector<at::Tensor> w16s,
     TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "fp8 w shape");
   }
   if (B == 0 || w16s.empty()) return;
-  launch_corners<float, false, true>(x, w16s, y, starts, B, I, O, Ftot, &scales);
+  launch_corners<float, false, true>(x, w16s, y, starts, B, I, O, Ftot, &amaxes);
   DFNO_CHECK_LAUNCH("spectral_fp8");
 }
 
 void spectral_corners_bwd_x_fp8(const at::Tensor& gy, std::vector<at::Tensor> w16s,
-                                std::vector<double> scales, at::Tensor& gx,
+                                std::vector<at::Tensor> amaxes, at::Tensor& gx,
                                 std::vector<std::vector<int64_t>> starts) {
   check_c(gy, "gy"); check_c(gx, "gx");
-  TORCH_CHECK(w16s.size() == starts.size() && scales.size() == w16s.size(),
-              "fp8 ws/starts/scales size mismatch");
+  TORCH_CHECK(w16s.size() == starts.size() && amaxes.size() == w16s.size(),
+              "fp8 ws/starts/amaxes size mismatch");
   TORCH_CHECK(gy.scalar_type() == at::kComplexFloat, "fp8 spectral: c64 only");
   int B = (int)gy.size(0), O = (int)gy.size(1), I = (int)gx.size(1);
   long Ftot = 1;
@@ -524,8 +574,53 @@ void spectral_corners_bwd_x_fp8(const at::Tensor& gy, std::vector<at::Tensor> w1
     TORCH_CHECK((int)w.size(0) == I && (int)w.size(1) == O, "fp8 w shape");
   }
   if (B == 0 || w16s.empty()) return;
-  launch_corners<float, true, true>(gy, w16s, gx, starts, B, I, O, Ftot, &scales);
+  launch_corners<float, true, true>(gy, w16s, gx, starts, B, I, O, Ftot, &amaxes);
   DFNO_CHECK_LAUNCH("spectral_fp8");
+}
+
+void fp8_quant_corners(std::vector<at::Tensor> ws, std::vector<at::Tensor> w16s,
+                       std::vector<at::Tensor> amaxes) {
+  // device-only re-quantization of complex64 masters into packed e4m3
+  // pairs: amax-reduce then encode, two launches for ALL corners, no sync
+  const int n = (int)ws.size();
+  TORCH_CHECK((int)w16s.size() == n && (int)amaxes.size() == n,
+              "fp8 quant: list size mismatch");
+  if (n == 0) return;
+  TORCH_CHECK(n <= 64, "fp8 quant: too many corners per call");
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+
+  std::vector<int64_t> host(4 * n);
+  long max_numel = 1;
+  for (int i = 0; i < n; ++i) {
+    TORCH_CHECK(ws[i].is_cuda() && ws[i].is_contiguous() &&
+                ws[i].scalar_type() == at::kComplexFloat, "fp8 quant: c64 masters");
+    check_w16(w16s[i]);
+    TORCH_CHECK(amaxes[i].is_cuda() && amaxes[i].scalar_type() == at::kFloat &&
+                amaxes[i].numel() == 1, "fp8 quant: amax slots");
+    long numel = ws[i].numel() * 2;   // real words
+    TORCH_CHECK(w16s[i].numel() == ws[i].numel(), "fp8 quant: shape mismatch");
+    host[i] = (int64_t)ws[i].data_ptr();
+    host[n + i] = numel;
+    host[2 * n + i] = (int64_t)amaxes[i].data_ptr();
+    host[3 * n + i] = (int64_t)w16s[i].data_ptr();
+    max_numel = std::max(max_numel, numel);
+    amaxes[i].zero_();
+  }
+  auto opts = at::TensorOptions().dtype(at::kLong);
+  auto dev_tab = at::from_blob(host.data(), {4 * (long)n}, opts)
+                     .to(ws[0].device(), /*non_blocking=*/false);
+  auto tab = dev_tab.data_ptr<int64_t>();
+  auto srcs = reinterpret_cast<const float* const*>(tab);
+  auto numels = reinterpret_cast<const long*>(tab + n);
+  auto amx = reinterpret_cast<float* const*>(tab + 2 * n);
+  auto dsts = reinterpret_cast<unsigned char* const*>(tab + 3 * n);
+
+  long gx = std::min((max_numel + kBlock - 1) / kBlock, 512L);
+  hipLaunchKernelGGL(fp8_amax_kernel, dim3((int)gx, n), dim3(kBlock), 0,
+                     stream, srcs, numels, amx);
+  hipLaunchKernelGGL(fp8_encode_kernel, dim3((int)gx, n), dim3(kBlock), 0,
+                     stream, srcs, numels, amx, dsts);
+  DFNO_CHECK_LAUNCH("fp8_quant");
 }
 
 void spectral_corners_bwd_w(const at::Tensor& x, const at::Tensor& gy,
