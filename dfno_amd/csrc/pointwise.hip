@@ -946,6 +946,98 @@ __global__ __launch_bounds__(kBlock) void gw_outer_glds3_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// fp32-MFMA grad-W for big-O shapes (the projection lift's gW3 [128, 20]).
+// gW[o,i] = sum_s gz[o,s] x[i,s] is GEMM-shaped with K = S ~ 10^7; the VALU
+// glds3 kernel sits at the round-1 "0.9 TB/s per resident block" ceiling
+// (2.8 ms at this shape).  v_mfma_f32_16x16x4_f32 runs the identical-
+// numerics contraction at the f32 matrix rate with one f32 operand register
+// per lane (cdna_hip_programming.md section 3: an untuned f32-MFMA tile
+// outruns a VALU f32 GEMM 122 vs 52 TF) — LDS-staged [rows][TS] tiles,
+// wave-per-(16x16)-output-tile, fp32 accumulate, atomic flush.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+typedef float f32x4_pw __attribute__((ext_vector_type(4)));
+
+template <int NPAIR>
+__global__ __launch_bounds__(kBlock) void gw_mfma_f32_kernel(
+    const float* __restrict__ gz, const float* __restrict__ x,
+    float* __restrict__ gW, int B, int O, int I, long S) {
+  constexpr int TS = 128;          // s per tile (32 MFMA K-steps)
+  constexpr int LD = TS + 4;       // float4-aligned row pad
+  constexpr int OSL = 64;          // o-rows per blockIdx.y slab
+  extern __shared__ __align__(16) char smem_raw[];
+  float* xs = reinterpret_cast<float*>(smem_raw);    // [I<=32][LD]
+  float* gs = xs + (size_t)32 * LD;                  // [O_sl][LD]
+
+  const int o0 = (int)blockIdx.y * OSL;
+  const int O_sl = min(OSL, O - o0);
+  const int iT = (I + 15) / 16;
+  const int npairs = ((O_sl + 15) / 16) * iT;
+
+  const int lane = (int)(threadIdx.x & 63);
+  const int wave = (int)(threadIdx.x >> 6);
+  const int l16 = lane & 15;
+  const int kg = lane >> 4;        // k-group within the MFMA K=4
+
+  f32x4_pw acc[NPAIR];
+#pragma unroll
+  for (int p = 0; p < NPAIR; ++p) acc[p] = f32x4_pw{0.f, 0.f, 0.f, 0.f};
+
+  const long ntiles = (S / TS) * (long)B;
+  for (long t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const int b = (int)(t / (S / TS));
+    const long s0 = (t % (S / TS)) * TS;
+    for (int r = threadIdx.x; r < (I + O_sl) * (TS / 4); r += kBlock) {
+      const int row = r / (TS / 4);
+      const int col = (r - row * (TS / 4)) * 4;
+      if (row < I) {
+        *reinterpret_cast<float4*>(&xs[row * LD + col]) =
+            *reinterpret_cast<const float4*>(x + ((long)b * I + row) * S + s0 + col);
+      } else {
+        const int ro = row - I;
+        *reinterpret_cast<float4*>(&gs[ro * LD + col]) =
+            *reinterpret_cast<const float4*>(gz + ((long)b * O + o0 + ro) * S + s0 + col);
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int pp = 0; pp < NPAIR; ++pp) {
+      const int p = wave + 4 * pp;
+      if (p < npairs) {
+        const int mt = p / iT, nt = p - mt * iT;
+        const float* gr = gs + (mt * 16 + l16) * LD;         // A row
+        const float* xr = xs + (nt * 16 + l16) * LD;         // B col's row
+        const bool av = (mt * 16 + l16) < O_sl;
+        const bool bv = (nt * 16 + l16) < I;
+#pragma unroll
+        for (int k = 0; k < TS; k += 4) {
+          const float a = av ? gr[k + kg] : 0.f;
+          const float bb = bv ? xr[k + kg] : 0.f;
+          acc[pp] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, acc[pp], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int pp = 0; pp < NPAIR; ++pp) {
+    const int p = wave + 4 * pp;
+    if (p >= npairs) continue;
+    const int mt = p / iT, nt = p - mt * iT;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int o = o0 + mt * 16 + kg * 4 + r;   // D row
+      const int i = nt * 16 + l16;               // D col
+      if (o < O && (mt * 16 + kg * 4 + r) < O_sl && i < I)
+        atomicAdd(&gW[(size_t)o * I + i], acc[pp][r]);
+    }
+  }
+}
+
 }  // namespace
 
 std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
@@ -1023,6 +1115,28 @@ std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor
           x.data_ptr<scalar_t>(), gW.data_ptr<scalar_t>(),                      \
           want_bias ? gb.data_ptr<scalar_t>() : nullptr,                        \
           B, O, I, S, n_schunk, want_bias);
+  static const bool no_mfma_gw = []() {
+    const char* e = getenv("DFNO_GW_NO_MFMA");  // A/B knob
+    return e && e[0] == '1';
+  }();
+  // big-O no-bias shapes (the projection lift's gW3 [128, 20]) take the
+  // f32-MFMA tile kernel; identical numerics (f32-in MFMA is a bitwise
+  // fmaf chain, cdna_hip_programming.md section 3)
+  if (is_f32 && vec && !no_mfma_gw && !want_bias && (long)O * I >= 1024 &&
+      I <= 32 && S % 128 == 0) {
+    long ntiles = (S / 128) * (long)B;
+    int nslab = (O + 63) / 64;
+    int gx_ = (int)std::min(ntiles, (long)(256 * 3 / std::max(nslab, 1)));
+    size_t smem = sizeof(float) * (size_t)(32 + std::min(O, 64)) * 132;
+    hipLaunchKernelGGL((gw_mfma_f32_kernel<2>), dim3(gx_, nslab),
+                       dim3(kBlock), smem, stream, gz.data_ptr<float>(),
+                       x.data_ptr<float>(), gW.data_ptr<float>(), B, O,
+                       (int)I, S);
+    hipError_t merr = hipGetLastError();
+    TORCH_CHECK(merr == hipSuccess, "gw_mfma launch failed: ",
+                hipGetErrorString(merr));
+    return {gW, gb};
+  }
   AT_DISPATCH_FLOATING_TYPES(gz.scalar_type(), "channel_mix_bwd_w", [&] {
     if (ow5) { GW_LDS(20, 5, 256, 3) }
     else if (vec && !no_glds && OW == 4) { GW_LDS(8, 4, 256, 3) }

@@ -63,6 +63,28 @@ def test_channel_mix_fwd_t(ext, dtype, B, I, O, S):
     assert torch.allclose(gx, gx_ref, **tol(dtype)), f"max {(gx - gx_ref).abs().max()}"
 
 
+@pytest.mark.parametrize("dtype", [torch.float32, torch.float64])
+@pytest.mark.parametrize("B,I,O,S,bias", [
+    (1, 20, 20, 4096, True),      # width-20 (glds3 OW=5 path)
+    (1, 20, 128, 7936, False),    # proj lift gW3: the f32-MFMA tile path
+    (2, 20, 128, 4096, False),    # batched MFMA path
+    (1, 16, 64, 4096, False),     # MFMA path, ragged o-slab
+    (1, 20, 128, 1000, False),    # S % 128 != 0 -> glds3 fallback
+])
+def test_channel_mix_bwd_w(ext, dtype, B, I, O, S, bias):
+    torch.manual_seed(3)
+    gz = torch.randn(B, O, S, device="cuda", dtype=dtype)
+    x = torch.randn(B, I, S, device="cuda", dtype=dtype)
+    gW, gb = ext.channel_mix_bwd_w(gz, x, bias)
+    refW = torch.einsum("bos,bis->oi", gz, x)
+    t = dict(rtol=1e-4, atol=float(refW.abs().max()) * 1e-5) \
+        if dtype == torch.float32 else dict(rtol=1e-11, atol=1e-9)
+    assert torch.allclose(gW, refW, **t), f"max {(gW - refW).abs().max()}"
+    if bias:
+        refb = gz.sum(dim=(0, 2))
+        assert torch.allclose(gb, refb, **t)
+
+
 # ---------------------------------------------------------------------------
 # gelu / add+gelu
 # ---------------------------------------------------------------------------
