@@ -946,6 +946,8 @@ __global__ __launch_bounds__(kBlock) void gw_outer_glds3_kernel(
   }
 }
 
+}  // namespace
+
 // ---------------------------------------------------------------------------
 // fp32-MFMA grad-W for big-O shapes (the projection lift's gW3 [128, 20]).
 // gW[o,i] = sum_s gz[o,s] x[i,s] is GEMM-shaped with K = S ~ 10^7; the VALU
