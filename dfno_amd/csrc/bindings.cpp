@@ -31,6 +31,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "adjoint of dft_rfft_trunc with fused accumulate addend");
   m.def("dft_pad_irfft", &dft_pad_irfft, "kept modes -> real inverse (last dim)");
   m.def("dft_pad_irfft_adj", &dft_pad_irfft_adj, "adjoint of dft_pad_irfft");
+  m.def("dft_pad_irfft_bf16", &dft_pad_irfft_bf16,
+        "pad_irfft with bf16 output (bf16 compute config)");
+  m.def("dft_rfft_trunc_adj_bf16", &dft_rfft_trunc_adj_bf16,
+        "rfft_trunc adjoint with bf16 output");
   m.def("spectral_corner_fwd", &spectral_corner_fwd,
         "corner-block complex spectral contraction (accumulate into y box)");
   m.def("spectral_corner_bwd_x", &spectral_corner_bwd_x,

@@ -20,6 +20,7 @@
 // per kept mode in registers (unrolled + predicated, no scratch).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <mutex>
 #include <unordered_map>
@@ -55,6 +56,18 @@ __device__ __forceinline__ void cmul_acc(T& cr, T& ci, T sr, T si) {
   T nr = cr * sr - ci * si;
   T ni = cr * si + ci * sr;
   cr = nr; ci = ni;
+}
+
+// bf16 <-> f32 helpers for the bf16-IO transform variants (the bf16 model's
+// real-side activations enter/leave the spectral path in bf16 storage;
+// arithmetic stays fp32 — see csrc/bf16.hip for the rationale)
+__device__ __forceinline__ float dft_b2f(unsigned short h) {
+  return __uint_as_float(((unsigned int)h) << 16);
+}
+
+__device__ __forceinline__ unsigned int dft_f2b2(float lo, float hi) {
+  __hip_bfloat162 h2 = __float22bfloat162_rn(float2{lo, hi});
+  return *reinterpret_cast<unsigned int*>(&h2);
 }
 
 // kept-mode k value for index ki
@@ -722,12 +735,14 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
 // register recurrences: the chain update was 4 of every 6 VALU ops (PMC:
 // VALU-bound at ~85% issue), and (j, k) are lane-uniform so the table reads
 // compile to scalar loads that the k-unrolled fma stream hides entirely.
-template <typename T, int MCAP, int LB = kBlock>
+// TI != T stages a bf16 input (convert at the LDS write; compute stays T)
+template <typename T, int MCAP, int LB = kBlock, typename TI = T>
 __global__ __launch_bounds__(LB) void dft_r2c_last_kernel(
-    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
+    const TI* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long lines, int N, int m, T scale, bool factors) {
   // LB = block size = lines per tile; 128/64 for N = 128/256 keep the
   // [LB * N] LDS tile within the 160 KiB budget (2 blocks/CU at 64 KiB).
+  constexpr bool kBf16In = std::is_same<TI, unsigned short>::value;
   extern __shared__ __align__(16) char smem_raw[];
   T* tile = reinterpret_cast<T*>(smem_raw);   // [LB * N]
 
@@ -736,20 +751,38 @@ __global__ __launch_bounds__(LB) void dft_r2c_last_kernel(
     long l0 = tb * LB;
     int nl = (int)min((long)LB, lines - l0);
     __syncthreads();
-    if constexpr (std::is_same<T, float>::value) {
+    if constexpr (kBf16In) {
+      const long base = l0 * N;
+      if ((nl * N) % 8 == 0 && (base % 8 == 0) &&
+          ((reinterpret_cast<uintptr_t>(in) & 15) == 0)) {
+        for (int idx = threadIdx.x * 8; idx < nl * N; idx += LB * 8) {
+          const uint4 raw = *reinterpret_cast<const uint4*>(in + base + idx);
+          const unsigned int w[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+          for (int k = 0; k < 4; ++k) {
+            tile[idx + 2 * k] = (T)dft_b2f((unsigned short)(w[k] & 0xffffu));
+            tile[idx + 2 * k + 1] = (T)dft_b2f((unsigned short)(w[k] >> 16));
+          }
+        }
+      } else {
+        for (int idx = threadIdx.x; idx < nl * N; idx += LB)
+          tile[idx] = (T)dft_b2f(in[base + idx]);
+      }
+    } else if constexpr (std::is_same<T, float>::value) {
       const long base = l0 * N;
       if ((nl * N) % 4 == 0 && (base % 4 == 0) &&
           ((reinterpret_cast<uintptr_t>(in) & 15) == 0)) {
         for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4)
           *reinterpret_cast<float4*>(&tile[idx]) =
-              *reinterpret_cast<const float4*>(in + base + idx);
+              *reinterpret_cast<const float4*>(
+                  reinterpret_cast<const float*>(in) + base + idx);
       } else {
         for (int idx = threadIdx.x; idx < nl * N; idx += LB)
-          tile[idx] = in[base + idx];
+          tile[idx] = reinterpret_cast<const float*>(in)[base + idx];
       }
     } else {
       for (int idx = threadIdx.x; idx < nl * N; idx += LB)
-        tile[idx] = in[l0 * N + idx];
+        tile[idx] = (T)in[l0 * N + idx];
     }
     __syncthreads();
     {
@@ -801,9 +834,10 @@ __global__ __launch_bounds__(LB) void dft_r2c_last_kernel(
 // acc != nullptr fuses a same-shape addend into the writeback (the block
 // residual's input-gradient accumulate: out = idft + acc in one pass instead
 // of a separate aten add over the full activation — docs/ROADMAP.md item 4).
-template <typename T, int MCAP, int NT = 0, int LB = kBlock>
+// TO != T emits a bf16 output (convert at the writeback; compute stays T).
+template <typename T, int MCAP, int NT = 0, int LB = kBlock, typename TO = T>
 __global__ __launch_bounds__(LB) void dft_c2r_last_kernel(
-    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
+    const T* __restrict__ in, TO* __restrict__ out, const T* __restrict__ tw,
     long lines, int N_, int m_, T scale, bool factors,
     const T* __restrict__ acc = nullptr) {
   // NT > 0 pins N and m (== MCAP) at compile time: full unroll + folded
@@ -858,8 +892,28 @@ __global__ __launch_bounds__(LB) void dft_c2r_last_kernel(
       }
     }
     __syncthreads();
-    if constexpr (std::is_same<T, float>::value) {
+    if constexpr (std::is_same<TO, unsigned short>::value) {
+      // bf16 output (no accumulate operand on this path)
       const long base = l0 * N;
+      if ((nl * N) % 8 == 0 && (base % 8 == 0) &&
+          ((reinterpret_cast<uintptr_t>(out) & 15) == 0)) {
+        for (int idx = threadIdx.x * 8; idx < nl * N; idx += LB * 8) {
+          uint4 raw;
+          raw.x = dft_f2b2((float)tile[idx], (float)tile[idx + 1]);
+          raw.y = dft_f2b2((float)tile[idx + 2], (float)tile[idx + 3]);
+          raw.z = dft_f2b2((float)tile[idx + 4], (float)tile[idx + 5]);
+          raw.w = dft_f2b2((float)tile[idx + 6], (float)tile[idx + 7]);
+          *reinterpret_cast<uint4*>(out + base + idx) = raw;
+        }
+      } else {
+        for (int idx = threadIdx.x; idx < nl * N; idx += LB) {
+          __hip_bfloat16 h = __float2bfloat16((float)tile[idx]);
+          out[base + idx] = *reinterpret_cast<unsigned short*>(&h);
+        }
+      }
+    } else if constexpr (std::is_same<T, float>::value) {
+      const long base = l0 * N;
+      float* outf = reinterpret_cast<float*>(out);
       if ((nl * N) % 4 == 0 && (base % 4 == 0) &&
           ((reinterpret_cast<uintptr_t>(out) & 15) == 0) &&
           (acc == nullptr || (reinterpret_cast<uintptr_t>(acc) & 15) == 0)) {
@@ -867,17 +921,17 @@ __global__ __launch_bounds__(LB) void dft_c2r_last_kernel(
           for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4) {
             const float4 t4 = *reinterpret_cast<const float4*>(&tile[idx]);
             const float4 a4 = *reinterpret_cast<const float4*>(acc + base + idx);
-            *reinterpret_cast<float4*>(out + base + idx) =
+            *reinterpret_cast<float4*>(outf + base + idx) =
                 make_float4(t4.x + a4.x, t4.y + a4.y, t4.z + a4.z, t4.w + a4.w);
           }
         } else {
           for (int idx = threadIdx.x * 4; idx < nl * N; idx += LB * 4)
-            *reinterpret_cast<float4*>(out + base + idx) =
+            *reinterpret_cast<float4*>(outf + base + idx) =
                 *reinterpret_cast<const float4*>(&tile[idx]);
         }
       } else {
         for (int idx = threadIdx.x; idx < nl * N; idx += LB)
-          out[base + idx] = tile[idx] + (acc ? acc[base + idx] : 0.f);
+          outf[base + idx] = tile[idx] + (acc ? acc[base + idx] : 0.f);
       }
     } else {
       for (int idx = threadIdx.x; idx < nl * N; idx += LB)
@@ -1061,17 +1115,44 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
   TORCH_CHECK(m <= 32, "dft_r2c: m > 32 unsupported natively");
   long lines = x.numel() / std::max(N, 1);
 
+  const bool bf16_in = x.scalar_type() == at::kBFloat16;
   auto sizes = x.sizes().vec();
   sizes[dim] = m;
   auto out = at::empty(sizes, x.options().dtype(
-      x.scalar_type() == at::kFloat ? at::kComplexFloat : at::kComplexDouble));
+      x.scalar_type() == at::kDouble ? at::kComplexDouble : at::kComplexFloat));
   if (x.numel() == 0) return out;
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   long ntiles = (lines + kBlock - 1) / kBlock;
   int grid = (int)std::min(ntiles, 4096L);
   auto tw = twiddle_table(N, (int)m, /*analysis=*/true,
-                          x.options().dtype(x.scalar_type()));
+                          x.options().dtype(bf16_in ? at::kFloat
+                                                    : x.scalar_type()));
+  if (bf16_in) {
+    // bf16-IO variant: bf16 staged + converted at the LDS write, fp32
+    // compute, complex64 out — the bf16 model's real-side transforms read
+    // half the bytes and skip the boundary-cast pass entirely
+    auto inp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+    auto op = reinterpret_cast<float*>(out.data_ptr());
+#define R2C_BF(MC, LBV)                                                        \
+    { long nt2 = (lines + LBV - 1) / LBV;                                      \
+      int grid2 = (int)std::min(nt2, 2048L);                                   \
+      size_t smem2 = sizeof(float) * (size_t)LBV * N;                          \
+      hipLaunchKernelGGL((dft_r2c_last_kernel<float, MC, LBV, unsigned short>),\
+                         dim3(grid2), dim3(LBV), smem2, stream, inp, op,       \
+                         tw.data_ptr<float>(), lines, N, (int)m,               \
+                         (float)scale, factors); }
+#define R2C_BF_M(LBV)                                                          \
+    if (m <= 8) { R2C_BF(8, LBV) } else if (m <= 16) { R2C_BF(16, LBV) }       \
+    else if (m <= 24) { R2C_BF(24, LBV) } else { R2C_BF(32, LBV) }
+    if (N <= 64) { R2C_BF_M(256) }
+    else if (N <= 128) { R2C_BF_M(128) }
+    else { R2C_BF_M(64) }
+#undef R2C_BF_M
+#undef R2C_BF
+    DFNO_CHECK_LAUNCH("dft_r2c_bf16");
+    return out;
+  }
 #define R2CG(MC, NGV, NTV)                                                     \
       hipLaunchKernelGGL((dft_r2c_glds_kernel<scalar_t, MC, NGV, NTV>),        \
                          dim3(grid), dim3(kBlock), smem2, stream, inp, op,     \
@@ -1130,7 +1211,8 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
 
 static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
                                double scale, bool factors,
-                               const at::Tensor& accum = at::Tensor()) {
+                               const at::Tensor& accum = at::Tensor(),
+                               bool out_bf16 = false) {
   TORCH_CHECK(y.is_cuda() && y.is_contiguous(), "dft_c2r: contiguous GPU input");
   TORCH_CHECK(dim == y.dim() - 1, "dft_c2r: last-dim only");
   TORCH_CHECK(y.scalar_type() == at::kComplexFloat || y.scalar_type() == at::kComplexDouble,
@@ -1143,15 +1225,49 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
 
   auto sizes = y.sizes().vec();
   sizes[dim] = N;
+  if (out_bf16) {
+    TORCH_CHECK(y.scalar_type() == at::kComplexFloat && !accum.defined(),
+                "dft_c2r: bf16 output needs c64 input, no accumulate");
+  }
   auto out = at::empty(sizes, y.options().dtype(
-      y.scalar_type() == at::kComplexFloat ? at::kFloat : at::kDouble));
+      out_bf16 ? at::kBFloat16
+               : (y.scalar_type() == at::kComplexFloat ? at::kFloat
+                                                       : at::kDouble)));
   if (y.numel() == 0) return out;
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   long ntiles = (lines + kBlock - 1) / kBlock;
   int grid = (int)std::min(ntiles, 4096L);
   auto tw = twiddle_table(N, m, /*analysis=*/false,
-                          out.options().dtype(out.scalar_type()));
+                          out.options().dtype(out_bf16 ? at::kFloat
+                                                       : out.scalar_type()));
+  if (out_bf16) {
+    // bf16-output writeback (fp32 compute/LDS tile), flagship NT folding kept
+    auto inp = reinterpret_cast<const float*>(y.data_ptr());
+    auto op = reinterpret_cast<unsigned short*>(out.data_ptr());
+#define C2RB(MC, NTV, LBV)                                                     \
+    { long nt2 = (lines + LBV - 1) / LBV;                                      \
+      int grid2 = (int)std::min(nt2, LBV == kBlock ? 4096L : 2048L);           \
+      size_t smem2 = sizeof(float) * (size_t)LBV * N;                          \
+      hipLaunchKernelGGL(                                                      \
+          (dft_c2r_last_kernel<float, MC, NTV, LBV, unsigned short>),          \
+          dim3(grid2), dim3(LBV), smem2, stream, inp, op,                      \
+          tw.data_ptr<float>(), lines, N, (int)m, (float)scale, factors,       \
+          nullptr); }
+#define C2RB_M(NTV, LBV)                                                       \
+    if (m <= 8) { C2RB(8, NTV, LBV) } else if (m <= 16) { C2RB(16, NTV, LBV) } \
+    else if (m <= 24) { C2RB(24, NTV, LBV) } else { C2RB(32, NTV, LBV) }
+    const bool mexact = (m == 8 || m == 16 || m == 24 || m == 32);
+    if (mexact && N == 30) { C2RB_M(30, kBlock) }
+    else if (mexact && N == 64) { C2RB_M(64, kBlock) }
+    else if (N <= 64) { C2RB_M(0, kBlock) }
+    else if (N <= 128) { C2RB_M(0, 128) }
+    else { C2RB_M(0, 64) }
+#undef C2RB_M
+#undef C2RB
+    DFNO_CHECK_LAUNCH("dft_c2r_bf16");
+    return out;
+  }
 #define C2RG(MC, NTV)                                                          \
       hipLaunchKernelGGL((dft_c2r_last_kernel<scalar_t, MC, NTV>),             \
                          dim3(grid), dim3(kBlock), smem, stream, inp, op,      \
@@ -1227,6 +1343,21 @@ at::Tensor dft_rfft_trunc_adj_acc(const at::Tensor& gy, int64_t dim, int64_t n,
 at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m) {
   TORCH_CHECK(y.size(dim) == m, "dft_pad_irfft: m mismatch");
   return dft_c2r_impl(y, dim, n_out, 1.0 / (double)n_out, /*factors=*/true);
+}
+
+at::Tensor dft_pad_irfft_bf16(const at::Tensor& y, int64_t dim, int64_t n_out,
+                              int64_t m) {
+  // bf16-output variant for the bf16 compute config (no boundary cast)
+  TORCH_CHECK(y.size(dim) == m, "dft_pad_irfft: m mismatch");
+  return dft_c2r_impl(y, dim, n_out, 1.0 / (double)n_out, /*factors=*/true,
+                      at::Tensor(), /*out_bf16=*/true);
+}
+
+at::Tensor dft_rfft_trunc_adj_bf16(const at::Tensor& gy, int64_t dim,
+                                   int64_t n) {
+  // bf16-output adjoint (the bf16 model's rfft input gradient)
+  return dft_c2r_impl(gy, dim, n, 1.0, /*factors=*/false, at::Tensor(),
+                      /*out_bf16=*/true);
 }
 
 at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m) {

@@ -72,6 +72,10 @@ at::Tensor dft_rfft_trunc_adj_acc(const at::Tensor& gy, int64_t dim, int64_t n,
                                   const at::Tensor& accum);
 at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m);
 at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m);
+// bf16-IO variants (bf16 real-side storage, fp32 compute, c64 spectrum):
+at::Tensor dft_pad_irfft_bf16(const at::Tensor& y, int64_t dim, int64_t n_out,
+                              int64_t m);
+at::Tensor dft_rfft_trunc_adj_bf16(const at::Tensor& gy, int64_t dim, int64_t n);
 
 // bf16-storage pointwise kernels (bf16.hip; fp32 arithmetic):
 std::vector<at::Tensor> bf16_channel_mix(const at::Tensor& x, const at::Tensor& W,

@@ -209,9 +209,14 @@ class DistributedFNOBlock(nn.Module):
             # rank, including ones whose local block is empty)
             return torch.empty(0, dtype=self.dtype_complex, device=x.device,
                                requires_grad=torch.is_grad_enabled())
-        if x.dtype == torch.bfloat16:
-            x = x.float()   # spectral path runs fp32/c64 (see dtype_complex)
         outermost = self.dim_m[-1]
+        if x.dtype == torch.bfloat16:
+            from ..ops.fft import rfft_bf16_native_ok
+            if not rfft_bf16_native_ok(x, outermost,
+                                       self.restrict_prefixes[outermost]):
+                # no bf16-IO kernel for this shape: boundary-cast to fp32
+                # (the spectral path runs fp32/c64 either way)
+                x = x.float()
         saved[outermost] = x.shape[outermost] // 2 + 1
         x = rfft_trunc(x, outermost, self.restrict_prefixes[outermost])
         for dim in reversed(self.dim_m[:-1]):
@@ -249,8 +254,10 @@ class DistributedFNOBlock(nn.Module):
                          self.restrict_prefixes[dim],
                          self.restrict_suffixes.get(dim, 0))
         y = pad_irfft(y, outermost, saved[outermost],
-                      self.in_shape[-1], self.restrict_prefixes[outermost])
-        # downcast BEFORE R4 so the real-activation exchange moves bf16
+                      self.in_shape[-1], self.restrict_prefixes[outermost],
+                      out_dtype=self.dtype)
+        # (bf16 models: the c2r kernel emits bf16 directly; the exchange in
+        # R4 then moves bf16.  Anything else lands here as a no-op.)
         return y.to(self.dtype) if y.dtype != self.dtype else y
 
     # ---- pipelined-chunk policy ------------------------------------------

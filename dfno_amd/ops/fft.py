@@ -99,12 +99,16 @@ class _RfftTruncFn(torch.autograd.Function):
     def forward(ctx, x, dim, m):
         ext = _ext.get(required=True)
         ctx.dim, ctx.m, ctx.n = dim, m, x.shape[dim]
+        ctx.bf16 = x.dtype == torch.bfloat16
         return ext.dft_rfft_trunc(x.contiguous(), dim, m)
 
     @staticmethod
     def backward(ctx, gy):
         ext = _ext.get(required=True)
-        gx = ext.dft_rfft_trunc_adj(gy.contiguous(), ctx.dim, ctx.n)
+        if ctx.bf16:
+            gx = ext.dft_rfft_trunc_adj_bf16(gy.contiguous(), ctx.dim, ctx.n)
+        else:
+            gx = ext.dft_rfft_trunc_adj(gy.contiguous(), ctx.dim, ctx.n)
         return gx, None, None
 
 
@@ -143,16 +147,20 @@ class _PadIfftFn(torch.autograd.Function):
 
 class _PadIrfftFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, y, dim, n_half, n_out, m):
+    def forward(ctx, y, dim, n_half, n_out, m, out_bf16):
         ext = _ext.get(required=True)
         ctx.dim, ctx.n_half, ctx.n_out, ctx.m = dim, n_half, n_out, m
+        if out_bf16:
+            return ext.dft_pad_irfft_bf16(y.contiguous(), dim, n_out, m)
         return ext.dft_pad_irfft(y.contiguous(), dim, n_out, m)
 
     @staticmethod
     def backward(ctx, gx):
+        # gx may be bf16 (bf16 output config): the r2c adjoint stages and
+        # converts it directly (no boundary cast)
         ext = _ext.get(required=True)
         gy = ext.dft_pad_irfft_adj(gx.contiguous(), ctx.dim, ctx.m)
-        return gy, None, None, None, None
+        return gy, None, None, None, None, None
 
 
 # ---------------------------------------------------------------------------
@@ -239,15 +247,26 @@ def stash_fusable(x, dim, m) -> bool:
 # public API
 # ---------------------------------------------------------------------------
 
+def rfft_bf16_native_ok(x, dim, m) -> bool:
+    """bf16-IO r2c coverage (last dim, staged+converted at the LDS write)."""
+    d = dim % x.dim()
+    m = min(m, x.shape[d] // 2 + 1)
+    return (d == x.dim() - 1 and x.is_cuda and x.dtype == torch.bfloat16
+            and x.shape[d] <= _MAX_N and m <= 32)
+
+
 def rfft_trunc(x, dim, m):
     d = dim % x.dim()
     # Clamp to the half-spectrum size: modes > n//2+1 keep the whole spectrum
     # (matches the reference's graceful [:m] slice, /root/reference/dfno/dfno.py:195).
     m = min(m, x.shape[d] // 2 + 1)
-    if d == x.dim() - 1 and _native_ok(x, x.shape[d], m, "rfft_trunc"):
+    if d == x.dim() - 1 and (_native_ok(x, x.shape[d], m, "rfft_trunc")
+                             or rfft_bf16_native_ok(x, dim, m)):
         return _RfftTruncFn.apply(x, d, m)
     if x.is_cuda and d != x.dim() - 1:
         note_fallback("rfft_trunc", f"non-last transform dim {d}")
+    if x.dtype == torch.bfloat16:
+        x = x.float()   # torch.fft has no bf16 path
     return _t_rfft_trunc(x, dim, m)
 
 
@@ -270,11 +289,16 @@ def pad_ifft(y, dim, n, m_lo, m_hi):
     return _t_pad_ifft(y, dim, n, m_lo, m_hi)
 
 
-def pad_irfft(y, dim, n_half, n_out, m):
+def pad_irfft(y, dim, n_half, n_out, m, out_dtype=None):
     d = dim % y.dim()
     m = min(m, n_half)
     if d == y.dim() - 1 and _native_ok(y, n_out, m, "pad_irfft"):
-        return _PadIrfftFn.apply(y, d, n_half, n_out, m)
+        out_bf16 = (out_dtype == torch.bfloat16
+                    and y.dtype == torch.complex64)
+        return _PadIrfftFn.apply(y, d, n_half, n_out, m, out_bf16)
     if y.is_cuda and d != y.dim() - 1:
         note_fallback("pad_irfft", f"non-last transform dim {d}")
-    return _t_pad_irfft(y, dim, n_half, n_out, m)
+    out = _t_pad_irfft(y, dim, n_half, n_out, m)
+    if out_dtype is not None and out.dtype != out_dtype:
+        out = out.to(out_dtype)
+    return out

@@ -123,6 +123,47 @@ def test_bf16_gelu_and_add_gelu():
     assert torch.allclose(gz.float(), ref, rtol=BT, atol=BT)
 
 
+@pytest.mark.parametrize("shape,m", [
+    ((1, 4, 9, 9, 30), 8),     # flagship t-dim
+    ((2, 3, 6, 64), 12),
+])
+def test_bf16_io_rfft_and_irfft(shape, m):
+    """bf16-IO transform kernels: bf16 in -> c64 spectrum -> bf16 out,
+    against the fp32 composition on the same (bf16-rounded) values."""
+    from dfno_amd.ops.fft import rfft_trunc, pad_irfft, _t_rfft_trunc, _t_pad_irfft
+    torch.manual_seed(9)
+    N = shape[-1]
+    xb = torch.randn(*shape, device="cuda").bfloat16().requires_grad_(True)
+    y = rfft_trunc(xb, -1, m)
+    assert y.dtype == torch.complex64
+    xr = xb.detach().float().requires_grad_(True)
+    yr = _t_rfft_trunc(xr, -1, m)
+    assert torch.allclose(y, yr, rtol=1e-4, atol=1e-4), \
+        f"r2c fwd {(y-yr).abs().max()}"
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g)
+    assert torch.allclose(xb.grad.float(), xr.grad, rtol=BT, atol=BT), \
+        f"r2c bwd {(xb.grad.float()-xr.grad).abs().max()}"
+
+    # inverse with bf16 output
+    n_half = N // 2 + 1
+    z = torch.randn(*shape[:-1], m, dtype=torch.complex64,
+                    device="cuda").requires_grad_(True)
+    w = pad_irfft(z, -1, n_half, N, m, out_dtype=torch.bfloat16)
+    assert w.dtype == torch.bfloat16
+    zr = z.detach().clone().requires_grad_(True)
+    wr = _t_pad_irfft(zr, -1, n_half, N, m)
+    assert torch.allclose(w.float(), wr, rtol=BT, atol=BT), \
+        f"c2r fwd {(w.float()-wr).abs().max()}"
+    g2 = torch.randn_like(w)
+    w.backward(g2)
+    wr.backward(g2.float())
+    # grads through a bf16-rounded gx: bf16 tolerance
+    assert torch.allclose(z.grad, zr.grad, rtol=BT, atol=BT * 4), \
+        f"c2r bwd {(z.grad-zr.grad).abs().max()}"
+
+
 def test_bf16_model_matches_fp32():
     """Full bf16 model fwd+bwd vs the fp32 model with the SAME weights
     (the 2D+time NS shape of BASELINE config #2, small extents)."""
