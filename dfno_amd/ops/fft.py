@@ -260,8 +260,8 @@ def rfft_trunc(x, dim, m):
     # Clamp to the half-spectrum size: modes > n//2+1 keep the whole spectrum
     # (matches the reference's graceful [:m] slice, /root/reference/dfno/dfno.py:195).
     m = min(m, x.shape[d] // 2 + 1)
-    if d == x.dim() - 1 and (_native_ok(x, x.shape[d], m, "rfft_trunc")
-                             or rfft_bf16_native_ok(x, dim, m)):
+    if d == x.dim() - 1 and (rfft_bf16_native_ok(x, dim, m)
+                             or _native_ok(x, x.shape[d], m, "rfft_trunc")):
         return _RfftTruncFn.apply(x, d, m)
     if x.is_cuda and d != x.dim() - 1:
         note_fallback("rfft_trunc", f"non-last transform dim {d}")
