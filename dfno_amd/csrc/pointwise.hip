@@ -17,6 +17,7 @@
 // round trips become 1.
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
@@ -1267,6 +1268,91 @@ __global__ __launch_bounds__(kBlock) void adam_multi_kernel(
   }
 }
 
+// bf16 multi-tensor variant (the bf16 config's linear parameters): bf16
+// storage for p/g/m/v (matching torch's state dtype for bf16 params), fp32
+// arithmetic, 8-wide vector IO.  Replaces the stock-torch fallback's ~100
+// tiny eager launches per step.
+struct AdamTabB {
+  unsigned short* p[kAdamMaxT];
+  const unsigned short* g[kAdamMaxT];
+  unsigned short* m[kAdamMaxT];
+  unsigned short* v[kAdamMaxT];
+  long end8[kAdamMaxT];   // exclusive prefix sum of ceil(n/8) per tensor
+  long nn[kAdamMaxT];     // exact element count (ragged-tail guard)
+  float c1[kAdamMaxT], c2[kAdamMaxT];
+  int nt;
+};
+
+__device__ __forceinline__ void adam_load8b(const unsigned short* p, float* o) {
+  const uint4 raw = *reinterpret_cast<const uint4*>(p);
+  const unsigned int w[4] = {raw.x, raw.y, raw.z, raw.w};
+#pragma unroll
+  for (int k = 0; k < 4; ++k) {
+    o[2 * k] = __uint_as_float((w[k] & 0xffffu) << 16);
+    o[2 * k + 1] = __uint_as_float((w[k] >> 16) << 16);
+  }
+}
+
+__device__ __forceinline__ void adam_store8b(unsigned short* p, const float* v) {
+  uint4 raw;
+  unsigned int w[4];
+#pragma unroll
+  for (int k = 0; k < 4; ++k) {
+    __hip_bfloat162 h2 = __float22bfloat162_rn(float2{v[2 * k], v[2 * k + 1]});
+    w[k] = *reinterpret_cast<unsigned int*>(&h2);
+  }
+  raw.x = w[0]; raw.y = w[1]; raw.z = w[2]; raw.w = w[3];
+  *reinterpret_cast<uint4*>(p) = raw;
+}
+
+__global__ __launch_bounds__(kBlock) void adam_multi_bf16_kernel(
+    AdamTabB tab, long total8, float lr, float b1, float b2, float eps,
+    float wd) {
+  long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  int t = 0;
+  for (long i = i0; i < total8; i += stride) {
+    while (i >= tab.end8[t]) ++t;
+    long base = (t == 0) ? 0 : tab.end8[t - 1];
+    long j = (i - base) * 8;
+    const float c1 = tab.c1[t], c2 = tab.c2[t];
+    if (j + 8 <= tab.nn[t]) {
+      float pr[8], gr[8], mr[8], vr[8];
+      adam_load8b(tab.p[t] + j, pr);
+      adam_load8b(tab.g[t] + j, gr);
+      adam_load8b(tab.m[t] + j, mr);
+      adam_load8b(tab.v[t] + j, vr);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) {
+        float gg = gr[k] + wd * pr[k];
+        mr[k] = b1 * mr[k] + (1.f - b1) * gg;
+        vr[k] = b2 * vr[k] + (1.f - b2) * gg * gg;
+        pr[k] -= lr * (mr[k] / c1) / (sqrtf(vr[k] / c2) + eps);
+      }
+      adam_store8b(tab.p[t] + j, pr);
+      adam_store8b(tab.m[t] + j, mr);
+      adam_store8b(tab.v[t] + j, vr);
+    } else {
+      for (long e = j; e < tab.nn[t]; ++e) {   // ragged tail, scalar
+        float pv = __uint_as_float(((unsigned int)tab.p[t][e]) << 16);
+        float gv = __uint_as_float(((unsigned int)tab.g[t][e]) << 16);
+        float mv = __uint_as_float(((unsigned int)tab.m[t][e]) << 16);
+        float vv = __uint_as_float(((unsigned int)tab.v[t][e]) << 16);
+        float gg = gv + wd * pv;
+        mv = b1 * mv + (1.f - b1) * gg;
+        vv = b2 * vv + (1.f - b2) * gg * gg;
+        pv -= lr * (mv / c1) / (sqrtf(vv / c2) + eps);
+        __hip_bfloat16 hp = __float2bfloat16(pv);
+        __hip_bfloat16 hm = __float2bfloat16(mv);
+        __hip_bfloat16 hv = __float2bfloat16(vv);
+        tab.p[t][e] = *reinterpret_cast<unsigned short*>(&hp);
+        tab.m[t][e] = *reinterpret_cast<unsigned short*>(&hm);
+        tab.v[t][e] = *reinterpret_cast<unsigned short*>(&hv);
+      }
+    }
+  }
+}
+
 // Uniform-size variant: a group of SAME-length tensors (the 32 spectral
 // corner weights dominate the parameter set) updates with blockIdx.y as the
 // tensor index — no per-iteration tensor search / end4 kernarg loads at all
@@ -1316,16 +1402,44 @@ void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 
   // bucket vec4-eligible tensors by length: same-size groups (>= 2) take
-  // the uniform kernel, the ragged rest the searched multi-tensor kernel
+  // the uniform kernel, the ragged rest the searched multi-tensor kernel;
+  // bf16 tensors batch into the bf16 multi-tensor kernel
   std::vector<size_t> vec_idx;
   std::map<long, std::vector<size_t>> by_len;
+  AdamTabB btab;
+  btab.nt = 0;
+  long btotal8 = 0;
+  auto flush_bf16 = [&]() {
+    if (btab.nt == 0) return;
+    int grid = grid_for(btotal8, kBlock);
+    hipLaunchKernelGGL(adam_multi_bf16_kernel, dim3(grid), dim3(kBlock), 0,
+                       stream, btab, btotal8, (float)lr, (float)beta1,
+                       (float)beta2, (float)eps, (float)weight_decay);
+    btab.nt = 0;
+    btotal8 = 0;
+  };
   for (size_t i = 0; i < ps.size(); ++i) {
     long n = ps[i].numel();
+    bool aligned = ((reinterpret_cast<uintptr_t>(ps[i].data_ptr()) & 15) == 0) &&
+                   ((reinterpret_cast<uintptr_t>(gs[i].data_ptr()) & 15) == 0) &&
+                   ((reinterpret_cast<uintptr_t>(ms[i].data_ptr()) & 15) == 0) &&
+                   ((reinterpret_cast<uintptr_t>(vs[i].data_ptr()) & 15) == 0);
+    if (ps[i].scalar_type() == at::kBFloat16 && n > 0 && aligned) {
+      int t = btab.nt++;
+      btab.p[t] = reinterpret_cast<unsigned short*>(ps[i].data_ptr());
+      btab.g[t] = reinterpret_cast<const unsigned short*>(gs[i].data_ptr());
+      btab.m[t] = reinterpret_cast<unsigned short*>(ms[i].data_ptr());
+      btab.v[t] = reinterpret_cast<unsigned short*>(vs[i].data_ptr());
+      btotal8 += (n + 7) / 8;
+      btab.end8[t] = btotal8;
+      btab.nn[t] = n;
+      btab.c1[t] = (float)(1.0 - std::pow(beta1, (double)steps[i]));
+      btab.c2[t] = (float)(1.0 - std::pow(beta2, (double)steps[i]));
+      if (btab.nt == kAdamMaxT) flush_bf16();
+      continue;
+    }
     bool vec = ps[i].scalar_type() == at::kFloat && (n % 4 == 0) && n > 0 &&
-               ((reinterpret_cast<uintptr_t>(ps[i].data_ptr()) & 15) == 0) &&
-               ((reinterpret_cast<uintptr_t>(gs[i].data_ptr()) & 15) == 0) &&
-               ((reinterpret_cast<uintptr_t>(ms[i].data_ptr()) & 15) == 0) &&
-               ((reinterpret_cast<uintptr_t>(vs[i].data_ptr()) & 15) == 0);
+               aligned;
     if (!vec) {  // rare: odd-sized or fp64 tensor keeps the single-tensor path
       adam_step_(ps[i], gs[i], ms[i], vs[i], lr, beta1, beta2, eps,
                  weight_decay, steps[i]);
@@ -1333,6 +1447,7 @@ void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
     }
     by_len[n].push_back(i);
   }
+  flush_bf16();
 
   auto fill = [&](AdamTab& tab, size_t i, int t) {
     tab.p[t] = ps[i].data_ptr<float>();

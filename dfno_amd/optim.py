@@ -64,7 +64,8 @@ class Adam(torch.optim.Adam):
                 if p.grad is None or p.numel() == 0:
                     continue
                 if not (p.is_cuda and p.dtype in (torch.float32, torch.complex64,
-                                                  torch.float64, torch.complex128)
+                                                  torch.float64, torch.complex128,
+                                                  torch.bfloat16)
                         and not p.grad.is_sparse and p.is_contiguous()):
                     slow.append(p)
                     continue

@@ -424,7 +424,13 @@ def test_fused_adam_matches_torch():
         return [torch.randn(1000, device="cuda", requires_grad=True),
                 torch.randn(20, 20, 123, device="cuda", dtype=torch.complex64,
                             requires_grad=True),
-                torch.randn(7, device="cuda", dtype=torch.float64, requires_grad=True)]
+                torch.randn(7, device="cuda", dtype=torch.float64, requires_grad=True),
+                # bf16 params take the fused bf16 multi-tensor kernel
+                # (8-wide + ragged tail)
+                torch.randn(20, 20, device="cuda", dtype=torch.bfloat16,
+                            requires_grad=True),
+                torch.randn(37, device="cuda", dtype=torch.bfloat16,
+                            requires_grad=True)]
 
     p1 = make_params()
     torch.manual_seed(20)
@@ -443,8 +449,10 @@ def test_fused_adam_matches_torch():
         o1.step()
         o2.step()
     for a, b in zip(p1, p2):
-        assert torch.allclose(a.detach(), b.detach(), rtol=1e-5, atol=1e-6), \
-            f"max {(a.detach()-b.detach()).abs().max()}"
+        tt = dict(rtol=1e-5, atol=1e-6) if a.dtype != torch.bfloat16 \
+            else dict(rtol=2e-2, atol=2e-2)   # bf16 state rounding differs
+        assert torch.allclose(a.detach().float(), b.detach().float(), **tt), \
+            f"{a.dtype}: max {(a.detach().float()-b.detach().float()).abs().max()}"
 
 
 # ---------------------------------------------------------------------------
