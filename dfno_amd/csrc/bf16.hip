@@ -84,12 +84,17 @@ __device__ __forceinline__ void store4(unsigned short* p, const float* v) {
 // x-resident channel mix (I <= IMAX): one thread owns kVec consecutive s.
 // ---------------------------------------------------------------------------
 
-template <int IMAX, int VEC, bool ACT>
+// IT/OT > 0 pin the channel counts at compile time (runtime trip counts
+// serialize on uniform-load waits — the round-1 folding lesson; the fp32
+// twins carry the same pins)
+template <int IMAX, int VEC, bool ACT, int IT = 0, int OT = 0>
 __global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ W,
     const unsigned short* __restrict__ bias, unsigned short* __restrict__ y,
-    unsigned short* __restrict__ z, int B, int I, int O, long S, bool wt,
+    unsigned short* __restrict__ z, int B, int I_, int O_, long S, bool wt,
     bool has_bias, bool write_z, const unsigned short* __restrict__ res) {
+  const int I = IT > 0 ? IT : I_;
+  const int O = OT > 0 ? OT : O_;
   // VEC=4 for the wide-IMAX variants (xr[24][8] = 192 VGPRs would spill)
   extern __shared__ __align__(16) char smem_raw[];
   float* Wl = reinterpret_cast<float*>(smem_raw);   // [O*I] fp32
@@ -124,8 +129,8 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
     unsigned short* yb = y + ((long)b * O) * S + s;
     unsigned short* zb = write_z ? z + ((long)b * O) * S + s : nullptr;
 #pragma unroll 4
-    for (int o = 0; o < 512; ++o) {
-      if (o >= O) break;
+    for (int o = 0; o < (OT > 0 ? OT : 512); ++o) {
+      if (OT == 0 && o >= O) break;
       float acc[VEC];
       const float bv = has_bias ? bl[o] : 0.f;
 #pragma unroll
@@ -158,12 +163,14 @@ __global__ __launch_bounds__(kBlock) void bf16_channel_mix_xres_kernel(
 // accumulator-resident channel mix (O <= OMAX, streams I) — the 128->1 head
 // ---------------------------------------------------------------------------
 
-template <int OMAX, int VEC, bool ACT>
+template <int OMAX, int VEC, bool ACT, int IT = 0, int OT = 0>
 __global__ __launch_bounds__(kBlock) void bf16_channel_mix_ores_kernel(
     const unsigned short* __restrict__ x, const unsigned short* __restrict__ W,
     const unsigned short* __restrict__ bias, unsigned short* __restrict__ y,
-    unsigned short* __restrict__ z, int B, int I, int O, long S, bool wt,
+    unsigned short* __restrict__ z, int B, int I_, int O_, long S, bool wt,
     bool has_bias, bool write_z, const unsigned short* __restrict__ res) {
+  const int I = IT > 0 ? IT : I_;
+  const int O = OT > 0 ? OT : O_;
   // VEC=4 for the wide-OMAX variants: acc[24][8] would eat 192 VGPRs and
   // spill; [24][4] keeps the whole accumulator set resident.
   extern __shared__ __align__(16) char smem_raw[];
@@ -526,25 +533,30 @@ std::vector<at::Tensor> bf16_channel_mix(const at::Tensor& x, const at::Tensor& 
                        stream, usp(x), usp(W), bp, usp_mut(y), zp, B, I, O, S, \
                        wt, has_bias, write_z, rp);                             \
   }
-#define BCMO(CAP, VEC)                                                         \
+#define BCMP(KER, CAP, VEC, ITV, OTV)                                          \
   if (act) {                                                                   \
-    hipLaunchKernelGGL((bf16_channel_mix_ores_kernel<CAP, VEC, true>),         \
-                       dim3(grid), dim3(kBlock), smem, stream, usp(x), usp(W), \
-                       bp, usp_mut(y), zp, B, I, O, S, wt, has_bias, write_z,  \
-                       rp);                                                    \
+    hipLaunchKernelGGL((KER<CAP, VEC, true, ITV, OTV>), dim3(grid),            \
+                       dim3(kBlock), smem, stream, usp(x), usp(W), bp,         \
+                       usp_mut(y), zp, B, I, O, S, wt, has_bias, write_z, rp); \
   } else {                                                                     \
-    hipLaunchKernelGGL((bf16_channel_mix_ores_kernel<CAP, VEC, false>),        \
-                       dim3(grid), dim3(kBlock), smem, stream, usp(x), usp(W), \
-                       bp, usp_mut(y), zp, B, I, O, S, wt, has_bias, write_z,  \
-                       rp);                                                    \
+    hipLaunchKernelGGL((KER<CAP, VEC, false, ITV, OTV>), dim3(grid),           \
+                       dim3(kBlock), smem, stream, usp(x), usp(W), bp,         \
+                       usp_mut(y), zp, B, I, O, S, wt, has_bias, write_z, rp); \
   }
-  if (I <= 8) { BCM(bf16_channel_mix_xres_kernel, 8, 8) }
+  // compile-time-pinned hot shapes first (block mixes, projection lift and
+  // its transpose, channel lift, projection head)
+  if (I == 20 && O == 20) { BCMP(bf16_channel_mix_xres_kernel, 24, 4, 20, 20) }
+  else if (I == 20 && O == 128) { BCMP(bf16_channel_mix_xres_kernel, 24, 4, 20, 128) }
+  else if (I == 2 && O == 20) { BCMP(bf16_channel_mix_xres_kernel, 8, 8, 2, 20) }
+  else if (I == 128 && O == 20) { BCMP(bf16_channel_mix_ores_kernel, 24, 4, 128, 20) }
+  else if (I == 128 && O == 1) { BCMP(bf16_channel_mix_ores_kernel, 8, 8, 128, 1) }
+  else if (I <= 8) { BCM(bf16_channel_mix_xres_kernel, 8, 8) }
   else if (I <= 24) { BCM(bf16_channel_mix_xres_kernel, 24, 4) }
   else if (I <= 32) { BCM(bf16_channel_mix_xres_kernel, 32, 4) }
-  else if (O <= 8) { BCMO(8, 8) }
-  else if (O <= 24) { BCMO(24, 4) }
+  else if (O <= 8) { BCMP(bf16_channel_mix_ores_kernel, 8, 8, 0, 0) }
+  else if (O <= 24) { BCMP(bf16_channel_mix_ores_kernel, 24, 4, 0, 0) }
   else { TORCH_CHECK(false, "bf16 channel mix: unsupported shape I=", I, " O=", O); }
-#undef BCMO
+#undef BCMP
 #undef BCM
   DFNO_CHECK_LAUNCH("bf16_channel_mix");
   return {y, z};
