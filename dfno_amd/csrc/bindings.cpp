@@ -24,6 +24,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "lift head backward: (gx, gW1, gb1, gW2, gb2)");
   m.def("adam_step_", &adam_step_, "fused Adam step (in-place)");
   m.def("adam_step_batch_", &adam_step_batch_, "fused Adam step over a tensor list");
+  m.def("adam_step_batch_fp8_", &adam_step_batch_fp8_,
+        "fused Adam step with in-kernel e4m3 requantization (delayed scaling)");
   m.def("dft_c2c", &dft_c2c, "truncated/padded complex DFT along a dim");
   m.def("dft_rfft_trunc", &dft_rfft_trunc, "real->kept-low-modes DFT (last dim)");
   m.def("dft_rfft_trunc_adj", &dft_rfft_trunc_adj, "adjoint of dft_rfft_trunc");

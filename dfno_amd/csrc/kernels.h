@@ -62,6 +62,15 @@ void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
                       std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
                       double lr, double beta1, double beta2, double eps,
                       double weight_decay, std::vector<int64_t> steps);
+// fp8 quantize-in-Adam (delayed scaling): qs/am_ins/am_outs parallel to ps
+// (undefined/empty tensors for non-fp8 params); returns the indices whose
+// e4m3 copies were refreshed by the update kernel itself.
+std::vector<int64_t> adam_step_batch_fp8_(
+    std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+    std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+    double lr, double beta1, double beta2, double eps, double weight_decay,
+    std::vector<int64_t> steps, std::vector<at::Tensor> qs,
+    std::vector<at::Tensor> am_ins, std::vector<at::Tensor> am_outs);
 
 // fused truncated-spectrum DFTs (see ops/fft.py):
 at::Tensor dft_c2c(const at::Tensor& x, int64_t dim, int64_t n,

@@ -1236,6 +1236,17 @@ struct AdamTab {
   int nt;
 };
 
+// fp8 quantize-in-Adam side table (the spectral-weight group): the update
+// pass emits the e4m3 copy for the NEXT forward using the PREVIOUS step's
+// amax (delayed scaling; e4m3 conversion saturates), and reduces the new
+// amax into am_out — re-quantization costs nothing beyond the bytes the
+// optimizer already streams (docs/ROADMAP.md item 5).
+struct AdamFp8Tab {
+  unsigned char* q[kAdamMaxT];
+  const float* am_in[kAdamMaxT];
+  float* am_out[kAdamMaxT];
+};
+
 __global__ __launch_bounds__(kBlock) void adam_multi_kernel(
     AdamTab tab, long total4, float lr, float b1, float b2, float eps,
     float wd) {
@@ -1357,14 +1368,18 @@ __global__ __launch_bounds__(kBlock) void adam_multi_bf16_kernel(
 // corner weights dominate the parameter set) updates with blockIdx.y as the
 // tensor index — no per-iteration tensor search / end4 kernarg loads at all
 // (docs/ROADMAP.md item 4: runtime-trip-count folding for the Adam loop).
+template <bool QOUT>
 __global__ __launch_bounds__(kBlock) void adam_multi_uniform_kernel(
-    AdamTab tab, long n4, float lr, float b1, float b2, float eps, float wd) {
+    AdamTab tab, AdamFp8Tab qt, long n4, float lr, float b1, float b2,
+    float eps, float wd) {
   const int t = blockIdx.y;
   float* __restrict__ p = tab.p[t];
   const float* __restrict__ g = tab.g[t];
   float* __restrict__ m = tab.m[t];
   float* __restrict__ v = tab.v[t];
   const float c1 = tab.c1[t], c2 = tab.c2[t];
+  float inv = 0.f, amax = 0.f;
+  if constexpr (QOUT) inv = 448.f / fmaxf(*qt.am_in[t], 1e-30f);
   long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long i = i0; i < n4; i += stride) {
@@ -1387,18 +1402,45 @@ __global__ __launch_bounds__(kBlock) void adam_multi_uniform_kernel(
     *reinterpret_cast<float4*>(p + j) = make_float4(pr[0], pr[1], pr[2], pr[3]);
     *reinterpret_cast<float4*>(m + j) = make_float4(mr[0], mr[1], mr[2], mr[3]);
     *reinterpret_cast<float4*>(v + j) = make_float4(vr[0], vr[1], vr[2], vr[3]);
+    if constexpr (QOUT) {
+      uchar4 qb;
+      __hip_fp8_e4m3 h0(pr[0] * inv), h1(pr[1] * inv), h2(pr[2] * inv),
+          h3(pr[3] * inv);
+      qb.x = h0.__x; qb.y = h1.__x; qb.z = h2.__x; qb.w = h3.__x;
+      *reinterpret_cast<uchar4*>(qt.q[t] + j) = qb;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) amax = fmaxf(amax, fabsf(pr[k]));
+    }
+  }
+  if constexpr (QOUT) {
+    // block-reduce the new amax, one device atomicMax per block
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      amax = fmaxf(amax, __shfl_down(amax, off, 64));
+    __shared__ float wred[4];
+    if ((threadIdx.x & 63) == 0) wred[threadIdx.x >> 6] = amax;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      amax = fmaxf(fmaxf(wred[0], wred[1]), fmaxf(wred[2], wred[3]));
+      atomicMax(reinterpret_cast<unsigned int*>(qt.am_out[t]),
+                __float_as_uint(amax));
+    }
   }
 }
 
 }  // namespace
 
-void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
-                      std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
-                      double lr, double beta1, double beta2, double eps,
-                      double weight_decay, std::vector<int64_t> steps) {
+static std::vector<int64_t> adam_step_batch_impl(
+    std::vector<at::Tensor>& ps, std::vector<at::Tensor>& gs,
+    std::vector<at::Tensor>& ms, std::vector<at::Tensor>& vs,
+    double lr, double beta1, double beta2, double eps,
+    double weight_decay, std::vector<int64_t>& steps,
+    std::vector<at::Tensor>* qs, std::vector<at::Tensor>* am_ins,
+    std::vector<at::Tensor>* am_outs) {
   TORCH_CHECK(ps.size() == gs.size() && ps.size() == ms.size() &&
               ps.size() == vs.size() && ps.size() == steps.size(),
               "adam batch: length mismatch");
+  std::vector<int64_t> quantized;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
 
   // bucket vec4-eligible tensors by length: same-size groups (>= 2) take
@@ -1485,23 +1527,65 @@ void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
     long n4 = n / 4;
     for (size_t off = 0; off < idxs.size(); off += kAdamMaxT) {
       AdamTab tab;
+      AdamFp8Tab qt{};
       int cnt = (int)std::min<size_t>(kAdamMaxT, idxs.size() - off);
-      for (int t = 0; t < cnt; ++t) fill(tab, idxs[off + t], t);
+      bool all_q = qs != nullptr;
+      for (int t = 0; t < cnt; ++t) {
+        size_t i = idxs[off + t];
+        fill(tab, i, t);
+        if (all_q && (*qs)[i].defined() && (*qs)[i].numel() > 0) {
+          qt.q[t] = reinterpret_cast<unsigned char*>((*qs)[i].data_ptr());
+          qt.am_in[t] = (*am_ins)[i].data_ptr<float>();
+          qt.am_out[t] = (*am_outs)[i].data_ptr<float>();
+        } else {
+          all_q = false;
+        }
+      }
       tab.nt = cnt;
       // size grid.x so that grid.x * cnt covers the chip at ~8 blocks/CU
       long gx = (n4 + kBlock - 1) / kBlock;
       long cap = std::max(1L, (256L * 8) / cnt);
       if (gx > cap) gx = cap;
-      hipLaunchKernelGGL(adam_multi_uniform_kernel, dim3((int)gx, cnt),
-                         dim3(kBlock), 0, stream, tab, n4, (float)lr,
-                         (float)beta1, (float)beta2, (float)eps,
-                         (float)weight_decay);
+      if (all_q) {
+        for (int t = 0; t < cnt; ++t) quantized.push_back((int64_t)idxs[off + t]);
+        hipLaunchKernelGGL(adam_multi_uniform_kernel<true>, dim3((int)gx, cnt),
+                           dim3(kBlock), 0, stream, tab, qt, n4, (float)lr,
+                           (float)beta1, (float)beta2, (float)eps,
+                           (float)weight_decay);
+      } else {
+        hipLaunchKernelGGL(adam_multi_uniform_kernel<false>, dim3((int)gx, cnt),
+                           dim3(kBlock), 0, stream, tab, qt, n4, (float)lr,
+                           (float)beta1, (float)beta2, (float)eps,
+                           (float)weight_decay);
+      }
     }
   }
   flush_ragged();
   hipError_t lerr = hipGetLastError();
   TORCH_CHECK(lerr == hipSuccess, "adam_multi launch failed: ",
               hipGetErrorString(lerr));
+  return quantized;
+}
+
+void adam_step_batch_(std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+                      std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+                      double lr, double beta1, double beta2, double eps,
+                      double weight_decay, std::vector<int64_t> steps) {
+  adam_step_batch_impl(ps, gs, ms, vs, lr, beta1, beta2, eps, weight_decay,
+                       steps, nullptr, nullptr, nullptr);
+}
+
+std::vector<int64_t> adam_step_batch_fp8_(
+    std::vector<at::Tensor> ps, std::vector<at::Tensor> gs,
+    std::vector<at::Tensor> ms, std::vector<at::Tensor> vs,
+    double lr, double beta1, double beta2, double eps, double weight_decay,
+    std::vector<int64_t> steps, std::vector<at::Tensor> qs,
+    std::vector<at::Tensor> am_ins, std::vector<at::Tensor> am_outs) {
+  TORCH_CHECK(qs.size() == ps.size() && am_ins.size() == ps.size() &&
+              am_outs.size() == ps.size(), "adam fp8: list size mismatch");
+  // returns the indices whose e4m3 copies were refreshed in-kernel
+  return adam_step_batch_impl(ps, gs, ms, vs, lr, beta1, beta2, eps,
+                              weight_decay, steps, &qs, &am_ins, &am_outs);
 }
 
 void adam_step_(at::Tensor& p, const at::Tensor& g, at::Tensor& m, at::Tensor& v,

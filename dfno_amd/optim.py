@@ -60,6 +60,12 @@ class Adam(torch.optim.Adam):
                 continue
             slow = []
             batch = ([], [], [], [], [])  # p, g, m, v, step
+            fp8_q, fp8_in, fp8_out, origs = [], [], [], []
+            try:
+                from .ops.spectral import _FP8_CACHE, _QUANT_EPOCH
+            except Exception:
+                _FP8_CACHE, _QUANT_EPOCH = {}, [0]
+            empty = None
             for p in group["params"]:
                 if p.grad is None or p.numel() == 0:
                     continue
@@ -83,7 +89,37 @@ class Adam(torch.optim.Adam):
                 batch[2].append(mv)
                 batch[3].append(vv)
                 batch[4].append(int(state["step"].item()))
-            if batch[0]:
+                # fp8 quantize-in-Adam (delayed scaling): spectral corner
+                # masters with a live e4m3 cache get their quantized copy
+                # refreshed by the update kernel itself — the per-step
+                # requantization pass disappears (docs/ROADMAP.md item 5)
+                origs.append(p)
+                ent = _FP8_CACHE.get(id(p)) if p.dtype == torch.complex64 else None
+                if ent is not None:
+                    if len(ent) == 3:   # bootstrap the measured-amax slot
+                        ent.append(ent[2].clone())
+                    ent[2].zero_()      # receives the NEW measured amax
+                    fp8_q.append(ent[1].view(torch.uint8))
+                    fp8_in.append(ent[3])   # delayed scale (prev measurement)
+                    fp8_out.append(ent[2])
+                else:
+                    if empty is None:
+                        empty = torch.empty(0, device=p.device)
+                    fp8_q.append(empty)
+                    fp8_in.append(empty)
+                    fp8_out.append(empty)
+            if batch[0] and any(q.numel() for q in fp8_q):
+                quantized = ext.adam_step_batch_fp8_(
+                    batch[0], batch[1], batch[2], batch[3], lr, beta1, beta2,
+                    eps, wd, batch[4], fp8_q, fp8_in, fp8_out)
+                for idx in quantized:
+                    ent = _FP8_CACHE[id(origs[idx])]
+                    # the copy was quantized with ent[3]'s scale; make that
+                    # the dequant scale and keep the fresh measurement as
+                    # the next delayed scale
+                    ent[2], ent[3] = ent[3], ent[2]
+                    ent[0] = _QUANT_EPOCH[0] + 1   # fresh for the next fwd
+            elif batch[0]:
                 ext.adam_step_batch_(batch[0], batch[1], batch[2], batch[3],
                                      lr, beta1, beta2, eps, wd, batch[4])
             if slow:

@@ -65,6 +65,52 @@ def test_fp8_spectral_conv_matches_dequantized_oracle():
             "straight-through grad-W"
 
 
+def test_fp8_quantize_in_adam():
+    """The fused Adam refreshes the e4m3 copies in-kernel (delayed scaling):
+    after a step, the cached quantization matches a re-quantization of the
+    UPDATED masters at the delayed scale, and the cache is marked fresh for
+    the next training forward (no standalone requant pass)."""
+    from dfno_amd.ops.spectral import (_FP8_CACHE, _QUANT_EPOCH, _fp8_weights,
+                                       bump_quant_epoch, dequantize_fp8)
+    from dfno_amd.optim import Adam
+    torch.manual_seed(3)
+    _, P_x, _ = dfno.create_standard_partitions((1, 1, 1, 1, 1, 1))
+    model = dfno.DistributedFNONd(P_x, [1, 2, 12, 12, 8, 1], 8, 8,
+                                  (3, 3, 2, 2), num_blocks=1,
+                                  device=torch.device("cuda"),
+                                  spectral_fp8=True)
+    opt = Adam(model.parameters(), lr=1e-2)
+    x = torch.rand(1, 2, 12, 12, 8, 1, device="cuda")
+
+    def step():
+        opt.zero_grad(set_to_none=True)
+        y = model(x)
+        dfno.DistributedRelativeLpLoss(P_x)(y, torch.rand_like(y)).backward()
+        opt.step()
+
+    step()   # first: fwd quantizes, adam bootstraps delayed slots
+    step()
+    w = model.blocks[0].weights[0]
+    ent = _FP8_CACHE[id(w)]
+    assert ent[0] == _QUANT_EPOCH[0] + 1, "cache not marked fresh by Adam"
+    # delayed-scale check: dequantizing the kernel's copy reproduces the
+    # UPDATED master within e4m3 resolution at the stored dequant scale
+    wq = dequantize_fp8(ent[1], ent[2])
+    scale = float(ent[2].clamp_min(1e-30)) / 448.0
+    err = (wq - w.detach()).abs().max().item()
+    assert err <= 4.0 * scale * 16 + 1e-7, f"fp8-in-adam quant err {err} scale {scale}"
+    # and the measured amax matches the updated master's amax
+    amax_meas = float(ent[3])
+    amax_true = torch.view_as_real(w.detach()).abs().amax().item()
+    assert abs(amax_meas - amax_true) < 1e-6 * max(1.0, amax_true)
+
+    # the NEXT training forward must not requantize (epochs line up)
+    bump_quant_epoch()
+    before = ent[1].clone()
+    _fp8_weights([w])
+    assert torch.equal(before, ent[1])
+
+
 def test_fp8_model_accuracy_vs_fp32():
     """End-to-end quantization error of the fp8-spectral model on the NS
     shape: output relative error must be small (e4m3 has a 2^-3 mantissa
