@@ -79,7 +79,10 @@ def test_fp8_quantize_in_adam():
                                   (3, 3, 2, 2), num_blocks=1,
                                   device=torch.device("cuda"),
                                   spectral_fp8=True)
-    opt = Adam(model.parameters(), lr=1e-2)
+    # realistic lr relative to the 1/width^2 init scale: delayed scaling
+    # assumes the amax moves slowly between steps (values past the previous
+    # step's amax saturate at 448*scale)
+    opt = Adam(model.parameters(), lr=1e-4)
     x = torch.rand(1, 2, 12, 12, 8, 1, device="cuda")
 
     def step():
@@ -96,9 +99,12 @@ def test_fp8_quantize_in_adam():
     # delayed-scale check: dequantizing the kernel's copy reproduces the
     # UPDATED master within e4m3 resolution at the stored dequant scale
     wq = dequantize_fp8(ent[1], ent[2])
-    scale = float(ent[2].clamp_min(1e-30)) / 448.0
+    amax = float(ent[2])
     err = (wq - w.detach()).abs().max().item()
-    assert err <= 4.0 * scale * 16 + 1e-7, f"fp8-in-adam quant err {err} scale {scale}"
+    # one top-binade e4m3 step is amax/16; allow that plus the per-step
+    # amax drift the delayed scale saturates against
+    assert err <= amax / 16 + 2e-4 * amax + 1e-7, \
+        f"fp8-in-adam quant err {err} amax {amax}"
     # and the measured amax matches the updated master's amax
     amax_meas = float(ent[3])
     amax_true = torch.view_as_real(w.detach()).abs().amax().item()
