@@ -53,3 +53,40 @@ def test_bench_json_contract_4rank():
     assert d["n_gpus"] == 4
     assert "2x2" in d["config"]["parallelism"]
     assert d["value"] > 0
+
+
+def test_bench_json_contract_8rank_driver_partition():
+    # the exact launch+partition the driver's 8-GPU SCALE run uses
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=8", "--master-addr", "127.0.0.1",
+         "--master-port", "29538", str(REPO / "bench.py"), "--gpus", "8",
+         "--steps", "1", "--warmup", "0", "--width", "4", "--num-blocks", "1"],
+        capture_output=True, text=True, timeout=900, env=env, cwd=str(REPO))
+    assert r.returncode == 0, r.stderr[-2000:]
+    d = json.loads(r.stdout.strip().splitlines()[-1])
+    assert d["n_gpus"] == 8
+    assert "4x2" in d["config"]["parallelism"]
+    assert d["value"] > 0
+
+
+def test_bench_json_contract_8rank_heavy_comm():
+    # trailing-axis partition: real R1/R4 all-to-alls + chunk pipeline at 8 ranks
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=8", "--master-addr", "127.0.0.1",
+         "--master-port", "29539", str(REPO / "bench.py"), "--gpus", "8",
+         "--heavy-comm", "--steps", "1", "--warmup", "0", "--width", "4",
+         "--num-blocks", "1"],
+        capture_output=True, text=True, timeout=900, env=env, cwd=str(REPO))
+    assert r.returncode == 0, r.stderr[-2000:]
+    d = json.loads(r.stdout.strip().splitlines()[-1])
+    assert d["n_gpus"] == 8
+    assert "2x2x2" in d["config"]["parallelism"]
+    assert d["value"] > 0
