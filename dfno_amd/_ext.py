@@ -32,6 +32,7 @@ SOURCES = [
     _CSRC / "lift_head.hip",
     _CSRC / "pack.hip",
     _CSRC / "bf16.hip",
+    _CSRC / "dft2d.hip",
 ]
 
 

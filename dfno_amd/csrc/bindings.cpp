@@ -33,6 +33,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "adjoint of dft_rfft_trunc with fused accumulate addend");
   m.def("dft_pad_irfft", &dft_pad_irfft, "kept modes -> real inverse (last dim)");
   m.def("dft_pad_irfft_adj", &dft_pad_irfft_adj, "adjoint of dft_pad_irfft");
+  m.def("dft_zt_fwd", &dft_zt_fwd,
+        "fused truncated (z,t) 2-D analysis (real -> kept-mode c64)");
+  m.def("dft_zt_inv", &dft_zt_inv,
+        "fused padded (z,t) 2-D synthesis (kept-mode c64 -> real)");
   m.def("dft_pad_irfft_bf16", &dft_pad_irfft_bf16,
         "pad_irfft with bf16 output (bf16 compute config)");
   m.def("dft_rfft_trunc_adj_bf16", &dft_rfft_trunc_adj_bf16,

@@ -1,0 +1,583 @@
+// Fused (z,t) boundary 2-D transform (docs/ROADMAP.md item: the pencil
+// chain's trailing pair).  The flagship's m-phase runs rfft_trunc(t) then
+// fft_trunc(z) as two full-activation passes with a [L, Z, mt] complex
+// intermediate (0.34 GB at the flagship, written+read 16 times per step
+// across fwd/inv/adjoints).  These kernels compute the truncated 2-D
+// transform plane-resident in LDS: each [Z x T] (or [mz x mt]) plane is
+// staged once and both dims are transformed before anything goes back to
+// HBM.
+//
+// Two kernels cover all four autograd uses (mirroring the 1-D impls):
+//   zt_fwd : real [L, Z, T] -> c64 [L, mz, mt]
+//            (a) forward analysis (scale 1, factors off)
+//            (b) adjoint of zt_inv (scale 1/(Z*T), factors on)
+//   zt_inv : c64 [L, mz, mt] -> real [L, Z, T]
+//            (a) forward inverse (scale 1/(Z*T), factors on)
+//            (b) adjoint of zt_fwd (scale 1, factors off)
+// mz = mz_lo + mz_hi kept z-modes (low block + high/negative block); the
+// t-dim keeps only the low mt half-spectrum modes (rfft).  "factors" is the
+// rfft half-spectrum convention: double every non-edge t-mode.  Edge (DC /
+// Nyquist) modes keep BOTH components — the kt=0 column of the 2-D spectrum
+// is hermitian across kz, not per-element real, so its imaginary part feeds
+// the real output through the z-transform; the synthesis side's final Re()
+// is what drops the truly unrepresentable component.
+//
+// TI/TO = unsigned short select bf16 real-side IO (fp32 compute), as in the
+// 1-D bf16-IO variants.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include <map>
+#include <tuple>
+
+#include "kernels.h"
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kMaxZ = 64;
+constexpr int kMaxT = 64;
+constexpr int kMaxMZ = 48;
+constexpr int kMaxMT = 32;
+
+__device__ __forceinline__ float zt_b2f(unsigned short h) {
+  return __uint_as_float(((unsigned int)h) << 16);
+}
+
+__device__ __forceinline__ unsigned short zt_f2b(float f) {
+  __hip_bfloat16 h = __float2bfloat16(f);
+  return *reinterpret_cast<unsigned short*>(&h);
+}
+
+template <typename TI>
+__device__ __forceinline__ float zt_ld(const TI* p) {
+  if constexpr (std::is_same<TI, unsigned short>::value) return zt_b2f(*p);
+  else return (float)*p;
+}
+
+// ---------------------------------------------------------------------------
+// zt_fwd: per plane, stage x [Z][T] in LDS, then
+//   B[z][kt] = sum_t x[z][t] wt[t][kt]           (Z*mt outputs, T MACs each)
+//   y[kz][kt] = sum_z B[z][kt] wz[z][kz]         (mz*mt outputs, Z cmacs)
+// one plane per wave round; a block processes PLW = 4 planes per pass.
+// ---------------------------------------------------------------------------
+
+// CZ/CT/CMZ/CMT pin the trip counts at compile time (0 = runtime): the
+// hot flagship configuration (Z=64, T=30, mz=24, mt=8) gets fully unrolled
+// inner loops; everything else takes the generic instantiation (PLW=2 so
+// the worst-case LDS footprint stays under 160 KB).  Twiddle tables are
+// staged into LDS once per block.
+template <typename TI, int PLW, int CZ, int CT, int CMZ, int CMT>
+__global__ __launch_bounds__(kBlock) void zt_fwd_kernel(
+    const TI* __restrict__ in, float* __restrict__ out,
+    const float* __restrict__ gtwt,  // [T, mt, 2]  w_T^{-t kt}
+    const float* __restrict__ gtwz,  // [Z, mz, 2]  w_Z^{-z kz(i)} (kept order)
+    long planes, int Z_, int T_, int mz_, int mt_, float scale, bool factors) {
+  const int Z = CZ ? CZ : Z_;
+  const int T = CT ? CT : T_;
+  const int mz = CMZ ? CMZ : mz_;
+  const int mt = CMT ? CMT : mt_;
+  extern __shared__ __align__(16) char smem_raw[];
+  float* twt = reinterpret_cast<float*>(smem_raw);      // [T*mt*2]
+  float* twz = twt + (size_t)T * mt * 2;                // [Z*mz*2]
+  float* xs = twz + (size_t)Z * mz * 2;                 // [PLW][Z*T]
+  float* bs = xs + (size_t)PLW * Z * T;                 // [PLW][Z*mt*2]
+  for (int i = threadIdx.x; i < T * mt * 2; i += kBlock) twt[i] = gtwt[i];
+  for (int i = threadIdx.x; i < Z * mz * 2; i += kBlock) twz[i] = gtwz[i];
+
+  const int zt = Z * T;
+  long ntile = (planes + PLW - 1) / PLW;
+  for (long tb = blockIdx.x; tb < ntile; tb += gridDim.x) {
+    const long p0 = tb * PLW;
+    const int np = (int)min((long)PLW, planes - p0);
+    __syncthreads();
+    // stage np planes
+    for (int idx = threadIdx.x; idx < np * zt; idx += kBlock) {
+      const int pl = idx / zt;
+      xs[pl * zt + (idx - pl * zt)] = zt_ld(in + (p0 + pl) * (long)zt +
+                                            (idx - pl * zt));
+    }
+    __syncthreads();
+    // stage 1: B[pl][z][kt]
+    for (int idx = threadIdx.x; idx < np * Z * mt; idx += kBlock) {
+      const int pl = idx / (Z * mt);
+      const int r = idx - pl * (Z * mt);
+      const int z = r / mt;
+      const int kt = r - z * mt;
+      const float* row = xs + pl * zt + z * T;
+      float ar = 0.f, ai = 0.f;
+#pragma unroll
+      for (int t = 0; t < (CT ? CT : 64); ++t) {
+        if (!CT && t >= T) break;
+        const float x = row[t];
+        ar += x * twt[(t * mt + kt) * 2];
+        ai += x * twt[(t * mt + kt) * 2 + 1];
+      }
+      float* b = bs + ((size_t)pl * Z + z) * mt * 2 + kt * 2;
+      b[0] = ar;
+      b[1] = ai;
+    }
+    __syncthreads();
+    // stage 2: y[pl][kz][kt] with optional rfft factors on kt.  NOTE: the
+    // edge (DC / Nyquist) modes keep their imaginary part — the kt=0 column
+    // of the 2-D half-spectrum is hermitian ACROSS kz, not per-element
+    // real, and the synthesis side drops the imaginary component only
+    // after its z-transform (see zt_inv stage 2's Re()).
+    for (int idx = threadIdx.x; idx < np * mz * mt; idx += kBlock) {
+      const int pl = idx / (mz * mt);
+      const int r = idx - pl * (mz * mt);
+      const int kz = r / mt;
+      const int kt = r - kz * mt;
+      float ar = 0.f, ai = 0.f;
+      const float* bp = bs + (size_t)pl * Z * mt * 2 + kt * 2;
+#pragma unroll
+      for (int z = 0; z < (CZ ? CZ : 64); ++z) {
+        if (!CZ && z >= Z) break;
+        const float wr = twz[(z * mz + kz) * 2];
+        const float wi = twz[(z * mz + kz) * 2 + 1];
+        const float br = bp[(size_t)z * mt * 2];
+        const float bi = bp[(size_t)z * mt * 2 + 1];
+        ar += br * wr - bi * wi;
+        ai += br * wi + bi * wr;
+      }
+      const bool edge = (kt == 0) || (T % 2 == 0 && 2 * kt == T);
+      const float f = (factors && !edge) ? 2.f * scale : scale;
+      float* dst = out + 2 * ((p0 + pl) * (long)mz * mt + kz * mt + kt);
+      dst[0] = f * ar;
+      dst[1] = f * ai;
+    }
+  }
+}
+
+// Specialised flagship analysis kernel (Z=64, T=30, mz=24, mt=8, PLW=2).
+// The generic kernel is LDS-issue bound (3 LDS reads per 2 FMAs in stage 1);
+// here each thread owns one fixed kt (kBlock = 32*mt), register-caches its
+// twiddle column, and reads plane rows as float2 — ~4x fewer LDS ops.
+template <typename TI>
+__global__ __launch_bounds__(kBlock) void zt_fwd_pin_kernel(
+    const TI* __restrict__ in, float* __restrict__ out,
+    const float* __restrict__ gtwt, const float* __restrict__ gtwz,
+    long planes, float scale, bool factors) {
+  constexpr int Z = 64, T = 30, MZ = 24, MT = 8, PLW = 2;
+  constexpr int ZT = Z * T;
+  extern __shared__ __align__(16) char smem_raw[];
+  float* twt = reinterpret_cast<float*>(smem_raw);   // [T*MT*2]   = 480
+  float* twz = twt + T * MT * 2;                     // [Z*MZ*2]   = 3072
+  float* xs = twz + Z * MZ * 2;                      // [PLW*ZT]   = 3840
+  float* bs = xs + PLW * ZT;                         // [PLW*Z*MT*2] = 2048
+  for (int i = threadIdx.x; i < T * MT * 2; i += kBlock) twt[i] = gtwt[i];
+  for (int i = threadIdx.x; i < Z * MZ * 2; i += kBlock) twz[i] = gtwz[i];
+
+  const int kt = threadIdx.x & (MT - 1);
+  const int g = threadIdx.x >> 3;                    // 32 row groups
+  // register-cache this thread's twiddle column w_T[t][kt]
+  float wtr[T], wti[T];
+  __syncthreads();
+#pragma unroll
+  for (int t = 0; t < T; ++t) {
+    const float2 w = *reinterpret_cast<const float2*>(twt + (t * MT + kt) * 2);
+    wtr[t] = w.x;
+    wti[t] = w.y;
+  }
+
+  long ntile = (planes + PLW - 1) / PLW;
+  for (long tb = blockIdx.x; tb < ntile; tb += gridDim.x) {
+    const long p0 = tb * PLW;
+    const int np = (int)min((long)PLW, planes - p0);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < np * ZT; idx += kBlock)
+      xs[idx] = zt_ld(in + p0 * (long)ZT + idx);
+    __syncthreads();
+    // stage 1: B[pl][z][kt] — rows as float2, twiddles from registers
+    const int nrow = np * Z;
+    for (int plz = g; plz < nrow; plz += kBlock / MT) {
+      const float2* row2 =
+          reinterpret_cast<const float2*>(xs + (plz >> 6) * ZT +
+                                          (plz & 63) * T);
+      float ar = 0.f, ai = 0.f;
+#pragma unroll
+      for (int i = 0; i < T / 2; ++i) {
+        const float2 x2 = row2[i];
+        ar += x2.x * wtr[2 * i] + x2.y * wtr[2 * i + 1];
+        ai += x2.x * wti[2 * i] + x2.y * wti[2 * i + 1];
+      }
+      float* b = bs + plz * (MT * 2) + kt * 2;
+      b[0] = ar;
+      b[1] = ai;
+    }
+    __syncthreads();
+    // stage 2: y[pl][kz][kt] = sum_z B[z][kt] wz[z][kz]
+    for (int idx = threadIdx.x; idx < np * MZ * MT; idx += kBlock) {
+      const int pl = idx / (MZ * MT);
+      const int r = idx - pl * (MZ * MT);
+      const int kz = r >> 3;
+      const int kt2 = r & (MT - 1);
+      const float2* bp2 =
+          reinterpret_cast<const float2*>(bs + pl * Z * MT * 2) + kt2;
+      float ar = 0.f, ai = 0.f;
+#pragma unroll
+      for (int z = 0; z < Z; ++z) {
+        const float2 w =
+            *reinterpret_cast<const float2*>(twz + (z * MZ + kz) * 2);
+        const float2 b = bp2[z * MT];
+        ar += b.x * w.x - b.y * w.y;
+        ai += b.x * w.y + b.y * w.x;
+      }
+      const float f = (factors && kt2 != 0) ? 2.f * scale : scale;
+      float* dst = out + 2 * ((p0 + pl) * (long)MZ * MT + r);
+      dst[0] = f * ar;
+      dst[1] = f * ai;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// zt_inv: per plane, stage y [mz][mt] c64 in LDS, then
+//   A[z][kt] = sum_kz y[kz][kt] wz[z][kz]        (Z*mt outputs, mz cmacs)
+//   x[z][t] = Re( sum_kt fac_kt A[z][kt] wt[t][kt] )
+// ---------------------------------------------------------------------------
+
+template <typename TO, int PLW, int CZ, int CT, int CMZ, int CMT>
+__global__ __launch_bounds__(kBlock) void zt_inv_kernel(
+    const float* __restrict__ in, TO* __restrict__ out,
+    const float* __restrict__ gtwt,  // [T, mt, 2]  w_T^{+t kt}
+    const float* __restrict__ gtwz,  // [Z, mz, 2]  w_Z^{+z kz(i)}
+    long planes, int Z_, int T_, int mz_, int mt_, float scale, bool factors,
+    const float* __restrict__ acc = nullptr) {
+  const int Z = CZ ? CZ : Z_;
+  const int T = CT ? CT : T_;
+  const int mz = CMZ ? CMZ : mz_;
+  const int mt = CMT ? CMT : mt_;
+  extern __shared__ __align__(16) char smem_raw[];
+  float* twt = reinterpret_cast<float*>(smem_raw);      // [T*mt*2]
+  float* twz = twt + (size_t)T * mt * 2;                // [Z*mz*2]
+  float* ys = twz + (size_t)Z * mz * 2;                 // [PLW][mz*mt*2]
+  float* as = ys + (size_t)PLW * mz * mt * 2;           // [PLW][Z*mt*2]
+  for (int i = threadIdx.x; i < T * mt * 2; i += kBlock) twt[i] = gtwt[i];
+  for (int i = threadIdx.x; i < Z * mz * 2; i += kBlock) twz[i] = gtwz[i];
+
+  const int ymt = mz * mt * 2;
+  long ntile = (planes + PLW - 1) / PLW;
+  for (long tb = blockIdx.x; tb < ntile; tb += gridDim.x) {
+    const long p0 = tb * PLW;
+    const int np = (int)min((long)PLW, planes - p0);
+    __syncthreads();
+    // stage (with the rfft factor applied on input; edge-mode imaginary
+    // parts are KEPT — stage 2's Re() drops what must be dropped, matching
+    // the z-then-t synthesis composition exactly)
+    for (int idx = threadIdx.x; idx < np * ymt; idx += kBlock) {
+      const int pl = idx / ymt;
+      const int r = idx - pl * ymt;
+      float v = in[(p0 + pl) * (long)ymt + r];
+      if (factors) {
+        const int kt = (r >> 1) % mt;
+        const bool edge = (kt == 0) || (T % 2 == 0 && 2 * kt == T);
+        if (!edge) v *= 2.f;
+      }
+      ys[pl * ymt + r] = v * scale;
+    }
+    __syncthreads();
+    // stage 1: A[pl][z][kt]
+    for (int idx = threadIdx.x; idx < np * Z * mt; idx += kBlock) {
+      const int pl = idx / (Z * mt);
+      const int r = idx - pl * (Z * mt);
+      const int z = r / mt;
+      const int kt = r - z * mt;
+      float ar = 0.f, ai = 0.f;
+      const float* yp = ys + (size_t)pl * ymt + kt * 2;
+#pragma unroll
+      for (int kz = 0; kz < (CMZ ? CMZ : 48); ++kz) {
+        if (!CMZ && kz >= mz) break;
+        const float wr = twz[(z * mz + kz) * 2];
+        const float wi = twz[(z * mz + kz) * 2 + 1];
+        const float yr = yp[(size_t)kz * mt * 2];
+        const float yi = yp[(size_t)kz * mt * 2 + 1];
+        ar += yr * wr - yi * wi;
+        ai += yr * wi + yi * wr;
+      }
+      float* a = as + ((size_t)pl * Z + z) * mt * 2 + kt * 2;
+      a[0] = ar;
+      a[1] = ai;
+    }
+    __syncthreads();
+    // stage 2: x[pl][z][t]
+    for (int idx = threadIdx.x; idx < np * Z * T; idx += kBlock) {
+      const int pl = idx / (Z * T);
+      const int r = idx - pl * (Z * T);
+      const int z = r / T;
+      const int t = r - z * T;
+      float acc0 = 0.f;
+      const float* ap = as + ((size_t)pl * Z + z) * mt * 2;
+#pragma unroll
+      for (int kt = 0; kt < (CMT ? CMT : 32); ++kt) {
+        if (!CMT && kt >= mt) break;
+        acc0 += ap[kt * 2] * twt[(t * mt + kt) * 2] -
+                ap[kt * 2 + 1] * twt[(t * mt + kt) * 2 + 1];
+      }
+      if constexpr (std::is_same<TO, float>::value) {
+        if (acc != nullptr)      // fused residual-grad accumulate (stash)
+          acc0 += acc[(p0 + pl) * (long)Z * T + r];
+      }
+      TO* dst = out + (p0 + pl) * (long)Z * T + r;
+      if constexpr (std::is_same<TO, unsigned short>::value)
+        *dst = zt_f2b(acc0);
+      else
+        *dst = (TO)acc0;
+    }
+  }
+}
+
+// Specialised flagship synthesis kernel (Z=64, T=30, mz=24, mt=8, PLW=4).
+// PLW=4 makes np*Z == kBlock so stage 2 maps one thread per output row:
+// the A row (8 complex) is register-cached via 4 float4 loads and every
+// thread reads the same broadcast twt row per t.
+template <typename TO>
+__global__ __launch_bounds__(kBlock) void zt_inv_pin_kernel(
+    const float* __restrict__ in, TO* __restrict__ out,
+    const float* __restrict__ gtwt, const float* __restrict__ gtwz,
+    long planes, float scale, bool factors,
+    const float* __restrict__ acc = nullptr) {
+  constexpr int Z = 64, T = 30, MZ = 24, MT = 8, PLW = 4;
+  constexpr int YMT = MZ * MT * 2;                   // 384 floats / plane
+  extern __shared__ __align__(16) char smem_raw[];
+  float* twt = reinterpret_cast<float*>(smem_raw);   // [T*MT*2]   = 480
+  float* twz = twt + T * MT * 2;                     // [Z*MZ*2]   = 3072
+  float* ys = twz + Z * MZ * 2;                      // [PLW*YMT]  = 1536
+  float* as = ys + PLW * YMT;                        // [PLW*Z*MT*2] = 4096
+  for (int i = threadIdx.x; i < T * MT * 2; i += kBlock) twt[i] = gtwt[i];
+  for (int i = threadIdx.x; i < Z * MZ * 2; i += kBlock) twz[i] = gtwz[i];
+
+  const int kt = threadIdx.x & (MT - 1);
+
+  long ntile = (planes + PLW - 1) / PLW;
+  for (long tb = blockIdx.x; tb < ntile; tb += gridDim.x) {
+    const long p0 = tb * PLW;
+    const int np = (int)min((long)PLW, planes - p0);
+    __syncthreads();
+    for (int idx = threadIdx.x; idx < np * YMT; idx += kBlock) {
+      float v = in[p0 * (long)YMT + idx];
+      if (factors && ((idx >> 1) & (MT - 1)) != 0) v *= 2.f;
+      ys[idx] = v * scale;
+    }
+    __syncthreads();
+    // stage 1: A[pl][z][kt] = sum_kz y[kz][kt] wz[z][kz]
+    for (int idx = threadIdx.x; idx < np * Z * MT; idx += kBlock) {
+      const int pl = idx / (Z * MT);
+      const int r = idx - pl * (Z * MT);
+      const int z = r >> 3;
+      const float2* yp2 =
+          reinterpret_cast<const float2*>(ys + pl * YMT) + kt;
+      float ar = 0.f, ai = 0.f;
+#pragma unroll
+      for (int kz = 0; kz < MZ; ++kz) {
+        const float2 w =
+            *reinterpret_cast<const float2*>(twz + (z * MZ + kz) * 2);
+        const float2 y2 = yp2[kz * MT];
+        ar += y2.x * w.x - y2.y * w.y;
+        ai += y2.x * w.y + y2.y * w.x;
+      }
+      float* a = as + (pl * Z + z) * (MT * 2) + kt * 2;
+      a[0] = ar;
+      a[1] = ai;
+    }
+    __syncthreads();
+    // stage 2: one thread per (pl, z) row (np*Z == kBlock when np == PLW);
+    // the A row lives in registers and every thread reads the same
+    // broadcast twt row per t.  The 30 outputs are held in registers and
+    // drained through an LDS staging pass (reusing the now-dead ys/as
+    // space) so the global stores — and the optional stash accumulate —
+    // stay fully coalesced.
+    const int plz = threadIdx.x;                     // == pl*Z + z
+    float xreg[T];
+    if (plz < np * Z) {
+      float4 a4[MT / 2];
+      const float4* ap4 = reinterpret_cast<const float4*>(as + plz * MT * 2);
+#pragma unroll
+      for (int i = 0; i < MT / 2; ++i) a4[i] = ap4[i];
+#pragma unroll 6
+      for (int t = 0; t < T; ++t) {
+        const float4* w4 = reinterpret_cast<const float4*>(twt + t * MT * 2);
+        float acc0 = 0.f;
+#pragma unroll
+        for (int i = 0; i < MT / 2; ++i) {
+          const float4 w = w4[i];
+          acc0 += a4[i].x * w.x - a4[i].y * w.y +
+                  a4[i].z * w.z - a4[i].w * w.w;
+        }
+        xreg[t] = acc0;
+      }
+    }
+    // drain in two half-tile chunks of 128 rows (128*T = 3840 floats fits
+    // the ys+as span, 5632 floats)
+    float* stagebuf = ys;
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      __syncthreads();                 // as reads / prior copy-out done
+      const int r0 = half * (kBlock / 2);
+      const int nrow = min(np * Z - r0, kBlock / 2);
+      if (nrow <= 0) continue;
+      if (plz >= r0 && plz < r0 + nrow) {
+        float* srow = stagebuf + (plz - r0) * T;
+#pragma unroll
+        for (int t = 0; t < T; ++t) srow[t] = xreg[t];
+      }
+      __syncthreads();
+      const long gbase = (p0 * Z + r0) * (long)T;
+      for (int i = threadIdx.x; i < nrow * T; i += kBlock) {
+        float v = stagebuf[i];
+        if constexpr (std::is_same<TO, unsigned short>::value) {
+          reinterpret_cast<unsigned short*>(out)[gbase + i] = zt_f2b(v);
+        } else {
+          if (acc != nullptr) v += acc[gbase + i];
+          out[gbase + i] = (TO)v;
+        }
+      }
+    }
+  }
+}
+
+// twiddle tables: [N, m, 2] with the KEPT-mode ordering on the z dim
+// (k(i) = i for i < mz_lo else Z - mz + i), sign by direction.
+// Cached per (N, m, m_lo, dir, device): the hot loop re-issues the same
+// handful of configurations every step.
+static at::Tensor zt_table(int N, int m, int m_lo, bool analysis,
+                           const at::TensorOptions& opt) {
+  static std::map<std::tuple<int, int, int, int, int>, at::Tensor> cache;
+  auto key = std::make_tuple(N, m, m_lo, (int)analysis,
+                             (int)opt.device().index());
+  auto it = cache.find(key);
+  if (it != cache.end()) return it->second;
+  auto j = at::arange(N, at::kDouble).reshape({N, 1});
+  auto ki = at::arange(m, at::kDouble).reshape({1, m});
+  // map kept index -> global frequency
+  auto k = at::where(ki < (double)m_lo, ki, ki - (double)m + (double)N);
+  double sgn = analysis ? -1.0 : 1.0;
+  auto ang = (sgn * 2.0 * M_PI / N) * j * k;
+  auto t = at::stack({ang.cos(), ang.sin()}, 2).to(opt).contiguous();
+  cache.emplace(key, t);
+  return t;
+}
+
+}  // namespace
+
+at::Tensor dft_zt_fwd(const at::Tensor& x, int64_t mz_lo, int64_t mz_hi,
+                      int64_t mt, double scale, bool factors) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() >= 2,
+              "zt_fwd: contiguous GPU input");
+  const int T = (int)x.size(-1);
+  const int Z = (int)x.size(-2);
+  const int mz = (int)(mz_lo + mz_hi);
+  TORCH_CHECK(Z <= kMaxZ && T <= kMaxT && mz <= kMaxMZ && mt <= kMaxMT &&
+              mz <= Z && mt <= T / 2 + 1, "zt_fwd: bad sizes");
+  const bool bf16_in = x.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf16_in || x.scalar_type() == at::kFloat, "zt_fwd: fp32/bf16");
+  long planes = x.numel() / ((long)Z * T);
+
+  auto sizes = x.sizes().vec();
+  sizes[x.dim() - 2] = mz;
+  sizes[x.dim() - 1] = mt;
+  auto out = at::empty(sizes, x.options().dtype(at::kComplexFloat));
+  if (x.numel() == 0) return out;
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  auto fopt = x.options().dtype(at::kFloat);
+  auto twt = zt_table(T, (int)mt, (int)mt, /*analysis=*/true, fopt);
+  auto twz = zt_table(Z, mz, (int)mz_lo, /*analysis=*/true, fopt);
+  // flagship pin (specialised kernel, PLW=2) vs generic (PLW=2, runtime trips)
+  const bool pin = (Z == 64 && T == 30 && mz == 24 && mt == 8);
+  const int plw = 2;
+  size_t smem = sizeof(float) * ((size_t)T * mt * 2 + (size_t)Z * mz * 2 +
+                                 (size_t)plw * (Z * T + (size_t)Z * mt * 2));
+  long ntile = (planes + plw - 1) / plw;
+  int grid = (int)std::min(ntile, 4096L);
+  auto op = reinterpret_cast<float*>(out.data_ptr());
+#define ZT_FWD_PIN(TI)                                                       \
+  hipLaunchKernelGGL((zt_fwd_pin_kernel<TI>), dim3(grid), dim3(kBlock),      \
+                     smem, stream,                                           \
+                     reinterpret_cast<const TI*>(x.data_ptr()), op,          \
+                     twt.data_ptr<float>(), twz.data_ptr<float>(), planes,   \
+                     (float)scale, factors)
+#define ZT_FWD(TI)                                                           \
+  hipLaunchKernelGGL((zt_fwd_kernel<TI, 2, 0, 0, 0, 0>), dim3(grid),         \
+                     dim3(kBlock), smem, stream,                             \
+                     reinterpret_cast<const TI*>(x.data_ptr()), op,          \
+                     twt.data_ptr<float>(), twz.data_ptr<float>(), planes,   \
+                     Z, T, mz, (int)mt, (float)scale, factors)
+  if (bf16_in) {
+    if (pin) ZT_FWD_PIN(unsigned short);
+    else     ZT_FWD(unsigned short);
+  } else {
+    if (pin) ZT_FWD_PIN(float);
+    else     ZT_FWD(float);
+  }
+#undef ZT_FWD
+#undef ZT_FWD_PIN
+  DFNO_CHECK_LAUNCH("zt_fwd");
+  return out;
+}
+
+at::Tensor dft_zt_inv(const at::Tensor& y, int64_t Z, int64_t T,
+                      int64_t mz_lo, int64_t mz_hi, double scale, bool factors,
+                      bool out_bf16, const at::Tensor& accum) {
+  TORCH_CHECK(y.is_cuda() && y.is_contiguous() && y.dim() >= 2 &&
+              y.scalar_type() == at::kComplexFloat, "zt_inv: c64 GPU input");
+  const int mt = (int)y.size(-1);
+  const int mz = (int)y.size(-2);
+  TORCH_CHECK(mz == (int)(mz_lo + mz_hi), "zt_inv: mz mismatch");
+  TORCH_CHECK(Z <= kMaxZ && T <= kMaxT && mz <= kMaxMZ && mt <= kMaxMT &&
+              mz <= Z && mt <= T / 2 + 1, "zt_inv: bad sizes");
+  long planes = y.numel() / ((long)mz * mt);
+
+  auto sizes = y.sizes().vec();
+  sizes[y.dim() - 2] = Z;
+  sizes[y.dim() - 1] = T;
+  auto out = at::empty(sizes, y.options().dtype(
+      out_bf16 ? at::kBFloat16 : at::kFloat));
+  if (y.numel() == 0) return out;
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  auto fopt = y.options().dtype(at::kFloat);
+  auto twt = zt_table((int)T, mt, mt, /*analysis=*/false, fopt);
+  auto twz = zt_table((int)Z, mz, (int)mz_lo, /*analysis=*/false, fopt);
+  const bool pin = (Z == 64 && T == 30 && mz == 24 && mt == 8);
+  const int plw = pin ? 4 : 2;      // pin: np*Z == kBlock, row-per-thread
+  size_t smem = sizeof(float) * ((size_t)T * mt * 2 + (size_t)Z * mz * 2 +
+                                 (size_t)plw * ((size_t)mz * mt * 2 +
+                                                (size_t)Z * mt * 2));
+  long ntile = (planes + plw - 1) / plw;
+  int grid = (int)std::min(ntile, 4096L);
+  auto inp = reinterpret_cast<const float*>(y.data_ptr());
+  const float* accp = nullptr;
+  if (accum.defined() && accum.numel() > 0) {
+    TORCH_CHECK(!out_bf16 && accum.is_cuda() && accum.is_contiguous() &&
+                accum.scalar_type() == at::kFloat &&
+                accum.numel() == out.numel(), "zt_inv: bad accumulate tensor");
+    accp = accum.data_ptr<float>();
+  }
+#define ZT_INV_PIN(TO, OUTP, ACCP)                                           \
+  hipLaunchKernelGGL((zt_inv_pin_kernel<TO>), dim3(grid), dim3(kBlock),      \
+                     smem, stream, inp, OUTP, twt.data_ptr<float>(),         \
+                     twz.data_ptr<float>(), planes, (float)scale, factors,   \
+                     ACCP)
+#define ZT_INV(TO, OUTP, ACCP)                                               \
+  hipLaunchKernelGGL((zt_inv_kernel<TO, 2, 0, 0, 0, 0>), dim3(grid),         \
+                     dim3(kBlock), smem, stream, inp, OUTP,                  \
+                     twt.data_ptr<float>(), twz.data_ptr<float>(), planes,   \
+                     (int)Z, (int)T, mz, mt, (float)scale, factors, ACCP)
+  if (out_bf16) {
+    auto o = reinterpret_cast<unsigned short*>(out.data_ptr());
+    if (pin) ZT_INV_PIN(unsigned short, o, nullptr);
+    else     ZT_INV(unsigned short, o, nullptr);
+  } else {
+    auto o = out.data_ptr<float>();
+    if (pin) ZT_INV_PIN(float, o, accp);
+    else     ZT_INV(float, o, accp);
+  }
+#undef ZT_INV
+#undef ZT_INV_PIN
+  DFNO_CHECK_LAUNCH("zt_inv");
+  return out;
+}

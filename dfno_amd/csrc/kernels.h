@@ -81,6 +81,15 @@ at::Tensor dft_rfft_trunc_adj_acc(const at::Tensor& gy, int64_t dim, int64_t n,
                                   const at::Tensor& accum);
 at::Tensor dft_pad_irfft(const at::Tensor& y, int64_t dim, int64_t n_out, int64_t m);
 at::Tensor dft_pad_irfft_adj(const at::Tensor& gx, int64_t dim, int64_t m);
+// fused (z,t) boundary 2-D transforms (dft2d.hip): plane-resident in LDS,
+// replaces rfft_trunc(t)+fft_trunc(z) (and the inverse pair) without the
+// [L, Z, mt] intermediate
+at::Tensor dft_zt_fwd(const at::Tensor& x, int64_t mz_lo, int64_t mz_hi,
+                      int64_t mt, double scale, bool factors);
+at::Tensor dft_zt_inv(const at::Tensor& y, int64_t Z, int64_t T,
+                      int64_t mz_lo, int64_t mz_hi, double scale, bool factors,
+                      bool out_bf16, const at::Tensor& accum);
+
 // bf16-IO variants (bf16 real-side storage, fp32 compute, c64 spectrum):
 at::Tensor dft_pad_irfft_bf16(const at::Tensor& y, int64_t dim, int64_t n_out,
                               int64_t m);

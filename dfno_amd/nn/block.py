@@ -210,6 +210,17 @@ class DistributedFNOBlock(nn.Module):
             return torch.empty(0, dtype=self.dtype_complex, device=x.device,
                                requires_grad=torch.is_grad_enabled())
         outermost = self.dim_m[-1]
+        if len(self.dim_m) == 2:
+            # fused (z,t) plane transform: both trailing dims in one kernel
+            from ..ops.fft import zt_enabled, zt_fwd, zt_native_ok
+            z_dim = self.dim_m[0]
+            mzl = self.restrict_prefixes[z_dim]
+            mzh = self.restrict_suffixes.get(z_dim, 0)
+            mt = self.restrict_prefixes[outermost]
+            if zt_enabled() and zt_native_ok(x, mzl, mzh, mt):
+                saved[outermost] = x.shape[outermost] // 2 + 1
+                saved[z_dim] = x.shape[z_dim]
+                return zt_fwd(x, mzl, mzh, mt)
         if x.dtype == torch.bfloat16:
             from ..ops.fft import rfft_bf16_native_ok
             if not rfft_bf16_native_ok(x, outermost,
@@ -249,6 +260,16 @@ class DistributedFNOBlock(nn.Module):
             return torch.empty(0, dtype=self.dtype, device=y.device,
                                requires_grad=torch.is_grad_enabled())
         outermost = self.dim_m[-1]
+        if len(self.dim_m) == 2 and y.is_cuda and y.dtype == torch.complex64:
+            from ..ops.fft import zt_enabled, zt_inv
+            z_dim = self.dim_m[0]
+            mzl = self.restrict_prefixes[z_dim]
+            mzh = self.restrict_suffixes.get(z_dim, 0)
+            Z, T = saved[z_dim], self.in_shape[-1]
+            if (zt_enabled() and Z <= 64 and T <= 64
+                    and mzl + mzh <= min(48, Z)
+                    and y.shape[-1] <= 32 and y.shape[-2] == mzl + mzh):
+                return zt_inv(y, Z, T, mzl, mzh, out_dtype=self.dtype)
         for dim in self.dim_m[:-1]:
             y = pad_ifft(y, dim, saved[dim],
                          self.restrict_prefixes[dim],
@@ -335,7 +356,27 @@ class DistributedFNOBlock(nn.Module):
 
         x_epi = x_orig
         outermost = self.dim_m[-1]
-        if (x is x_orig
+        zt_stash = False
+        if (x is x_orig and len(self.dim_m) == 2 and x.is_cuda
+                and x.numel() > 0 and x.dtype == torch.float32
+                and torch.is_grad_enabled() and x.requires_grad):
+            from ..ops.fft import zt_enabled, zt_fwd, zt_native_ok
+            z_dim = self.dim_m[0]
+            mzl = self.restrict_prefixes[z_dim]
+            mzh = self.restrict_suffixes.get(z_dim, 0)
+            mt = self.restrict_prefixes[outermost]
+            if zt_enabled() and zt_native_ok(x, mzl, mzh, mt):
+                # fused (z,t) transform + residual-grad stash in its adjoint
+                key = new_stash_key()
+                saved[outermost] = x.shape[outermost] // 2 + 1
+                saved[z_dim] = x.shape[z_dim]
+                xm, tok = zt_fwd(x, mzl, mzh, mt, stash_key=key)
+                x_epi = StashGradFn.apply(x_orig, tok, key)
+                x = xm
+                zt_stash = True
+        if zt_stash:
+            pass
+        elif (x is x_orig
                 and stash_fusable(x, outermost, self.restrict_prefixes[outermost])):
             # R1 is an identity here: the chain's input-grad producer is the
             # rfft adjoint — fold the epilogue's gradient into its writeback

@@ -22,6 +22,8 @@ torch.fft ops with identical semantics.
 
 from __future__ import annotations
 
+import os
+
 import torch
 
 from .. import _ext
@@ -241,6 +243,109 @@ def stash_fusable(x, dim, m) -> bool:
             and torch.is_grad_enabled() and x.requires_grad
             and x.shape[d] <= _MAX_N and m <= 32
             and x.dtype in (torch.float32, torch.float64))
+
+
+# ---------------------------------------------------------------------------
+# fused (z,t) boundary 2-D transform (csrc/dft2d.hip)
+# ---------------------------------------------------------------------------
+#
+# zt_fwd = fft_trunc(z) . rfft_trunc(t) and zt_inv = pad_irfft(t) .
+# pad_ifft(z) in one plane-resident kernel each: the [L, Z, mt] complex
+# intermediate of the two-pass chain (0.34 GB at the flagship, 16
+# write+read passes per step over fwd/inv/adjoints) never exists.  The
+# stash (residual-grad accumulate) folds into the adjoint's writeback like
+# the 1-D c2r path.
+#
+# OFF BY DEFAULT (opt in with DFNO_ZT=1): measured on MI355X the fused
+# kernels run ~600 us/call x 16 calls/step at the flagship while the 1-D
+# chain they replace (r2c_glds 178 us + c2r_last 297 us + z-axis radix-8
+# c2c, all traffic-bound) totals ~5.2 ms/step — a net loss of ~4 ms.  The
+# direct z-contraction is VALU-issue-bound; the fusion only wins (~1 ms
+# ceiling) with MFMA-tiled z-stages.  profiles/optimization_log.md has the
+# numbers; the kernels stay correct + tested for that follow-up.
+
+_ZT_EMPTY = {}
+
+
+def _zt_empty(device):
+    t = _ZT_EMPTY.get(device)
+    if t is None:
+        t = torch.empty(0, device=device)
+        _ZT_EMPTY[device] = t
+    return t
+
+
+def zt_enabled() -> bool:
+    """Fused (z,t) model-path gate (env DFNO_ZT=1; default off, see above)."""
+    return os.environ.get("DFNO_ZT", "0") == "1"
+
+
+def zt_native_ok(x, mz_lo, mz_hi, mt) -> bool:
+    if not x.is_cuda or x.dim() < 2:
+        return False
+    Z, T = x.shape[-2], x.shape[-1]
+    mt = min(mt, T // 2 + 1)
+    return (x.dtype in (torch.float32, torch.bfloat16)
+            and Z <= 64 and T <= 64 and mz_lo + mz_hi <= min(48, Z)
+            and mt <= 32)
+
+
+class _ZtFwdFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, mz_lo, mz_hi, mt, stash_key):
+        ext = _ext.get(required=True)
+        ctx.dims = (x.shape[-2], x.shape[-1], mz_lo, mz_hi)
+        ctx.bf16 = x.dtype == torch.bfloat16
+        ctx.key = stash_key
+        y = ext.dft_zt_fwd(x.contiguous(), mz_lo, mz_hi, mt, 1.0, False)
+        if stash_key is not None:
+            tok = torch.empty(0, device=x.device)
+            return y, tok
+        return y
+
+    @staticmethod
+    def backward(ctx, gy, *rest):
+        ext = _ext.get(required=True)
+        Z, T, mz_lo, mz_hi = ctx.dims
+        acc = _zt_empty(gy.device)
+        if ctx.key is not None:
+            acc = _GRAD_STASH.pop(ctx.key, None)
+            if acc is None:
+                raise RuntimeError("zt stash: epilogue gradient missing")
+        gx = ext.dft_zt_inv(gy.contiguous(), Z, T, mz_lo, mz_hi, 1.0, False,
+                            ctx.bf16, acc if not ctx.bf16 else
+                            _zt_empty(gy.device))
+        return gx, None, None, None, None
+
+
+class _ZtInvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y, Z, T, mz_lo, mz_hi, out_bf16):
+        ext = _ext.get(required=True)
+        ctx.dims = (Z, T, mz_lo, mz_hi, y.shape[-1])
+        x = ext.dft_zt_inv(y.contiguous(), Z, T, mz_lo, mz_hi,
+                           1.0 / (Z * T), True, out_bf16, _zt_empty(y.device))
+        return x
+
+    @staticmethod
+    def backward(ctx, gx):
+        ext = _ext.get(required=True)
+        Z, T, mz_lo, mz_hi, mt = ctx.dims
+        gy = ext.dft_zt_fwd(gx.contiguous(), mz_lo, mz_hi, mt,
+                            1.0 / (Z * T), True)
+        return gy, None, None, None, None, None
+
+
+def zt_fwd(x, mz_lo, mz_hi, mt, stash_key=None):
+    """Fused truncated 2-D analysis over the trailing (z, t) dims."""
+    mt = min(mt, x.shape[-1] // 2 + 1)
+    return _ZtFwdFn.apply(x, mz_lo, mz_hi, mt, stash_key)
+
+
+def zt_inv(y, Z, T, mz_lo, mz_hi, out_dtype=None):
+    """Fused padded 2-D synthesis back to the (Z, T) extents."""
+    out_bf16 = out_dtype == torch.bfloat16
+    return _ZtInvFn.apply(y, Z, T, mz_lo, mz_hi, out_bf16)
 
 
 # ---------------------------------------------------------------------------

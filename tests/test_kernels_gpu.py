@@ -413,6 +413,72 @@ def test_dft_pad_irfft_bigN(dtype, tt, shape, n_half, n_out, m):
 
 
 # ---------------------------------------------------------------------------
+# fused (z,t) boundary 2-D transform vs the 1-D composition
+# ---------------------------------------------------------------------------
+
+@pytest.mark.parametrize("shape,mzl,mzh,mt,bf16", [
+    ((1, 5, 9, 64, 30), 12, 12, 8, False),   # flagship (z,t)
+    ((2, 3, 7, 48, 32), 8, 8, 9, False),     # even T edge mode in range
+    ((1, 4, 6, 64, 30), 12, 12, 8, True),    # bf16 IO
+])
+def test_zt_fused_transform(shape, mzl, mzh, mt, bf16):
+    from dfno_amd.ops.fft import (zt_fwd, zt_inv, _t_rfft_trunc, _t_fft_trunc,
+                                  _t_pad_ifft, _t_pad_irfft)
+    torch.manual_seed(13)
+    Z, T = shape[-2], shape[-1]
+    dt = torch.bfloat16 if bf16 else torch.float32
+    tol = 2e-2 if bf16 else 3e-4
+    x = torch.randn(*shape, device="cuda").to(dt).requires_grad_(True)
+    y = zt_fwd(x, mzl, mzh, mt)
+    xr = x.detach().float().requires_grad_(True)
+    yr = _t_fft_trunc(_t_rfft_trunc(xr, -1, mt), -2, mzl, mzh)
+    assert torch.allclose(y, yr, rtol=tol, atol=tol * 10), \
+        f"zt fwd {(y-yr).abs().max()}"
+    g = torch.randn_like(y)
+    y.backward(g)
+    yr.backward(g)
+    assert torch.allclose(x.grad.float(), xr.grad, rtol=tol, atol=tol * 10), \
+        f"zt fwd bwd {(x.grad.float()-xr.grad).abs().max()}"
+
+    # inverse: kept modes -> (Z, T) real
+    z = torch.randn(*shape[:-2], mzl + mzh, mt, dtype=torch.complex64,
+                    device="cuda").requires_grad_(True)
+    w = zt_inv(z, Z, T, mzl, mzh, out_dtype=dt)
+    assert w.dtype == dt
+    zr = z.detach().clone().requires_grad_(True)
+    wr = _t_pad_irfft(_t_pad_ifft(zr, -2, Z, mzl, mzh), -1, T // 2 + 1, T, mt)
+    assert torch.allclose(w.float(), wr, rtol=tol, atol=tol * 10), \
+        f"zt inv {(w.float()-wr).abs().max()}"
+    g2 = torch.randn_like(w)
+    w.backward(g2)
+    wr.backward(g2.float())
+    assert torch.allclose(z.grad, zr.grad, rtol=tol, atol=tol * 10), \
+        f"zt inv bwd {(z.grad-zr.grad).abs().max()}"
+
+
+def test_zt_stash_accumulate():
+    """The zt adjoint's fused accumulate (residual-grad stash) matches the
+    unfused add."""
+    from dfno_amd.ops.fft import zt_fwd, StashGradFn, new_stash_key
+    torch.manual_seed(14)
+    shape, mzl, mzh, mt = (1, 4, 8, 64, 30), 12, 12, 8
+    x = torch.randn(*shape, device="cuda", requires_grad=True)
+    key = new_stash_key()
+    y, tok = zt_fwd(x, mzl, mzh, mt, stash_key=key)
+    x_epi = StashGradFn.apply(x, tok, key)
+    out = y.abs().square().sum() + (x_epi * 3.0).sum()
+    out.backward()
+    g_fused = x.grad.clone()
+
+    xr = x.detach().clone().requires_grad_(True)
+    yr = zt_fwd(xr, mzl, mzh, mt)
+    outr = yr.abs().square().sum() + (xr * 3.0).sum()
+    outr.backward()
+    assert torch.allclose(g_fused, xr.grad, rtol=1e-5, atol=1e-5), \
+        f"stash acc {(g_fused-xr.grad).abs().max()}"
+
+
+# ---------------------------------------------------------------------------
 # fused Adam vs torch.optim.Adam
 # ---------------------------------------------------------------------------
 

@@ -162,3 +162,39 @@ def test_pack_unpack_boxes_roundtrip_gpu():
             sl = tuple(slice(a, c) for a, c in b)
             y_ref[sl] = x[sl]
         assert torch.equal(y, y_ref)
+
+
+@pytest.mark.parametrize("in_shape,t_out,modes", [
+    ([1, 2, 8, 8, 64, 1], 30, (3, 3, 12, 8)),   # flagship (z,t): pinned kernels
+    ([1, 2, 10, 10, 32, 1], 24, (3, 3, 8, 6)),  # generic-shape kernels
+])
+def test_fno_zt_fused_path_matches_default(monkeypatch, in_shape, t_out, modes):
+    """DFNO_ZT=1 routes the trailing (z, t) pair through the fused 2-D
+    kernels (csrc/dft2d.hip, incl. the stash); output and grads must match
+    the default 1-D chain."""
+    torch.manual_seed(3)
+    _, P_x, _ = dfno.create_standard_partitions((1,) * len(in_shape))
+    dev = torch.device("cuda")
+
+    def run(zt):
+        monkeypatch.setenv("DFNO_ZT", "1" if zt else "0")
+        torch.manual_seed(7)
+        model = dfno.DistributedFNONd(P_x, in_shape, t_out, 8, modes,
+                                      num_blocks=2, device=dev)
+        x = torch.rand(*in_shape, device=dev, requires_grad=True)
+        y = model(x)
+        loss = y.square().mean()
+        loss.backward()
+        return (y.detach().cpu(), x.grad.cpu(),
+                {n: p.grad.cpu() for n, p in model.named_parameters()
+                 if p.grad is not None})
+
+    y0, gx0, gp0 = run(False)
+    y1, gx1, gp1 = run(True)
+    assert torch.allclose(y1, y0, rtol=1e-4, atol=1e-4), \
+        f"y {(y1-y0).abs().max()}"
+    assert torch.allclose(gx1, gx0, rtol=1e-4, atol=1e-4), \
+        f"gx {(gx1-gx0).abs().max()}"
+    for n in gp0:
+        assert torch.allclose(gp1[n], gp0[n], rtol=1e-4, atol=1e-3), \
+            f"{n}: {(gp1[n]-gp0[n]).abs().max()}"
