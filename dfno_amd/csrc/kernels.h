@@ -40,6 +40,13 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
 std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& W3, const at::Tensor& b3,
                                       const at::Tensor& W4);
+// fully-fused flagship backward (I=20, M=128, O2<=2, fp32): returns
+// {gx, gW3, gb3, gW4, gb4} without materializing gz3
+std::vector<at::Tensor> proj_head_bwd_fused(const at::Tensor& gy,
+                                            const at::Tensor& x,
+                                            const at::Tensor& W3,
+                                            const at::Tensor& b3,
+                                            const at::Tensor& W4);
 
 // split-s outer-product reduction: gW[o,i] = sum_{b,s} gz[b,o,s] x[b,i,s]
 // (+ gb[o] = sum gz when want_bias); requires I <= 32.

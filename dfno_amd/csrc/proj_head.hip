@@ -319,6 +319,244 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fully-fused flagship backward (I=20, M=128, O2<=2, fp32): one kernel
+// produces gx, gW3, gb3, gW4, gb4 with the [M, S] hidden-grad living only
+// as a [M x 64] LDS tile — the 4 GB gz3 intermediate of the three-kernel
+// chain (proj_head_bwd 1.67 ms + channel_mix_fwd_t 1.22 ms + gw_mfma
+// 1.85 ms, kernel_stats_r02_final.csv) is never written to HBM.  Traffic
+// drops from ~13 GB to read x + gy, write gx (~1.3 GB).  grad-W3 runs as
+// v_mfma_f32_16x16x4 on the LDS tile with per-wave fragment accumulators
+// carried across tiles (identical numerics to the library chain up to
+// fp32 atomic reduction order, like every gw kernel here).
+// ---------------------------------------------------------------------------
+
+typedef float f32x4_ph __attribute__((ext_vector_type(4)));
+
+// sum across each 16-lane segment of the wave
+__device__ __forceinline__ float seg_sum16(float v) {
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+template <int IT, int MT, int O2T>
+__global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
+    const float* __restrict__ gy, const float* __restrict__ x,
+    const float* __restrict__ W3, const float* __restrict__ b3,
+    const float* __restrict__ W4,
+    float* __restrict__ gx, float* __restrict__ gW3,
+    float* __restrict__ gb3, float* __restrict__ gW4,
+    float* __restrict__ gb4, int B, int O2, long S) {
+  constexpr int TS = 64;           // s-columns per tile (16 MFMA K-steps)
+  constexpr int LD = TS + 4;       // row pad (float4-aligned, 4-bank skew)
+  extern __shared__ __align__(16) char smem_raw[];
+  float* gzt = reinterpret_cast<float*>(smem_raw);   // [MT][LD]
+  float* xt = gzt + (size_t)MT * LD;                 // [IT][LD]
+  float* gyt = xt + (size_t)IT * LD;                 // [O2T][TS]
+  float* W3l = gyt + O2T * TS;                       // [MT*IT]
+  float* b3l = W3l + (size_t)MT * IT;                // [MT]
+  float* W4l = b3l + MT;                             // [O2T*MT]
+  float* gb3s = W4l + O2T * MT;                      // [MT]
+  float* gW4s = gb3s + MT;                           // [O2T*MT]
+#pragma clang loop unroll(disable)
+  for (int k = threadIdx.x; k < MT * IT; k += kBlock) W3l[k] = W3[k];
+  for (int k = threadIdx.x; k < MT; k += kBlock) {
+    b3l[k] = b3[k];
+    gb3s[k] = 0.f;
+  }
+  for (int k = threadIdx.x; k < O2T * MT; k += kBlock) {
+    W4l[k] = W4[k];
+    gW4s[k] = 0.f;
+  }
+
+  const int lane = (int)(threadIdx.x & 63);
+  const int wave = (int)(threadIdx.x >> 6);
+  const int l16 = lane & 15;
+  const int kg = lane >> 4;
+
+  f32x4_ph wacc[4];                // gW3 frags: 8 mt-tiles x 2 nt-tiles
+#pragma unroll
+  for (int p = 0; p < 4; ++p) wacc[p] = f32x4_ph{0.f, 0.f, 0.f, 0.f};
+  float gb4a0 = 0.f, gb4a1 = 0.f;
+
+  const long stiles = (S + TS - 1) / TS;
+  const long tend = (long)B * stiles;
+  // software-pipelined staging: tile t+grid's x/gy columns are loaded into
+  // registers while tile t computes (the PMC counters put 45% of wave
+  // cycles parked on the staging round-trip when it sits between the
+  // barriers).
+  constexpr int NPF = ((IT + 2) * TS + kBlock - 1) / kBlock;
+  float pf[NPF];
+  auto prefetch = [&](long tt) {
+    if (tt >= tend) return;
+    const int b = (int)(tt / stiles);
+    const long s0 = (tt % stiles) * TS;
+    const int nv = (int)min((long)TS, S - s0);
+#pragma unroll
+    for (int q = 0; q < NPF; ++q) {
+      const int r = (int)threadIdx.x + q * kBlock;
+      if (r >= (IT + O2T) * TS) break;
+      const int row = r / TS;
+      const int c = r - row * TS;
+      float v = 0.f;
+      if (c < nv) {
+        v = (row < IT) ? x[((long)b * IT + row) * S + s0 + c]
+                       : gy[((long)b * O2T + (row - IT)) * S + s0 + c];
+      }
+      pf[q] = v;
+    }
+  };
+  prefetch(blockIdx.x);
+  for (long t = blockIdx.x; t < tend; t += gridDim.x) {
+    const int b = (int)(t / stiles);
+    const long s0 = (t % stiles) * TS;
+    const int nv = (int)min((long)TS, S - s0);
+    __syncthreads();               // prior tile's phase-2 reads done
+#pragma unroll
+    for (int q = 0; q < NPF; ++q) {
+      const int r = (int)threadIdx.x + q * kBlock;
+      if (r >= (IT + O2T) * TS) break;
+      const int row = r / TS;
+      const int c = r - row * TS;
+      const float v = pf[q];
+      if (row < IT) {
+        xt[row * LD + c] = v;
+      } else {
+        gyt[(row - IT) * TS + c] = v;
+        if (row == IT) gb4a0 += v; else gb4a1 += v;
+      }
+    }
+    prefetch(t + gridDim.x);       // next tile's loads fly over this tile
+    __syncthreads();
+    // phase 1: gz tile.  Each PAIR of threads owns one hidden channel j
+    // (2 x 4 s-columns per pass, TS/8 passes): the W3 row and the gb3/gW4
+    // partials live in registers across the tile, so the only cross-lane
+    // reduction is one shfl_xor(1) per tile (the per-iteration 16-lane
+    // shuffle chains were the dominant latency at 3 waves/SIMD).
+    {
+      const int jj = (int)(threadIdx.x >> 1);        // 0..127
+      const int half = (int)(threadIdx.x & 1) * 4;   // 0 or 4
+      const float bj = b3l[jj];
+      const float w40 = W4l[jj];
+      const float w41 = (O2T == 2) ? W4l[MT + jj] : 0.f;
+      float w3r[IT];
+#pragma unroll
+      for (int i = 0; i < IT; ++i) w3r[i] = W3l[jj * IT + i];
+      float pb = 0.f, pw0 = 0.f, pw1 = 0.f;
+#pragma unroll
+      for (int k = 0; k < TS / 8; ++k) {
+        const int c4 = half + k * 8;
+        float z[4] = {bj, bj, bj, bj};
+#pragma unroll
+        for (int i = 0; i < IT; ++i) {
+          const float4 xv =
+              *reinterpret_cast<const float4*>(xt + i * LD + c4);
+          z[0] += w3r[i] * xv.x; z[1] += w3r[i] * xv.y;
+          z[2] += w3r[i] * xv.z; z[3] += w3r[i] * xv.w;
+        }
+        float g[4], dg[4];
+#pragma unroll
+        for (int q = 0; q < 4; ++q)
+          dfno_gelu::gelu_and_grad(z[q], g[q], dg[q]);
+        const float4 gy4 = *reinterpret_cast<const float4*>(gyt + c4);
+        float gz[4] = {w40 * gy4.x, w40 * gy4.y, w40 * gy4.z, w40 * gy4.w};
+        pw0 += gy4.x * g[0] + gy4.y * g[1] + gy4.z * g[2] + gy4.w * g[3];
+        if (O2T == 2) {
+          const float4 gy4b =
+              *reinterpret_cast<const float4*>(gyt + TS + c4);
+          gz[0] += w41 * gy4b.x; gz[1] += w41 * gy4b.y;
+          gz[2] += w41 * gy4b.z; gz[3] += w41 * gy4b.w;
+          pw1 += gy4b.x * g[0] + gy4b.y * g[1] + gy4b.z * g[2] +
+                 gy4b.w * g[3];
+        }
+#pragma unroll
+        for (int q = 0; q < 4; ++q) gz[q] *= dg[q];
+        *reinterpret_cast<float4*>(gzt + jj * LD + c4) =
+            make_float4(gz[0], gz[1], gz[2], gz[3]);
+        pb += gz[0] + gz[1] + gz[2] + gz[3];
+      }
+      pb += __shfl_xor(pb, 1, 64);
+      pw0 += __shfl_xor(pw0, 1, 64);
+      if (O2T == 2) pw1 += __shfl_xor(pw1, 1, 64);
+      if ((threadIdx.x & 1) == 0) {
+        atomicAdd(&gb3s[jj], pb);
+        atomicAdd(&gW4s[jj], pw0);
+        if (O2T == 2) atomicAdd(&gW4s[MT + jj], pw1);
+      }
+    }
+    __syncthreads();
+    // phase 2a: gx tile = W3^T @ gz via MFMA.  D[i][c] = sum_j A[i][j]B[c][j]
+    // with A[m][k] = W3l[k*IT + m], B[n][k] = gzt[k*LD + n]; 2 m-tiles x
+    // 4 n-tiles split across the 4 waves (2 pairs each), K = MT.
+#pragma unroll
+    for (int pp = 0; pp < 2; ++pp) {
+      const int p = wave + 4 * pp;
+      const int mt = p >> 2, nt = p & 3;
+      const int m = mt * 16 + l16;
+      const int n = nt * 16 + l16;
+      const bool av = m < IT;
+      f32x4_ph c4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 8
+      for (int k = 0; k < MT; k += 4) {
+        const float a = av ? W3l[(k + kg) * IT + m] : 0.f;
+        const float bb = gzt[(k + kg) * LD + n];
+        c4 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, c4, 0, 0, 0);
+      }
+      const long sc = s0 + nt * 16 + l16;
+      if (sc < S) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int i = mt * 16 + kg * 4 + r;
+          if (i < IT) gx[((long)b * IT + i) * S + sc] = c4[r];
+        }
+      }
+    }
+    // phase 2b: gW3 fragments += gz_tile @ x_tile^T (reads only; the tile
+    // loop's top sync fences the next staging pass)
+#pragma unroll
+    for (int pp = 0; pp < 4; ++pp) {
+      const int p = wave + 4 * pp;
+      const int mt = p >> 1, nt = p & 1;
+      const float* gr = gzt + (mt * 16 + l16) * LD;
+      const float* xr = xt + (nt * 16 + l16) * LD;
+      const bool bv = (nt * 16 + l16) < IT;
+#pragma unroll 8
+      for (int k = 0; k < TS; k += 4) {
+        const float a = gr[k + kg];
+        const float bb = bv ? xr[k + kg] : 0.f;
+        wacc[pp] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, wacc[pp],
+                                                        0, 0, 0);
+      }
+    }
+  }
+
+  // flush: gW3 fragments, gb4 wave sums, gb3/gW4 LDS partials
+#pragma unroll
+  for (int pp = 0; pp < 4; ++pp) {
+    const int p = wave + 4 * pp;
+    const int mt = p >> 1, nt = p & 1;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int j = mt * 16 + kg * 4 + r;
+      const int i = nt * 16 + l16;
+      if (i < IT && wacc[pp][r] != 0.f)
+        atomicAdd(&gW3[(size_t)j * IT + i], wacc[pp][r]);
+    }
+  }
+  float v0 = wave_sum(gb4a0);
+  if (lane == 0 && v0 != 0.f) atomicAdd(&gb4[0], v0);
+  if (O2T == 2) {
+    float v1 = wave_sum(gb4a1);
+    if (lane == 0 && v1 != 0.f) atomicAdd(&gb4[1], v1);
+  }
+  __syncthreads();
+  for (int k = threadIdx.x; k < MT; k += kBlock)
+    if (gb3s[k] != 0.f) atomicAdd(&gb3[k], gb3s[k]);
+  for (int k = threadIdx.x; k < O2T * MT; k += kBlock)
+    if (gW4s[k] != 0.f) atomicAdd(&gW4[k], gW4s[k]);
+}
+
 int grid_for_p(long work) {
   long g = (work + kBlock - 1) / kBlock;
   long cap = 256L * 8;
@@ -451,4 +689,47 @@ std::vector<at::Tensor> proj_head_bwd(const at::Tensor& gy, const at::Tensor& x,
   DFNO_CHECK_LAUNCH("proj_head");
 #undef PH_LAUNCH_B
   return {gz3, gb3, gW4, gb4};
+}
+
+std::vector<at::Tensor> proj_head_bwd_fused(const at::Tensor& gy,
+                                            const at::Tensor& x,
+                                            const at::Tensor& W3,
+                                            const at::Tensor& b3,
+                                            const at::Tensor& W4) {
+  check_pf(gy, "gy"); check_pf(x, "x"); check_pf(W3, "W3");
+  check_pf(b3, "b3"); check_pf(W4, "W4");
+  int B = (int)x.size(0), I = (int)x.size(1);
+  long S = x.size(2);
+  int M = (int)W3.size(0), O2 = (int)W4.size(0);
+  TORCH_CHECK(I == 20 && M == 128 && O2 <= 2 &&
+              x.scalar_type() == at::kFloat,
+              "proj_head_bwd_fused: flagship shape only");
+
+  auto gx = at::empty({B, I, S}, x.options());
+  auto gW3 = at::zeros({M, I}, x.options());
+  auto gb3 = at::zeros({M}, x.options());
+  auto gW4 = at::zeros({O2, M}, x.options());
+  auto gb4 = at::zeros({O2}, x.options());
+  if (x.numel() == 0) return {gx, gW3, gb3, gW4, gb4};
+
+  auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+  constexpr int TS = 64, LD = TS + 4;
+  // gzt + xt + gyt + W3l + b3l + W4l + gb3s + gW4s (kernel layout order)
+  size_t smem = sizeof(float) *
+      ((size_t)128 * LD + 20 * LD + (size_t)O2 * TS + (size_t)128 * 20 +
+       128 + 2 * (size_t)O2 * 128 + 128);
+  long stiles = (S + TS - 1) / TS;
+  int grid = (int)std::min((long)B * stiles, 768L);
+#define PH_FUSED(O2T)                                                        \
+  hipLaunchKernelGGL((proj_head_bwd_fused_kernel<20, 128, O2T>), dim3(grid), \
+                     dim3(kBlock), smem, stream, gy.data_ptr<float>(),       \
+                     x.data_ptr<float>(), W3.data_ptr<float>(),              \
+                     b3.data_ptr<float>(), W4.data_ptr<float>(),             \
+                     gx.data_ptr<float>(), gW3.data_ptr<float>(),            \
+                     gb3.data_ptr<float>(), gW4.data_ptr<float>(),           \
+                     gb4.data_ptr<float>(), B, O2, S)
+  if (O2 == 2) PH_FUSED(2); else PH_FUSED(1);
+#undef PH_FUSED
+  DFNO_CHECK_LAUNCH("proj_head_bwd_fused");
+  return {gx, gW3, gb3, gW4, gb4};
 }

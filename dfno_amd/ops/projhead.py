@@ -37,11 +37,18 @@ class _ProjHeadFn(torch.autograd.Function):
     def backward(ctx, gy):
         x3, W3, b3, W4 = ctx.saved_tensors
         ext = _ext.get(required=True)
-        gz3, gb3, gW4, gb4 = ext.proj_head_bwd(
+        if (x3.shape[1] == 20 and W3.shape[0] == 128 and W4.shape[0] <= 2
+                and x3.dtype == torch.float32):
+            # flagship: one kernel, no [B,128,S] gz3 intermediate in HBM
+            gx, gW3, gb3, gW4g, gb4 = ext.proj_head_bwd_fused(
+                gy.contiguous(), x3, W3.contiguous(), b3.contiguous(),
+                W4.contiguous())
+            return gx.reshape(ctx.x_shape), gW3, gb3, gW4g, gb4
+        gz3, gb3, gW4g, gb4 = ext.proj_head_bwd(
             gy.contiguous(), x3, W3.contiguous(), b3.contiguous(), W4.contiguous())
         gx = ext.channel_mix_fwd_t(gz3, W3.contiguous())   # W3^T @ gz3
         gW3, _ = ext.channel_mix_bwd_w(gz3, x3, False)     # gz3 @ x^T
-        return gx.reshape(ctx.x_shape), gW3, gb3, gW4, gb4
+        return gx.reshape(ctx.x_shape), gW3, gb3, gW4g, gb4
 
 
 def proj_head(x: torch.Tensor, W3, b3, W4, b4) -> torch.Tensor:

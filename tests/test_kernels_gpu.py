@@ -219,8 +219,10 @@ def test_op_spectral_autograd_gpu():
 
 @pytest.mark.parametrize("dtype,tt", [(torch.float32, 3e-4), (torch.float64, 1e-10)])
 @pytest.mark.parametrize("B,I,M,O2,S", [
-    (1, 20, 128, 1, 4096),   # flagship head shape
-    (2, 12, 64, 2, 1001),    # odd S (scalar path), multi-batch, O2=2
+    (1, 20, 128, 1, 4096),    # flagship head shape -> fused backward (fp32)
+    (2, 12, 64, 2, 1001),     # odd S (scalar path), multi-batch, O2=2
+    (2, 20, 128, 2, 4099),    # fused backward: O2=2, tail tile, B=2
+    (1, 20, 128, 1, 51233),   # fused backward: > 768 tiles (grid wrap) + tail
 ])
 def test_proj_head(dtype, tt, B, I, M, O2, S):
     from dfno_amd.ops import proj_head
