@@ -40,11 +40,18 @@ __device__ __forceinline__ T lh_wave_sum(T v) {
 
 // x: [B, C, S] (T_in == 1 folded away), out: [B, W, S, T]
 // W1: [T, 1], b1: [T], W2: [W, C], b2: [W]
+//
+// Both kernels chunk the time axis (KC columns at a time): the first cut
+// held h/z1/gh[CCAP][TCAP] per thread (384 floats -> 256 VGPRs + ~600
+// spilled, 1 wave/SIMD, every element round-tripping scratch).  Chunked,
+// the live set is ~2*CCAP*KC + temps and z1 is recomputed (2 FMA) where
+// its gradient factor is needed.
 template <typename T, int CCAP, int TCAP, int WCAP, int WT = 0>
-__global__ __launch_bounds__(kBlock) void lift_head_fwd_kernel(
+__global__ __launch_bounds__(kBlock, 3) void lift_head_fwd_kernel(
     const T* __restrict__ x, const T* __restrict__ W1, const T* __restrict__ b1,
     const T* __restrict__ W2, const T* __restrict__ b2, T* __restrict__ out,
     int B, int C, int W, int Tn, long S) {
+  constexpr int KC = 10;
   __shared__ T w1[TCAP], bb1[TCAP], w2[WCAP * CCAP], bb2[WCAP];
   for (int k = threadIdx.x; k < Tn; k += kBlock) { w1[k] = W1[k]; bb1[k] = b1[k]; }
   for (int k = threadIdx.x; k < W * C; k += kBlock) w2[k] = W2[k];
@@ -62,50 +69,53 @@ __global__ __launch_bounds__(kBlock) void lift_head_fwd_kernel(
     for (int c = 0; c < CCAP; ++c)
       if (c < C) xv[c] = x[(b * C + c) * S + s];
 
-    // h[c][t] = gelu(W1[t] * x[c] + b1[t])
-    T h[CCAP][TCAP];
-#pragma unroll
-    for (int c = 0; c < CCAP; ++c) {
-      if (c < C) {
-#pragma unroll
-        for (int k = 0; k < TCAP; ++k)
-          if (k < Tn) h[c][k] = gelu_(w1[k] * xv[c] + bb1[k]);
-      }
-    }
-
-    #pragma unroll 4
-    for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
-      if (WT == 0 && w >= W) break;
-      T acc[TCAP];
-#pragma unroll
-      for (int k = 0; k < TCAP; ++k)
-        if (k < Tn) acc[k] = bb2[w];
+    for (int k0 = 0; k0 < Tn; k0 += KC) {
+      const int kn = min(KC, Tn - k0);
+      // h[c][k] = gelu(W1[k] * x[c] + b1[k]) for this chunk
+      T h[CCAP][KC];
 #pragma unroll
       for (int c = 0; c < CCAP; ++c) {
         if (c < C) {
-          T wv = w2[(size_t)w * C + c];
 #pragma unroll
-          for (int k = 0; k < TCAP; ++k)
-            if (k < Tn) acc[k] += wv * h[c][k];
+          for (int k = 0; k < KC; ++k)
+            if (k < kn) h[c][k] = gelu_(w1[k0 + k] * xv[c] + bb1[k0 + k]);
         }
       }
-      T* dst = out + ((b * W + w) * S + s) * Tn;
+#pragma unroll 4
+      for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
+        if (WT == 0 && w >= W) break;
+        T acc[KC];
 #pragma unroll
-      for (int k = 0; k < TCAP; ++k)
-        if (k < Tn) dst[k] = gelu_(acc[k]);
+        for (int k = 0; k < KC; ++k)
+          if (k < kn) acc[k] = bb2[w];
+#pragma unroll
+        for (int c = 0; c < CCAP; ++c) {
+          if (c < C) {
+            T wv = w2[(size_t)w * C + c];
+#pragma unroll
+            for (int k = 0; k < KC; ++k)
+              if (k < kn) acc[k] += wv * h[c][k];
+          }
+        }
+        T* dst = out + ((b * W + w) * S + s) * Tn + k0;
+#pragma unroll
+        for (int k = 0; k < KC; ++k)
+          if (k < kn) dst[k] = gelu_(acc[k]);
+      }
     }
   }
 }
 
 // gy: [B, W, S, T]; outputs gx [B, C, S] and gW1/gb1/gW2/gb2 via atomics.
 template <typename T, int CCAP, int TCAP, int WCAP, int WT = 0>
-__global__ __launch_bounds__(kBlock) void lift_head_bwd_kernel(
+__global__ __launch_bounds__(kBlock, 3) void lift_head_bwd_kernel(
     const T* __restrict__ gy, const T* __restrict__ x,
     const T* __restrict__ W1, const T* __restrict__ b1,
     const T* __restrict__ W2, const T* __restrict__ b2,
     T* __restrict__ gx, T* __restrict__ gW1, T* __restrict__ gb1,
     T* __restrict__ gW2, T* __restrict__ gb2,
     int B, int C, int W, int Tn, long S) {
+  constexpr int KC = 10;
   __shared__ T w1[TCAP], bb1[TCAP], w2[WCAP * CCAP], bb2[WCAP];
   // per-wave partial accumulators
   __shared__ T a_gW1[4][TCAP], a_gb1[4][TCAP], a_gW2[4][WCAP * CCAP], a_gb2[4][WCAP];
@@ -129,98 +139,106 @@ __global__ __launch_bounds__(kBlock) void lift_head_bwd_kernel(
     long b = t / S;
     long s = t % S;
 
-    T xv[CCAP];
-#pragma unroll
-    for (int c = 0; c < CCAP; ++c)
-      if (c < C) xv[c] = x[(b * C + c) * S + s];
-
-    T z1[CCAP][TCAP], h[CCAP][TCAP], gh[CCAP][TCAP];
+    T xv[CCAP], gxa[CCAP];
 #pragma unroll
     for (int c = 0; c < CCAP; ++c) {
       if (c < C) {
-#pragma unroll
-        for (int k = 0; k < TCAP; ++k) {
-          if (k < Tn) {
-            z1[c][k] = w1[k] * xv[c] + bb1[k];
-            h[c][k] = gelu_(z1[c][k]);
-            gh[c][k] = T(0);
-          }
-        }
+        xv[c] = x[(b * C + c) * S + s];
+        gxa[c] = T(0);
       }
     }
 
-    #pragma unroll 4
-    for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
-      if (WT == 0 && w >= W) break;
-      const T* gyp = gy + ((b * W + w) * S + s) * Tn;
-      T gz2[TCAP];
-      T gb2p = T(0);
-#pragma unroll
-      for (int k = 0; k < TCAP; ++k) {
-        if (k < Tn) {
-          // recompute z2[w][k]
-          T z2 = bb2[w];
-#pragma unroll
-          for (int c = 0; c < CCAP; ++c)
-            if (c < C) z2 += w2[(size_t)w * C + c] * h[c][k];
-          gz2[k] = gyp[k] * gelu_g_(z2);
-          gb2p += gz2[k];
-        }
-      }
-      // accumulate gh and wave-reduced gW2 partials
+    for (int k0 = 0; k0 < Tn; k0 += KC) {
+      const int kn = min(KC, Tn - k0);
+      T h[CCAP][KC], gh[CCAP][KC];
 #pragma unroll
       for (int c = 0; c < CCAP; ++c) {
         if (c < C) {
-          T wv = w2[(size_t)w * C + c];
-          T gw2p = T(0);
 #pragma unroll
-          for (int k = 0; k < TCAP; ++k) {
-            if (k < Tn) {
-              gh[c][k] += wv * gz2[k];
-              gw2p += gz2[k] * h[c][k];
+          for (int k = 0; k < KC; ++k) {
+            if (k < kn) {
+              h[c][k] = gelu_(w1[k0 + k] * xv[c] + bb1[k0 + k]);
+              gh[c][k] = T(0);
             }
           }
-          gw2p = lh_wave_sum(gw2p);
-          if (lane == 0) a_gW2[wave][w * C + c] += gw2p;
         }
       }
-      gb2p = lh_wave_sum(gb2p);
-      if (lane == 0) a_gb2[wave][w] += gb2p;
-    }
 
-    // gz1 = gh * gelu'(z1); gx[c] = sum_k W1[k] gz1[c][k];
-    // gW1[k] += sum_c gz1[c][k] * x[c]; gb1[k] += sum_c gz1[c][k]
+#pragma unroll 4
+      for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
+        if (WT == 0 && w >= W) break;
+        const T* gyp = gy + ((b * W + w) * S + s) * Tn + k0;
+        T gz2[KC];
+        T gb2p = T(0);
 #pragma unroll
-    for (int c = 0; c < CCAP; ++c) {
-      if (c < C) {
-        T gxa = T(0);
+        for (int k = 0; k < KC; ++k) {
+          if (k < kn) {
+            // recompute z2[w][k]
+            T z2 = bb2[w];
 #pragma unroll
-        for (int k = 0; k < TCAP; ++k) {
-          if (k < Tn) {
-            T gz1 = gh[c][k] * gelu_g_(z1[c][k]);
-            gh[c][k] = gz1;  // reuse as gz1 for the reductions below
-            gxa += w1[k] * gz1;
+            for (int c = 0; c < CCAP; ++c)
+              if (c < C) z2 += w2[(size_t)w * C + c] * h[c][k];
+            gz2[k] = gyp[k] * gelu_g_(z2);
+            gb2p += gz2[k];
           }
         }
-        gx[(b * C + c) * S + s] = gxa;
-      }
-    }
-#pragma unroll
-    for (int k = 0; k < TCAP; ++k) {
-      if (k < Tn) {
-        T pw = T(0), pb = T(0);
+        // accumulate gh and wave-reduced gW2 partials
 #pragma unroll
         for (int c = 0; c < CCAP; ++c) {
           if (c < C) {
-            pw += gh[c][k] * xv[c];
-            pb += gh[c][k];
+            T wv = w2[(size_t)w * C + c];
+            T gw2p = T(0);
+#pragma unroll
+            for (int k = 0; k < KC; ++k) {
+              if (k < kn) {
+                gh[c][k] += wv * gz2[k];
+                gw2p += gz2[k] * h[c][k];
+              }
+            }
+            gw2p = lh_wave_sum(gw2p);
+            if (lane == 0) a_gW2[wave][w * C + c] += gw2p;
           }
         }
-        pw = lh_wave_sum(pw);
-        pb = lh_wave_sum(pb);
-        if (lane == 0) { a_gW1[wave][k] += pw; a_gb1[wave][k] += pb; }
+        gb2p = lh_wave_sum(gb2p);
+        if (lane == 0) a_gb2[wave][w] += gb2p;
+      }
+
+      // gz1 = gh * gelu'(z1); gx[c] += sum_k W1[k] gz1[c][k];
+      // gW1[k] += sum_c gz1[c][k] * x[c]; gb1[k] += sum_c gz1[c][k]
+#pragma unroll
+      for (int c = 0; c < CCAP; ++c) {
+        if (c < C) {
+#pragma unroll
+          for (int k = 0; k < KC; ++k) {
+            if (k < kn) {
+              T gz1 = gh[c][k] * gelu_g_(w1[k0 + k] * xv[c] + bb1[k0 + k]);
+              gh[c][k] = gz1;  // reuse as gz1 for the reductions below
+              gxa[c] += w1[k0 + k] * gz1;
+            }
+          }
+        }
+      }
+#pragma unroll
+      for (int k = 0; k < KC; ++k) {
+        if (k < kn) {
+          T pw = T(0), pb = T(0);
+#pragma unroll
+          for (int c = 0; c < CCAP; ++c) {
+            if (c < C) {
+              pw += gh[c][k] * xv[c];
+              pb += gh[c][k];
+            }
+          }
+          pw = lh_wave_sum(pw);
+          pb = lh_wave_sum(pb);
+          if (lane == 0) { a_gW1[wave][k0 + k] += pw; a_gb1[wave][k0 + k] += pb; }
+        }
       }
     }
+
+#pragma unroll
+    for (int c = 0; c < CCAP; ++c)
+      if (c < C) gx[(b * C + c) * S + s] = gxa[c];
   }
 
   __syncthreads();
@@ -237,6 +255,193 @@ __global__ __launch_bounds__(kBlock) void lift_head_bwd_kernel(
   for (int k = threadIdx.x; k < W; k += kBlock) {
     T v = a_gb2[0][k] + a_gb2[1][k] + a_gb2[2][k] + a_gb2[3][k];
     if (v != T(0)) atomicAdd(&gb2[k], v);
+  }
+}
+
+// fp32 lane-per-k variants: each 64-lane wave covers TWO s-points with
+// lanes 0..Tn-1 / 32..32+Tn-1 owning one time column each, so every gy
+// read and out write is a contiguous Tn-dword run (the per-thread layout
+// above streams 30-float runs per lane -> fully scattered wave accesses;
+// measured 699/843 us vs ~110/130 us of traffic).  Per-lane state is
+// ~40 VGPRs; gW1/gb1 partials accumulate in registers (k fixed per lane).
+template <int WT>
+__global__ __launch_bounds__(kBlock) void lift_head_fwd_lk_kernel(
+    const float* __restrict__ x, const float* __restrict__ W1,
+    const float* __restrict__ b1, const float* __restrict__ W2,
+    const float* __restrict__ b2, float* __restrict__ out,
+    int B, int C, int W, int Tn, long S) {
+  constexpr int CC = 4;
+  __shared__ float w1[32], bb1[32], w2[24 * CC], bb2[24];
+  for (int k = threadIdx.x; k < Tn; k += kBlock) { w1[k] = W1[k]; bb1[k] = b1[k]; }
+  for (int k = threadIdx.x; k < W * C; k += kBlock) w2[k] = W2[k];
+  for (int k = threadIdx.x; k < W; k += kBlock) bb2[k] = b2[k];
+  __syncthreads();
+
+  const int lane = (int)(threadIdx.x & 63);
+  const int half = lane >> 5;                 // 0 or 1: which s of the pair
+  const int k = lane & 31;
+  const bool kv = k < Tn;
+  const long npair = ((long)B * S + 1) / 2;
+  long t0 = (long)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+  long stride = (long)gridDim.x * (blockDim.x / 64);
+  for (long t = t0; t < npair; t += stride) {
+    const long e = 2 * t + half;              // element index
+    const bool ev = e < (long)B * S;
+    const long b = e / S;
+    const long s = e - b * S;
+
+    float h[CC];
+#pragma unroll
+    for (int c = 0; c < CC; ++c) {
+      if (c < C) {
+        const float xv = ev ? x[(b * C + c) * S + s] : 0.f;
+        h[c] = kv ? gelu_(w1[k] * xv + bb1[k]) : 0.f;
+      }
+    }
+#pragma unroll 4
+    for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
+      if (WT == 0 && w >= W) break;
+      float acc = bb2[w];
+#pragma unroll
+      for (int c = 0; c < CC; ++c)
+        if (c < C) acc += w2[w * C + c] * h[c];
+      if (ev && kv)
+        out[((b * W + w) * S + s) * Tn + k] = gelu_(acc);
+    }
+  }
+}
+
+typedef float f32x4_lh __attribute__((ext_vector_type(4)));
+
+template <int WT>
+__global__ __launch_bounds__(kBlock) void lift_head_bwd_lk_kernel(
+    const float* __restrict__ gy, const float* __restrict__ x,
+    const float* __restrict__ W1, const float* __restrict__ b1,
+    const float* __restrict__ W2, const float* __restrict__ b2,
+    float* __restrict__ gx, float* __restrict__ gW1, float* __restrict__ gb1,
+    float* __restrict__ gW2, float* __restrict__ gb2,
+    int B, int C, int W, int Tn, long S) {
+  constexpr int CC = 4;
+  constexpr int GLD = 68;                     // gz/h tile row pad (banks)
+  __shared__ float w1[32], bb1[32], w2[24 * CC], bb2[24];
+  // per-wave tiles: gz [24 w][64 lanes] and h [4 c][64 lanes]; the gW2 and
+  // gb2 reductions run as v_mfma_f32_16x16x4 over these (B-operand column
+  // 4 is a constant ones column giving gb2) with the C fragments carried
+  // in registers across pairs — the per-pair 6-shuffle wave_sum chains of
+  // the first cut were the whole kernel's latency.
+  __shared__ float gzt[4][24 * GLD];
+  __shared__ float ht[4][CC * GLD];
+  for (int k = threadIdx.x; k < Tn; k += kBlock) { w1[k] = W1[k]; bb1[k] = b1[k]; }
+  for (int k = threadIdx.x; k < W * C; k += kBlock) w2[k] = W2[k];
+  for (int k = threadIdx.x; k < W; k += kBlock) bb2[k] = b2[k];
+  __syncthreads();
+
+  const int wave = (int)(threadIdx.x >> 6);
+  const int lane = (int)(threadIdx.x & 63);
+  const int half = lane >> 5;
+  const int k = lane & 31;
+  const bool kv = k < Tn;
+  const int l16 = lane & 15;
+  const int kg = lane >> 4;
+  float pgW1 = 0.f, pgb1 = 0.f;               // per-lane: k is fixed
+  f32x4_lh wacc[2];                           // gW2/gb2 frags (2 m-tiles)
+  wacc[0] = f32x4_lh{0.f, 0.f, 0.f, 0.f};
+  wacc[1] = f32x4_lh{0.f, 0.f, 0.f, 0.f};
+  float* gzw = &gzt[wave][0];
+  float* hw = &ht[wave][0];
+
+  const long npair = ((long)B * S + 1) / 2;
+  long t0 = (long)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+  long stride = (long)gridDim.x * (blockDim.x / 64);
+  for (long t = t0; t < npair; t += stride) {
+    const long e = 2 * t + half;
+    const bool ev = e < (long)B * S;
+    const long b = e / S;
+    const long s = e - b * S;
+
+    float xv[CC], h[CC], gh[CC];
+#pragma unroll
+    for (int c = 0; c < CC; ++c) {
+      if (c < C) {
+        xv[c] = ev ? x[(b * C + c) * S + s] : 0.f;
+        h[c] = kv ? gelu_(w1[k] * xv[c] + bb1[k]) : 0.f;
+        gh[c] = 0.f;
+        hw[c * GLD + lane] = h[c];
+      }
+    }
+    // all W gy columns preloaded back-to-back: one memory round trip per
+    // pair instead of one per unroll batch (the loads were the kernel's
+    // whole latency — all reduction variants measured the same ~900 us)
+    float gyv[WT > 0 ? WT : 24];
+#pragma unroll
+    for (int w = 0; w < (WT > 0 ? WT : 24); ++w) {
+      gyv[w] = (w < W && ev && kv) ? gy[((b * W + w) * S + s) * Tn + k]
+                                   : 0.f;
+    }
+#pragma unroll 4
+    for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
+      if (WT == 0 && w >= W) break;
+      float z2 = bb2[w];
+#pragma unroll
+      for (int c = 0; c < CC; ++c)
+        if (c < C) z2 += w2[w * C + c] * h[c];
+      const float gz2 = gyv[w] * gelu_g_(z2);
+      gzw[w * GLD + lane] = gz2;
+#pragma unroll
+      for (int c = 0; c < CC; ++c)
+        if (c < C) gh[c] += w2[w * C + c] * gz2;
+    }
+    // gW2/gb2 fragment update from this pair's tiles (wave-synchronous:
+    // same wave wrote gzt/ht just above)
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      const int wrow = mt * 16 + l16;
+      const bool av = wrow < W;
+#pragma unroll 4
+      for (int k0 = 0; k0 < 64; k0 += 4) {
+        const float a = av ? gzw[wrow * GLD + k0 + kg] : 0.f;
+        const float bb = (l16 < C) ? hw[l16 * GLD + k0 + kg]
+                                   : (l16 == C ? 1.f : 0.f);
+        wacc[mt] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, wacc[mt],
+                                                        0, 0, 0);
+      }
+    }
+    // gz1 = gh * gelu'(z1); per-lane gW1/gb1 partials; gx via 32-lane
+    // segmented reduce (k lanes of each s-half)
+#pragma unroll
+    for (int c = 0; c < CC; ++c) {
+      if (c < C) {
+        const float gz1 = kv ? gh[c] * gelu_g_(w1[k] * xv[c] + bb1[k]) : 0.f;
+        pgW1 += gz1 * xv[c];
+        pgb1 += gz1;
+        float gxa = kv ? w1[k] * gz1 : 0.f;
+#pragma unroll
+        for (int off = 16; off > 0; off >>= 1)
+          gxa += __shfl_xor(gxa, off, 64);
+        if (ev && k == 0) gx[(b * C + c) * S + s] = gxa;
+      }
+    }
+  }
+
+  // flush: pair the two s-halves for gW1/gb1, then one atomic per (wave, k);
+  // gW2/gb2 from the MFMA fragments (D[m=w][n]: n<C -> gW2, n==C -> gb2)
+  pgW1 += __shfl_xor(pgW1, 32, 64);
+  pgb1 += __shfl_xor(pgb1, 32, 64);
+  if (half == 0 && kv) {
+    atomicAdd(&gW1[k], pgW1);
+    atomicAdd(&gb1[k], pgb1);
+  }
+#pragma unroll
+  for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int w = mt * 16 + kg * 4 + r;
+      const float v = wacc[mt][r];
+      if (w < W && v != 0.f) {
+        if (l16 < C) atomicAdd(&gW2[w * C + l16], v);
+        else if (l16 == C) atomicAdd(&gb2[w], v);
+      }
+    }
   }
 }
 
@@ -273,6 +478,24 @@ at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
   if (x.numel() == 0) return out;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_l((long)B * S);
+  if (x.scalar_type() == at::kFloat) {
+    int grid2 = grid_for_l(2 * (((long)B * S + 1) / 2) * 32);
+    if (W == 20) {
+      hipLaunchKernelGGL((lift_head_fwd_lk_kernel<20>), dim3(grid2),
+                         dim3(kBlock), 0, stream, x.data_ptr<float>(),
+                         W1.data_ptr<float>(), b1.data_ptr<float>(),
+                         W2.data_ptr<float>(), b2.data_ptr<float>(),
+                         out.data_ptr<float>(), B, C, W, Tn, S);
+    } else {
+      hipLaunchKernelGGL((lift_head_fwd_lk_kernel<0>), dim3(grid2),
+                         dim3(kBlock), 0, stream, x.data_ptr<float>(),
+                         W1.data_ptr<float>(), b1.data_ptr<float>(),
+                         W2.data_ptr<float>(), b2.data_ptr<float>(),
+                         out.data_ptr<float>(), B, C, W, Tn, S);
+    }
+    DFNO_CHECK_LAUNCH("lift_head");
+    return out;
+  }
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "lift_head_fwd", [&] {
     if (W == 20) {
       hipLaunchKernelGGL((lift_head_fwd_kernel<scalar_t, 4, 32, 24, 20>), dim3(grid),
@@ -310,6 +533,30 @@ std::vector<at::Tensor> lift_head_bwd(const at::Tensor& gy, const at::Tensor& x,
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_l((long)B * S);
+  if (x.scalar_type() == at::kFloat) {
+    int grid2 = grid_for_l(2 * (((long)B * S + 1) / 2) * 32);
+    if (W == 20) {
+      hipLaunchKernelGGL((lift_head_bwd_lk_kernel<20>), dim3(grid2),
+                         dim3(kBlock), 0, stream, gy.data_ptr<float>(),
+                         x.data_ptr<float>(), W1.data_ptr<float>(),
+                         b1.data_ptr<float>(), W2.data_ptr<float>(),
+                         b2.data_ptr<float>(), gx.data_ptr<float>(),
+                         gW1.data_ptr<float>(), gb1.data_ptr<float>(),
+                         gW2.data_ptr<float>(), gb2.data_ptr<float>(),
+                         B, C, W, Tn, S);
+    } else {
+      hipLaunchKernelGGL((lift_head_bwd_lk_kernel<0>), dim3(grid2),
+                         dim3(kBlock), 0, stream, gy.data_ptr<float>(),
+                         x.data_ptr<float>(), W1.data_ptr<float>(),
+                         b1.data_ptr<float>(), W2.data_ptr<float>(),
+                         b2.data_ptr<float>(), gx.data_ptr<float>(),
+                         gW1.data_ptr<float>(), gb1.data_ptr<float>(),
+                         gW2.data_ptr<float>(), gb2.data_ptr<float>(),
+                         B, C, W, Tn, S);
+    }
+    DFNO_CHECK_LAUNCH("lift_head");
+    return {gx, gW1, gb1, gW2, gb2};
+  }
   AT_DISPATCH_FLOATING_TYPES(x.scalar_type(), "lift_head_bwd", [&] {
     if (W == 20) {
       hipLaunchKernelGGL((lift_head_bwd_kernel<scalar_t, 4, 32, 24, 20>), dim3(grid),
