@@ -33,6 +33,7 @@ SOURCES = [
     _CSRC / "pack.hip",
     _CSRC / "bf16.hip",
     _CSRC / "dft2d.hip",
+    _CSRC / "mix_bwd.hip",
 ]
 
 

@@ -19,6 +19,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("proj_head_fwd", &proj_head_fwd, "fused linear->gelu->linear head");
   m.def("proj_head_bwd", &proj_head_bwd,
         "fused head backward: returns (gz3, gb3, gW4, gb4)");
+  m.def("channel_mix_bwd_fused", &channel_mix_bwd_fused,
+        "fused trunk mix backward: (gx, gW, gb, gz)");
   m.def("proj_head_bwd_fused", &proj_head_bwd_fused,
         "fully-fused flagship head backward: (gx, gW3, gb3, gW4, gb4)");
   m.def("lift_head_fwd", &lift_head_fwd, "fused time-lift + channel-lift + gelus");

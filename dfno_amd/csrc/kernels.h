@@ -53,6 +53,15 @@ std::vector<at::Tensor> proj_head_bwd_fused(const at::Tensor& gy,
 std::vector<at::Tensor> channel_mix_bwd_w(const at::Tensor& gz, const at::Tensor& x,
                                           bool want_bias);
 
+// fused trunk 20x20 mix backward (mix_bwd.hip): gz = gy * gelu'(z) lives
+// only as an LDS tile; returns {gx, gW, gb, gz} (gb empty unless
+// want_bias, gz empty unless want_gz — the linear_res_gelu residual grad)
+std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
+                                              const at::Tensor& z,
+                                              const at::Tensor& x,
+                                              const at::Tensor& W,
+                                              bool want_bias, bool want_gz);
+
 // fused lift head (T_in == 1): y = gelu(W2 @ gelu(W1 *t x + b1) + b2)
 at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
                          const at::Tensor& b1, const at::Tensor& W2,
