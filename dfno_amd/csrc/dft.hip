@@ -631,9 +631,11 @@ __global__ __launch_bounds__(kBlock) void dft_c2c_radix8_syn_kernel(
 // changes nothing).  Staging tile t+1 by global_load_lds DMA while t is
 // being consumed, with a counted s_waitcnt vmcnt(NG) + raw barriers, runs
 // the same shape at 5.4 TB/s -- 2.3x the staged version (r2cprobe.hip).
-template <typename T, int MCAP, int NG, int NT = 0>
+// TI = unsigned short DMAs the raw bf16 tile (half the LDS and HBM bytes)
+// and converts at the compute read; T stays the fp32 math type.
+template <typename T, int MCAP, int NG, int NT = 0, typename TI = T>
 __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
-    const T* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
+    const TI* __restrict__ in, T* __restrict__ out, const T* __restrict__ tw,
     long lines, int N_, int m, T scale, bool factors) {
   // NT > 0 pins the transform length AND the mode count (m == MCAP, host-
   // checked) at compile time: the j-loop fully unrolls with all twiddle
@@ -643,10 +645,11 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
   // the folded version runs 0.18 ms at the flagship shape (r2cprobe.hip).
   const int N = NT > 0 ? NT : N_;
   const int mm = NT > 0 ? MCAP : m;
-  if constexpr (!std::is_same<T, float>::value) return;  // float-only path
+  if constexpr (!std::is_same<T, float>::value) return;  // float-math path
+  constexpr int EPV = 16 / (int)sizeof(TI);          // elements per 16B DMA
   extern __shared__ __align__(16) char smem_raw[];
-  float* ring = reinterpret_cast<float*>(smem_raw);  // [2][kBlock * N]
-  const int tilef = kBlock * N;                      // floats per tile
+  TI* ring = reinterpret_cast<TI*>(smem_raw);        // [2][kBlock * N]
+  const int tilef = kBlock * N;                      // elements per tile
   // the counted s_waitcnt needs a compile-time-exact per-wave instruction
   // count: host picks NG = ceil(N/4) rounded up to a supported value and
   // issues are padded to exactly NG glds (pad slots re-stage the tile's
@@ -657,15 +660,15 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
   long ntiles = (lines + kBlock - 1) / kBlock;
 
   auto issue = [&](int buf, long tb) {
-    const float* src = reinterpret_cast<const float*>(in);
+    const TI* src = in;
     const long base = tb * (long)tilef;
-    float* dst = ring + (long)buf * tilef;
+    TI* dst = ring + (long)buf * tilef;
 #pragma unroll
     for (int k = 0; k < NG; ++k) {
-      long fo = (long)(wave * 64 + k * kBlock + lane) * 4;
-      if (fo + 4 > tilef) fo = tilef - 4;            // pad: clamp in-tile
+      long fo = (long)(wave * 64 + k * kBlock + lane) * EPV;
+      if (fo + EPV > tilef) fo = tilef - EPV;        // pad: clamp in-tile
       long gfo = base + fo;
-      if (gfo + 4 > nfl) gfo = nfl - 4;              // clamp last tile
+      if (gfo + EPV > nfl) gfo = nfl - EPV;          // clamp last tile
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)(src + gfo),
           (__attribute__((address_space(3))) void*)&dst[fo], 16, 0, 0);
@@ -687,8 +690,8 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
 
-    const float* tile = ring + (long)(c & 1) * tilef;
-    const float* src = tile + threadIdx.x * N;
+    const TI* tile = ring + (long)(c & 1) * tilef;
+    const TI* src = tile + threadIdx.x * N;
     T ar[MCAP], ai[MCAP];
 #pragma unroll
     for (int k = 0; k < MCAP; ++k) {
@@ -697,7 +700,11 @@ __global__ __launch_bounds__(kBlock) void dft_r2c_glds_kernel(
 #pragma unroll
     for (int j = 0; j < (NT > 0 ? NT : 64); ++j) {
       if (NT == 0 && j >= N) break;
-      const T x = (T)src[j];
+      T x;
+      if constexpr (std::is_same<TI, unsigned short>::value)
+        x = dft_b2f(src[j]);
+      else
+        x = (T)src[j];
       auto twj = (const __attribute__((address_space(4))) T*)
           (tw + (long)(j * 2) * mm);
 #pragma unroll
@@ -1134,6 +1141,31 @@ static at::Tensor dft_r2c_impl(const at::Tensor& x, int64_t dim, int64_t m,
     // half the bytes and skip the boundary-cast pass entirely
     auto inp = reinterpret_cast<const unsigned short*>(x.data_ptr());
     auto op = reinterpret_cast<float*>(out.data_ptr());
+    bool glds_ok = N <= 64 &&
+                   ((reinterpret_cast<uintptr_t>(inp) & 15) == 0) &&
+                   ((long)kBlock * N) % 8 == 0 && lines * N >= 8;
+    bool mexact = (m == 8 || m == 16 || m == 24 || m == 32);
+    if (glds_ok && mexact && (N == 30 || N == 64 || N == 40 || N == 32)) {
+      // DMA the raw bf16 tile (glds ring at half the fp32 footprint)
+      size_t smem2 = 2 * sizeof(unsigned short) * (size_t)kBlock * N;
+#define R2CGB(MC, NGV, NTV)                                                    \
+      hipLaunchKernelGGL((dft_r2c_glds_kernel<float, MC, NGV, NTV,             \
+                                              unsigned short>),                \
+                         dim3(grid), dim3(kBlock), smem2, stream, inp, op,     \
+                         tw.data_ptr<float>(), lines, N, (int)m,               \
+                         (float)scale, factors);
+#define R2CGB_M(NGV, NTV)                                                      \
+      if (m == 8) { R2CGB(8, NGV, NTV) } else if (m == 16) { R2CGB(16, NGV, NTV) } \
+      else if (m == 24) { R2CGB(24, NGV, NTV) } else { R2CGB(32, NGV, NTV) }
+      if (N == 30) { R2CGB_M(4, 30) }
+      else if (N == 64) { R2CGB_M(8, 64) }
+      else if (N == 40) { R2CGB_M(8, 40) }
+      else { R2CGB_M(4, 32) }
+#undef R2CGB_M
+#undef R2CGB
+      DFNO_CHECK_LAUNCH("dft_r2c_bf16");
+      return out;
+    }
 #define R2C_BF(MC, LBV)                                                        \
     { long nt2 = (lines + LBV - 1) / LBV;                                      \
       int grid2 = (int)std::min(nt2, 2048L);                                   \

@@ -15,6 +15,7 @@
 // one atomic flush per block (proj_head pattern).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
@@ -30,6 +31,16 @@ __device__ __forceinline__ T gelu_(T z) { return dfno_gelu::gelu(z); }
 
 template <typename T>
 __device__ __forceinline__ T gelu_g_(T z) { return dfno_gelu::gelu_grad(z); }
+
+__device__ __forceinline__ float lh_ld(const float* p) { return *p; }
+__device__ __forceinline__ float lh_ld(const unsigned short* p) {
+  return __uint_as_float(((unsigned int)*p) << 16);
+}
+__device__ __forceinline__ void lh_st(float* p, float v) { *p = v; }
+__device__ __forceinline__ void lh_st(unsigned short* p, float v) {
+  __hip_bfloat16 h = __float2bfloat16(v);
+  *p = *reinterpret_cast<unsigned short*>(&h);
+}
 
 template <typename T>
 __device__ __forceinline__ T lh_wave_sum(T v) {
@@ -264,11 +275,11 @@ __global__ __launch_bounds__(kBlock, 3) void lift_head_bwd_kernel(
 // above streams 30-float runs per lane -> fully scattered wave accesses;
 // measured 699/843 us vs ~110/130 us of traffic).  Per-lane state is
 // ~40 VGPRs; gW1/gb1 partials accumulate in registers (k fixed per lane).
-template <int WT>
+template <int WT, typename TIO = float>
 __global__ __launch_bounds__(kBlock) void lift_head_fwd_lk_kernel(
-    const float* __restrict__ x, const float* __restrict__ W1,
+    const TIO* __restrict__ x, const float* __restrict__ W1,
     const float* __restrict__ b1, const float* __restrict__ W2,
-    const float* __restrict__ b2, float* __restrict__ out,
+    const float* __restrict__ b2, TIO* __restrict__ out,
     int B, int C, int W, int Tn, long S) {
   constexpr int CC = 4;
   __shared__ float w1[32], bb1[32], w2[24 * CC], bb2[24];
@@ -294,7 +305,7 @@ __global__ __launch_bounds__(kBlock) void lift_head_fwd_lk_kernel(
 #pragma unroll
     for (int c = 0; c < CC; ++c) {
       if (c < C) {
-        const float xv = ev ? x[(b * C + c) * S + s] : 0.f;
+        const float xv = ev ? lh_ld(x + (b * C + c) * S + s) : 0.f;
         h[c] = kv ? gelu_(w1[k] * xv + bb1[k]) : 0.f;
       }
     }
@@ -306,19 +317,19 @@ __global__ __launch_bounds__(kBlock) void lift_head_fwd_lk_kernel(
       for (int c = 0; c < CC; ++c)
         if (c < C) acc += w2[w * C + c] * h[c];
       if (ev && kv)
-        out[((b * W + w) * S + s) * Tn + k] = gelu_(acc);
+        lh_st(out + ((b * W + w) * S + s) * Tn + k, gelu_(acc));
     }
   }
 }
 
 typedef float f32x4_lh __attribute__((ext_vector_type(4)));
 
-template <int WT>
+template <int WT, typename TIO = float>
 __global__ __launch_bounds__(kBlock) void lift_head_bwd_lk_kernel(
-    const float* __restrict__ gy, const float* __restrict__ x,
+    const TIO* __restrict__ gy, const TIO* __restrict__ x,
     const float* __restrict__ W1, const float* __restrict__ b1,
     const float* __restrict__ W2, const float* __restrict__ b2,
-    float* __restrict__ gx, float* __restrict__ gW1, float* __restrict__ gb1,
+    TIO* __restrict__ gx, float* __restrict__ gW1, float* __restrict__ gb1,
     float* __restrict__ gW2, float* __restrict__ gb2,
     int B, int C, int W, int Tn, long S) {
   constexpr int CC = 4;
@@ -363,7 +374,7 @@ __global__ __launch_bounds__(kBlock) void lift_head_bwd_lk_kernel(
 #pragma unroll
     for (int c = 0; c < CC; ++c) {
       if (c < C) {
-        xv[c] = ev ? x[(b * C + c) * S + s] : 0.f;
+        xv[c] = ev ? lh_ld(x + (b * C + c) * S + s) : 0.f;
         h[c] = kv ? gelu_(w1[k] * xv[c] + bb1[k]) : 0.f;
         gh[c] = 0.f;
         hw[c * GLD + lane] = h[c];
@@ -375,8 +386,8 @@ __global__ __launch_bounds__(kBlock) void lift_head_bwd_lk_kernel(
     float gyv[WT > 0 ? WT : 24];
 #pragma unroll
     for (int w = 0; w < (WT > 0 ? WT : 24); ++w) {
-      gyv[w] = (w < W && ev && kv) ? gy[((b * W + w) * S + s) * Tn + k]
-                                   : 0.f;
+      gyv[w] = (w < W && ev && kv)
+                   ? lh_ld(gy + ((b * W + w) * S + s) * Tn + k) : 0.f;
     }
 #pragma unroll 4
     for (int w = 0; w < (WT > 0 ? WT : 512); ++w) {
@@ -418,7 +429,7 @@ __global__ __launch_bounds__(kBlock) void lift_head_bwd_lk_kernel(
 #pragma unroll
         for (int off = 16; off > 0; off >>= 1)
           gxa += __shfl_xor(gxa, off, 64);
-        if (ev && k == 0) gx[(b * C + c) * S + s] = gxa;
+        if (ev && k == 0) lh_st(gx + (b * C + c) * S + s, gxa);
       }
     }
   }
@@ -465,8 +476,8 @@ void check_lf(const at::Tensor& t, const char* name) {
 at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
                          const at::Tensor& b1, const at::Tensor& W2,
                          const at::Tensor& b2) {
-  check_lf(x, "x"); check_lf(W1, "W1"); check_lf(b1, "b1");
-  check_lf(W2, "W2"); check_lf(b2, "b2");
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous GPU");
+  if (x.scalar_type() != at::kBFloat16) { check_lf(x, "x"); }
   TORCH_CHECK(x.dim() == 3, "x must be [B,C,S]");
   int B = (int)x.size(0), C = (int)x.size(1);
   long S = x.size(2);
@@ -478,21 +489,24 @@ at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
   if (x.numel() == 0) return out;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_l((long)B * S);
-  if (x.scalar_type() == at::kFloat) {
+  if (x.scalar_type() == at::kFloat || x.scalar_type() == at::kBFloat16) {
     int grid2 = grid_for_l(2 * (((long)B * S + 1) / 2) * 32);
-    if (W == 20) {
-      hipLaunchKernelGGL((lift_head_fwd_lk_kernel<20>), dim3(grid2),
-                         dim3(kBlock), 0, stream, x.data_ptr<float>(),
-                         W1.data_ptr<float>(), b1.data_ptr<float>(),
-                         W2.data_ptr<float>(), b2.data_ptr<float>(),
-                         out.data_ptr<float>(), B, C, W, Tn, S);
-    } else {
-      hipLaunchKernelGGL((lift_head_fwd_lk_kernel<0>), dim3(grid2),
-                         dim3(kBlock), 0, stream, x.data_ptr<float>(),
-                         W1.data_ptr<float>(), b1.data_ptr<float>(),
-                         W2.data_ptr<float>(), b2.data_ptr<float>(),
-                         out.data_ptr<float>(), B, C, W, Tn, S);
-    }
+    const bool bf16 = x.scalar_type() == at::kBFloat16;
+    auto W1f = bf16 ? W1.to(at::kFloat).contiguous() : W1;
+    auto b1f = bf16 ? b1.to(at::kFloat).contiguous() : b1;
+    auto W2f = bf16 ? W2.to(at::kFloat).contiguous() : W2;
+    auto b2f = bf16 ? b2.to(at::kFloat).contiguous() : b2;
+#define LH_FWD(WT, TIO)                                                       \
+    hipLaunchKernelGGL((lift_head_fwd_lk_kernel<WT, TIO>), dim3(grid2),       \
+                       dim3(kBlock), 0, stream,                               \
+                       reinterpret_cast<const TIO*>(x.data_ptr()),            \
+                       W1f.data_ptr<float>(), b1f.data_ptr<float>(),          \
+                       W2f.data_ptr<float>(), b2f.data_ptr<float>(),          \
+                       reinterpret_cast<TIO*>(out.data_ptr()), B, C, W, Tn, S)
+    if (bf16) { if (W == 20) LH_FWD(20, unsigned short);
+                else LH_FWD(0, unsigned short); }
+    else { if (W == 20) LH_FWD(20, float); else LH_FWD(0, float); }
+#undef LH_FWD
     DFNO_CHECK_LAUNCH("lift_head");
     return out;
   }
@@ -518,42 +532,48 @@ at::Tensor lift_head_fwd(const at::Tensor& x, const at::Tensor& W1,
 std::vector<at::Tensor> lift_head_bwd(const at::Tensor& gy, const at::Tensor& x,
                                       const at::Tensor& W1, const at::Tensor& b1,
                                       const at::Tensor& W2, const at::Tensor& b2) {
-  check_lf(gy, "gy"); check_lf(x, "x");
+  TORCH_CHECK(gy.is_cuda() && gy.is_contiguous() && x.is_contiguous() &&
+              gy.scalar_type() == x.scalar_type(), "lift_head_bwd IO");
+  if (x.scalar_type() != at::kBFloat16) { check_lf(gy, "gy"); check_lf(x, "x"); }
   int B = (int)x.size(0), C = (int)x.size(1);
   long S = x.size(2);
   int Tn = (int)W1.size(0), W = (int)W2.size(0);
   TORCH_CHECK(C <= 4 && Tn <= 32 && W <= 24, "lift_head: unsupported dims");
 
   auto gx = at::empty_like(x);
-  auto gW1 = at::zeros({Tn, 1}, x.options());
-  auto gb1 = at::zeros({Tn}, x.options());
-  auto gW2 = at::zeros({W, C}, x.options());
-  auto gb2 = at::zeros({W}, x.options());
+  // weight grads accumulate fp32 even for bf16 activations
+  auto fopt = x.options().dtype(x.scalar_type() == at::kDouble
+                                    ? at::kDouble : at::kFloat);
+  auto gW1 = at::zeros({Tn, 1}, fopt);
+  auto gb1 = at::zeros({Tn}, fopt);
+  auto gW2 = at::zeros({W, C}, fopt);
+  auto gb2 = at::zeros({W}, fopt);
   if (x.numel() == 0) return {gx, gW1, gb1, gW2, gb2};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_l((long)B * S);
-  if (x.scalar_type() == at::kFloat) {
+  if (x.scalar_type() == at::kFloat || x.scalar_type() == at::kBFloat16) {
     int grid2 = grid_for_l(2 * (((long)B * S + 1) / 2) * 32);
-    if (W == 20) {
-      hipLaunchKernelGGL((lift_head_bwd_lk_kernel<20>), dim3(grid2),
-                         dim3(kBlock), 0, stream, gy.data_ptr<float>(),
-                         x.data_ptr<float>(), W1.data_ptr<float>(),
-                         b1.data_ptr<float>(), W2.data_ptr<float>(),
-                         b2.data_ptr<float>(), gx.data_ptr<float>(),
-                         gW1.data_ptr<float>(), gb1.data_ptr<float>(),
-                         gW2.data_ptr<float>(), gb2.data_ptr<float>(),
-                         B, C, W, Tn, S);
-    } else {
-      hipLaunchKernelGGL((lift_head_bwd_lk_kernel<0>), dim3(grid2),
-                         dim3(kBlock), 0, stream, gy.data_ptr<float>(),
-                         x.data_ptr<float>(), W1.data_ptr<float>(),
-                         b1.data_ptr<float>(), W2.data_ptr<float>(),
-                         b2.data_ptr<float>(), gx.data_ptr<float>(),
-                         gW1.data_ptr<float>(), gb1.data_ptr<float>(),
-                         gW2.data_ptr<float>(), gb2.data_ptr<float>(),
-                         B, C, W, Tn, S);
-    }
+    const bool bf16 = x.scalar_type() == at::kBFloat16;
+    auto W1f = bf16 ? W1.to(at::kFloat).contiguous() : W1;
+    auto b1f = bf16 ? b1.to(at::kFloat).contiguous() : b1;
+    auto W2f = bf16 ? W2.to(at::kFloat).contiguous() : W2;
+    auto b2f = bf16 ? b2.to(at::kFloat).contiguous() : b2;
+#define LH_BWD(WT, TIO)                                                       \
+    hipLaunchKernelGGL((lift_head_bwd_lk_kernel<WT, TIO>), dim3(grid2),       \
+                       dim3(kBlock), 0, stream,                               \
+                       reinterpret_cast<const TIO*>(gy.data_ptr()),           \
+                       reinterpret_cast<const TIO*>(x.data_ptr()),            \
+                       W1f.data_ptr<float>(), b1f.data_ptr<float>(),          \
+                       W2f.data_ptr<float>(), b2f.data_ptr<float>(),          \
+                       reinterpret_cast<TIO*>(gx.data_ptr()),                 \
+                       gW1.data_ptr<float>(), gb1.data_ptr<float>(),          \
+                       gW2.data_ptr<float>(), gb2.data_ptr<float>(),          \
+                       B, C, W, Tn, S)
+    if (bf16) { if (W == 20) LH_BWD(20, unsigned short);
+                else LH_BWD(0, unsigned short); }
+    else { if (W == 20) LH_BWD(20, float); else LH_BWD(0, float); }
+#undef LH_BWD
     DFNO_CHECK_LAUNCH("lift_head");
     return {gx, gW1, gb1, gW2, gb2};
   }

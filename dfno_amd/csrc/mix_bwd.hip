@@ -16,6 +16,7 @@
 // around the spectral conv).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
@@ -28,12 +29,24 @@ constexpr int kBlock = 256;
 
 typedef float f32x4_mb __attribute__((ext_vector_type(4)));
 
-template <bool WG>
+__device__ __forceinline__ float mb_ld(const float* p) { return *p; }
+__device__ __forceinline__ float mb_ld(const unsigned short* p) {
+  return __uint_as_float(((unsigned int)*p) << 16);
+}
+__device__ __forceinline__ void mb_st(float* p, float v) { *p = v; }
+__device__ __forceinline__ void mb_st(unsigned short* p, float v) {
+  __hip_bfloat16 h = __float2bfloat16(v);
+  *p = *reinterpret_cast<unsigned short*>(&h);
+}
+
+// TIO = unsigned short selects bf16 activations (fp32 compute; W and the
+// gW/gb accumulators stay fp32)
+template <typename TIO, bool WG>
 __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
-    const float* __restrict__ gy, const float* __restrict__ z,
-    const float* __restrict__ x, const float* __restrict__ W,
-    float* __restrict__ gx, float* __restrict__ gW, float* __restrict__ gb,
-    float* __restrict__ gzout, int B, long S) {
+    const TIO* __restrict__ gy, const TIO* __restrict__ z,
+    const TIO* __restrict__ x, const float* __restrict__ W,
+    TIO* __restrict__ gx, float* __restrict__ gW, float* __restrict__ gb,
+    TIO* __restrict__ gzout, int B, long S) {
   constexpr int C = 20;            // in = out channels (trunk width)
   constexpr int TS = 64;           // s-columns per tile (16 MFMA K-steps)
   constexpr int LD = TS + 4;
@@ -69,11 +82,11 @@ __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
       float a = 0.f, bb = 0.f;
       if (c < nv) {
         if (row < C) {
-          a = x[((long)b * C + row) * S + s0 + c];
+          a = mb_ld(x + ((long)b * C + row) * S + s0 + c);
         } else {
           const long off = ((long)b * C + (row - C)) * S + s0 + c;
-          a = gy[off];
-          bb = z[off];
+          a = mb_ld(gy + off);
+          bb = mb_ld(z + off);
         }
       }
       pfa[q] = a;
@@ -98,7 +111,7 @@ __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
         const float gzv = pfa[q] * dfno_gelu::gelu_grad(pfb[q]);
         gzt[(row - C) * LD + c] = gzv;
         if (WG && c < nv)
-          gzout[((long)b * C + (row - C)) * S + s0 + c] = gzv;
+          mb_st(gzout + ((long)b * C + (row - C)) * S + s0 + c, gzv);
       }
     }
     prefetch(t + gridDim.x);
@@ -125,7 +138,7 @@ __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int i = mt * 16 + kg * 4 + r;
-          if (i < C) gx[((long)b * C + i) * S + sc] = c4[r];
+          if (i < C) mb_st(gx + ((long)b * C + i) * S + sc, c4[r]);
         }
       }
     }
@@ -172,10 +185,14 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
                                               const at::Tensor& x,
                                               const at::Tensor& W,
                                               bool want_bias, bool want_gz) {
+  const bool bf16 = gy.scalar_type() == at::kBFloat16;
   TORCH_CHECK(gy.is_cuda() && gy.is_contiguous() && z.is_contiguous() &&
               x.is_contiguous() && W.is_contiguous() &&
-              gy.scalar_type() == at::kFloat &&
-              x.scalar_type() == at::kFloat, "mix_bwd_fused: fp32 GPU only");
+              (bf16 || gy.scalar_type() == at::kFloat) &&
+              x.scalar_type() == gy.scalar_type() &&
+              z.scalar_type() == gy.scalar_type() &&
+              W.scalar_type() == at::kFloat,
+              "mix_bwd_fused: fp32/bf16 activations, fp32 W");
   int B = (int)x.size(0), I = (int)x.size(1);
   long S = x.size(2);
   int O = (int)W.size(0);
@@ -184,9 +201,9 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
               "mix_bwd_fused: shape mismatch");
 
   auto gx = at::empty({B, I, S}, x.options());
-  auto gW = at::zeros({O, I}, x.options());
-  auto gb = want_bias ? at::zeros({O}, x.options())
-                      : at::empty({0}, x.options());
+  auto fopt = x.options().dtype(at::kFloat);   // gW/gb accumulate fp32
+  auto gW = at::zeros({O, I}, fopt);
+  auto gb = want_bias ? at::zeros({O}, fopt) : at::empty({0}, fopt);
   auto gz = want_gz ? at::empty({B, O, S}, x.options())
                     : at::empty({0}, x.options());
   if (x.numel() == 0) return {gx, gW, gb, gz};
@@ -197,19 +214,26 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
   long stiles = (S + TS - 1) / TS;
   int grid = (int)std::min((long)B * stiles, 1024L);
   float* gbp = want_bias ? gb.data_ptr<float>() : nullptr;
-  if (want_gz) {
-    hipLaunchKernelGGL((mix_bwd_fused_kernel<true>), dim3(grid), dim3(kBlock),
-                       smem, stream, gy.data_ptr<float>(), z.data_ptr<float>(),
-                       x.data_ptr<float>(), W.data_ptr<float>(),
-                       gx.data_ptr<float>(), gW.data_ptr<float>(), gbp,
-                       gz.data_ptr<float>(), B, S);
+#define MB_LAUNCH(TIO, WG, GZP)                                              \
+  hipLaunchKernelGGL((mix_bwd_fused_kernel<TIO, WG>), dim3(grid),            \
+                     dim3(kBlock), smem, stream,                             \
+                     reinterpret_cast<const TIO*>(gy.data_ptr()),            \
+                     reinterpret_cast<const TIO*>(z.data_ptr()),             \
+                     reinterpret_cast<const TIO*>(x.data_ptr()),             \
+                     W.data_ptr<float>(),                                    \
+                     reinterpret_cast<TIO*>(gx.data_ptr()),                  \
+                     gW.data_ptr<float>(), gbp, GZP, B, S)
+  if (bf16) {
+    auto gzp = want_gz ? reinterpret_cast<unsigned short*>(gz.data_ptr())
+                       : nullptr;
+    if (want_gz) MB_LAUNCH(unsigned short, true, gzp);
+    else         MB_LAUNCH(unsigned short, false, nullptr);
   } else {
-    hipLaunchKernelGGL((mix_bwd_fused_kernel<false>), dim3(grid), dim3(kBlock),
-                       smem, stream, gy.data_ptr<float>(), z.data_ptr<float>(),
-                       x.data_ptr<float>(), W.data_ptr<float>(),
-                       gx.data_ptr<float>(), gW.data_ptr<float>(), gbp,
-                       nullptr, B, S);
+    auto gzp = want_gz ? reinterpret_cast<float*>(gz.data_ptr()) : nullptr;
+    if (want_gz) MB_LAUNCH(float, true, gzp);
+    else         MB_LAUNCH(float, false, nullptr);
   }
+#undef MB_LAUNCH
   DFNO_CHECK_LAUNCH("mix_bwd_fused");
   return {gx, gW, gb, gz};
 }

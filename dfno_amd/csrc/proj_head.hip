@@ -16,6 +16,7 @@
 // library-GEMM grad-W3 reduction.
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
@@ -29,6 +30,18 @@ constexpr int kBlock = 256;
 template <typename T>
 __device__ __forceinline__ T gelu_erf_(T z) { return dfno_gelu::gelu(z); }
 
+__device__ __forceinline__ float ph_ld(const float* p) { return *p; }
+__device__ __forceinline__ double ph_ld(const double* p) { return *p; }
+__device__ __forceinline__ float ph_ld(const unsigned short* p) {
+  return __uint_as_float(((unsigned int)*p) << 16);
+}
+__device__ __forceinline__ void ph_st(float* p, float v) { *p = v; }
+__device__ __forceinline__ void ph_st(double* p, double v) { *p = v; }
+__device__ __forceinline__ void ph_st(unsigned short* p, float v) {
+  __hip_bfloat16 h = __float2bfloat16(v);
+  *p = *reinterpret_cast<unsigned short*>(&h);
+}
+
 template <typename T>
 __device__ __forceinline__ T gelu_grad_erf_(T z) { return dfno_gelu::gelu_grad(z); }
 
@@ -41,10 +54,10 @@ __device__ __forceinline__ T wave_sum(T v) {
 }
 
 template <typename T, int IMAX, int O2MAX, int VEC, bool VECTOR,
-          int MT = 0, int IT = 0>
+          int MT = 0, int IT = 0, typename TIO = T>
 __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
-    const T* __restrict__ x, const T* __restrict__ W3l, const T* __restrict__ b3l,
-    const T* __restrict__ W4l, const T* __restrict__ b4l, T* __restrict__ out,
+    const TIO* __restrict__ x, const T* __restrict__ W3l, const T* __restrict__ b3l,
+    const T* __restrict__ W4l, const T* __restrict__ b4l, TIO* __restrict__ out,
     int B, int I_, int M_, int O2, long S) {
   // MT/IT > 0 pin the hidden width and input channels at compile time: the
   // M-loop fully unrolls with weight offsets folded into the scalar loads
@@ -67,12 +80,14 @@ __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
     int nv = full ? VEC : (int)(S - s);
 
     T xr[IMAX][VEC];
-    const T* xb = x + ((long)b * I) * S + s;
-    if constexpr (VECTOR && std::is_same<T, float>::value) {
+    const TIO* xb = x + ((long)b * I) * S + s;
+    if constexpr (VECTOR && std::is_same<T, float>::value &&
+                  std::is_same<TIO, float>::value) {
 #pragma unroll
       for (int i = 0; i < IMAX; ++i) {
         if (i < I) {
-        const float4 v = *reinterpret_cast<const float4*>(xb + (long)i * S);
+        const float4 v = *reinterpret_cast<const float4*>(
+            reinterpret_cast<const float*>(xb) + (long)i * S);
         xr[i][0] = v.x; xr[i][1] = v.y; xr[i][2] = v.z; xr[i][3] = v.w;
               }
       }
@@ -82,7 +97,7 @@ __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
         if (i < I) {
 #pragma unroll
         for (int k = 0; k < VEC; ++k)
-          xr[i][k] = (full || k < nv) ? xb[(long)i * S + k] : T(0);
+          xr[i][k] = (full || k < nv) ? (T)ph_ld(xb + (long)i * S + k) : T(0);
               }
       }
     }
@@ -123,18 +138,20 @@ __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
       }
     }
 
-    T* ob = out + ((long)b * O2) * S + s;
+    TIO* ob = out + ((long)b * O2) * S + s;
 #pragma unroll
     for (int o = 0; o < O2MAX; ++o) {
       if (o < O2) {
-      if constexpr (VECTOR && std::is_same<T, float>::value) {
-        *reinterpret_cast<float4*>(ob + (long)o * S) =
+      if constexpr (VECTOR && std::is_same<T, float>::value &&
+                    std::is_same<TIO, float>::value) {
+        *reinterpret_cast<float4*>(
+            reinterpret_cast<float*>(ob) + (long)o * S) =
             make_float4(acc[o][0], acc[o][1], acc[o][2], acc[o][3]);
       } else {
 #pragma unroll
         for (int k = 0; k < VEC; ++k) {
           if (!full && k >= nv) break;
-          ob[(long)o * S + k] = acc[o][k];
+          ph_st(ob + (long)o * S + k, acc[o][k]);
         }
       }
           }
@@ -340,16 +357,19 @@ __device__ __forceinline__ float seg_sum16(float v) {
   return v;
 }
 
-template <int IT, int MT, int O2T>
+template <int IT, int MT, int O2T, typename TIO = float>
 __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
-    const float* __restrict__ gy, const float* __restrict__ x,
+    const TIO* __restrict__ gy, const TIO* __restrict__ x,
     const float* __restrict__ W3, const float* __restrict__ b3,
     const float* __restrict__ W4,
-    float* __restrict__ gx, float* __restrict__ gW3,
+    TIO* __restrict__ gx, float* __restrict__ gW3,
     float* __restrict__ gb3, float* __restrict__ gW4,
     float* __restrict__ gb4, int B, int O2, long S) {
   constexpr int TS = 64;           // s-columns per tile (16 MFMA K-steps)
   constexpr int LD = TS + 4;       // row pad (float4-aligned, 4-bank skew)
+  // gz tile + staged x tile + weights in LDS (52 KB -> 3 blocks/CU; an
+  // x/W3-from-global variant at 4 blocks/CU measured 2x slower: the cold
+  // per-lane B-operand loads serialize inside the MFMA chains)
   extern __shared__ __align__(16) char smem_raw[];
   float* gzt = reinterpret_cast<float*>(smem_raw);   // [MT][LD]
   float* xt = gzt + (size_t)MT * LD;                 // [IT][LD]
@@ -382,10 +402,8 @@ __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
 
   const long stiles = (S + TS - 1) / TS;
   const long tend = (long)B * stiles;
-  // software-pipelined staging: tile t+grid's x/gy columns are loaded into
-  // registers while tile t computes (the PMC counters put 45% of wave
-  // cycles parked on the staging round-trip when it sits between the
-  // barriers).
+  // software-pipelined staging: tile t+grid's x/gy columns load into
+  // registers while tile t computes
   constexpr int NPF = ((IT + 2) * TS + kBlock - 1) / kBlock;
   float pf[NPF];
   auto prefetch = [&](long tt) {
@@ -401,8 +419,8 @@ __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
       const int c = r - row * TS;
       float v = 0.f;
       if (c < nv) {
-        v = (row < IT) ? x[((long)b * IT + row) * S + s0 + c]
-                       : gy[((long)b * O2T + (row - IT)) * S + s0 + c];
+        v = (row < IT) ? ph_ld(x + ((long)b * IT + row) * S + s0 + c)
+                       : ph_ld(gy + ((long)b * O2T + (row - IT)) * S + s0 + c);
       }
       pf[q] = v;
     }
@@ -412,7 +430,7 @@ __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
     const int b = (int)(t / stiles);
     const long s0 = (t % stiles) * TS;
     const int nv = (int)min((long)TS, S - s0);
-    __syncthreads();               // prior tile's phase-2 reads done
+    __syncthreads();               // prior tile's phase reads done
 #pragma unroll
     for (int q = 0; q < NPF; ++q) {
       const int r = (int)threadIdx.x + q * kBlock;
@@ -427,34 +445,43 @@ __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
         if (row == IT) gb4a0 += v; else gb4a1 += v;
       }
     }
-    prefetch(t + gridDim.x);       // next tile's loads fly over this tile
+    prefetch(t + gridDim.x);
+    __syncthreads();               // xt/gyt staged
+    // phase 1a: z3 tile = W3 @ x via MFMA (bias-free, into gzt).
+    // 8 m-tiles x 4 n-tiles over K = IT = 20; 8 pairs per wave.
+#pragma unroll
+    for (int pp = 0; pp < 8; ++pp) {
+      const int p = wave + 4 * pp;
+      const int mt = p >> 2, nt = p & 3;
+      const int m = mt * 16 + l16;
+      const int n = nt * 16 + l16;
+      f32x4_ph z4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int k = 0; k < IT; k += 4) {
+        const float a = W3l[m * IT + k + kg];
+        const float bb = xt[(k + kg) * LD + n];
+        z4 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, z4, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        gzt[(mt * 16 + kg * 4 + r) * LD + nt * 16 + l16] = z4[r];
+    }
     __syncthreads();
-    // phase 1: gz tile.  Each PAIR of threads owns one hidden channel j
-    // (2 x 4 s-columns per pass, TS/8 passes): the W3 row and the gb3/gW4
-    // partials live in registers across the tile, so the only cross-lane
-    // reduction is one shfl_xor(1) per tile (the per-iteration 16-lane
-    // shuffle chains were the dominant latency at 3 waves/SIMD).
+    // phase 1b: gz = (W4^T gy) * gelu'(z3 + b3) in place over the tile.
+    // Each PAIR of threads owns one hidden channel j, so the gb3/gW4
+    // partials accumulate in registers with one shfl_xor(1) per tile.
     {
       const int jj = (int)(threadIdx.x >> 1);        // 0..127
       const int half = (int)(threadIdx.x & 1) * 4;   // 0 or 4
       const float bj = b3l[jj];
       const float w40 = W4l[jj];
       const float w41 = (O2T == 2) ? W4l[MT + jj] : 0.f;
-      float w3r[IT];
-#pragma unroll
-      for (int i = 0; i < IT; ++i) w3r[i] = W3l[jj * IT + i];
       float pb = 0.f, pw0 = 0.f, pw1 = 0.f;
 #pragma unroll
       for (int k = 0; k < TS / 8; ++k) {
         const int c4 = half + k * 8;
-        float z[4] = {bj, bj, bj, bj};
-#pragma unroll
-        for (int i = 0; i < IT; ++i) {
-          const float4 xv =
-              *reinterpret_cast<const float4*>(xt + i * LD + c4);
-          z[0] += w3r[i] * xv.x; z[1] += w3r[i] * xv.y;
-          z[2] += w3r[i] * xv.z; z[3] += w3r[i] * xv.w;
-        }
+        float4 zv = *reinterpret_cast<float4*>(gzt + jj * LD + c4);
+        float z[4] = {zv.x + bj, zv.y + bj, zv.z + bj, zv.w + bj};
         float g[4], dg[4];
 #pragma unroll
         for (int q = 0; q < 4; ++q)
@@ -479,16 +506,15 @@ __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
       pb += __shfl_xor(pb, 1, 64);
       pw0 += __shfl_xor(pw0, 1, 64);
       if (O2T == 2) pw1 += __shfl_xor(pw1, 1, 64);
-      if ((threadIdx.x & 1) == 0) {
+      if ((lane & 1) == 0) {
         atomicAdd(&gb3s[jj], pb);
         atomicAdd(&gW4s[jj], pw0);
         if (O2T == 2) atomicAdd(&gW4s[MT + jj], pw1);
       }
     }
     __syncthreads();
-    // phase 2a: gx tile = W3^T @ gz via MFMA.  D[i][c] = sum_j A[i][j]B[c][j]
-    // with A[m][k] = W3l[k*IT + m], B[n][k] = gzt[k*LD + n]; 2 m-tiles x
-    // 4 n-tiles split across the 4 waves (2 pairs each), K = MT.
+    // phase 2a: gx tile = W3^T @ gz via MFMA.  A[m=i][k=j] = W3[k*IT+m]
+    // (per-lane global), B[n=c][k=j] = gzt[k*LD+n]; K = MT.
 #pragma unroll
     for (int pp = 0; pp < 2; ++pp) {
       const int p = wave + 4 * pp;
@@ -508,12 +534,12 @@ __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int i = mt * 16 + kg * 4 + r;
-          if (i < IT) gx[((long)b * IT + i) * S + sc] = c4[r];
+          if (i < IT) ph_st(gx + ((long)b * IT + i) * S + sc, c4[r]);
         }
       }
     }
-    // phase 2b: gW3 fragments += gz_tile @ x_tile^T (reads only; the tile
-    // loop's top sync fences the next staging pass)
+    // phase 2b: gW3 fragments += gz_tile @ x_tile^T (reads only; the
+    // tile loop's top sync fences the next staging pass)
 #pragma unroll
     for (int pp = 0; pp < 4; ++pp) {
       const int p = wave + 4 * pp;
@@ -586,8 +612,9 @@ bool vec_ok(long S, std::initializer_list<const void*> ptrs) {
 at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
                          const at::Tensor& b3, const at::Tensor& W4,
                          const at::Tensor& b4) {
-  check_pf(x, "x"); check_pf(W3, "W3"); check_pf(b3, "b3");
-  check_pf(W4, "W4"); check_pf(b4, "b4");
+  const bool bf16 = x.scalar_type() == at::kBFloat16;
+  if (!bf16) { check_pf(x, "x"); }
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "x must be contiguous GPU");
   TORCH_CHECK(x.dim() == 3, "x must be [B,I,S]");
   int B = (int)x.size(0), I = (int)x.size(1);
   long S = x.size(2);
@@ -599,6 +626,30 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
   if (x.numel() == 0) return out;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_p((long)B * ((S + 3) / 4));
+  if (bf16) {
+    // bf16 activations, fp32 weights/math (host-casts the tiny weights)
+    auto W3f = W3.to(at::kFloat).contiguous();
+    auto b3f = b3.to(at::kFloat).contiguous();
+    auto W4f = W4.to(at::kFloat).contiguous();
+    auto b4f = b4.to(at::kFloat).contiguous();
+    auto inp = reinterpret_cast<const unsigned short*>(x.data_ptr());
+    auto op = reinterpret_cast<unsigned short*>(out.data_ptr());
+#define PH_LAUNCH_BF(IM, OM, MTV, ITV)                                        \
+    hipLaunchKernelGGL((proj_head_fwd_kernel<float, IM, OM, 4, false, MTV,    \
+                                             ITV, unsigned short>),           \
+                       dim3(grid), dim3(kBlock), 0, stream, inp,              \
+                       W3f.data_ptr<float>(), b3f.data_ptr<float>(),          \
+                       W4f.data_ptr<float>(), b4f.data_ptr<float>(), op,      \
+                       B, I, M, O2, S);
+    if (I == 20 && M == 128 && O2 <= 2) { PH_LAUNCH_BF(24, 2, 128, 20) }
+    else if (I <= 24 && O2 <= 2) { PH_LAUNCH_BF(24, 2, 0, 0) }
+    else { PH_LAUNCH_BF(32, 8, 0, 0) }
+#undef PH_LAUNCH_BF
+    DFNO_CHECK_LAUNCH("proj_head");
+    return out;
+  }
+  check_pf(W3, "W3"); check_pf(b3, "b3");
+  check_pf(W4, "W4"); check_pf(b4, "b4");
 
 #define PH_LAUNCH_FT(V)                                                       \
     hipLaunchKernelGGL((proj_head_fwd_kernel<scalar_t, IM, OM, 4, V, 128, 20>), \
@@ -696,20 +747,26 @@ std::vector<at::Tensor> proj_head_bwd_fused(const at::Tensor& gy,
                                             const at::Tensor& W3,
                                             const at::Tensor& b3,
                                             const at::Tensor& W4) {
-  check_pf(gy, "gy"); check_pf(x, "x"); check_pf(W3, "W3");
-  check_pf(b3, "b3"); check_pf(W4, "W4");
+  const bool bf16 = x.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(gy.is_cuda() && gy.is_contiguous() && x.is_contiguous() &&
+              gy.scalar_type() == x.scalar_type(), "proj_head_bwd_fused IO");
   int B = (int)x.size(0), I = (int)x.size(1);
   long S = x.size(2);
   int M = (int)W3.size(0), O2 = (int)W4.size(0);
   TORCH_CHECK(I == 20 && M == 128 && O2 <= 2 &&
-              x.scalar_type() == at::kFloat,
+              (bf16 || x.scalar_type() == at::kFloat),
               "proj_head_bwd_fused: flagship shape only");
 
+  auto W3f = bf16 ? W3.to(at::kFloat).contiguous() : W3;
+  auto b3f = bf16 ? b3.to(at::kFloat).contiguous() : b3;
+  auto W4f = bf16 ? W4.to(at::kFloat).contiguous() : W4;
+  check_pf(W3f, "W3"); check_pf(b3f, "b3"); check_pf(W4f, "W4");
+  auto fopt = x.options().dtype(at::kFloat);   // weight grads accumulate fp32
   auto gx = at::empty({B, I, S}, x.options());
-  auto gW3 = at::zeros({M, I}, x.options());
-  auto gb3 = at::zeros({M}, x.options());
-  auto gW4 = at::zeros({O2, M}, x.options());
-  auto gb4 = at::zeros({O2}, x.options());
+  auto gW3 = at::zeros({M, I}, fopt);
+  auto gb3 = at::zeros({M}, fopt);
+  auto gW4 = at::zeros({O2, M}, fopt);
+  auto gb4 = at::zeros({O2}, fopt);
   if (x.numel() == 0) return {gx, gW3, gb3, gW4, gb4};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
@@ -720,15 +777,22 @@ std::vector<at::Tensor> proj_head_bwd_fused(const at::Tensor& gy,
        128 + 2 * (size_t)O2 * 128 + 128);
   long stiles = (S + TS - 1) / TS;
   int grid = (int)std::min((long)B * stiles, 768L);
-#define PH_FUSED(O2T)                                                        \
-  hipLaunchKernelGGL((proj_head_bwd_fused_kernel<20, 128, O2T>), dim3(grid), \
-                     dim3(kBlock), smem, stream, gy.data_ptr<float>(),       \
-                     x.data_ptr<float>(), W3.data_ptr<float>(),              \
-                     b3.data_ptr<float>(), W4.data_ptr<float>(),             \
-                     gx.data_ptr<float>(), gW3.data_ptr<float>(),            \
-                     gb3.data_ptr<float>(), gW4.data_ptr<float>(),           \
-                     gb4.data_ptr<float>(), B, O2, S)
-  if (O2 == 2) PH_FUSED(2); else PH_FUSED(1);
+#define PH_FUSED(O2T, TIO)                                                   \
+  hipLaunchKernelGGL((proj_head_bwd_fused_kernel<20, 128, O2T, TIO>),        \
+                     dim3(grid), dim3(kBlock), smem, stream,                 \
+                     reinterpret_cast<const TIO*>(gy.data_ptr()),            \
+                     reinterpret_cast<const TIO*>(x.data_ptr()),             \
+                     W3f.data_ptr<float>(), b3f.data_ptr<float>(),           \
+                     W4f.data_ptr<float>(),                                  \
+                     reinterpret_cast<TIO*>(gx.data_ptr()),                  \
+                     gW3.data_ptr<float>(), gb3.data_ptr<float>(),           \
+                     gW4.data_ptr<float>(), gb4.data_ptr<float>(),           \
+                     B, O2, S)
+  if (bf16) {
+    if (O2 == 2) PH_FUSED(2, unsigned short); else PH_FUSED(1, unsigned short);
+  } else {
+    if (O2 == 2) PH_FUSED(2, float); else PH_FUSED(1, float);
+  }
 #undef PH_FUSED
   DFNO_CHECK_LAUNCH("proj_head_bwd_fused");
   return {gx, gW3, gb3, gW4, gb4};

@@ -117,8 +117,6 @@ class DistributedFNONd(nn.Module):
             return lift_head(x, W1, b1, W2, b2)
 
         if x.is_cuda and x.dtype != torch.bfloat16:
-            # (bf16 deliberately composes the native bf16 channel-mix
-            # kernels; only non-bf16 misses are real fallbacks)
             note_fallback("lift_head", "unsupported lift shape/dtype for the "
                           "fused kernel (composed linears instead)")
         x = l1(x, activation="gelu")
@@ -138,6 +136,12 @@ class DistributedFNONd(nn.Module):
         supported = (x.is_cuda and x.dtype in (torch.float32, torch.float64)
                      and l3.in_features <= 32 and l3.out_features <= 512
                      and l4.out_features <= 8)
+        # bf16: the fused pair (bf16-IO fwd + fused backward) covers the
+        # flagship head shape only
+        supported = supported or (
+            x.is_cuda and x.dtype == torch.bfloat16
+            and l3.in_features == 20 and l3.out_features == 128
+            and l4.out_features <= 2)
         if supported:
             with comm_region() as r:
                 W3 = l3.W_bcast(l3.W)

@@ -15,7 +15,8 @@ __all__ = ["lift_head", "lift_head_supported"]
 
 
 def lift_head_supported(x, t_in, t_out, c_in, width) -> bool:
-    return (x.is_cuda and x.dtype in (torch.float32, torch.float64)
+    return (x.is_cuda
+            and x.dtype in (torch.float32, torch.float64, torch.bfloat16)
             and t_in == 1 and t_out <= 32 and c_in <= 4 and width <= 24)
 
 
@@ -35,7 +36,9 @@ class _LiftHeadFn(torch.autograd.Function):
         gx, gW1, gb1, gW2, gb2 = ext.lift_head_bwd(
             gy.contiguous(), x3, W1.contiguous(), b1.contiguous(),
             W2.contiguous(), b2.contiguous())
-        return gx, gW1, gb1, gW2, gb2
+        # bf16 activations: weight grads come back fp32-accumulated
+        return (gx, gW1.to(W1.dtype), gb1.to(b1.dtype), gW2.to(W2.dtype),
+                gb2.to(b2.dtype))
 
 
 def lift_head(x, W1, b1, W2, b2):

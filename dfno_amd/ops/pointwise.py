@@ -108,15 +108,16 @@ class _ChannelMixFn(torch.autograd.Function):
         x3, W, z3 = ctx.saved_tensors
         act = ctx.act
         gy = gy.contiguous()
-        if (act and gy.is_cuda and gy.dtype == torch.float32
-                and x3.shape[1] == 20 and W.shape[0] == 20):
+        if (act and gy.is_cuda and gy.dtype in (torch.float32, torch.bfloat16)
+                and x3.shape[1] == 20 and W.shape[0] == 20
+                and gy.dtype == x3.dtype):
             # trunk 20x20: one kernel, gz never materialized (mix_bwd.hip)
             ext = _ext.get(required=True)
             gx, gW, gb, _ = ext.channel_mix_bwd_fused(
-                gy, z3.contiguous(), x3.contiguous(), W.contiguous(),
-                ctx.has_bias, False)
-            return (gx.reshape(ctx.x_shape), gW,
-                    gb if ctx.has_bias else None, None)
+                gy, z3.contiguous(), x3.contiguous(),
+                W.contiguous().float(), ctx.has_bias, False)
+            return (gx.reshape(ctx.x_shape), gW.to(W.dtype),
+                    gb.to(W.dtype) if ctx.has_bias else None, None)
         is_bf16 = gy.is_cuda and gy.dtype == torch.bfloat16
         bf16_ok = is_bf16 and _bf16_mix_ok(gy.reshape(x3.shape[0], W.shape[0], -1),
                                            W.shape[0], x3.shape[1], "channel_mix_bwd")
@@ -237,14 +238,16 @@ class _LinearResGeluFn(torch.autograd.Function):
     def backward(ctx, gy):
         x3, W, z3 = ctx.saved_tensors
         gy = gy.contiguous()
-        if (gy.is_cuda and gy.dtype == torch.float32
-                and x3.shape[1] == 20 and W.shape[0] == 20):
+        if (gy.is_cuda and gy.dtype in (torch.float32, torch.bfloat16)
+                and x3.shape[1] == 20 and W.shape[0] == 20
+                and gy.dtype == x3.dtype):
             # trunk 20x20: one kernel; gz comes back as the residual grad
             ext = _ext.get(required=True)
             gx, gW, _, gz = ext.channel_mix_bwd_fused(
-                gy, z3.contiguous(), x3.contiguous(), W.contiguous(),
-                False, True)
-            return gx.reshape(ctx.x_shape), gW, gz.reshape(ctx.res_shape)
+                gy, z3.contiguous(), x3.contiguous(),
+                W.contiguous().float(), False, True)
+            return (gx.reshape(ctx.x_shape), gW.to(W.dtype),
+                    gz.reshape(ctx.res_shape))
         if gy.is_cuda and _native_dt(gy):
             ext = _ext.get(required=True)
             gz = ext.gelu_bwd(gy, z3)

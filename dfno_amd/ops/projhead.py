@@ -16,6 +16,9 @@ __all__ = ["proj_head", "proj_head_supported"]
 
 
 def proj_head_supported(x: torch.Tensor, W3, W4) -> bool:
+    if (x.is_cuda and x.dtype == torch.bfloat16 and W3.shape[1] == 20
+            and W3.shape[0] == 128 and W4.shape[0] <= 2):
+        return True    # bf16-IO fwd + fused backward (flagship head shape)
     return (x.is_cuda and x.dtype in (torch.float32, torch.float64)
             and W3.shape[1] <= 32 and W3.shape[0] <= 512 and W4.shape[0] <= 8)
 
@@ -38,12 +41,13 @@ class _ProjHeadFn(torch.autograd.Function):
         x3, W3, b3, W4 = ctx.saved_tensors
         ext = _ext.get(required=True)
         if (x3.shape[1] == 20 and W3.shape[0] == 128 and W4.shape[0] <= 2
-                and x3.dtype == torch.float32):
+                and x3.dtype in (torch.float32, torch.bfloat16)):
             # flagship: one kernel, no [B,128,S] gz3 intermediate in HBM
             gx, gW3, gb3, gW4g, gb4 = ext.proj_head_bwd_fused(
                 gy.contiguous(), x3, W3.contiguous(), b3.contiguous(),
                 W4.contiguous())
-            return gx.reshape(ctx.x_shape), gW3, gb3, gW4g, gb4
+            return (gx.reshape(ctx.x_shape), gW3.to(W3.dtype),
+                    gb3.to(W3.dtype), gW4g.to(W3.dtype), gb4.to(W3.dtype))
         gz3, gb3, gW4g, gb4 = ext.proj_head_bwd(
             gy.contiguous(), x3, W3.contiguous(), b3.contiguous(), W4.contiguous())
         gx = ext.channel_mix_fwd_t(gz3, W3.contiguous())   # W3^T @ gz3
@@ -58,7 +62,9 @@ def proj_head(x: torch.Tensor, W3, b3, W4, b4) -> torch.Tensor:
     """
     b3f = b3.reshape(-1)
     b4f = b4.reshape(-1)
-    if x.is_cuda and x.dtype in (torch.float32, torch.float64):
+    bf16_ok = (x.is_cuda and x.dtype == torch.bfloat16 and W3.shape[1] == 20
+               and W3.shape[0] == 128 and W4.shape[0] <= 2)
+    if bf16_ok or (x.is_cuda and x.dtype in (torch.float32, torch.float64)):
         out3 = _ProjHeadFn.apply(x, W3, b3f, W4, b4f)
         out_shape = list(x.shape)
         out_shape[1] = W4.shape[0]

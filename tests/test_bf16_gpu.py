@@ -230,3 +230,77 @@ def test_bf16_model_all_native():
     y.float().square().sum().backward()
     torch.cuda.synchronize()
     dispatch.assert_all_native()
+
+
+def test_bf16_proj_head_fused():
+    """bf16-IO proj head (fwd kernel + fused backward) vs the fp32
+    composition at the flagship head shape."""
+    import torch.nn.functional as F
+    from dfno_amd.ops import proj_head
+    torch.manual_seed(11)
+    B, I, M, O2, S = 1, 20, 128, 1, 4099
+    x = torch.randn(B, I, S, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    W3 = (torch.randn(M, I, device="cuda", dtype=torch.bfloat16) / I
+          ).requires_grad_(True)
+    b3 = torch.randn(M, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    W4 = (torch.randn(O2, M, device="cuda", dtype=torch.bfloat16) / M
+          ).requires_grad_(True)
+    b4 = torch.randn(O2, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    y = proj_head(x, W3, b3, W4, b4)
+    assert y.dtype == torch.bfloat16
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    xr = x.detach().float().requires_grad_(True)
+    W3r = W3.detach().float().requires_grad_(True)
+    b3r = b3.detach().float().requires_grad_(True)
+    W4r = W4.detach().float().requires_grad_(True)
+    b4r = b4.detach().float().requires_grad_(True)
+    h = F.gelu(torch.einsum("mi,bis->bms", W3r, xr) + b3r.view(1, -1, 1))
+    yr = torch.einsum("om,bms->bos", W4r, h) + b4r.view(1, -1, 1)
+    yr.backward(gy.float())
+    assert torch.allclose(y.float(), yr, rtol=2e-2, atol=2e-2), \
+        f"fwd {(y.float()-yr).abs().max()}"
+    for a, b in [(x, xr), (b3, b3r), (b4, b4r)]:
+        assert torch.allclose(a.grad.float(), b.grad, rtol=5e-2, atol=5e-2), \
+            f"grad {(a.grad.float()-b.grad).abs().max()}"
+    # weight grads accumulate over S: looser tolerance
+    for a, b in [(W3, W3r), (W4, W4r)]:
+        assert torch.allclose(a.grad.float(), b.grad, rtol=5e-2, atol=5e-1), \
+            f"wgrad {(a.grad.float()-b.grad).abs().max()}"
+
+
+def test_bf16_lift_head_fused():
+    """bf16-IO lift head (lane-per-k kernels) vs the fp32 composition."""
+    import torch.nn.functional as F
+    from dfno_amd.ops import lift_head
+    torch.manual_seed(12)
+    B, C, Wd, Tn, S = 1, 4, 20, 30, 1000
+    x = torch.randn(B, C, S, 1, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    W1 = torch.randn(Tn, 1, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    b1 = torch.randn(Tn, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    W2 = (torch.randn(Wd, C, device="cuda", dtype=torch.bfloat16) / C
+          ).requires_grad_(True)
+    b2 = torch.randn(Wd, device="cuda", dtype=torch.bfloat16).requires_grad_(True)
+    y = lift_head(x, W1, b1, W2, b2)
+    assert y.dtype == torch.bfloat16
+    gy = torch.randn_like(y)
+    y.backward(gy)
+
+    xr = x.detach().float().requires_grad_(True)
+    W1r = W1.detach().float().requires_grad_(True)
+    b1r = b1.detach().float().requires_grad_(True)
+    W2r = W2.detach().float().requires_grad_(True)
+    b2r = b2.detach().float().requires_grad_(True)
+    h = F.gelu(torch.einsum("to,bcso->bcst", W1r, xr) + b1r.view(1, 1, 1, -1))
+    yr = F.gelu(torch.einsum("wc,bcst->bwst", W2r, h) + b2r.view(1, -1, 1, 1))
+    yr.backward(gy.float())
+    assert torch.allclose(y.float(), yr, rtol=2e-2, atol=2e-2), \
+        f"fwd {(y.float()-yr).abs().max()}"
+    assert torch.allclose(x.grad.float(), xr.grad, rtol=5e-2, atol=5e-2), \
+        f"gx {(x.grad.float()-xr.grad).abs().max()}"
+    for a, b in [(W1, W1r), (b1, b1r), (W2, W2r), (b2, b2r)]:
+        assert torch.allclose(a.grad.float(), b.grad, rtol=5e-2, atol=5e-1), \
+            f"wgrad {(a.grad.float()-b.grad).abs().max()}"
