@@ -64,6 +64,9 @@ def _fp8_weights(weights):
     stale, stale16, stale_am = [], [], []
     for w in weights:
         ent = _FP8_CACHE.get(id(w))
+        if ent is not None and (ent[1].shape != w.shape
+                                or ent[1].device != w.device):
+            ent = None   # id() reuse after a freed master (fresh models)
         if ent is None:
             w16 = torch.empty(w.shape, dtype=torch.uint16, device=w.device)
             amax = torch.zeros(1, dtype=torch.float32, device=w.device)

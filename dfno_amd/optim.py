@@ -95,6 +95,9 @@ class Adam(torch.optim.Adam):
                 # requantization pass disappears (docs/ROADMAP.md item 5)
                 origs.append(p)
                 ent = _FP8_CACHE.get(id(p)) if p.dtype == torch.complex64 else None
+                if ent is not None and (ent[1].shape != p.shape
+                                        or ent[1].device != p.device):
+                    ent = None   # id() reuse after a freed master
                 if ent is not None:
                     if len(ent) == 3:   # bootstrap the measured-amax slot
                         ent.append(ent[2].clone())
