@@ -40,8 +40,9 @@ __device__ __forceinline__ void mb_st(unsigned short* p, float v) {
 }
 
 // TIO = unsigned short selects bf16 activations (fp32 compute; W and the
-// gW/gb accumulators stay fp32)
-template <typename TIO, bool WG>
+// gW/gb accumulators stay fp32).  PH: phase mask for perf bisection
+// (1 staging+gz, +2 gx, +4 gW); production uses 7.
+template <typename TIO, bool WG, int PH = 7>
 __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
     const TIO* __restrict__ gy, const TIO* __restrict__ z,
     const TIO* __restrict__ x, const float* __restrict__ W,
@@ -116,10 +117,11 @@ __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
     }
     prefetch(t + gridDim.x);
     __syncthreads();
+    if constexpr ((PH & 2) == 0 && (PH & 4) == 0) continue;
     // phase A: gx = W^T @ gz.  A[m=i][k=o] = W[o*C+i] (per-lane global,
     // L1-resident), B[n=c][k=o] = gzt[o][c]; K = C padded to 24.
 #pragma unroll
-    for (int pp = 0; pp < 2; ++pp) {
+    for (int pp = 0; pp < ((PH & 2) ? 2 : 0); ++pp) {
       const int p = wave + 4 * pp;
       const int mt = p >> 2, nt = p & 3;
       const int m = mt * 16 + l16;
@@ -145,7 +147,7 @@ __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
     // phase B: gW fragments += gz @ [x; ones]^T.  One (mt, nt) pair per
     // wave: mt = wave>>1 (o-tiles), nt = wave&1 (i-tiles; col C = ones
     // for gb).
-    {
+    if constexpr ((PH & 4) != 0) {
       const int mt = wave >> 1, nt = wave & 1;
       const float* gr = gzt + (mt * 16 + l16) * LD;
       const int ncol = nt * 16 + l16;
@@ -214,8 +216,13 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
   long stiles = (S + TS - 1) / TS;
   int grid = (int)std::min((long)B * stiles, 1024L);
   float* gbp = want_bias ? gb.data_ptr<float>() : nullptr;
-#define MB_LAUNCH(TIO, WG, GZP)                                              \
-  hipLaunchKernelGGL((mix_bwd_fused_kernel<TIO, WG>), dim3(grid),            \
+  static const int ph = []() {
+    const char* e = getenv("DFNO_MIX_PHASES");   // perf-bisect knob
+    return e ? atoi(e) : 7;
+  }();
+#define MB_LAUNCH(TIO, WG, GZP) MB_LAUNCH_P(TIO, WG, GZP, 7)
+#define MB_LAUNCH_P(TIO, WG, GZP, PHV)                                              \
+  hipLaunchKernelGGL((mix_bwd_fused_kernel<TIO, WG, PHV>), dim3(grid),      \
                      dim3(kBlock), smem, stream,                             \
                      reinterpret_cast<const TIO*>(gy.data_ptr()),            \
                      reinterpret_cast<const TIO*>(z.data_ptr()),             \
@@ -230,9 +237,15 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
     else         MB_LAUNCH(unsigned short, false, nullptr);
   } else {
     auto gzp = want_gz ? reinterpret_cast<float*>(gz.data_ptr()) : nullptr;
-    if (want_gz) MB_LAUNCH(float, true, gzp);
+    if (want_gz) {
+      if (ph == 1) { MB_LAUNCH_P(float, true, gzp, 1); }
+      else if (ph == 3) { MB_LAUNCH_P(float, true, gzp, 3); }
+      else if (ph == 5) { MB_LAUNCH_P(float, true, gzp, 5); }
+      else { MB_LAUNCH_P(float, true, gzp, 7); }
+    }
     else         MB_LAUNCH(float, false, nullptr);
   }
+#undef MB_LAUNCH_P
 #undef MB_LAUNCH
   DFNO_CHECK_LAUNCH("mix_bwd_fused");
   return {gx, gW, gb, gz};
