@@ -71,7 +71,8 @@ characterize("two-phase 3D 64^3x30, w20, m(12,12,12,8)", 6,
              [1, 2, 64, 64, 64, 1], 30, 20, (12, 12, 12, 8), 4,
              [1, 1, 64, 64, 64, 30])
 print("""
-bf16 = bf16 activation storage + fp32 arithmetic in the native pointwise
-kernels (csrc/bf16.hip) with MFMA grad-W; the spectral/transform core runs
-fp32/complex64 behind boundary casts (docs/ROADMAP.md item 1 is the
-bf16-IO transform step to parity+).""")
+bf16 = bf16 activation storage + fp32 arithmetic everywhere: native
+pointwise kernels (csrc/bf16.hip) with MFMA grad-W, bf16-IO transforms
+(incl. the raw-bf16 glds DMA r2c), bf16-IO fused heads and the one-kernel
+trunk mix backward; the spectral core stays complex64.  At the flagship
+this config now runs AHEAD of fp32 (BASELINE.md).""")

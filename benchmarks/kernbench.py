@@ -76,6 +76,18 @@ def main():
     dt = bench(lambda: ext.proj_head_bwd(gy, x, W3, b3, W4))
     report("proj_head_bwd (gz3 write)", dt, (1 + 20 + 128) * S * 4)
 
+    # fully-fused head backward (no gz3 in HBM; reads x+gy, writes gx)
+    dt = bench(lambda: ext.proj_head_bwd_fused(gy, x, W3, b3, W4))
+    report("proj_head_bwd_fused", dt, (20 + 1 + 20) * S * 4)
+
+    # fused trunk mix backward (gz as an LDS tile; gz streamed = res grad)
+    zmix = torch.randn(1, 20, S, device=dev)
+    gymix = torch.randn(1, 20, S, device=dev)
+    Wmix = torch.randn(20, 20, device=dev) / 20
+    dt = bench(lambda: ext.channel_mix_bwd_fused(gymix, zmix, x, Wmix,
+                                                 False, True))
+    report("channel_mix_bwd_fused (+gz out)", dt, (3 * 20 + 2 * 20) * S * 4)
+
     # add gelu
     a = torch.randn(1, 20, S, device=dev)
     bb = torch.randn(1, 20, S, device=dev)
