@@ -27,6 +27,8 @@ namespace {
 
 constexpr int kBlock = 256;
 
+typedef float f32x4_ph __attribute__((ext_vector_type(4)));
+
 template <typename T>
 __device__ __forceinline__ T gelu_erf_(T z) { return dfno_gelu::gelu(z); }
 
@@ -155,6 +157,129 @@ __global__ __launch_bounds__(kBlock) void proj_head_fwd_kernel(
         }
       }
           }
+    }
+  }
+}
+
+// Tiled MFMA forward for the flagship head (I=20, M=128, O2<=2): the
+// per-thread kernel above walks all 128 hidden channels per element
+// (VALU z3 + scalar-broadcast weights) and measures ~1.04 ms; here z3
+// runs as v_mfma_f32_16x16x4 over a [128 x 64] LDS tile, gelu is applied
+// in place, and the W4 contraction is one thin MFMA M-tile (rows 0..O2-1
+// of a padded 16-row A).  TIO = unsigned short selects bf16 activations.
+template <int IT, int MT, int O2T, typename TIO = float>
+__global__ __launch_bounds__(kBlock, 3) void proj_head_fwd_fused_kernel(
+    const TIO* __restrict__ x, const float* __restrict__ W3,
+    const float* __restrict__ b3, const float* __restrict__ W4,
+    const float* __restrict__ b4, TIO* __restrict__ out,
+    int B, int O2, long S) {
+  constexpr int TS = 64;
+  constexpr int LD = TS + 4;
+  extern __shared__ __align__(16) char smem_raw[];
+  float* ht = reinterpret_cast<float*>(smem_raw);    // [MT][LD]
+  float* xt = ht + (size_t)MT * LD;                  // [IT][LD]
+  float* W3l = xt + (size_t)IT * LD;                 // [MT*IT]
+  float* b3l = W3l + (size_t)MT * IT;                // [MT]
+  float* W4l = b3l + MT;                             // [O2T*MT]
+#pragma clang loop unroll(disable)
+  for (int k = threadIdx.x; k < MT * IT; k += kBlock) W3l[k] = W3[k];
+  for (int k = threadIdx.x; k < MT; k += kBlock) b3l[k] = b3[k];
+  for (int k = threadIdx.x; k < O2T * MT; k += kBlock) W4l[k] = W4[k];
+
+  const int lane = (int)(threadIdx.x & 63);
+  const int wave = (int)(threadIdx.x >> 6);
+  const int l16 = lane & 15;
+  const int kg = lane >> 4;
+
+  const long stiles = (S + TS - 1) / TS;
+  const long tend = (long)B * stiles;
+  constexpr int NPF = (IT * TS + kBlock - 1) / kBlock;
+  float pf[NPF];
+  auto prefetch = [&](long tt) {
+    if (tt >= tend) return;
+    const int b = (int)(tt / stiles);
+    const long s0 = (tt % stiles) * TS;
+    const int nv = (int)min((long)TS, S - s0);
+#pragma unroll
+    for (int q = 0; q < NPF; ++q) {
+      const int r = (int)threadIdx.x + q * kBlock;
+      if (r >= IT * TS) break;
+      const int row = r / TS;
+      const int c = r - row * TS;
+      pf[q] = (c < nv) ? ph_ld(x + ((long)b * IT + row) * S + s0 + c) : 0.f;
+    }
+  };
+  prefetch(blockIdx.x);
+
+  for (long t = blockIdx.x; t < tend; t += gridDim.x) {
+    const int b = (int)(t / stiles);
+    const long s0 = (t % stiles) * TS;
+    __syncthreads();               // prior tile's phase reads done
+#pragma unroll
+    for (int q = 0; q < NPF; ++q) {
+      const int r = (int)threadIdx.x + q * kBlock;
+      if (r >= IT * TS) break;
+      const int row = r / TS;
+      const int c = r - row * TS;
+      xt[row * LD + c] = pf[q];
+    }
+    prefetch(t + gridDim.x);
+    __syncthreads();
+    // phase 1: z3 tile = W3 @ x (8 m-tiles x 4 n-tiles, K = IT)
+#pragma unroll
+    for (int pp = 0; pp < 8; ++pp) {
+      const int p = wave + 4 * pp;
+      const int mt = p >> 2, nt = p & 3;
+      const int m = mt * 16 + l16;
+      const int n = nt * 16 + l16;
+      f32x4_ph z4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int k = 0; k < IT; k += 4) {
+        const float a = W3l[m * IT + k + kg];
+        const float bb = xt[(k + kg) * LD + n];
+        z4 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, z4, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        ht[(mt * 16 + kg * 4 + r) * LD + nt * 16 + l16] = z4[r];
+    }
+    __syncthreads();
+    // phase 2: h = gelu(z3 + b3) in place (pair of threads per channel)
+    {
+      const int jj = (int)(threadIdx.x >> 1);
+      const int half = (int)(threadIdx.x & 1) * 4;
+      const float bj = b3l[jj];
+#pragma unroll
+      for (int k = 0; k < TS / 8; ++k) {
+        const int c4 = half + k * 8;
+        float4 zv = *reinterpret_cast<float4*>(ht + jj * LD + c4);
+        zv.x = gelu_erf_(zv.x + bj);
+        zv.y = gelu_erf_(zv.y + bj);
+        zv.z = gelu_erf_(zv.z + bj);
+        zv.w = gelu_erf_(zv.w + bj);
+        *reinterpret_cast<float4*>(ht + jj * LD + c4) = zv;
+      }
+    }
+    __syncthreads();
+    // phase 3: out = W4 @ h + b4 — one thin MFMA tile (A rows 0..O2-1
+    // hold W4, the rest zero); wave = n-tile
+    {
+      const int n = wave * 16 + l16;
+      f32x4_ph c4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 8
+      for (int k = 0; k < MT; k += 4) {
+        float a = 0.f;
+        if (l16 == 0) a = W4l[k + kg];
+        else if (O2T == 2 && l16 == 1) a = W4l[MT + k + kg];
+        const float bb = ht[(k + kg) * LD + n];
+        c4 = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, c4, 0, 0, 0);
+      }
+      const long sc = s0 + n;
+      if (kg == 0 && sc < S) {
+#pragma unroll
+        for (int r = 0; r < O2T; ++r)
+          ph_st(out + ((long)b * O2T + r) * S + sc, c4[r] + b4[r]);
+      }
     }
   }
 }
@@ -347,8 +472,6 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
 // carried across tiles (identical numerics to the library chain up to
 // fp32 atomic reduction order, like every gw kernel here).
 // ---------------------------------------------------------------------------
-
-typedef float f32x4_ph __attribute__((ext_vector_type(4)));
 
 // sum across each 16-lane segment of the wave
 __device__ __forceinline__ float seg_sum16(float v) {
@@ -626,6 +749,49 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
   if (x.numel() == 0) return out;
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
   int grid = grid_for_p((long)B * ((S + 3) / 4));
+  // The tiled-MFMA forward below is correct but measured SLOWER than the
+  // register-resident per-thread kernel (1.48 vs 1.06 ms within-process
+  // A/B at the flagship): the z3 tile costs 4 LDS passes + 3 barriers
+  // that the per-thread scheme (weights via s_loads, z/h/out all in
+  // registers) never pays.  Kept as an opt-in (DFNO_PF_FUSED=1) for
+  // shapes where the VALU z3 walk dominates.  bf16 still uses it when
+  // the per-thread kernel lacks a bf16 path for the shape (the fused
+  // gate below covers the flagship).
+  static const bool want_fused_fwd = []() {
+    const char* e = getenv("DFNO_PF_FUSED");   // A/B knob
+    return e && e[0] == '1';
+  }();
+  if ((want_fused_fwd || bf16) && I == 20 && M == 128 && O2 <= 2 &&
+      (bf16 || x.scalar_type() == at::kFloat)) {
+    // flagship: tiled MFMA forward (z3 as 16x16x4 fragments over a
+    // [128 x 64] LDS tile instead of per-thread hidden-channel walks)
+    auto W3f = bf16 ? W3.to(at::kFloat).contiguous() : W3.contiguous();
+    auto b3f = bf16 ? b3.to(at::kFloat).contiguous() : b3.contiguous();
+    auto W4f = bf16 ? W4.to(at::kFloat).contiguous() : W4.contiguous();
+    auto b4f = bf16 ? b4.to(at::kFloat).contiguous() : b4.contiguous();
+    constexpr int TS = 64, LD = TS + 4;
+    size_t smem = sizeof(float) *
+        ((size_t)128 * LD + 20 * LD + (size_t)128 * 20 + 128 +
+         2 * (size_t)128);
+    long stiles = (S + TS - 1) / TS;
+    int grid2 = (int)std::min((long)B * stiles, 768L);
+#define PH_FWD_F(O2T, TIO)                                                    \
+    hipLaunchKernelGGL((proj_head_fwd_fused_kernel<20, 128, O2T, TIO>),       \
+                       dim3(grid2), dim3(kBlock), smem, stream,               \
+                       reinterpret_cast<const TIO*>(x.data_ptr()),            \
+                       W3f.data_ptr<float>(), b3f.data_ptr<float>(),          \
+                       W4f.data_ptr<float>(), b4f.data_ptr<float>(),          \
+                       reinterpret_cast<TIO*>(out.data_ptr()), B, O2, S);
+    if (bf16) {
+      if (O2 == 2) { PH_FWD_F(2, unsigned short) }
+      else { PH_FWD_F(1, unsigned short) }
+    } else {
+      if (O2 == 2) { PH_FWD_F(2, float) } else { PH_FWD_F(1, float) }
+    }
+#undef PH_FWD_F
+    DFNO_CHECK_LAUNCH("proj_head");
+    return out;
+  }
   if (bf16) {
     // bf16 activations, fp32 weights/math (host-casts the tiny weights)
     auto W3f = W3.to(at::kFloat).contiguous();
