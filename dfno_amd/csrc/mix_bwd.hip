@@ -42,14 +42,13 @@ __device__ __forceinline__ void mb_st(unsigned short* p, float v) {
 // TIO = unsigned short selects bf16 activations (fp32 compute; W and the
 // gW/gb accumulators stay fp32).  PH: phase mask for perf bisection
 // (1 staging+gz, +2 gx, +4 gW); production uses 7.
-template <typename TIO, bool WG, int PH = 7>
+template <typename TIO, bool WG, int PH = 7, int TS = 64>
 __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
     const TIO* __restrict__ gy, const TIO* __restrict__ z,
     const TIO* __restrict__ x, const float* __restrict__ W,
     TIO* __restrict__ gx, float* __restrict__ gW, float* __restrict__ gb,
     TIO* __restrict__ gzout, int B, long S) {
   constexpr int C = 20;            // in = out channels (trunk width)
-  constexpr int TS = 64;           // s-columns per tile (16 MFMA K-steps)
   constexpr int LD = TS + 4;
   extern __shared__ __align__(16) char smem_raw[];
   float* gzt = reinterpret_cast<float*>(smem_raw);   // [C][LD]
@@ -121,9 +120,9 @@ __global__ __launch_bounds__(kBlock, 4) void mix_bwd_fused_kernel(
     // phase A: gx = W^T @ gz.  A[m=i][k=o] = W[o*C+i] (per-lane global,
     // L1-resident), B[n=c][k=o] = gzt[o][c]; K = C padded to 24.
 #pragma unroll
-    for (int pp = 0; pp < ((PH & 2) ? 2 : 0); ++pp) {
+    for (int pp = 0; pp < ((PH & 2) ? (2 * (TS / 16)) / 4 : 0); ++pp) {
       const int p = wave + 4 * pp;
-      const int mt = p >> 2, nt = p & 3;
+      const int mt = p / (TS / 16), nt = p % (TS / 16);
       const int m = mt * 16 + l16;
       const int n = nt * 16 + l16;
       const bool av = m < C;
@@ -211,8 +210,12 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
   if (x.numel() == 0) return {gx, gW, gb, gz};
 
   auto stream = at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
-  constexpr int TS = 64, LD = TS + 4;
-  size_t smem = sizeof(float) * 2 * 20 * LD;
+  static const bool ts128 = []() {
+    const char* e = getenv("DFNO_MIX_TS128");    // tile-size A/B knob
+    return e && e[0] == '1';
+  }();
+  const int TS = ts128 ? 128 : 64;
+  size_t smem = sizeof(float) * 2 * 20 * (TS + 4);
   long stiles = (S + TS - 1) / TS;
   int grid = (int)std::min((long)B * stiles, 1024L);
   float* gbp = want_bias ? gb.data_ptr<float>() : nullptr;
@@ -220,9 +223,10 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
     const char* e = getenv("DFNO_MIX_PHASES");   // perf-bisect knob
     return e ? atoi(e) : 7;
   }();
-#define MB_LAUNCH(TIO, WG, GZP) MB_LAUNCH_P(TIO, WG, GZP, 7)
-#define MB_LAUNCH_P(TIO, WG, GZP, PHV)                                              \
-  hipLaunchKernelGGL((mix_bwd_fused_kernel<TIO, WG, PHV>), dim3(grid),      \
+#define MB_LAUNCH(TIO, WG, GZP) MB_LAUNCH_T(TIO, WG, GZP, 7, 64)
+#define MB_LAUNCH_P(TIO, WG, GZP, PHV) MB_LAUNCH_T(TIO, WG, GZP, PHV, 64)
+#define MB_LAUNCH_T(TIO, WG, GZP, PHV, TSV)                                  \
+  hipLaunchKernelGGL((mix_bwd_fused_kernel<TIO, WG, PHV, TSV>), dim3(grid), \
                      dim3(kBlock), smem, stream,                             \
                      reinterpret_cast<const TIO*>(gy.data_ptr()),            \
                      reinterpret_cast<const TIO*>(z.data_ptr()),             \
@@ -238,13 +242,15 @@ std::vector<at::Tensor> channel_mix_bwd_fused(const at::Tensor& gy,
   } else {
     auto gzp = want_gz ? reinterpret_cast<float*>(gz.data_ptr()) : nullptr;
     if (want_gz) {
-      if (ph == 1) { MB_LAUNCH_P(float, true, gzp, 1); }
-      else if (ph == 3) { MB_LAUNCH_P(float, true, gzp, 3); }
-      else if (ph == 5) { MB_LAUNCH_P(float, true, gzp, 5); }
-      else { MB_LAUNCH_P(float, true, gzp, 7); }
+      if (ts128) { MB_LAUNCH_T(float, true, gzp, 7, 128); }
+      else if (ph == 1) { MB_LAUNCH_T(float, true, gzp, 1, 64); }
+      else if (ph == 3) { MB_LAUNCH_T(float, true, gzp, 3, 64); }
+      else if (ph == 5) { MB_LAUNCH_T(float, true, gzp, 5, 64); }
+      else { MB_LAUNCH_T(float, true, gzp, 7, 64); }
     }
     else         MB_LAUNCH(float, false, nullptr);
   }
+#undef MB_LAUNCH_T
 #undef MB_LAUNCH_P
 #undef MB_LAUNCH
   DFNO_CHECK_LAUNCH("mix_bwd_fused");
