@@ -900,21 +900,40 @@ __global__ __launch_bounds__(LB) void dft_c2r_last_kernel(
     }
     __syncthreads();
     if constexpr (std::is_same<TO, unsigned short>::value) {
-      // bf16 output (no accumulate operand on this path)
+      // bf16 output; acc (when set) is a packed-bf16 tensor of the output
+      // shape reinterpreted through the fp32-typed parameter — the stash
+      // fused residual-grad accumulate for the bf16 model
+      const unsigned short* ab = reinterpret_cast<const unsigned short*>(acc);
       const long base = l0 * N;
       if ((nl * N) % 8 == 0 && (base % 8 == 0) &&
-          ((reinterpret_cast<uintptr_t>(out) & 15) == 0)) {
+          ((reinterpret_cast<uintptr_t>(out) & 15) == 0) &&
+          (ab == nullptr ||
+           ((reinterpret_cast<uintptr_t>(ab) & 15) == 0))) {
         for (int idx = threadIdx.x * 8; idx < nl * N; idx += LB * 8) {
+          float v[8];
+#pragma unroll
+          for (int e = 0; e < 8; ++e) v[e] = (float)tile[idx + e];
+          if (ab != nullptr) {
+            const uint4 a8 = *reinterpret_cast<const uint4*>(ab + base + idx);
+            const unsigned int w[4] = {a8.x, a8.y, a8.z, a8.w};
+#pragma unroll
+            for (int q = 0; q < 4; ++q) {
+              v[2 * q] += dft_b2f((unsigned short)(w[q] & 0xffffu));
+              v[2 * q + 1] += dft_b2f((unsigned short)(w[q] >> 16));
+            }
+          }
           uint4 raw;
-          raw.x = dft_f2b2((float)tile[idx], (float)tile[idx + 1]);
-          raw.y = dft_f2b2((float)tile[idx + 2], (float)tile[idx + 3]);
-          raw.z = dft_f2b2((float)tile[idx + 4], (float)tile[idx + 5]);
-          raw.w = dft_f2b2((float)tile[idx + 6], (float)tile[idx + 7]);
+          raw.x = dft_f2b2(v[0], v[1]);
+          raw.y = dft_f2b2(v[2], v[3]);
+          raw.z = dft_f2b2(v[4], v[5]);
+          raw.w = dft_f2b2(v[6], v[7]);
           *reinterpret_cast<uint4*>(out + base + idx) = raw;
         }
       } else {
         for (int idx = threadIdx.x; idx < nl * N; idx += LB) {
-          __hip_bfloat16 h = __float2bfloat16((float)tile[idx]);
+          float v = (float)tile[idx];
+          if (ab != nullptr) v += dft_b2f(ab[base + idx]);
+          __hip_bfloat16 h = __float2bfloat16(v);
           out[base + idx] = *reinterpret_cast<unsigned short*>(&h);
         }
       }
@@ -1258,8 +1277,12 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
   auto sizes = y.sizes().vec();
   sizes[dim] = N;
   if (out_bf16) {
-    TORCH_CHECK(y.scalar_type() == at::kComplexFloat && !accum.defined(),
-                "dft_c2r: bf16 output needs c64 input, no accumulate");
+    TORCH_CHECK(y.scalar_type() == at::kComplexFloat,
+                "dft_c2r: bf16 output needs c64 input");
+    TORCH_CHECK(!accum.defined() || accum.numel() == 0 ||
+                (accum.scalar_type() == at::kBFloat16 &&
+                 accum.is_contiguous()),
+                "dft_c2r: bf16 accumulate must be contiguous bf16");
   }
   auto out = at::empty(sizes, y.options().dtype(
       out_bf16 ? at::kBFloat16
@@ -1277,6 +1300,11 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
     // bf16-output writeback (fp32 compute/LDS tile), flagship NT folding kept
     auto inp = reinterpret_cast<const float*>(y.data_ptr());
     auto op = reinterpret_cast<unsigned short*>(out.data_ptr());
+    const float* accb = nullptr;               // packed bf16 through T*
+    if (accum.defined() && accum.numel() > 0) {
+      TORCH_CHECK(accum.numel() == out.numel(), "dft_c2r: accumulate shape");
+      accb = reinterpret_cast<const float*>(accum.data_ptr());
+    }
 #define C2RB(MC, NTV, LBV)                                                     \
     { long nt2 = (lines + LBV - 1) / LBV;                                      \
       int grid2 = (int)std::min(nt2, LBV == kBlock ? 4096L : 2048L);           \
@@ -1285,7 +1313,7 @@ static at::Tensor dft_c2r_impl(const at::Tensor& y, int64_t dim, int64_t n_out,
           (dft_c2r_last_kernel<float, MC, NTV, LBV, unsigned short>),          \
           dim3(grid2), dim3(LBV), smem2, stream, inp, op,                      \
           tw.data_ptr<float>(), lines, N, (int)m, (float)scale, factors,       \
-          nullptr); }
+          accb); }
 #define C2RB_M(NTV, LBV)                                                       \
     if (m <= 8) { C2RB(8, NTV, LBV) } else if (m <= 16) { C2RB(16, NTV, LBV) } \
     else if (m <= 24) { C2RB(24, NTV, LBV) } else { C2RB(32, NTV, LBV) }
@@ -1386,9 +1414,10 @@ at::Tensor dft_pad_irfft_bf16(const at::Tensor& y, int64_t dim, int64_t n_out,
 }
 
 at::Tensor dft_rfft_trunc_adj_bf16(const at::Tensor& gy, int64_t dim,
-                                   int64_t n) {
-  // bf16-output adjoint (the bf16 model's rfft input gradient)
-  return dft_c2r_impl(gy, dim, n, 1.0, /*factors=*/false, at::Tensor(),
+                                   int64_t n, const at::Tensor& accum) {
+  // bf16-output adjoint (the bf16 model's rfft input gradient); accum is
+  // the optional stashed residual gradient fused into the writeback
+  return dft_c2r_impl(gy, dim, n, 1.0, /*factors=*/false, accum,
                       /*out_bf16=*/true);
 }
 

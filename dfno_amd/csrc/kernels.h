@@ -109,7 +109,8 @@ at::Tensor dft_zt_inv(const at::Tensor& y, int64_t Z, int64_t T,
 // bf16-IO variants (bf16 real-side storage, fp32 compute, c64 spectrum):
 at::Tensor dft_pad_irfft_bf16(const at::Tensor& y, int64_t dim, int64_t n_out,
                               int64_t m);
-at::Tensor dft_rfft_trunc_adj_bf16(const at::Tensor& gy, int64_t dim, int64_t n);
+at::Tensor dft_rfft_trunc_adj_bf16(const at::Tensor& gy, int64_t dim,
+                                   int64_t n, const at::Tensor& accum);
 
 // bf16-storage pointwise kernels (bf16.hip; fp32 arithmetic):
 std::vector<at::Tensor> bf16_channel_mix(const at::Tensor& x, const at::Tensor& W,

@@ -108,7 +108,8 @@ class _RfftTruncFn(torch.autograd.Function):
     def backward(ctx, gy):
         ext = _ext.get(required=True)
         if ctx.bf16:
-            gx = ext.dft_rfft_trunc_adj_bf16(gy.contiguous(), ctx.dim, ctx.n)
+            gx = ext.dft_rfft_trunc_adj_bf16(gy.contiguous(), ctx.dim,
+                                             ctx.n, _zt_empty(gy.device))
         else:
             gx = ext.dft_rfft_trunc_adj(gy.contiguous(), ctx.dim, ctx.n)
         return gx, None, None
@@ -211,6 +212,7 @@ class _RfftTruncStashFn(torch.autograd.Function):
     def forward(ctx, x, dim, m, key):
         ext = _ext.get(required=True)
         ctx.dim, ctx.m, ctx.n, ctx.key = dim, m, x.shape[dim], key
+        ctx.bf16 = x.dtype == torch.bfloat16
         y = ext.dft_rfft_trunc(x.contiguous(), dim, m)
         tok = torch.empty(0, device=x.device)
         return y, tok
@@ -223,7 +225,12 @@ class _RfftTruncStashFn(torch.autograd.Function):
             raise RuntimeError(
                 "rfft stash: epilogue gradient missing (stash backward did "
                 "not run before the rfft adjoint)")
-        gx = ext.dft_rfft_trunc_adj_acc(gy.contiguous(), ctx.dim, ctx.n, acc)
+        if ctx.bf16:
+            gx = ext.dft_rfft_trunc_adj_bf16(gy.contiguous(), ctx.dim,
+                                             ctx.n, acc)
+        else:
+            gx = ext.dft_rfft_trunc_adj_acc(gy.contiguous(), ctx.dim,
+                                            ctx.n, acc)
         return gx, None, None, None
 
 
@@ -239,10 +246,15 @@ def rfft_trunc_stash(x, dim, m, key):
 def stash_fusable(x, dim, m) -> bool:
     d = dim % x.dim()
     m = min(m, x.shape[d] // 2 + 1)
-    return (d == x.dim() - 1 and x.is_cuda and x.numel() > 0
+    if not (d == x.dim() - 1 and x.is_cuda and x.numel() > 0
             and torch.is_grad_enabled() and x.requires_grad
-            and x.shape[d] <= _MAX_N and m <= 32
-            and x.dtype in (torch.float32, torch.float64))
+            and x.shape[d] <= _MAX_N and m <= 32):
+        return False
+    if x.dtype == torch.bfloat16:
+        # bf16 stash: bf16-IO r2c forward + bf16 adjoint with the packed
+        # bf16 accumulate operand (the epilogue grad is bf16)
+        return rfft_bf16_native_ok(x, d, m)
+    return x.dtype in (torch.float32, torch.float64)
 
 
 # ---------------------------------------------------------------------------

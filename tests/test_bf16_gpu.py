@@ -308,3 +308,55 @@ def test_bf16_lift_head_fused():
     for a, b in [(W1, W1r), (b1, b1r), (W2, W2r), (b2, b2r)]:
         assert torch.allclose(a.grad.float(), b.grad, rtol=5e-2, atol=5e-1), \
             f"wgrad {(a.grad.float()-b.grad).abs().max()}"
+
+
+def test_bf16_rfft_adj_accumulate():
+    """bf16 c2r adjoint with the packed-bf16 accumulate operand (the bf16
+    stash path) vs unfused adjoint + add."""
+    from dfno_amd import _ext
+    ext = _ext.get(required=True)
+    torch.manual_seed(13)
+    for shape, n in [((2, 6, 64, 30), 30), ((1, 3, 17, 31), 31)]:
+        m = 8
+        gy = torch.randn(*shape[:-1], m, dtype=torch.complex64, device="cuda")
+        acc = torch.randn(*shape, device="cuda", dtype=torch.bfloat16)
+        fused = ext.dft_rfft_trunc_adj_bf16(gy, len(shape) - 1, n, acc)
+        base = ext.dft_rfft_trunc_adj_bf16(
+            gy, len(shape) - 1, n, torch.empty(0, dtype=torch.bfloat16,
+                                               device="cuda"))
+        ref = (base.float() + acc.float()).bfloat16()
+        assert fused.dtype == torch.bfloat16
+        assert torch.allclose(fused.float(), ref.float(), rtol=2e-2,
+                              atol=2e-2), \
+            f"{(fused.float()-ref.float()).abs().max()}"
+
+
+def test_bf16_model_stash_grads():
+    """bf16 serial model grads with the stash active match fp32 (the
+    residual-grad add now folds into the bf16 c2r adjoint writeback)."""
+    import dfno_amd as dfno
+    torch.manual_seed(21)
+    _, P_x, _ = dfno.create_standard_partitions((1, 1, 1, 1, 1, 1))
+    in_shape = [1, 2, 8, 8, 32, 1]
+    ref = dfno.DistributedFNONd(P_x, in_shape, 16, 8, (3, 3, 8, 6),
+                                num_blocks=2, device=torch.device("cuda"))
+    bf = dfno.DistributedFNONd(P_x, in_shape, 16, 8, (3, 3, 8, 6),
+                               num_blocks=2, device=torch.device("cuda"),
+                               dtype=torch.bfloat16)
+    sd = {k: (v.bfloat16() if v.is_floating_point() and v.dtype == torch.float32
+              else v) for k, v in ref.state_dict().items()}
+    bf.load_state_dict(sd)
+    x = torch.rand(*in_shape, device="cuda")
+    yr = ref(x)
+    yb = bf(x.bfloat16())
+    yr.square().mean().backward()
+    yb.float().square().mean().backward()
+    for (n, pr), (_, pb) in zip(ref.named_parameters(), bf.named_parameters()):
+        if pr.grad is None or pr.grad.numel() == 0:
+            continue
+        a, b = pr.grad.float(), pb.grad.float()
+        if a.is_complex():
+            a, b = torch.view_as_real(a), torch.view_as_real(b)
+        scale = a.abs().max().clamp_min(1e-6)
+        assert (a - b).abs().max() / scale < 0.12, \
+            f"{n}: rel {(a-b).abs().max()/scale}"
