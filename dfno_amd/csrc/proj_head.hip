@@ -753,15 +753,14 @@ at::Tensor proj_head_fwd(const at::Tensor& x, const at::Tensor& W3,
   // register-resident per-thread kernel (1.48 vs 1.06 ms within-process
   // A/B at the flagship): the z3 tile costs 4 LDS passes + 3 barriers
   // that the per-thread scheme (weights via s_loads, z/h/out all in
-  // registers) never pays.  Kept as an opt-in (DFNO_PF_FUSED=1) for
-  // shapes where the VALU z3 walk dominates.  bf16 still uses it when
-  // the per-thread kernel lacks a bf16 path for the shape (the fused
-  // gate below covers the flagship).
+  // registers) never pays; the per-thread bf16-IO variant likewise wins
+  // for bf16 (1.15 vs ~1.6 ms).  Kept as an opt-in (DFNO_PF_FUSED=1)
+  // for shapes where the VALU z3 walk dominates.
   static const bool want_fused_fwd = []() {
     const char* e = getenv("DFNO_PF_FUSED");   // A/B knob
     return e && e[0] == '1';
   }();
-  if ((want_fused_fwd || bf16) && I == 20 && M == 128 && O2 <= 2 &&
+  if (want_fused_fwd && I == 20 && M == 128 && O2 <= 2 &&
       (bf16 || x.scalar_type() == at::kFloat)) {
     // flagship: tiled MFMA forward (z3 as 16x16x4 fragments over a
     // [128 x 64] LDS tile instead of per-thread hidden-channel walks)
