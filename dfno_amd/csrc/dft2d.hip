@@ -1,5 +1,8 @@
 // Fused (z,t) boundary 2-D transform (docs/ROADMAP.md item: the pencil
-// chain's trailing pair).  The flagship's m-phase runs rfft_trunc(t) then
+// chain's trailing pair).  Reference semantics: the torch.fft rfftn/fft
+// pair over the P_m-local trailing dims of the pencil block
+// (/root/reference/dfno/dfno.py:224-241, the "compute spectral
+// representation" stage).  The flagship's m-phase runs rfft_trunc(t) then
 // fft_trunc(z) as two full-activation passes with a [L, Z, mt] complex
 // intermediate (0.34 GB at the flagship, written+read 16 times per step
 // across fwd/inv/adjoints).  These kernels compute the truncated 2-D

@@ -1,5 +1,9 @@
 // Repartition pack/unpack kernels ("pack/unpack via HIP
-// kernels").  A repartition plan's send side packs every block-intersection
+// kernels").  Reference counterpart: DistDL's Repartition moves each
+// block-intersection as an individual tensor slice through MPI (used at
+// /root/reference/dfno/dfno.py:51-60); here every box is gathered into
+// ONE flat staging buffer per peer (and scattered back) so the RCCL
+// exchange sends a single contiguous message per rank pair.  A repartition plan's send side packs every block-intersection
 // box of the (contiguous) local tensor into ONE flat staging buffer (per-peer
 // contiguous ranges are sliced off for grouped ncclSend), and the recv side
 // scatters one flat buffer into the destination block — one kernel launch per
