@@ -495,8 +495,12 @@ class _PackedMeta:
 
     def __init__(self, plan: "_RepartitionPlan", dtype: torch.dtype, device):
         wpe = 2 if dtype.is_complex else 1
-        self.word_dtype = (torch.float64 if dtype in (torch.complex128, torch.float64)
-                           else torch.float32)
+        if dtype in (torch.complex128, torch.float64):
+            self.word_dtype = torch.float64
+        elif dtype == torch.bfloat16:
+            self.word_dtype = torch.bfloat16   # 2-byte words (pack.hip ushort)
+        else:
+            self.word_dtype = torch.float32
         src_shape = None
         if plan.P_src.active:
             sb = block_bounds(plan.P_src, plan.gshape, plan.P_src.rank)
@@ -536,9 +540,9 @@ def _issue_plan_packed(plan: "_RepartitionPlan", x: torch.Tensor,
     ext = _ext.get(required=False)
     if ext is None:
         return None
-    if dtype not in (torch.float32, torch.float64,
+    if dtype not in (torch.float32, torch.float64, torch.bfloat16,
                      torch.complex64, torch.complex128):
-        return None   # bf16 exchanges use the slicing path (2-byte words)
+        return None   # unsupported word size: slicing path
     if not x.is_contiguous():
         x = x.contiguous()
 

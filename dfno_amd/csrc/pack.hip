@@ -90,8 +90,17 @@ void pack_boxes(const at::Tensor& src, at::Tensor& flat,
     hipLaunchKernelGGL((copy_boxes_kernel<double, true>), grid, dim3(kTPB), 0, stream,
                        src.data_ptr<double>(), nullptr, nullptr,
                        flat.data_ptr<double>(), desc.data_ptr<long>());
+  } else if (src.scalar_type() == at::kBFloat16) {
+    // bf16 exchanges: 2-byte words (the descriptors count elements)
+    hipLaunchKernelGGL((copy_boxes_kernel<unsigned short, true>), grid,
+                       dim3(kTPB), 0, stream,
+                       reinterpret_cast<const unsigned short*>(src.data_ptr()),
+                       nullptr, nullptr,
+                       reinterpret_cast<unsigned short*>(flat.data_ptr()),
+                       desc.data_ptr<long>());
   } else {
-    TORCH_CHECK(src.scalar_type() == at::kFloat, "pack_boxes: fp32/fp64 words only");
+    TORCH_CHECK(src.scalar_type() == at::kFloat,
+                "pack_boxes: fp32/fp64/bf16 words only");
     hipLaunchKernelGGL((copy_boxes_kernel<float, true>), grid, dim3(kTPB), 0, stream,
                        src.data_ptr<float>(), nullptr, nullptr,
                        flat.data_ptr<float>(), desc.data_ptr<long>());
@@ -111,8 +120,15 @@ void unpack_boxes(const at::Tensor& flat, at::Tensor& dst,
     hipLaunchKernelGGL((copy_boxes_kernel<double, false>), grid, dim3(kTPB), 0, stream,
                        nullptr, dst.data_ptr<double>(),
                        flat.data_ptr<double>(), nullptr, desc.data_ptr<long>());
+  } else if (dst.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((copy_boxes_kernel<unsigned short, false>), grid,
+                       dim3(kTPB), 0, stream, nullptr,
+                       reinterpret_cast<unsigned short*>(dst.data_ptr()),
+                       reinterpret_cast<const unsigned short*>(flat.data_ptr()),
+                       nullptr, desc.data_ptr<long>());
   } else {
-    TORCH_CHECK(dst.scalar_type() == at::kFloat, "unpack_boxes: fp32/fp64 words only");
+    TORCH_CHECK(dst.scalar_type() == at::kFloat,
+                "unpack_boxes: fp32/fp64/bf16 words only");
     hipLaunchKernelGGL((copy_boxes_kernel<float, false>), grid, dim3(kTPB), 0, stream,
                        nullptr, dst.data_ptr<float>(),
                        flat.data_ptr<float>(), nullptr, desc.data_ptr<long>());

@@ -114,13 +114,14 @@ def test_nccl_backend_single_rank_paths():
 
 
 def test_pack_unpack_boxes_roundtrip_gpu():
-    """HIP pack/unpack kernels against aten slicing (fp32 + complex64)."""
+    """HIP pack/unpack kernels against aten slicing (fp32/c64/fp64/bf16)."""
     from dfno_amd import _ext
     from dfno_amd.comm import _box_record, _PACK_REC
 
     ext = _ext.get(required=True)
     torch.manual_seed(5)
-    for dtype in (torch.float32, torch.complex64, torch.float64):
+    for dtype in (torch.float32, torch.complex64, torch.float64,
+                  torch.bfloat16):
         wpe = 2 if dtype.is_complex else 1
         shape = (2, 5, 9, 7, 6)
         x = (torch.randn(*shape, dtype=dtype, device="cuda")
@@ -139,7 +140,9 @@ def test_pack_unpack_boxes_roundtrip_gpu():
             mx = max(mx, n)
         desc = torch.tensor(recs, dtype=torch.int64, device="cuda")
         assert desc.shape[1] == _PACK_REC
-        word = torch.float64 if dtype in (torch.float64,) else torch.float32
+        word = (torch.float64 if dtype == torch.float64
+                else torch.bfloat16 if dtype == torch.bfloat16
+                else torch.float32)
         flat = torch.empty(off, dtype=word, device="cuda")
         xw = (torch.view_as_real(x) if dtype.is_complex else x).reshape(-1)
         ext.pack_boxes(xw.contiguous(), flat, desc, mx)
