@@ -71,3 +71,56 @@ def test_standard_partitions_serial():
 def test_oversubscribed_raises():
     with pytest.raises(ValueError):
         create_standard_partitions((1, 2, 2))
+
+
+# ---------------------------------------------------------------------------
+# packed-repartition descriptor property test (CPU emulation of the
+# csrc/pack.hip index walk vs direct slicing; hypothesis-driven shapes)
+# ---------------------------------------------------------------------------
+
+from hypothesis import given, settings, strategies as st
+
+
+def _emulate_pack(x_words, rec):
+    """Replicate copy_boxes_kernel's mixed-radix walk for one descriptor."""
+    flat_off, tens_off, numel, nd = rec[0], rec[1], rec[2], rec[3]
+    mdims = rec[4:12]
+    mstrs = rec[12:20]
+    out = torch.empty(numel, dtype=x_words.dtype)
+    for e in range(numel):
+        rem, off = e, tens_off
+        for k in range(7, 0, -1):
+            if k < nd:
+                idx = rem % mdims[k]
+                rem //= mdims[k]
+                off += idx * mstrs[k]
+        off += rem * mstrs[0]
+        out[e] = x_words[off]
+    return out, flat_off
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.data())
+def test_box_record_matches_slicing(data):
+    """_box_record's merged-dim descriptors address exactly the elements
+    direct slicing produces, for random shapes/boxes/word widths."""
+    from dfno_amd.comm import _box_record
+
+    ndim = data.draw(st.integers(2, 5))
+    shape = tuple(data.draw(st.integers(1, 6)) for _ in range(ndim))
+    box = []
+    for d in shape:
+        a = data.draw(st.integers(0, d - 1))
+        b = data.draw(st.integers(a + 1, d))
+        box.append((a, b))
+    wpe = data.draw(st.sampled_from([1, 2]))
+
+    torch.manual_seed(0)
+    x = torch.randn(*shape, 2)[..., :wpe].contiguous()   # [..., wpe] words
+    x_words = x.reshape(-1)
+    rec, n = _box_record(shape, box, wpe, flat_off=0)
+    assert len(rec) == 20
+    got, _ = _emulate_pack(x_words, rec)
+    ref = x[tuple(slice(a, b) for a, b in box)].reshape(-1)
+    assert n == ref.numel()
+    assert torch.equal(got, ref)
