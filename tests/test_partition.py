@@ -124,3 +124,23 @@ def test_box_record_matches_slicing(data):
     ref = x[tuple(slice(a, b) for a, b in box)].reshape(-1)
     assert n == ref.numel()
     assert torch.equal(got, ref)
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.integers(0, 200), st.integers(1, 16))
+def test_balanced_splits_tile_exactly(n, p):
+    """The balanced split must cover [0, n) contiguously with p pieces
+    whose sizes differ by at most one — every repartition plan, block
+    bound and shard shape derives from this."""
+    from dfno_amd.partition import _balanced_splits
+
+    s = _balanced_splits(n, p)
+    assert len(s) == p
+    cur = 0
+    sizes = []
+    for a, b in s:
+        assert a == cur and b >= a
+        sizes.append(b - a)
+        cur = b
+    assert cur == n
+    assert max(sizes) - min(sizes) <= 1
