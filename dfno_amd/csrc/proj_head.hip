@@ -473,13 +473,6 @@ __global__ __launch_bounds__(kBlock) void proj_head_bwd_kernel(
 // fp32 atomic reduction order, like every gw kernel here).
 // ---------------------------------------------------------------------------
 
-// sum across each 16-lane segment of the wave
-__device__ __forceinline__ float seg_sum16(float v) {
-#pragma unroll
-  for (int off = 8; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
-  return v;
-}
-
 template <int IT, int MT, int O2T, typename TIO = float>
 __global__ __launch_bounds__(kBlock, 3) void proj_head_bwd_fused_kernel(
     const TIO* __restrict__ gy, const TIO* __restrict__ x,
