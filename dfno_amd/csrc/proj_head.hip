@@ -8,12 +8,17 @@
 //
 // Here the whole head is one pass: each thread holds its x column (I<=32
 // channels) in registers, walks the M<=512 hidden channels recomputing
-// z3 = W3 x + b3 on the fly (weights in LDS, wave-uniform broadcast reads),
-// and accumulates the O2<=8 outputs.  Forward traffic = read x + write out.
-// Backward recomputes z3 from x (cheap: x is I channels) in one kernel that
-// produces grad-x, grad-b3, grad-W4, grad-b4 directly (wave-shuffle
-// reductions + one LDS/global atomic per block) and materializes gz3 for the
-// library-GEMM grad-W3 reduction.
+// z3 = W3 x + b3 on the fly (weights via wave-uniform scalar loads), and
+// accumulates the O2<=8 outputs.  Forward traffic = read x + write out.
+//
+// Backward, flagship shape (I=20, M=128, O2<=2, fp32/bf16 IO):
+// proj_head_bwd_fused_kernel — ONE kernel over 64-column S-tiles that
+// recomputes z3 via v_mfma_f32_16x16x4, produces grad-x (second MFMA
+// pair), grad-W3 (fragments carried across tiles), grad-b3/grad-W4
+// (register partials, one shfl per tile) and grad-b4; the [B,128,S]
+// hidden-grad tensor (~4 GB at the flagship) never exists in HBM.
+// Generic shapes keep the original proj_head_bwd_kernel, which DOES
+// materialize gz3 for the channel_mix grad-x / grad-W3 reductions.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
