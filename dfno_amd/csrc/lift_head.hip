@@ -10,9 +10,13 @@
 // column per grid point.  Restricted to T_in == 1 (the two-phase flagship;
 // other configs use the unfused path).
 //
-// Backward (T_in == 1) recomputes the hidden activations from x and emits
-// grad-x plus all four weight/bias grads via per-wave LDS accumulators and
-// one atomic flush per block (proj_head pattern).
+// fp32/bf16 forward+backward run the lane-per-k variants: each 64-lane
+// wave covers TWO s-points with one time column per lane, so every out
+// write / gy read is a contiguous Tn-dword run; backward's gW2/gb2
+// reduce as v_mfma_f32_16x16x4 over per-wave LDS tiles (ones-column
+// trick for gb2) with fragments carried across pairs, and gW1/gb1
+// accumulate per-lane (k is fixed).  The chunked per-thread kernels
+// below them serve fp64/odd shapes.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
