@@ -54,6 +54,13 @@ _FP8_CACHE = {}   # id(w) -> (epoch, w16, scale)
 
 def bump_quant_epoch() -> None:
     _QUANT_EPOCH[0] += 1
+    # evict entries whose master stopped refreshing (model freed): live
+    # masters touch their entry every epoch, so a lag > 8 means the id is
+    # dead and its e4m3/amax buffers would otherwise pin GPU memory forever
+    if _QUANT_EPOCH[0] % 64 == 0:
+        cutoff = _QUANT_EPOCH[0] - 8
+        for k in [k for k, ent in _FP8_CACHE.items() if ent[0] < cutoff]:
+            del _FP8_CACHE[k]
 
 
 def _fp8_weights(weights):
