@@ -55,6 +55,10 @@ def main():
     p.add_argument("--shape", type=int, nargs=4, default=(60, 60, 64, 30),
                    help="global X Y Z T")
     p.add_argument("--lr", type=float, default=1e-3)
+    p.add_argument("--heartbeat-timeout", type=float, default=0.0,
+                   metavar="SEC", help="enable store-heartbeat rank-failure "
+                   "detection: abort with RankFailure when a peer misses "
+                   "heartbeats for SEC seconds (0 = off)")
     p.add_argument("--out-dir", type=str, default="data/")
     p.add_argument("--cache-dir", type=str, default=None)
     args = p.parse_args()
@@ -126,10 +130,18 @@ def main():
         print(f"rank = {P_x.rank}, resumed epoch {start_epoch} from {cands[-1]}")
     P_x.barrier()
 
+    hb = None
+    if args.heartbeat_timeout > 0:
+        from dfno_amd.health import HeartbeatMonitor
+        hb = HeartbeatMonitor(interval=max(args.heartbeat_timeout / 8, 0.5),
+                              timeout=args.heartbeat_timeout).start()
+
     for i in range(start_epoch, args.num_epochs):
         model.train()
         train_loss, n_train_batch = 0.0, 0
         for j, (x, y) in enumerate(train_loader):
+            if hb is not None:
+                hb.check()     # fail fast (RankFailure) on a dead peer
             optimizer.zero_grad(set_to_none=True)
             t0 = time.time()
             x = x.to(device)
@@ -193,6 +205,8 @@ def main():
     print(f"rank = {P_x.rank}, saved model after final iteration: {path}")
     if P_root.active:
         print("training finished.")
+    if hb is not None:
+        hb.stop()
 
 
 if __name__ == "__main__":
